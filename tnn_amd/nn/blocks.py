@@ -1,0 +1,275 @@
+"""Blocks: layer compositions (reference include/nn/blocks_impl/).
+
+``Sequential`` is the unit of pipeline-stage deployment, exactly as in the
+reference (include/nn/blocks_impl/sequential.hpp:21): the partitioner slices
+a Sequential into per-rank stage Sequentials, which round-trip through config.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Any, Dict, List, Optional, Tuple
+
+import torch
+import torch.nn as nn
+
+from .. import ops
+from .layer import Layer, register_layer, layer_from_config
+
+
+@register_layer("sequential")
+class Sequential(Layer):
+    def __init__(self, layers: Optional[List[Layer]] = None, name: str = "sequential",
+                 dtype: torch.dtype = torch.float32):
+        super().__init__(name, dtype)
+        self.layers = nn.ModuleList(layers or [])
+
+    def forward(self, x):
+        for layer in self.layers:
+            x = layer(x)
+        return x
+
+    def output_shape(self, in_shape):
+        for layer in self.layers:
+            in_shape = layer.output_shape(in_shape)
+        return tuple(in_shape)
+
+    def flops_per_item(self, in_shape):
+        total = 0
+        for layer in self.layers:
+            total += layer.flops_per_item(in_shape)
+            in_shape = layer.output_shape(in_shape)
+        return total
+
+    def slice(self, start: int, end: int, name: Optional[str] = None) -> "Sequential":
+        """Sub-Sequential over layers [start, end) — used by the partitioner."""
+        return Sequential(list(self.layers[start:end]),
+                          name=name or f"{self.name}[{start}:{end}]")
+
+    def __len__(self):
+        return len(self.layers)
+
+    def __iter__(self):
+        return iter(self.layers)
+
+    def extra_config(self):
+        return {"layers": [l.get_config() for l in self.layers]}
+
+    @classmethod
+    def from_config(cls, cfg):
+        cfg = dict(cfg)
+        cfg.pop("type", None)
+        cfg.pop("dtype", None)
+        layers = [layer_from_config(c) for c in cfg.pop("layers", [])]
+        return cls(layers, **cfg)
+
+
+@register_layer("residual_block")
+class ResidualBlock(Layer):
+    """main + optional shortcut, add-join, final activation
+    (reference src/nn/blocks_impl/residual_block.cpp:22-80)."""
+
+    def __init__(self, main: Layer, shortcut: Optional[Layer] = None,
+                 final_activation: str = "relu", name: str = "residual_block",
+                 dtype: torch.dtype = torch.float32):
+        super().__init__(name, dtype)
+        self.main = main
+        self.shortcut = shortcut
+        self.final_activation = final_activation
+
+    def forward(self, x):
+        y = self.main(x)
+        s = self.shortcut(x) if self.shortcut is not None else x
+        y = y + s
+        if self.final_activation not in ("linear", "none"):
+            y = ops.activation(y, self.final_activation)
+        return y
+
+    def output_shape(self, in_shape):
+        return self.main.output_shape(in_shape)
+
+    def flops_per_item(self, in_shape):
+        f = self.main.flops_per_item(in_shape)
+        if self.shortcut is not None:
+            f += self.shortcut.flops_per_item(in_shape)
+        return f + math.prod(self.main.output_shape(in_shape))
+
+    def extra_config(self):
+        return {"main": self.main.get_config(),
+                "shortcut": self.shortcut.get_config() if self.shortcut else None,
+                "final_activation": self.final_activation}
+
+    @classmethod
+    def from_config(cls, cfg):
+        cfg = dict(cfg)
+        cfg.pop("type", None)
+        cfg.pop("dtype", None)
+        main = layer_from_config(cfg.pop("main"))
+        sc = cfg.pop("shortcut", None)
+        shortcut = layer_from_config(sc) if sc else None
+        return cls(main, shortcut, **cfg)
+
+
+@register_layer("msequential")
+class MSequential(Layer):
+    """Parallel branches + elementwise join (reference msequential.cpp:40-89).
+
+    The reference's memory-aware branch ordering (sort by peak-workspace −
+    retained bytes) is a workspace-allocator concern; with the caching
+    allocator we run branches in declaration order.
+    """
+
+    JOINS = {"add": torch.add, "sub": torch.sub, "mul": torch.mul,
+             "div": torch.div}
+
+    def __init__(self, branches: List[Layer], join: str = "add",
+                 name: str = "msequential", dtype: torch.dtype = torch.float32):
+        super().__init__(name, dtype)
+        self.branches = nn.ModuleList(branches)
+        self.join = join
+
+    def forward(self, x):
+        outs = [b(x) for b in self.branches]
+        if self.join == "concat":
+            return torch.cat(outs, dim=-1)
+        fn = self.JOINS[self.join]
+        y = outs[0]
+        for o in outs[1:]:
+            y = fn(y, o)
+        return y
+
+    def output_shape(self, in_shape):
+        shapes = [b.output_shape(in_shape) for b in self.branches]
+        if self.join == "concat":
+            last = sum(s[-1] for s in shapes)
+            return (*shapes[0][:-1], last)
+        return shapes[0]
+
+    def flops_per_item(self, in_shape):
+        return sum(b.flops_per_item(in_shape) for b in self.branches)
+
+    def extra_config(self):
+        return {"branches": [b.get_config() for b in self.branches], "join": self.join}
+
+    @classmethod
+    def from_config(cls, cfg):
+        cfg = dict(cfg)
+        cfg.pop("type", None)
+        cfg.pop("dtype", None)
+        branches = [layer_from_config(c) for c in cfg.pop("branches")]
+        return cls(branches, **cfg)
+
+
+class _MHABase(Layer):
+    """Shared q/k/v/o projection plumbing for attention blocks
+    (reference src/nn/blocks_impl/attention_block.cpp:144)."""
+
+    def __init__(self, dim: int, num_heads: int, causal: bool = True,
+                 name: str = "attention", dtype: torch.dtype = torch.float32):
+        super().__init__(name, dtype)
+        assert dim % num_heads == 0
+        self.dim, self.num_heads, self.causal = dim, num_heads, causal
+        self.head_dim = dim // num_heads
+        std = 0.02
+        self.wq = nn.Parameter(torch.randn(dim, dim, dtype=dtype) * std)
+        self.wk = nn.Parameter(torch.randn(dim, dim, dtype=dtype) * std)
+        self.wv = nn.Parameter(torch.randn(dim, dim, dtype=dtype) * std)
+        self.wo = nn.Parameter(torch.randn(dim, dim, dtype=dtype) * std)
+        self.bq = nn.Parameter(torch.zeros(dim, dtype=dtype))
+        self.bk = nn.Parameter(torch.zeros(dim, dtype=dtype))
+        self.bv = nn.Parameter(torch.zeros(dim, dtype=dtype))
+        self.bo = nn.Parameter(torch.zeros(dim, dtype=dtype))
+
+    def _project(self, x):
+        b, s, d = x.shape
+        h, hd = self.num_heads, self.head_dim
+        q = ops.linear(x, self.wq, self.bq).view(b, s, h, hd).transpose(1, 2)
+        k = ops.linear(x, self.wk, self.bk).view(b, s, h, hd).transpose(1, 2)
+        v = ops.linear(x, self.wv, self.bv).view(b, s, h, hd).transpose(1, 2)
+        return q, k, v
+
+    def _merge(self, o, b, s):
+        o = o.transpose(1, 2).reshape(b, s, self.dim)
+        return ops.linear(o, self.wo, self.bo)
+
+    def output_shape(self, in_shape):
+        return tuple(in_shape)
+
+    def flops_per_item(self, in_shape):
+        s, d = in_shape
+        return 8 * s * d * d + 4 * s * s * d
+
+    def extra_config(self):
+        return {"dim": self.dim, "num_heads": self.num_heads, "causal": self.causal}
+
+
+@register_layer("attention_block")
+class AttentionBlock(_MHABase):
+    """Materialized-scores attention (reference AttentionBlock)."""
+
+    def forward(self, x):
+        b, s, _ = x.shape
+        q, k, v = self._project(x)
+        scale = self.head_dim ** -0.5
+        scores = torch.matmul(q.float(), k.float().transpose(-1, -2)) * scale
+        if self.causal:
+            mask = torch.ones(s, s, dtype=torch.bool, device=x.device).triu(1)
+            scores = scores.masked_fill(mask, float("-inf"))
+        p = torch.softmax(scores, dim=-1).to(v.dtype)
+        o = torch.matmul(p, v)
+        return self._merge(o, b, s)
+
+
+@register_layer("flash_attention_block")
+class FlashAttentionBlock(_MHABase):
+    """Flash attention via the hand-written CDNA4 kernel on GPU
+    (reference FlashAttentionBlock / cudnn SDPA graph)."""
+
+    def forward(self, x):
+        b, s, _ = x.shape
+        q, k, v = self._project(x)
+        o = ops.attention(q.contiguous(), k.contiguous(), v.contiguous(),
+                          causal=self.causal)
+        return self._merge(o, b, s)
+
+
+@register_layer("gpt_block")
+class GPTBlock(Layer):
+    """Pre-LN transformer block: ln→attn→res, ln→mlp(gelu)→res
+    (reference gpt models, src/nn/example_models.cpp:384)."""
+
+    def __init__(self, dim: int, num_heads: int, mlp_ratio: int = 4,
+                 flash: bool = True, dropout: float = 0.0,
+                 name: str = "gpt_block", dtype: torch.dtype = torch.float32):
+        super().__init__(name, dtype)
+        self.dim, self.num_heads = dim, num_heads
+        self.mlp_ratio, self.flash, self.dropout_p = mlp_ratio, flash, dropout
+        from .layers import LayerNorm, Dropout
+        cls = FlashAttentionBlock if flash else AttentionBlock
+        self.ln1 = LayerNorm(dim, name=f"{name}_ln1", dtype=dtype)
+        self.attn = cls(dim, num_heads, causal=True, name=f"{name}_attn", dtype=dtype)
+        self.ln2 = LayerNorm(dim, name=f"{name}_ln2", dtype=dtype)
+        hidden = dim * mlp_ratio
+        std = 0.02
+        self.w1 = nn.Parameter(torch.randn(dim, hidden, dtype=dtype) * std)
+        self.b1 = nn.Parameter(torch.zeros(hidden, dtype=dtype))
+        self.w2 = nn.Parameter(torch.randn(hidden, dim, dtype=dtype) * std)
+        self.b2 = nn.Parameter(torch.zeros(dim, dtype=dtype))
+        self.drop = Dropout(dropout, name=f"{name}_drop", dtype=dtype) if dropout > 0 else None
+
+    def forward(self, x):
+        x = x + self.attn(self.ln1(x))
+        h = ops.linear(self.ln2(x), self.w1, self.b1, act="gelu")
+        h = ops.linear(h, self.w2, self.b2)
+        if self.drop is not None:
+            h = self.drop(h)
+        return x + h
+
+    def flops_per_item(self, in_shape):
+        s, d = in_shape
+        return self.attn.flops_per_item(in_shape) + 4 * s * d * d * self.mlp_ratio
+
+    def extra_config(self):
+        return {"dim": self.dim, "num_heads": self.num_heads,
+                "mlp_ratio": self.mlp_ratio, "flash": self.flash,
+                "dropout": self.dropout_p}

@@ -1,0 +1,151 @@
+"""Optimizers with fused CDNA4 update kernels
+(reference include/nn/optimizers.hpp:70,149; GPU kernels
+src/nn/optimizers_impl/cuda/sgd_kernels.cu:17, adam_kernels.cu:19).
+
+Low-precision params (bf16) automatically get an fp32 master copy in
+optimizer state; the fused kernel updates the master and writes the cast
+parameter in the same pass (the reference trains fp32 only — this is the
+MI355X-native mixed-precision upgrade).
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Any, Dict, Iterable, List, Optional
+
+import torch
+
+from .. import _C
+
+
+class Optimizer:
+    _type = "optimizer"
+
+    def __init__(self, params: Iterable[torch.nn.Parameter], lr: float):
+        self.params: List[torch.nn.Parameter] = [p for p in params if p.requires_grad]
+        self.lr = lr
+        self.state: Dict[int, Dict[str, Any]] = {}
+        self.step_count = 0
+
+    def zero_grad(self):
+        for p in self.params:
+            p.grad = None
+
+    @torch.no_grad()
+    def step(self):
+        self.step_count += 1
+        for i, p in enumerate(self.params):
+            if p.grad is None:
+                continue
+            st = self.state.setdefault(i, {})
+            if p.dtype != torch.float32 and "master" not in st:
+                st["master"] = p.detach().float().clone()
+            self._update(p, p.grad, st)
+
+    def _update(self, p, g, st):
+        raise NotImplementedError
+
+    def get_config(self) -> Dict[str, Any]:
+        return {"type": self._type, "lr": self.lr, **self.extra_config()}
+
+    def extra_config(self) -> Dict[str, Any]:
+        return {}
+
+    # -- state round-trip for checkpoint/stage-redeploy ----------------------
+    def state_tensors(self):
+        out = []
+        for i in sorted(self.state):
+            for k in sorted(self.state[i]):
+                v = self.state[i][k]
+                if torch.is_tensor(v):
+                    out.append((f"{i}.{k}", v))
+        return out
+
+
+class SGD(Optimizer):
+    _type = "sgd"
+
+    def __init__(self, params, lr=0.01, momentum=0.0, weight_decay=0.0,
+                 nesterov=False):
+        super().__init__(params, lr)
+        self.momentum, self.weight_decay, self.nesterov = momentum, weight_decay, nesterov
+
+    def _update(self, p, g, st):
+        if p.is_cuda:
+            ext = _C.ext()
+            master = st.get("master", p.data)
+            if self.momentum != 0.0 and "buf" not in st:
+                st["buf"] = torch.zeros_like(master)
+            ext.sgd_step(p.data, master, g, st.get("buf"),
+                         self.lr, self.momentum, self.weight_decay, self.nesterov)
+            return
+        master = st.get("master", p.data)
+        gf = g.float()
+        if self.weight_decay:
+            gf = gf.add(master, alpha=self.weight_decay)
+        if self.momentum != 0.0:
+            buf = st.setdefault("buf", torch.zeros_like(master))
+            buf.mul_(self.momentum).add_(gf)
+            gf = gf.add(buf, alpha=self.momentum) if self.nesterov else buf
+        master.add_(gf, alpha=-self.lr)
+        if master.data_ptr() != p.data.data_ptr():
+            p.data.copy_(master)
+
+    def extra_config(self):
+        return {"momentum": self.momentum, "weight_decay": self.weight_decay,
+                "nesterov": self.nesterov}
+
+
+class Adam(Optimizer):
+    _type = "adam"
+    _adamw = False
+
+    def __init__(self, params, lr=1e-3, beta1=0.9, beta2=0.999, eps=1e-8,
+                 weight_decay=0.0):
+        super().__init__(params, lr)
+        self.beta1, self.beta2, self.eps, self.weight_decay = beta1, beta2, eps, weight_decay
+
+    def _update(self, p, g, st):
+        if "m" not in st:
+            ref = st.get("master", p.data)
+            st["m"] = torch.zeros_like(ref, dtype=torch.float32)
+            st["v"] = torch.zeros_like(ref, dtype=torch.float32)
+        if p.is_cuda:
+            ext = _C.ext()
+            master = st.get("master", p.data)
+            ext.adam_step(p.data, master, g, st["m"], st["v"], self.step_count,
+                          self.lr, self.beta1, self.beta2, self.eps,
+                          self.weight_decay, self._adamw)
+            return
+        master = st.get("master", p.data)
+        gf = g.float()
+        if self.weight_decay and not self._adamw:
+            gf = gf.add(master, alpha=self.weight_decay)
+        st["m"].mul_(self.beta1).add_(gf, alpha=1 - self.beta1)
+        st["v"].mul_(self.beta2).addcmul_(gf, gf, value=1 - self.beta2)
+        bc1 = 1 - self.beta1 ** self.step_count
+        bc2 = 1 - self.beta2 ** self.step_count
+        if self._adamw and self.weight_decay:
+            master.mul_(1 - self.lr * self.weight_decay)
+        denom = (st["v"] / bc2).sqrt_().add_(self.eps)
+        master.addcdiv_(st["m"], denom, value=-self.lr / bc1)
+        if master.data_ptr() != p.data.data_ptr():
+            p.data.copy_(master)
+
+    def extra_config(self):
+        return {"beta1": self.beta1, "beta2": self.beta2, "eps": self.eps,
+                "weight_decay": self.weight_decay}
+
+
+class AdamW(Adam):
+    _type = "adamw"
+    _adamw = True
+
+
+_OPTS = {"sgd": SGD, "adam": Adam, "adamw": AdamW}
+
+
+def optimizer_from_config(cfg: Dict[str, Any], params) -> Optimizer:
+    cfg = dict(cfg)
+    cls = _OPTS[cfg.pop("type")]
+    return cls(params, **cfg)

@@ -1,0 +1,487 @@
+"""Functional ops with CPU(torch-composition) / GPU(HIP kernel) dispatch.
+
+GPU autograd Functions call into the in-tree HIP extension; CPU paths are
+pure differentiable torch compositions used as the numerics oracle
+(reference unit_tests/layer_device_agnosticity_test.cpp:25 pattern).
+
+All image ops are NHWC: x is a contiguous ``[N, H, W, C]`` tensor and conv
+weights are ``[KH, KW, Cin, Cout]`` (the implicit-GEMM B-operand layout).
+"""
+
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+import torch.nn.functional as F
+
+from .. import _C
+
+# ---------------------------------------------------------------------------
+# helpers
+# ---------------------------------------------------------------------------
+
+ACT_KINDS = {
+    "linear": 0,
+    "relu": 1,
+    "gelu": 2,       # tanh approximation (reference gelu_kernels.cu:18)
+    "sigmoid": 3,
+    "tanh": 4,
+    "elu": 5,
+    "leaky_relu": 6,
+    "silu": 7,
+}
+
+
+def _nhwc_to_nchw(x: torch.Tensor) -> torch.Tensor:
+    return x.permute(0, 3, 1, 2)
+
+
+def _nchw_to_nhwc(x: torch.Tensor) -> torch.Tensor:
+    return x.permute(0, 2, 3, 1).contiguous()
+
+
+def _use_hip(*tensors: torch.Tensor) -> bool:
+    return any(t is not None and t.is_cuda for t in tensors)
+
+
+# ---------------------------------------------------------------------------
+# conv2d NHWC (implicit GEMM on GPU; reference legacy_conv2d_layer.cpp:137)
+# ---------------------------------------------------------------------------
+
+
+class _Conv2dNHWC(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, bias, stride, padding, fuse_relu):
+        ext = _C.ext()
+        y = ext.conv2d_fwd(x, w, bias, stride[0], stride[1], padding[0], padding[1], fuse_relu)
+        ctx.save_for_backward(x, w, y if fuse_relu else None)
+        ctx.stride, ctx.padding, ctx.fuse_relu = stride, padding, fuse_relu
+        ctx.has_bias = bias is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w, y = ctx.saved_tensors
+        ext = _C.ext()
+        dy = dy.contiguous()
+        if ctx.fuse_relu:
+            dy = ext.relu_bwd_mask(dy, y)
+        dx = dw = db = None
+        if ctx.needs_input_grad[0]:
+            dx = ext.conv2d_dgrad(dy, w, x.shape[1], x.shape[2],
+                                  ctx.stride[0], ctx.stride[1], ctx.padding[0], ctx.padding[1])
+        if ctx.needs_input_grad[1]:
+            dw = ext.conv2d_wgrad(x, dy, w.shape[0], w.shape[1],
+                                  ctx.stride[0], ctx.stride[1], ctx.padding[0], ctx.padding[1])
+            dw = dw.to(w.dtype)
+        if ctx.has_bias and ctx.needs_input_grad[2]:
+            db = ext.colsum(dy.reshape(-1, dy.shape[-1]))
+        return dx, dw, db, None, None, None
+
+
+def conv2d_nhwc(
+    x: torch.Tensor,
+    w: torch.Tensor,
+    bias: Optional[torch.Tensor] = None,
+    stride: Tuple[int, int] = (1, 1),
+    padding: Tuple[int, int] = (0, 0),
+    fuse_relu: bool = False,
+) -> torch.Tensor:
+    """2-D convolution, NHWC activations, ``[KH,KW,Cin,Cout]`` weights."""
+    if _use_hip(x):
+        return _Conv2dNHWC.apply(x.contiguous(), w.contiguous(),
+                                 None if bias is None else bias.contiguous(),
+                                 stride, padding, fuse_relu)
+    wn = w.permute(3, 2, 0, 1)  # -> [Cout, Cin, KH, KW]
+    y = F.conv2d(_nhwc_to_nchw(x), wn, bias, stride=stride, padding=padding)
+    y = _nchw_to_nhwc(y)
+    if fuse_relu:
+        y = F.relu(y)
+    return y
+
+
+# ---------------------------------------------------------------------------
+# batch norm (+ fused ReLU) NHWC; stats always fp32
+# (reference src/nn/layers_impl/batchnorm_layer.cpp:133-149)
+# ---------------------------------------------------------------------------
+
+
+class _BatchNormAct(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, gamma, beta, eps, relu):
+        ext = _C.ext()
+        y, mean, invstd = ext.bn_fwd_train(x, gamma, beta, eps, relu)
+        ctx.save_for_backward(x, gamma, mean, invstd, y if relu else None)
+        ctx.relu = relu
+        return y, mean, invstd
+
+    @staticmethod
+    def backward(ctx, dy, _dmean, _dinvstd):
+        x, gamma, mean, invstd, y = ctx.saved_tensors
+        ext = _C.ext()
+        dx, dgamma, dbeta = ext.bn_bwd(x, dy.contiguous(), gamma, mean, invstd,
+                                       y if ctx.relu else None)
+        return dx, dgamma, dbeta, None, None
+
+
+def batch_norm_act(
+    x: torch.Tensor,
+    gamma: torch.Tensor,
+    beta: torch.Tensor,
+    running_mean: torch.Tensor,
+    running_var: torch.Tensor,
+    training: bool,
+    momentum: float = 0.1,
+    eps: float = 1e-5,
+    relu: bool = False,
+):
+    """BatchNorm over NHWC channels-last with optional fused ReLU.
+
+    Returns y. Updates running stats in-place when training (fp32, unbiased
+    variance, matching torch semantics). gamma/beta/running stats are fp32
+    regardless of x dtype (reference batchnorm_layer.cpp:140-148).
+    """
+    C = x.shape[-1]
+    n = x.numel() // C
+    if _use_hip(x):
+        if training:
+            y, mean, invstd = _BatchNormAct.apply(x.contiguous(), gamma, beta, eps, relu)
+            with torch.no_grad():
+                var = (1.0 / (invstd * invstd)) - eps
+                unbiased = var * (n / max(n - 1, 1))
+                running_mean.mul_(1 - momentum).add_(mean.float(), alpha=momentum)
+                running_var.mul_(1 - momentum).add_(unbiased.float(), alpha=momentum)
+            return y
+        ext = _C.ext()
+        return ext.bn_fwd_infer(x.contiguous(), gamma, beta, running_mean, running_var,
+                                eps, relu)
+    # CPU reference path
+    xf = x.reshape(-1, C).float()
+    if training:
+        mean = xf.mean(0)
+        var = xf.var(0, unbiased=False)
+        with torch.no_grad():
+            running_mean.mul_(1 - momentum).add_(mean.detach(), alpha=momentum)
+            running_var.mul_(1 - momentum).add_(var.detach() * (n / max(n - 1, 1)),
+                                                alpha=momentum)
+    else:
+        mean, var = running_mean, running_var
+    xhat = (x.float() - mean) / torch.sqrt(var + eps)
+    y = xhat * gamma + beta
+    if relu:
+        y = F.relu(y)
+    return y.to(x.dtype)
+
+
+# ---------------------------------------------------------------------------
+# dense / GEMM (reference src/nn/layers_impl/cuda/dense_ops.cu:17-117)
+# ---------------------------------------------------------------------------
+
+
+class _Linear(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, bias, act):
+        ext = _C.ext()
+        y = ext.gemm(x, w, bias, ACT_KINDS[act])
+        ctx.save_for_backward(x, w, y if act != "linear" else None)
+        ctx.act = act
+        ctx.has_bias = bias is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w, y = ctx.saved_tensors
+        ext = _C.ext()
+        dy = dy.contiguous()
+        if ctx.act == "relu":
+            dy = ext.relu_bwd_mask(dy, y)
+        elif ctx.act != "linear":
+            raise NotImplementedError(f"fused act bwd for {ctx.act}")
+        dx = dw = db = None
+        if ctx.needs_input_grad[0]:
+            dx = ext.gemm_nt(dy, w)       # [M,N] @ [K,N]^T -> [M,K]
+        if ctx.needs_input_grad[1]:
+            dw = ext.gemm_tn(x, dy)       # [M,K]^T @ [M,N] -> [K,N]
+            dw = dw.to(w.dtype)
+        if ctx.has_bias and ctx.needs_input_grad[2]:
+            db = ext.colsum(dy)
+        return dx, dw, db, None
+
+
+def linear(x: torch.Tensor, w: torch.Tensor, bias: Optional[torch.Tensor] = None,
+           act: str = "linear") -> torch.Tensor:
+    """y = act(x @ w + bias); x ``[..., K]``, w ``[K, N]`` row-major."""
+    lead = x.shape[:-1]
+    x2 = x.reshape(-1, x.shape[-1])
+    if _use_hip(x):
+        y = _Linear.apply(x2.contiguous(), w.contiguous(),
+                          None if bias is None else bias.contiguous(), act)
+    else:
+        y = x2 @ w
+        if bias is not None:
+            y = y + bias
+        if act == "relu":
+            y = F.relu(y)
+        elif act == "gelu":
+            y = F.gelu(y, approximate="tanh")
+        elif act != "linear":
+            y = activation(y, act)
+    return y.reshape(*lead, w.shape[1])
+
+
+def matmul(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    """Plain 2-D/batched matmul through the HIP GEMM on GPU."""
+    if _use_hip(a, b):
+        if a.dim() == 2 and b.dim() == 2:
+            return _Linear.apply(a.contiguous(), b.contiguous(), None, "linear")
+        ext = _C.ext()
+        return ext.bmm(a.contiguous(), b.contiguous())
+    return a @ b
+
+
+# ---------------------------------------------------------------------------
+# pooling NHWC (reference maxpool_ops.cu / avgpool_ops.cu)
+# ---------------------------------------------------------------------------
+
+
+class _MaxPool2d(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, kernel, stride, padding):
+        ext = _C.ext()
+        y, idx = ext.maxpool_fwd(x, kernel[0], kernel[1], stride[0], stride[1],
+                                 padding[0], padding[1])
+        ctx.save_for_backward(idx)
+        ctx.in_shape = x.shape
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (idx,) = ctx.saved_tensors
+        ext = _C.ext()
+        dx = ext.maxpool_bwd(dy.contiguous(), idx, ctx.in_shape[1], ctx.in_shape[2])
+        return dx, None, None, None
+
+
+class _AvgPool2d(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, kernel, stride, padding):
+        ext = _C.ext()
+        y = ext.avgpool_fwd(x, kernel[0], kernel[1], stride[0], stride[1],
+                            padding[0], padding[1])
+        ctx.in_shape = x.shape
+        ctx.kernel, ctx.stride, ctx.padding = kernel, stride, padding
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = _C.ext()
+        dx = ext.avgpool_bwd(dy.contiguous(), ctx.in_shape[1], ctx.in_shape[2],
+                             ctx.kernel[0], ctx.kernel[1], ctx.stride[0], ctx.stride[1],
+                             ctx.padding[0], ctx.padding[1])
+        return dx, None, None, None
+
+
+def max_pool2d_nhwc(x, kernel, stride=None, padding=(0, 0)):
+    stride = stride or kernel
+    if _use_hip(x):
+        return _MaxPool2d.apply(x.contiguous(), kernel, stride, padding)
+    y = F.max_pool2d(_nhwc_to_nchw(x), kernel, stride, padding)
+    return _nchw_to_nhwc(y)
+
+
+def avg_pool2d_nhwc(x, kernel, stride=None, padding=(0, 0)):
+    stride = stride or kernel
+    if _use_hip(x):
+        return _AvgPool2d.apply(x.contiguous(), kernel, stride, padding)
+    y = F.avg_pool2d(_nhwc_to_nchw(x), kernel, stride, padding)
+    return _nchw_to_nhwc(y)
+
+
+# ---------------------------------------------------------------------------
+# dropout (Philox mask on GPU; reference dropout.cu:33)
+# ---------------------------------------------------------------------------
+
+
+class _Dropout(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, p, seed):
+        ext = _C.ext()
+        y, mask = ext.dropout_fwd(x, p, seed)
+        ctx.save_for_backward(mask)
+        ctx.p = p
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (mask,) = ctx.saved_tensors
+        ext = _C.ext()
+        return ext.dropout_bwd(dy.contiguous(), mask, ctx.p), None, None
+
+
+def dropout(x: torch.Tensor, p: float, training: bool) -> torch.Tensor:
+    if not training or p <= 0.0:
+        return x
+    if _use_hip(x):
+        seed = int(torch.randint(0, 2 ** 62, (1,)).item())
+        return _Dropout.apply(x.contiguous(), p, seed)
+    return F.dropout(x, p, training=True)
+
+
+# ---------------------------------------------------------------------------
+# standalone activations (reference src/nn/activations_impl/)
+# ---------------------------------------------------------------------------
+
+
+class _Activation(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, kind):
+        ext = _C.ext()
+        y = ext.act_fwd(x, ACT_KINDS[kind])
+        ctx.save_for_backward(x, y)
+        ctx.kind = kind
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, y = ctx.saved_tensors
+        ext = _C.ext()
+        return ext.act_bwd(dy.contiguous(), x, y, ACT_KINDS[ctx.kind]), None
+
+
+def activation(x: torch.Tensor, kind: str) -> torch.Tensor:
+    if kind == "linear":
+        return x
+    if kind == "softmax":
+        return F.softmax(x, dim=-1)
+    if _use_hip(x):
+        return _Activation.apply(x.contiguous(), kind)
+    if kind == "relu":
+        return F.relu(x)
+    if kind == "gelu":
+        return F.gelu(x, approximate="tanh")
+    if kind == "sigmoid":
+        return torch.sigmoid(x)
+    if kind == "tanh":
+        return torch.tanh(x)
+    if kind == "elu":
+        return F.elu(x)
+    if kind == "leaky_relu":
+        return F.leaky_relu(x, 0.01)
+    if kind == "silu":
+        return F.silu(x)
+    raise ValueError(f"unknown activation {kind!r}")
+
+
+# ---------------------------------------------------------------------------
+# fused logsoftmax + cross-entropy (reference loss_ops.cu:76,150)
+# ---------------------------------------------------------------------------
+
+
+class _SoftmaxXent(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, targets):
+        ext = _C.ext()
+        loss, lse = ext.ce_fwd(logits, targets)
+        ctx.save_for_backward(logits, targets, lse)
+        return loss
+
+    @staticmethod
+    def backward(ctx, dloss):
+        logits, targets, lse = ctx.saved_tensors
+        ext = _C.ext()
+        dlogits = ext.ce_bwd(logits, targets, lse, dloss.contiguous())
+        return dlogits, None
+
+
+def softmax_cross_entropy(logits: torch.Tensor, targets: torch.Tensor) -> torch.Tensor:
+    """Mean cross-entropy from raw logits; ``targets`` are int64 class ids."""
+    l2 = logits.reshape(-1, logits.shape[-1])
+    t = targets.reshape(-1)
+    if _use_hip(logits):
+        per = _SoftmaxXent.apply(l2.contiguous(), t.contiguous())
+        return per.mean()
+    return F.cross_entropy(l2.float(), t)
+
+
+# ---------------------------------------------------------------------------
+# layer norm (reference layer_norm_ops.cu:21,51)
+# ---------------------------------------------------------------------------
+
+
+class _LayerNorm(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, gamma, beta, eps):
+        ext = _C.ext()
+        y, mean, invstd = ext.ln_fwd(x, gamma, beta, eps)
+        ctx.save_for_backward(x, gamma, mean, invstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, gamma, mean, invstd = ctx.saved_tensors
+        ext = _C.ext()
+        dx, dgamma, dbeta = ext.ln_bwd(x, dy.contiguous(), gamma, mean, invstd)
+        return dx, dgamma, dbeta, None
+
+
+def layer_norm(x: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
+               eps: float = 1e-5) -> torch.Tensor:
+    if _use_hip(x):
+        lead = x.shape[:-1]
+        y = _LayerNorm.apply(x.reshape(-1, x.shape[-1]).contiguous(), gamma, beta, eps)
+        return y.reshape(*lead, x.shape[-1])
+    xf = x.float()
+    y = F.layer_norm(xf, (x.shape[-1],), gamma.float(), beta.float(), eps)
+    return y.to(x.dtype)
+
+
+# ---------------------------------------------------------------------------
+# embedding (reference embedding_ops.cu:17,48)
+# ---------------------------------------------------------------------------
+
+
+class _Embedding(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, ids, table):
+        ext = _C.ext()
+        y = ext.embedding_fwd(ids, table)
+        ctx.save_for_backward(ids)
+        ctx.rows = table.shape[0]
+        ctx.dtype = table.dtype
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (ids,) = ctx.saved_tensors
+        ext = _C.ext()
+        dtab = ext.embedding_bwd(ids, dy.contiguous(), ctx.rows)
+        return None, dtab.to(ctx.dtype)
+
+
+def embedding(ids: torch.Tensor, table: torch.Tensor) -> torch.Tensor:
+    if _use_hip(table):
+        return _Embedding.apply(ids.contiguous(), table.contiguous())
+    return F.embedding(ids, table)
+
+
+# ---------------------------------------------------------------------------
+# scaled-dot-product attention (flash kernel on GPU, Phase 3;
+# reference src/nn/blocks_impl/cuda/cudnn_flash_attention_ops.cu:81)
+# ---------------------------------------------------------------------------
+
+
+def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+              causal: bool = True) -> torch.Tensor:
+    """q/k/v: [B, H, S, D] -> [B, H, S, D]."""
+    if _use_hip(q):
+        from .attention import flash_attention
+        return flash_attention(q, k, v, causal=causal)
+    scale = q.shape[-1] ** -0.5
+    s = (q.float() @ k.float().transpose(-1, -2)) * scale
+    if causal:
+        S = q.shape[-2]
+        mask = torch.ones(S, S, dtype=torch.bool, device=q.device).triu(1)
+        s = s.masked_fill(mask, float("-inf"))
+    p = F.softmax(s, dim=-1)
+    return (p @ v.float()).to(q.dtype)

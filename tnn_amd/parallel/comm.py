@@ -1,0 +1,120 @@
+"""Communicator over torch.distributed (reference include/distributed/
+communicator.hpp:30, tcp_communicator.hpp:63, roce_communicator.hpp:39).
+
+The reference moves serialized Messages over 4 TCP sockets or RoCE
+RDMA-WRITE rendezvous, staging through host memory. Here the data plane is
+RCCL P2P send/recv over xGMI: activations stay device-resident end to end
+(what the reference's IbvAllocator only approximated), shapes are
+negotiated once from shape inference instead of per-packet headers, and
+the tiny control plane (config deploy, barriers, profiler fetch) is
+``broadcast_object_list`` on a gloo side-group — mirroring the reference's
+split between Job payloads and CommandType control messages.
+"""
+
+from __future__ import annotations
+
+import datetime
+import os
+from typing import Any, List, Optional, Sequence
+
+import torch
+import torch.distributed as dist
+
+
+def init_distributed(backend: Optional[str] = None,
+                     timeout_s: int = 600) -> "Communicator":
+    """Initialize from torchrun env vars (RANK/WORLD_SIZE/MASTER_ADDR...)."""
+    if not dist.is_initialized():
+        if backend is None:
+            backend = "nccl" if torch.cuda.is_available() else "gloo"
+        kwargs = {}
+        if "RANK" not in os.environ:
+            os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+            os.environ.setdefault("MASTER_PORT", "29571")
+            kwargs = {"rank": 0, "world_size": 1}
+        dist.init_process_group(backend=backend,
+                                timeout=datetime.timedelta(seconds=timeout_s),
+                                **kwargs)
+    if torch.cuda.is_available():
+        local = int(os.environ.get("LOCAL_RANK", dist.get_rank()))
+        torch.cuda.set_device(local % torch.cuda.device_count())
+    return Communicator()
+
+
+def is_initialized() -> bool:
+    return dist.is_initialized()
+
+
+def rank() -> int:
+    return dist.get_rank() if dist.is_initialized() else 0
+
+
+def world_size() -> int:
+    return dist.get_world_size() if dist.is_initialized() else 1
+
+
+class Communicator:
+    """P2P data plane + object control plane for one rank."""
+
+    def __init__(self):
+        self.rank = rank()
+        self.world_size = world_size()
+        self.backend = dist.get_backend() if dist.is_initialized() else None
+        # control plane on gloo so tiny python objects never touch RCCL
+        self._ctrl = None
+        if dist.is_initialized() and self.world_size > 1 and self.backend == "nccl":
+            self._ctrl = dist.new_group(backend="gloo")
+        # dedicated comm streams: send and recv overlap with compute
+        self._send_stream = (torch.cuda.Stream() if torch.cuda.is_available()
+                             else None)
+        self._recv_stream = (torch.cuda.Stream() if torch.cuda.is_available()
+                             else None)
+
+    # -- data plane ----------------------------------------------------------
+    def isend(self, t: torch.Tensor, dst: int):
+        return dist.isend(t.contiguous(), dst)
+
+    def irecv(self, t: torch.Tensor, src: int):
+        return dist.irecv(t, src)
+
+    def send(self, t: torch.Tensor, dst: int):
+        dist.send(t.contiguous(), dst)
+
+    def recv(self, t: torch.Tensor, src: int) -> torch.Tensor:
+        dist.recv(t, src)
+        return t
+
+    def batch_p2p(self, ops: List[dist.P2POp]):
+        """Issue a batched send/recv group (rcclGroupStart/End underneath) —
+        the deadlock-free way to cross-send on RCCL."""
+        if not ops:
+            return []
+        return dist.batch_isend_irecv(ops)
+
+    # -- control plane -------------------------------------------------------
+    def broadcast_object(self, obj: Any, src: int = 0) -> Any:
+        if self.world_size == 1:
+            return obj
+        buf = [obj if self.rank == src else None]
+        dist.broadcast_object_list(buf, src=src, group=self._ctrl)
+        return buf[0]
+
+    def gather_objects(self, obj: Any, dst: int = 0) -> Optional[List[Any]]:
+        if self.world_size == 1:
+            return [obj]
+        out = [None] * self.world_size if self.rank == dst else None
+        dist.gather_object(obj, out, dst=dst, group=self._ctrl)
+        return out
+
+    def allreduce_(self, t: torch.Tensor, op=None):
+        if self.world_size > 1:
+            dist.all_reduce(t, op or dist.ReduceOp.SUM)
+        return t
+
+    def barrier(self):
+        if self.world_size > 1:
+            dist.barrier()
+
+    def destroy(self):
+        if dist.is_initialized():
+            dist.destroy_process_group()

@@ -1,0 +1,89 @@
+"""Data-parallel engine with bucketed gradient all-reduce.
+
+The reference declared DATA mode but never implemented gradient
+aggregation (reference include/nn/reducer.hpp is a stub).
+This implements it the MI355X way: bucketed RCCL all-reduce launched from
+autograd hooks as soon as a bucket's grads are ready, overlapping the
+reduction with the rest of backward. Bucket size is tuned for xGMI ring
+collectives (per-link-bound ⇒ fewer, larger buckets than on NVSwitch).
+"""
+
+from __future__ import annotations
+
+from typing import Dict, List, Optional
+
+import torch
+import torch.distributed as dist
+
+from ..nn.optim import Optimizer
+from .comm import Communicator
+
+
+class _Bucket:
+    def __init__(self, params: List[torch.nn.Parameter]):
+        self.params = params
+        self.pending = 0
+        self.work = None
+
+
+class DataParallelEngine:
+    def __init__(self, model: torch.nn.Module, comm: Communicator,
+                 bucket_bytes: int = 64 * 1024 * 1024):
+        self.model = model
+        self.comm = comm
+        self.world = comm.world_size
+        self._buckets: List[_Bucket] = []
+        self._param_bucket: Dict[int, _Bucket] = {}
+        if self.world > 1:
+            self._build_buckets(bucket_bytes)
+            self._sync_initial_state()
+
+    def _build_buckets(self, bucket_bytes: int):
+        # reverse order: grads become ready roughly from the last layer back
+        params = [p for p in self.model.parameters() if p.requires_grad][::-1]
+        cur, size = [], 0
+        for p in params:
+            cur.append(p)
+            size += p.numel() * p.element_size()
+            if size >= bucket_bytes:
+                self._buckets.append(_Bucket(cur))
+                cur, size = [], 0
+        if cur:
+            self._buckets.append(_Bucket(cur))
+        for b in self._buckets:
+            for p in b.params:
+                self._param_bucket[id(p)] = b
+                p.register_post_accumulate_grad_hook(self._hook)
+
+    def _sync_initial_state(self):
+        for t in list(self.model.parameters()) + list(self.model.buffers()):
+            dist.broadcast(t.data, src=0)
+
+    def _hook(self, p: torch.nn.Parameter):
+        b = self._param_bucket[id(p)]
+        b.pending += 1
+        if b.pending == len(b.params):
+            grads = [q.grad for q in b.params if q.grad is not None]
+            flat = torch._utils._flatten_dense_tensors(grads)
+            flat.div_(self.world)
+            b.work = (dist.all_reduce(flat, async_op=True), flat, grads)
+            b.pending = 0
+
+    def finish_backward(self):
+        """Wait for in-flight reductions and scatter results back."""
+        if self.world == 1:
+            return
+        for b in self._buckets:
+            if b.work is not None:
+                work, flat, grads = b.work
+                work.wait()
+                for g, r in zip(grads,
+                                torch._utils._unflatten_dense_tensors(flat, grads)):
+                    g.copy_(r)
+                b.work = None
+
+    def train_step(self, loss: torch.Tensor, optimizer: Optimizer):
+        loss.backward()
+        self.finish_backward()
+        optimizer.step()
+        optimizer.zero_grad()
