@@ -1,0 +1,125 @@
+"""Flagship benchmark: WRN-16-8 pipeline-parallel training throughput
+(images/sec, whole node) on synthetic 32x32x3 data — BASELINE.json's
+headline metric. Reference numbers: ~1,900-2,530 img/s for the same model
+at batch 256 on 2 consumer GPUs over RoCE.
+
+Single process:   python bench.py --gpus 1 --steps 30 --warmup 10
+Multi-GPU (driver): python -m torch.distributed.run --nnodes=1
+    --nproc-per-node N --master-addr 127.0.0.1 bench.py --gpus N ...
+
+Weak scaling: global batch = 256 * N (per-GPU work constant as the model
+is split N ways while the batch grows N ways).
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+BASELINE_IMG_S = 2530.0  # top of the reference band (BASELINE.md per-batch)
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=int(os.environ.get("WORLD_SIZE", 1)))
+    p.add_argument("--steps", type=int, default=30)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--model", default="cifar100_wrn16_8")
+    p.add_argument("--batch-per-gpu", type=int, default=256)
+    p.add_argument("--microbatch", type=int, default=64)
+    p.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
+    p.add_argument("--lr", type=float, default=1e-3)
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    from tnn_amd import models
+    from tnn_amd.nn import CrossEntropyLoss
+    from tnn_amd.nn.layer import cast_compute_dtype
+    from tnn_amd.parallel import init_distributed, Communicator, PipelineEngine
+
+    have_gpu = torch.cuda.is_available()
+    device = torch.device("cuda" if have_gpu else "cpu")
+    dtype = torch.bfloat16 if (args.dtype == "bf16" and have_gpu) else torch.float32
+
+    comm = init_distributed()
+    world = comm.world_size
+    rank = comm.rank
+    assert world == args.gpus or args.gpus == 1, \
+        f"launched with WORLD_SIZE={world} but --gpus {args.gpus}"
+
+    global_batch = args.batch_per_gpu * world
+    num_micro = max(1, global_batch // args.microbatch) if world > 1 else 1
+
+    model = models.create_model(args.model) if rank == 0 else None
+    in_shape = (32, 32, 3)
+    num_classes = {"cifar100_wrn16_8": 100, "cifar10_resnet9": 10}.get(
+        args.model, 100)
+
+    engine = PipelineEngine(
+        model, comm, input_shape=in_shape, num_microbatches=num_micro,
+        criterion=CrossEntropyLoss(),
+        optimizer_config={"type": "adamw", "lr": args.lr},
+        device=device, io_dtype=dtype)
+
+    # fixed synthetic batch, random-init weights (no network for datasets)
+    g = torch.Generator().manual_seed(1234 + 0)
+    x = torch.randn(global_batch, *in_shape, generator=g).to(device)
+    y = torch.randint(0, num_classes, (global_batch,), generator=g).to(device)
+
+    for _ in range(args.warmup):
+        engine.train_batch(x, y)
+
+    comm.barrier()
+    if have_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        engine.train_batch(x, y)
+    comm.barrier()
+    if have_gpu:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    t = torch.tensor([elapsed], dtype=torch.float64)
+    if world > 1:
+        import torch.distributed as dist
+        tt = t.to(device) if comm.backend == "nccl" else t
+        dist.all_reduce(tt, dist.ReduceOp.MAX)
+        elapsed = float(tt.cpu().item())
+
+    if rank == 0:
+        value = global_batch * args.steps / elapsed
+        print(json.dumps({
+            "metric": "images/sec (whole node) WRN-16-8 on 32x32x3, "
+                      "pipeline-parallel",
+            "value": round(value, 1),
+            "unit": "img/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1e3, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": round(value / BASELINE_IMG_S, 2),
+            "dtype": "bf16" if dtype == torch.bfloat16 else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": global_batch,
+                "input": "32x32x3",
+                "microbatches": num_micro,
+                "parallelism": f"pp{world}" if world > 1 else "single",
+            },
+        }))
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,246 @@
+// Fused NHWC BatchNorm (+ReLU) kernels
+// (reference src/nn/layers_impl/cuda/batchnorm_nchw_ops.cu:119-449 and the
+// cuDNN-fe BN+ReLU graphs, cudnn_batchnorm_ops.cu:159).
+//
+// NHWC: channels are the fast dim, so thread t of a block covers channel
+// (t % CPB) — global loads are perfectly coalesced across channels. All
+// statistics accumulate in fp32 regardless of io dtype (SURVEY §7 hard
+// part 5). Grid-level reduction: per-block partial sums combined with
+// fp32 atomics, then a tiny finalize kernel.
+
+#include "common.h"
+#include "kernels.h"
+
+namespace tnn {
+
+// pass 1a: partial sum / sumsq per channel, atomics into f32 buffers
+template <typename T>
+__global__ void k_bn_partial(const T* __restrict__ x, float* __restrict__ sum,
+                             float* __restrict__ sumsq, int64_t rows,
+                             int cols) {
+  const int cpb = min(cols, 256);
+  const int rows_per_iter = 256 / cpb;
+  const int c = threadIdx.x % cpb + blockIdx.x * cpb;
+  const int r_off = threadIdx.x / cpb;
+  if (c >= cols || r_off >= rows_per_iter) return;
+  const int64_t r0 = rows * blockIdx.y / gridDim.y;
+  const int64_t r1 = rows * (blockIdx.y + 1) / gridDim.y;
+  float s = 0.0f, ss = 0.0f;
+  for (int64_t r = r0 + r_off; r < r1; r += rows_per_iter) {
+    float v = VecIO<T>::to_f32(x[r * cols + c]);
+    s += v;
+    ss += v * v;
+  }
+  // combine the rows_per_iter partials for one channel via LDS
+  __shared__ float sh_s[256], sh_ss[256];
+  sh_s[threadIdx.x] = s;
+  sh_ss[threadIdx.x] = ss;
+  __syncthreads();
+  if (r_off == 0) {
+    for (int j = 1; j < rows_per_iter; ++j) {
+      s += sh_s[threadIdx.x + j * cpb];
+      ss += sh_ss[threadIdx.x + j * cpb];
+    }
+    atomicAdd(&sum[c], s);
+    atomicAdd(&sumsq[c], ss);
+  }
+}
+
+__global__ void k_bn_finalize(float* mean, float* invstd, const float* sum,
+                              const float* sumsq, int64_t rows, int cols,
+                              float eps) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= cols) return;
+  float m = sum[c] / rows;
+  float var = fmaxf(sumsq[c] / rows - m * m, 0.0f);
+  mean[c] = m;
+  invstd[c] = rsqrtf(var + eps);
+}
+
+template <typename T>
+__global__ void k_bn_apply(const T* __restrict__ x, const float* mean,
+                           const float* invstd, const float* gamma,
+                           const float* beta, T* __restrict__ y, int64_t n,
+                           int cols, bool relu) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    int c = i % cols;
+    float v = (VecIO<T>::to_f32(x[i]) - mean[c]) * invstd[c] * gamma[c] + beta[c];
+    if (relu) v = fmaxf(v, 0.0f);
+    y[i] = VecIO<T>::from_f32(v);
+  }
+}
+
+template <typename T>
+__global__ void k_bn_infer(const T* __restrict__ x, const float* rmean,
+                           const float* rvar, const float* gamma,
+                           const float* beta, T* __restrict__ y, int64_t n,
+                           int cols, float eps, bool relu) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    int c = i % cols;
+    float inv = rsqrtf(rvar[c] + eps);
+    float v = (VecIO<T>::to_f32(x[i]) - rmean[c]) * inv * gamma[c] + beta[c];
+    if (relu) v = fmaxf(v, 0.0f);
+    y[i] = VecIO<T>::from_f32(v);
+  }
+}
+
+// backward reduce: sum_dy[c], sum_dy_xhat[c] (dy masked by relu output)
+template <typename T>
+__global__ void k_bn_bwd_reduce(const T* __restrict__ x, const T* __restrict__ dy,
+                                const T* __restrict__ y_relu, const float* mean,
+                                const float* invstd, float* __restrict__ sum_dy,
+                                float* __restrict__ sum_dy_xhat, int64_t rows,
+                                int cols) {
+  const int cpb = min(cols, 256);
+  const int rows_per_iter = 256 / cpb;
+  const int c = threadIdx.x % cpb + blockIdx.x * cpb;
+  const int r_off = threadIdx.x / cpb;
+  if (c >= cols || r_off >= rows_per_iter) return;
+  const int64_t r0 = rows * blockIdx.y / gridDim.y;
+  const int64_t r1 = rows * (blockIdx.y + 1) / gridDim.y;
+  const float m = mean[c], is = invstd[c];
+  float s = 0.0f, sx = 0.0f;
+  for (int64_t r = r0 + r_off; r < r1; r += rows_per_iter) {
+    int64_t i = r * cols + c;
+    float g = VecIO<T>::to_f32(dy[i]);
+    if (y_relu && VecIO<T>::to_f32(y_relu[i]) <= 0.0f) g = 0.0f;
+    float xhat = (VecIO<T>::to_f32(x[i]) - m) * is;
+    s += g;
+    sx += g * xhat;
+  }
+  __shared__ float sh_s[256], sh_sx[256];
+  sh_s[threadIdx.x] = s;
+  sh_sx[threadIdx.x] = sx;
+  __syncthreads();
+  if (r_off == 0) {
+    for (int j = 1; j < rows_per_iter; ++j) {
+      s += sh_s[threadIdx.x + j * cpb];
+      sx += sh_sx[threadIdx.x + j * cpb];
+    }
+    atomicAdd(&sum_dy[c], s);
+    atomicAdd(&sum_dy_xhat[c], sx);
+  }
+}
+
+template <typename T>
+__global__ void k_bn_bwd_apply(const T* __restrict__ x, const T* __restrict__ dy,
+                               const T* __restrict__ y_relu, const float* mean,
+                               const float* invstd, const float* gamma,
+                               const float* sum_dy, const float* sum_dy_xhat,
+                               T* __restrict__ dx, int64_t n, int64_t rows,
+                               int cols) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const float inv_n = 1.0f / (float)rows;
+  for (; i < n; i += stride) {
+    int c = i % cols;
+    float g = VecIO<T>::to_f32(dy[i]);
+    if (y_relu && VecIO<T>::to_f32(y_relu[i]) <= 0.0f) g = 0.0f;
+    float xhat = (VecIO<T>::to_f32(x[i]) - mean[c]) * invstd[c];
+    float v = gamma[c] * invstd[c] *
+              (g - inv_n * (sum_dy[c] + xhat * sum_dy_xhat[c]));
+    dx[i] = VecIO<T>::from_f32(v);
+  }
+}
+
+// ---------------------------------------------------------------------------
+static inline dim3 bn_reduce_grid(int64_t rows, int cols) {
+  int cpb = cols < 256 ? cols : 256;
+  int cblocks = (cols + cpb - 1) / cpb;
+  int rows_per_iter = 256 / cpb;
+  // aim for >=512 blocks to fill 256 CUs
+  int64_t iters = (rows + rows_per_iter - 1) / rows_per_iter;
+  int rslices = (int)std::min<int64_t>(std::max<int64_t>(512 / cblocks, 1),
+                                  std::max<int64_t>(iters / 16, 1));
+  return dim3(cblocks, rslices);
+}
+
+void bn_stats_launch(DT dt, const void* x, float* mean, float* invstd,
+                     int64_t rows, int cols, float eps, hipStream_t s) {
+  // mean/invstd double as the scratch sum/sumsq buffers (finalized in place);
+  // they must be zeroed first.
+  hipMemsetAsync(mean, 0, cols * sizeof(float), s);
+  hipMemsetAsync(invstd, 0, cols * sizeof(float), s);
+  dim3 grid = bn_reduce_grid(rows, cols);
+  if (dt == DT::F32)
+    hipLaunchKernelGGL(k_bn_partial<float>, grid, dim3(256), 0, s,
+                       (const float*)x, mean, invstd, rows, cols);
+  else
+    hipLaunchKernelGGL(k_bn_partial<bf16>, grid, dim3(256), 0, s,
+                       (const bf16*)x, mean, invstd, rows, cols);
+  hipLaunchKernelGGL(k_bn_finalize, dim3((cols + 255) / 256), dim3(256), 0, s,
+                     mean, invstd, mean, invstd, rows, cols, eps);
+}
+
+void bn_apply_launch(DT dt, const void* x, const float* mean,
+                     const float* invstd, const float* gamma, const float* beta,
+                     void* y, int64_t rows, int cols, bool relu,
+                     hipStream_t s) {
+  int64_t n = rows * cols;
+  int blocks = (int)std::min<int64_t>((n + 255) / 256, (int64_t)2048);
+  if (dt == DT::F32)
+    hipLaunchKernelGGL(k_bn_apply<float>, dim3(blocks), dim3(256), 0, s,
+                       (const float*)x, mean, invstd, gamma, beta, (float*)y, n,
+                       cols, relu);
+  else
+    hipLaunchKernelGGL(k_bn_apply<bf16>, dim3(blocks), dim3(256), 0, s,
+                       (const bf16*)x, mean, invstd, gamma, beta, (bf16*)y, n,
+                       cols, relu);
+}
+
+void bn_infer_launch(DT dt, const void* x, const float* rmean,
+                     const float* rvar, const float* gamma, const float* beta,
+                     void* y, int64_t rows, int cols, float eps, bool relu,
+                     hipStream_t s) {
+  int64_t n = rows * cols;
+  int blocks = (int)std::min<int64_t>((n + 255) / 256, (int64_t)2048);
+  if (dt == DT::F32)
+    hipLaunchKernelGGL(k_bn_infer<float>, dim3(blocks), dim3(256), 0, s,
+                       (const float*)x, rmean, rvar, gamma, beta, (float*)y, n,
+                       cols, eps, relu);
+  else
+    hipLaunchKernelGGL(k_bn_infer<bf16>, dim3(blocks), dim3(256), 0, s,
+                       (const bf16*)x, rmean, rvar, gamma, beta, (bf16*)y, n,
+                       cols, eps, relu);
+}
+
+void bn_bwd_reduce_launch(DT dt, const void* x, const void* dy,
+                          const void* y_relu, const float* mean,
+                          const float* invstd, float* sum_dy,
+                          float* sum_dy_xhat, int64_t rows, int cols,
+                          hipStream_t s) {
+  dim3 grid = bn_reduce_grid(rows, cols);
+  if (dt == DT::F32)
+    hipLaunchKernelGGL(k_bn_bwd_reduce<float>, grid, dim3(256), 0, s,
+                       (const float*)x, (const float*)dy, (const float*)y_relu,
+                       mean, invstd, sum_dy, sum_dy_xhat, rows, cols);
+  else
+    hipLaunchKernelGGL(k_bn_bwd_reduce<bf16>, grid, dim3(256), 0, s,
+                       (const bf16*)x, (const bf16*)dy, (const bf16*)y_relu,
+                       mean, invstd, sum_dy, sum_dy_xhat, rows, cols);
+}
+
+void bn_bwd_apply_launch(DT dt, const void* x, const void* dy,
+                         const void* y_relu, const float* mean,
+                         const float* invstd, const float* gamma,
+                         const float* sum_dy, const float* sum_dy_xhat,
+                         void* dx, int64_t rows, int cols, hipStream_t s) {
+  int64_t n = rows * cols;
+  int blocks = (int)std::min<int64_t>((n + 255) / 256, (int64_t)2048);
+  if (dt == DT::F32)
+    hipLaunchKernelGGL(k_bn_bwd_apply<float>, dim3(blocks), dim3(256), 0, s,
+                       (const float*)x, (const float*)dy, (const float*)y_relu,
+                       mean, invstd, gamma, sum_dy, sum_dy_xhat, (float*)dx, n,
+                       rows, cols);
+  else
+    hipLaunchKernelGGL(k_bn_bwd_apply<bf16>, dim3(blocks), dim3(256), 0, s,
+                       (const bf16*)x, (const bf16*)dy, (const bf16*)y_relu,
+                       mean, invstd, gamma, sum_dy, sum_dy_xhat, (bf16*)dx, n,
+                       rows, cols);
+}
+
+}  // namespace tnn

@@ -1,0 +1,508 @@
+// Torch bindings for the tnn_amd gfx950 kernels.
+//
+// Tensor-level checks live here; the .hip translation units only see raw
+// pointers + shapes + the stream. Every entry point CHECKs device/layout —
+// a CPU tensor reaching these is a dispatch bug in the python layer.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "kernels.h"
+
+namespace tnn {
+
+static hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+}
+
+static DT dt_of(const at::Tensor& t) {
+  if (t.scalar_type() == at::kFloat) return DT::F32;
+  if (t.scalar_type() == at::kBFloat16) return DT::BF16;
+  TORCH_CHECK(false, "tnn_amd kernels support fp32/bf16, got ", t.scalar_type());
+}
+
+#define CHECK_IN(t)                                             \
+  TORCH_CHECK((t).is_cuda(), #t " must be on GPU");             \
+  TORCH_CHECK((t).is_contiguous(), #t " must be contiguous")
+
+// ---- elementwise -----------------------------------------------------------
+at::Tensor act_fwd(const at::Tensor& x, int64_t kind) {
+  CHECK_IN(x);
+  auto y = at::empty_like(x);
+  act_fwd_launch(dt_of(x), x.data_ptr(), y.data_ptr(), x.numel(), (int)kind,
+                 cur_stream());
+  return y;
+}
+
+at::Tensor act_bwd(const at::Tensor& dy, const at::Tensor& x,
+                   const at::Tensor& y, int64_t kind) {
+  CHECK_IN(dy);
+  auto dx = at::empty_like(dy);
+  act_bwd_launch(dt_of(dy), dy.data_ptr(), x.data_ptr(), y.data_ptr(),
+                 dx.data_ptr(), dy.numel(), (int)kind, cur_stream());
+  return dx;
+}
+
+at::Tensor relu_bwd_mask(const at::Tensor& dy, const at::Tensor& y) {
+  CHECK_IN(dy);
+  CHECK_IN(y);
+  auto dx = at::empty_like(dy);
+  relu_bwd_mask_launch(dt_of(dy), dy.data_ptr(), y.data_ptr(), dx.data_ptr(),
+                       dy.numel(), cur_stream());
+  return dx;
+}
+
+std::vector<at::Tensor> dropout_fwd(const at::Tensor& x, double p,
+                                    int64_t seed) {
+  CHECK_IN(x);
+  auto y = at::empty_like(x);
+  auto mask = at::empty(x.sizes(), x.options().dtype(at::kByte));
+  dropout_fwd_launch(dt_of(x), x.data_ptr(), y.data_ptr(),
+                     mask.data_ptr<uint8_t>(), x.numel(), (float)p,
+                     (uint64_t)seed, cur_stream());
+  return {y, mask};
+}
+
+at::Tensor dropout_bwd(const at::Tensor& dy, const at::Tensor& mask, double p) {
+  CHECK_IN(dy);
+  auto dx = at::empty_like(dy);
+  dropout_bwd_launch(dt_of(dy), dy.data_ptr(), mask.data_ptr<uint8_t>(),
+                     dx.data_ptr(), dy.numel(), (float)p, cur_stream());
+  return dx;
+}
+
+at::Tensor colsum(const at::Tensor& x) {
+  CHECK_IN(x);
+  TORCH_CHECK(x.dim() == 2, "colsum wants [rows, cols]");
+  auto out = at::zeros({x.size(1)}, x.options().dtype(at::kFloat));
+  colsum_launch(dt_of(x), x.data_ptr(), out.data_ptr(), x.size(0), x.size(1),
+                cur_stream());
+  if (x.scalar_type() != at::kFloat) {
+    auto out_t = at::empty({x.size(1)}, x.options());
+    cast_f32_launch(dt_of(x), out.data_ptr<float>(), out_t.data_ptr(),
+                    out.numel(), cur_stream());
+    return out_t;
+  }
+  return out;
+}
+
+// ---- gemm ------------------------------------------------------------------
+at::Tensor gemm(const at::Tensor& a, const at::Tensor& b,
+                const c10::optional<at::Tensor>& bias, int64_t act_kind) {
+  CHECK_IN(a);
+  CHECK_IN(b);
+  TORCH_CHECK(a.dim() == 2 && b.dim() == 2 && a.size(1) == b.size(0),
+              "gemm shapes ", a.sizes(), " @ ", b.sizes());
+  int M = a.size(0), K = a.size(1), N = b.size(1);
+  auto c = at::empty({M, N}, a.options());
+  const void* bp = bias.has_value() ? bias->data_ptr() : nullptr;
+  gemm_launch(dt_of(a), a.data_ptr(), b.data_ptr(), bp, c.data_ptr(), M, N, K,
+              /*trans_b=*/false, (int)act_kind, cur_stream());
+  return c;
+}
+
+at::Tensor gemm_nt(const at::Tensor& a, const at::Tensor& b) {
+  // c[m, j] = sum_k a[m, k] * b[j, k]
+  CHECK_IN(a);
+  CHECK_IN(b);
+  TORCH_CHECK(a.dim() == 2 && b.dim() == 2 && a.size(1) == b.size(1),
+              "gemm_nt shapes ", a.sizes(), " @ ", b.sizes(), "^T");
+  int M = a.size(0), K = a.size(1), N = b.size(0);
+  auto c = at::empty({M, N}, a.options());
+  gemm_launch(dt_of(a), a.data_ptr(), b.data_ptr(), nullptr, c.data_ptr(), M, N,
+              K, /*trans_b=*/true, 0, cur_stream());
+  return c;
+}
+
+at::Tensor gemm_tn(const at::Tensor& a, const at::Tensor& b) {
+  // c[k, n] = sum_m a[m, k] * b[m, n]  (fp32 accumulate + output)
+  CHECK_IN(a);
+  CHECK_IN(b);
+  TORCH_CHECK(a.dim() == 2 && b.dim() == 2 && a.size(0) == b.size(0),
+              "gemm_tn shapes ", a.sizes(), "^T @ ", b.sizes());
+  int M = a.size(0), K = a.size(1), N = b.size(1);
+  auto c = at::zeros({K, N}, a.options().dtype(at::kFloat));
+  gemm_tn_launch(dt_of(a), a.data_ptr(), b.data_ptr(), c.data_ptr<float>(), M,
+                 N, K, cur_stream());
+  return c;
+}
+
+at::Tensor mfma_selftest(const at::Tensor& a, const at::Tensor& b) {
+  // a: [16, 32] bf16, b: [32, 16] bf16 -> d [16, 16] f32 (layout check)
+  CHECK_IN(a);
+  auto d = at::zeros({16, 16}, a.options().dtype(at::kFloat));
+  mfma_selftest_launch(a.data_ptr(), b.data_ptr(), d.data_ptr<float>(),
+                       cur_stream());
+  return d;
+}
+
+at::Tensor mfma_selftest_f32(const at::Tensor& a, const at::Tensor& b) {
+  // a: [16, 4] f32, b: [4, 16] f32 -> d [16, 16] f32
+  CHECK_IN(a);
+  auto d = at::zeros({16, 16}, a.options().dtype(at::kFloat));
+  mfma_selftest_f32_launch(a.data_ptr<float>(), b.data_ptr<float>(),
+                           d.data_ptr<float>(), cur_stream());
+  return d;
+}
+
+// ---- conv2d ----------------------------------------------------------------
+static ConvShape conv_shape(const at::Tensor& x_like, int Cin, int Cout, int KH,
+                            int KW, int SH, int SW, int PH, int PW) {
+  ConvShape cs;
+  cs.N = x_like.size(0);
+  cs.H = x_like.size(1);
+  cs.W = x_like.size(2);
+  cs.Cin = Cin;
+  cs.Cout = Cout;
+  cs.KH = KH;
+  cs.KW = KW;
+  cs.SH = SH;
+  cs.SW = SW;
+  cs.PH = PH;
+  cs.PW = PW;
+  cs.OH = (cs.H + 2 * PH - KH) / SH + 1;
+  cs.OW = (cs.W + 2 * PW - KW) / SW + 1;
+  return cs;
+}
+
+at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
+                      const c10::optional<at::Tensor>& bias, int64_t sh,
+                      int64_t sw, int64_t ph, int64_t pw, bool relu) {
+  CHECK_IN(x);
+  CHECK_IN(w);
+  TORCH_CHECK(x.dim() == 4 && w.dim() == 4 && x.size(3) == w.size(2),
+              "conv2d_fwd x ", x.sizes(), " w ", w.sizes());
+  auto cs = conv_shape(x, w.size(2), w.size(3), w.size(0), w.size(1), sh, sw,
+                       ph, pw);
+  auto y = at::empty({cs.N, cs.OH, cs.OW, cs.Cout}, x.options());
+  const void* bp = bias.has_value() ? bias->data_ptr() : nullptr;
+  conv2d_fwd_launch(dt_of(x), x.data_ptr(), w.data_ptr(), bp, y.data_ptr(), cs,
+                    relu, cur_stream());
+  return y;
+}
+
+at::Tensor conv2d_dgrad(const at::Tensor& dy, const at::Tensor& w, int64_t H,
+                        int64_t W, int64_t sh, int64_t sw, int64_t ph,
+                        int64_t pw) {
+  CHECK_IN(dy);
+  CHECK_IN(w);
+  int KH = w.size(0), KW = w.size(1), Cin = w.size(2), Cout = w.size(3);
+  ConvShape cs;
+  cs.N = dy.size(0);
+  cs.H = H;
+  cs.W = W;
+  cs.Cin = Cin;
+  cs.Cout = Cout;
+  cs.KH = KH;
+  cs.KW = KW;
+  cs.SH = sh;
+  cs.SW = sw;
+  cs.PH = ph;
+  cs.PW = pw;
+  cs.OH = dy.size(1);
+  cs.OW = dy.size(2);
+  // dgrad consumes the weight as B[(kh,kw,co)][ci]: transpose once per call
+  auto w_t = at::empty({KH, KW, Cout, Cin}, w.options());
+  transpose_w_launch(dt_of(w), w.data_ptr(), w_t.data_ptr(), KH, KW, Cin, Cout,
+                     cur_stream());
+  auto dx = at::empty({cs.N, H, W, Cin}, dy.options());
+  conv2d_dgrad_launch(dt_of(dy), dy.data_ptr(), w_t.data_ptr(), dx.data_ptr(),
+                      cs, cur_stream());
+  return dx;
+}
+
+at::Tensor conv2d_wgrad(const at::Tensor& x, const at::Tensor& dy, int64_t KH,
+                        int64_t KW, int64_t sh, int64_t sw, int64_t ph,
+                        int64_t pw) {
+  CHECK_IN(x);
+  CHECK_IN(dy);
+  auto cs = conv_shape(x, x.size(3), dy.size(3), KH, KW, sh, sw, ph, pw);
+  TORCH_CHECK(dy.size(1) == cs.OH && dy.size(2) == cs.OW, "wgrad shape");
+  auto dw = at::zeros({KH, KW, cs.Cin, cs.Cout},
+                      x.options().dtype(at::kFloat));
+  conv2d_wgrad_launch(dt_of(x), x.data_ptr(), dy.data_ptr(),
+                      dw.data_ptr<float>(), cs, cur_stream());
+  return dw;
+}
+
+// ---- batch norm ------------------------------------------------------------
+std::vector<at::Tensor> bn_fwd_train(const at::Tensor& x,
+                                     const at::Tensor& gamma,
+                                     const at::Tensor& beta, double eps,
+                                     bool relu) {
+  CHECK_IN(x);
+  int C = x.size(-1);
+  int64_t rows = x.numel() / C;
+  auto mean = at::empty({C}, x.options().dtype(at::kFloat));
+  auto invstd = at::empty({C}, x.options().dtype(at::kFloat));
+  auto y = at::empty_like(x);
+  bn_stats_launch(dt_of(x), x.data_ptr(), mean.data_ptr<float>(),
+                  invstd.data_ptr<float>(), rows, C, (float)eps, cur_stream());
+  bn_apply_launch(dt_of(x), x.data_ptr(), mean.data_ptr<float>(),
+                  invstd.data_ptr<float>(), gamma.data_ptr<float>(),
+                  beta.data_ptr<float>(), y.data_ptr(), rows, C, relu,
+                  cur_stream());
+  return {y, mean, invstd};
+}
+
+at::Tensor bn_fwd_infer(const at::Tensor& x, const at::Tensor& gamma,
+                        const at::Tensor& beta, const at::Tensor& rmean,
+                        const at::Tensor& rvar, double eps, bool relu) {
+  CHECK_IN(x);
+  int C = x.size(-1);
+  int64_t rows = x.numel() / C;
+  auto y = at::empty_like(x);
+  bn_infer_launch(dt_of(x), x.data_ptr(), rmean.data_ptr<float>(),
+                  rvar.data_ptr<float>(), gamma.data_ptr<float>(),
+                  beta.data_ptr<float>(), y.data_ptr(), rows, C, (float)eps,
+                  relu, cur_stream());
+  return y;
+}
+
+std::vector<at::Tensor> bn_bwd(const at::Tensor& x, const at::Tensor& dy,
+                               const at::Tensor& gamma, const at::Tensor& mean,
+                               const at::Tensor& invstd,
+                               const c10::optional<at::Tensor>& y_relu) {
+  CHECK_IN(x);
+  CHECK_IN(dy);
+  int C = x.size(-1);
+  int64_t rows = x.numel() / C;
+  auto sum_dy = at::zeros({C}, x.options().dtype(at::kFloat));
+  auto sum_dy_xhat = at::zeros({C}, x.options().dtype(at::kFloat));
+  auto dx = at::empty_like(x);
+  const void* yr = y_relu.has_value() ? y_relu->data_ptr() : nullptr;
+  bn_bwd_reduce_launch(dt_of(x), x.data_ptr(), dy.data_ptr(), yr,
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       sum_dy.data_ptr<float>(), sum_dy_xhat.data_ptr<float>(),
+                       rows, C, cur_stream());
+  bn_bwd_apply_launch(dt_of(x), x.data_ptr(), dy.data_ptr(), yr,
+                      mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                      gamma.data_ptr<float>(), sum_dy.data_ptr<float>(),
+                      sum_dy_xhat.data_ptr<float>(), dx.data_ptr(), rows, C,
+                      cur_stream());
+  // dgamma = sum_dy_xhat, dbeta = sum_dy (fp32, matching fp32 gamma/beta)
+  return {dx, sum_dy_xhat, sum_dy};
+}
+
+// ---- pooling ---------------------------------------------------------------
+static PoolShape pool_shape(const at::Tensor& x, int KH, int KW, int SH, int SW,
+                            int PH, int PW) {
+  PoolShape ps;
+  ps.N = x.size(0);
+  ps.H = x.size(1);
+  ps.W = x.size(2);
+  ps.C = x.size(3);
+  ps.KH = KH;
+  ps.KW = KW;
+  ps.SH = SH;
+  ps.SW = SW;
+  ps.PH = PH;
+  ps.PW = PW;
+  ps.OH = (ps.H + 2 * PH - KH) / SH + 1;
+  ps.OW = (ps.W + 2 * PW - KW) / SW + 1;
+  return ps;
+}
+
+std::vector<at::Tensor> maxpool_fwd(const at::Tensor& x, int64_t kh, int64_t kw,
+                                    int64_t sh, int64_t sw, int64_t ph,
+                                    int64_t pw) {
+  CHECK_IN(x);
+  auto ps = pool_shape(x, kh, kw, sh, sw, ph, pw);
+  auto y = at::empty({ps.N, ps.OH, ps.OW, ps.C}, x.options());
+  auto idx = at::empty({ps.N, ps.OH, ps.OW, ps.C}, x.options().dtype(at::kInt));
+  maxpool_fwd_launch(dt_of(x), x.data_ptr(), y.data_ptr(), idx.data_ptr<int>(),
+                     ps, cur_stream());
+  return {y, idx};
+}
+
+at::Tensor maxpool_bwd(const at::Tensor& dy, const at::Tensor& idx, int64_t H,
+                       int64_t W) {
+  CHECK_IN(dy);
+  PoolShape ps;
+  ps.N = dy.size(0);
+  ps.OH = dy.size(1);
+  ps.OW = dy.size(2);
+  ps.C = dy.size(3);
+  ps.H = H;
+  ps.W = W;
+  auto dxf = at::zeros({ps.N, H, W, ps.C}, dy.options().dtype(at::kFloat));
+  maxpool_bwd_launch(dt_of(dy), dy.data_ptr(), idx.data_ptr<int>(),
+                     dxf.data_ptr<float>(), ps, cur_stream());
+  if (dy.scalar_type() == at::kFloat) return dxf;
+  auto dx = at::empty({ps.N, H, W, ps.C}, dy.options());
+  cast_f32_launch(dt_of(dy), dxf.data_ptr<float>(), dx.data_ptr(), dx.numel(),
+                  cur_stream());
+  return dx;
+}
+
+at::Tensor avgpool_fwd(const at::Tensor& x, int64_t kh, int64_t kw, int64_t sh,
+                       int64_t sw, int64_t ph, int64_t pw) {
+  CHECK_IN(x);
+  auto ps = pool_shape(x, kh, kw, sh, sw, ph, pw);
+  auto y = at::empty({ps.N, ps.OH, ps.OW, ps.C}, x.options());
+  avgpool_fwd_launch(dt_of(x), x.data_ptr(), y.data_ptr(), ps, cur_stream());
+  return y;
+}
+
+at::Tensor avgpool_bwd(const at::Tensor& dy, int64_t H, int64_t W, int64_t kh,
+                       int64_t kw, int64_t sh, int64_t sw, int64_t ph,
+                       int64_t pw) {
+  CHECK_IN(dy);
+  PoolShape ps;
+  ps.N = dy.size(0);
+  ps.OH = dy.size(1);
+  ps.OW = dy.size(2);
+  ps.C = dy.size(3);
+  ps.H = H;
+  ps.W = W;
+  ps.KH = kh;
+  ps.KW = kw;
+  ps.SH = sh;
+  ps.SW = sw;
+  ps.PH = ph;
+  ps.PW = pw;
+  auto dx = at::empty({ps.N, H, W, ps.C}, dy.options());
+  avgpool_bwd_launch(dt_of(dy), dy.data_ptr(), dx.data_ptr(), ps, cur_stream());
+  return dx;
+}
+
+// ---- loss ------------------------------------------------------------------
+std::vector<at::Tensor> ce_fwd(const at::Tensor& logits,
+                               const at::Tensor& targets) {
+  CHECK_IN(logits);
+  CHECK_IN(targets);
+  int64_t rows = logits.size(0);
+  int cols = logits.size(1);
+  auto loss = at::empty({rows}, logits.options().dtype(at::kFloat));
+  auto lse = at::empty({rows}, logits.options().dtype(at::kFloat));
+  ce_fwd_launch(dt_of(logits), logits.data_ptr(), targets.data_ptr<int64_t>(),
+                loss.data_ptr<float>(), lse.data_ptr<float>(), rows, cols,
+                cur_stream());
+  return {loss, lse};
+}
+
+at::Tensor ce_bwd(const at::Tensor& logits, const at::Tensor& targets,
+                  const at::Tensor& lse, const at::Tensor& dloss) {
+  CHECK_IN(logits);
+  auto dl = at::empty_like(logits);
+  ce_bwd_launch(dt_of(logits), logits.data_ptr(), targets.data_ptr<int64_t>(),
+                lse.data_ptr<float>(), dloss.contiguous().data_ptr<float>(),
+                dl.data_ptr(), logits.size(0), logits.size(1), cur_stream());
+  return dl;
+}
+
+// ---- layer norm ------------------------------------------------------------
+std::vector<at::Tensor> ln_fwd(const at::Tensor& x, const at::Tensor& gamma,
+                               const at::Tensor& beta, double eps) {
+  CHECK_IN(x);
+  int cols = x.size(-1);
+  int64_t rows = x.numel() / cols;
+  auto y = at::empty_like(x);
+  auto mean = at::empty({rows}, x.options().dtype(at::kFloat));
+  auto invstd = at::empty({rows}, x.options().dtype(at::kFloat));
+  ln_fwd_launch(dt_of(x), x.data_ptr(), gamma.data_ptr<float>(),
+                beta.data_ptr<float>(), y.data_ptr(), mean.data_ptr<float>(),
+                invstd.data_ptr<float>(), rows, cols, (float)eps, cur_stream());
+  return {y, mean, invstd};
+}
+
+std::vector<at::Tensor> ln_bwd(const at::Tensor& x, const at::Tensor& dy,
+                               const at::Tensor& gamma, const at::Tensor& mean,
+                               const at::Tensor& invstd) {
+  CHECK_IN(x);
+  CHECK_IN(dy);
+  int cols = x.size(-1);
+  int64_t rows = x.numel() / cols;
+  auto dx = at::empty_like(x);
+  auto dgamma = at::zeros({cols}, x.options().dtype(at::kFloat));
+  auto dbeta = at::zeros({cols}, x.options().dtype(at::kFloat));
+  ln_bwd_launch(dt_of(x), x.data_ptr(), dy.data_ptr(), gamma.data_ptr<float>(),
+                mean.data_ptr<float>(), invstd.data_ptr<float>(), dx.data_ptr(),
+                dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), rows, cols,
+                cur_stream());
+  return {dx, dgamma, dbeta};
+}
+
+// ---- embedding -------------------------------------------------------------
+at::Tensor embedding_fwd(const at::Tensor& ids, const at::Tensor& table) {
+  CHECK_IN(ids);
+  CHECK_IN(table);
+  auto sizes = ids.sizes().vec();
+  sizes.push_back(table.size(1));
+  auto y = at::empty(sizes, table.options());
+  embedding_fwd_launch(dt_of(table), ids.data_ptr<int64_t>(), table.data_ptr(),
+                       y.data_ptr(), ids.numel(), table.size(1), cur_stream());
+  return y;
+}
+
+at::Tensor embedding_bwd(const at::Tensor& ids, const at::Tensor& dy,
+                         int64_t rows) {
+  CHECK_IN(ids);
+  CHECK_IN(dy);
+  int dim = dy.size(-1);
+  auto dt = at::zeros({rows, dim}, dy.options().dtype(at::kFloat));
+  embedding_bwd_launch(dt_of(dy), ids.data_ptr<int64_t>(), dy.data_ptr(),
+                       dt.data_ptr<float>(), ids.numel(), dim, cur_stream());
+  return dt;
+}
+
+// ---- optimizers ------------------------------------------------------------
+void sgd_step(at::Tensor param, at::Tensor master, const at::Tensor& grad,
+              const c10::optional<at::Tensor>& momentum_buf, double lr,
+              double momentum, double weight_decay, bool nesterov) {
+  CHECK_IN(param);
+  bool has_master = master.data_ptr() != param.data_ptr();
+  float* mb = momentum_buf.has_value() ? momentum_buf->data_ptr<float>()
+                                       : nullptr;
+  sgd_step_launch(dt_of(param), grad.data_ptr(), dt_of(grad), param.data_ptr(),
+                  master.data_ptr<float>(), mb, param.numel(), (float)lr,
+                  (float)momentum, (float)weight_decay, nesterov, has_master,
+                  cur_stream());
+}
+
+void adam_step(at::Tensor param, at::Tensor master, const at::Tensor& grad,
+               at::Tensor m, at::Tensor v, int64_t step, double lr,
+               double beta1, double beta2, double eps, double weight_decay,
+               bool adamw) {
+  CHECK_IN(param);
+  bool has_master = master.data_ptr() != param.data_ptr();
+  adam_step_launch(dt_of(param), grad.data_ptr(), dt_of(grad), param.data_ptr(),
+                   master.data_ptr<float>(), m.data_ptr<float>(),
+                   v.data_ptr<float>(), param.numel(), (int)step, (float)lr,
+                   (float)beta1, (float)beta2, (float)eps, (float)weight_decay,
+                   adamw, has_master, cur_stream());
+}
+
+}  // namespace tnn
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("act_fwd", &tnn::act_fwd);
+  m.def("act_bwd", &tnn::act_bwd);
+  m.def("relu_bwd_mask", &tnn::relu_bwd_mask);
+  m.def("dropout_fwd", &tnn::dropout_fwd);
+  m.def("dropout_bwd", &tnn::dropout_bwd);
+  m.def("colsum", &tnn::colsum);
+  m.def("gemm", &tnn::gemm);
+  m.def("gemm_nt", &tnn::gemm_nt);
+  m.def("gemm_tn", &tnn::gemm_tn);
+  m.def("mfma_selftest", &tnn::mfma_selftest);
+  m.def("mfma_selftest_f32", &tnn::mfma_selftest_f32);
+  m.def("conv2d_fwd", &tnn::conv2d_fwd);
+  m.def("conv2d_dgrad", &tnn::conv2d_dgrad);
+  m.def("conv2d_wgrad", &tnn::conv2d_wgrad);
+  m.def("bn_fwd_train", &tnn::bn_fwd_train);
+  m.def("bn_fwd_infer", &tnn::bn_fwd_infer);
+  m.def("bn_bwd", &tnn::bn_bwd);
+  m.def("maxpool_fwd", &tnn::maxpool_fwd);
+  m.def("maxpool_bwd", &tnn::maxpool_bwd);
+  m.def("avgpool_fwd", &tnn::avgpool_fwd);
+  m.def("avgpool_bwd", &tnn::avgpool_bwd);
+  m.def("ce_fwd", &tnn::ce_fwd);
+  m.def("ce_bwd", &tnn::ce_bwd);
+  m.def("ln_fwd", &tnn::ln_fwd);
+  m.def("ln_bwd", &tnn::ln_bwd);
+  m.def("embedding_fwd", &tnn::embedding_fwd);
+  m.def("embedding_bwd", &tnn::embedding_bwd);
+  m.def("sgd_step", &tnn::sgd_step);
+  m.def("adam_step", &tnn::adam_step);
+}

@@ -1,0 +1,388 @@
+// NHWC implicit-GEMM convolution on MFMA: fwd / dgrad / wgrad.
+//
+// Replaces the reference's cuDNN-frontend conv graphs
+// (src/nn/layers_impl/cuda/cudnn_conv2d_ops.cu) and the legacy
+// im2col->GEMM path (src/nn/layers_impl/legacy_conv2d_layer.cpp:137) with
+// the fused form SURVEY §7 calls for: the im2col gather is folded into the
+// GEMM's global->LDS staging, so no im2col buffer ever exists.
+//
+// GEMM views (row-major):
+//   fwd  : y[M=N*OH*OW, Cout] = A[M, K=KH*KW*Cin] @ w[K, Cout]
+//   dgrad: dx[M=N*H*W, Cin]   = A'[M, K=KH*KW*Cout] @ w_t[K, Cin]
+//   wgrad: dw[K=KH*KW*Cin, Cout] += im2col(x)^T @ dy  (split-M atomics)
+//
+// A-tiles are gathered on the fly: a 16B k-chunk stays within one (kh,kw)
+// slice whenever Cin (fwd) / Cout (dgrad) is a multiple of the vector
+// width, which holds for every hot layer; otherwise a scalar gather path
+// handles stem convs (Cin=3).
+
+#include "common.h"
+#include "kernels.h"
+#include "tile_gemm.h"
+
+namespace tnn {
+
+using namespace tile;
+
+// ---------------------------------------------------------------------------
+template <typename T>
+__launch_bounds__(THREADS)
+__global__ void k_conv_fwd(const T* __restrict__ X, const T* __restrict__ Wt,
+                           const float* __restrict__ bias_f32,
+                           const T* __restrict__ bias_t, T* __restrict__ Y,
+                           ConvShape cs, int act_kind) {
+  constexpr int P = BKP<T>();
+  constexpr int V = 16 / sizeof(T);
+  __shared__ T As[BM * P];
+  __shared__ T Bs[BN * P];
+
+  const int M = cs.N * cs.OH * cs.OW;
+  const int K = cs.KH * cs.KW * cs.Cin;
+  const int m0 = blockIdx.x * BM;
+  const int n0 = blockIdx.y * BN;
+  const WaveCoord wc;
+  f32x4 acc[FM][FN] = {};
+  using VecT = Pack16<T>;
+
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    // ---- stage A: im2col gather ----
+#pragma unroll
+    for (int c = threadIdx.x; c < BM * (BK / V); c += THREADS) {
+      int row = c / (BK / V);
+      int kk = (c % (BK / V)) * V;
+      int gm = m0 + row, gk = k0 + kk;
+      VecT v = {};
+      if (gm < M && gk < K) {
+        int n = gm / (cs.OH * cs.OW);
+        int rem = gm % (cs.OH * cs.OW);
+        int oh = rem / cs.OW, ow = rem % cs.OW;
+        int ci = gk % cs.Cin;
+        int kidx = gk / cs.Cin;
+        int kw = kidx % cs.KW, kh = kidx / cs.KW;
+        if (ci + V <= cs.Cin && gk + V <= K) {
+          int ih = oh * cs.SH - cs.PH + kh;
+          int iw = ow * cs.SW - cs.PW + kw;
+          if (ih >= 0 && ih < cs.H && iw >= 0 && iw < cs.W)
+            v = *(const VecT*)&X[(((int64_t)n * cs.H + ih) * cs.W + iw) *
+                                     cs.Cin + ci];
+        } else {
+#pragma unroll
+          for (int j = 0; j < V; ++j) {
+            int k = gk + j;
+            if (k < K) {
+              int cij = k % cs.Cin;
+              int kj = k / cs.Cin;
+              int kwj = kj % cs.KW, khj = kj / cs.KW;
+              int ih = oh * cs.SH - cs.PH + khj;
+              int iw = ow * cs.SW - cs.PW + kwj;
+              if (ih >= 0 && ih < cs.H && iw >= 0 && iw < cs.W)
+                v.e[j] = X[(((int64_t)n * cs.H + ih) * cs.W + iw) * cs.Cin + cij];
+            }
+          }
+        }
+      }
+      *(VecT*)&As[row * P + kk] = v;
+    }
+    // ---- stage B (weights [K, Cout]) -> Bs[n][k] scatter-transpose ----
+#pragma unroll
+    for (int c = threadIdx.x; c < BK * (BN / V); c += THREADS) {
+      int kk = c / (BN / V);
+      int nn = (c % (BN / V)) * V;
+      int gk = k0 + kk, gn = n0 + nn;
+      VecT v = {};
+      if (gk < K) {
+        if (gn + V <= cs.Cout) {
+          v = *(const VecT*)&Wt[(int64_t)gk * cs.Cout + gn];
+        } else {
+#pragma unroll
+          for (int j = 0; j < V; ++j)
+            if (gn + j < cs.Cout) v.e[j] = Wt[(int64_t)gk * cs.Cout + gn + j];
+        }
+      }
+#pragma unroll
+      for (int j = 0; j < V; ++j) Bs[(nn + j) * P + kk] = v.e[j];
+    }
+    __syncthreads();
+    mfma_compute_tile(As, Bs, wc, acc);
+    __syncthreads();
+  }
+
+  epilogue_visit(wc, acc, m0, n0, [&](int row, int col, float v) {
+    if (row < M && col < cs.Cout) {
+      if (bias_f32) v += bias_f32[col];
+      if (bias_t) v += VecIO<T>::to_f32(bias_t[col]);
+      if (act_kind != ACT_LINEAR) v = act_apply(v, act_kind);
+      Y[(int64_t)row * cs.Cout + col] = VecIO<T>::from_f32(v);
+    }
+  });
+}
+
+// ---------------------------------------------------------------------------
+// dgrad: dx[n,ih,iw,ci] = sum_{kh,kw,co} dy[n,oh,ow,co] * w[kh,kw,ci,co]
+// with oh = (ih+PH-kh)/SH when divisible. Wt here is the transposed weight
+// [KH,KW,Cout,Cin] so B rows are k=(kh,kw,co) with ci contiguous.
+// ---------------------------------------------------------------------------
+template <typename T>
+__launch_bounds__(THREADS)
+__global__ void k_conv_dgrad(const T* __restrict__ DY, const T* __restrict__ WT,
+                             T* __restrict__ DX, ConvShape cs) {
+  constexpr int P = BKP<T>();
+  constexpr int V = 16 / sizeof(T);
+  __shared__ T As[BM * P];
+  __shared__ T Bs[BN * P];
+
+  const int M = cs.N * cs.H * cs.W;
+  const int K = cs.KH * cs.KW * cs.Cout;
+  const int m0 = blockIdx.x * BM;
+  const int n0 = blockIdx.y * BN;
+  const WaveCoord wc;
+  f32x4 acc[FM][FN] = {};
+  using VecT = Pack16<T>;
+
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    // ---- stage A: gather dy with stride/padding inversion ----
+#pragma unroll
+    for (int c = threadIdx.x; c < BM * (BK / V); c += THREADS) {
+      int row = c / (BK / V);
+      int kk = (c % (BK / V)) * V;
+      int gm = m0 + row, gk = k0 + kk;
+      VecT v = {};
+      if (gm < M && gk < K) {
+        int n = gm / (cs.H * cs.W);
+        int rem = gm % (cs.H * cs.W);
+        int ih = rem / cs.W, iw = rem % cs.W;
+        int co = gk % cs.Cout;
+        int kidx = gk / cs.Cout;
+        int kw = kidx % cs.KW, kh = kidx / cs.KW;
+        auto gather_one = [&](int khj, int kwj, int coj) -> T {
+          int th = ih + cs.PH - khj, tw = iw + cs.PW - kwj;
+          if (th < 0 || tw < 0 || th % cs.SH || tw % cs.SW) return T(0.0f);
+          int oh = th / cs.SH, ow = tw / cs.SW;
+          if (oh >= cs.OH || ow >= cs.OW) return T(0.0f);
+          return DY[(((int64_t)n * cs.OH + oh) * cs.OW + ow) * cs.Cout + coj];
+        };
+        if (co + V <= cs.Cout && gk + V <= K) {
+          int th = ih + cs.PH - kh, tw = iw + cs.PW - kw;
+          if (th >= 0 && tw >= 0 && th % cs.SH == 0 && tw % cs.SW == 0) {
+            int oh = th / cs.SH, ow = tw / cs.SW;
+            if (oh < cs.OH && ow < cs.OW)
+              v = *(const VecT*)&DY[(((int64_t)n * cs.OH + oh) * cs.OW + ow) *
+                                        cs.Cout + co];
+          }
+        } else {
+#pragma unroll
+          for (int j = 0; j < V; ++j) {
+            int k = gk + j;
+            if (k < K) {
+              int coj = k % cs.Cout;
+              int kj = k / cs.Cout;
+              v.e[j] = gather_one(kj / cs.KW, kj % cs.KW, coj);
+            }
+          }
+        }
+      }
+      *(VecT*)&As[row * P + kk] = v;
+    }
+    // ---- stage B (w_t [K, Cin]) -> Bs[ci][k] ----
+#pragma unroll
+    for (int c = threadIdx.x; c < BK * (BN / V); c += THREADS) {
+      int kk = c / (BN / V);
+      int nn = (c % (BN / V)) * V;
+      int gk = k0 + kk, gn = n0 + nn;
+      VecT v = {};
+      if (gk < K) {
+        if (gn + V <= cs.Cin) {
+          v = *(const VecT*)&WT[(int64_t)gk * cs.Cin + gn];
+        } else {
+#pragma unroll
+          for (int j = 0; j < V; ++j)
+            if (gn + j < cs.Cin) v.e[j] = WT[(int64_t)gk * cs.Cin + gn + j];
+        }
+      }
+#pragma unroll
+      for (int j = 0; j < V; ++j) Bs[(nn + j) * P + kk] = v.e[j];
+    }
+    __syncthreads();
+    mfma_compute_tile(As, Bs, wc, acc);
+    __syncthreads();
+  }
+
+  epilogue_visit(wc, acc, m0, n0, [&](int row, int col, float v) {
+    if (row < M && col < cs.Cin)
+      DX[(int64_t)row * cs.Cin + col] = VecIO<T>::from_f32(v);
+  });
+}
+
+// ---------------------------------------------------------------------------
+// wgrad: dw[(kh,kw,ci), co] += sum_m x_gather * dy ; fp32 atomics over
+// grid.z m-slices.
+// ---------------------------------------------------------------------------
+template <typename T>
+__launch_bounds__(THREADS)
+__global__ void k_conv_wgrad(const T* __restrict__ X, const T* __restrict__ DY,
+                             float* __restrict__ DW, ConvShape cs) {
+  constexpr int P = BKP<T>();
+  constexpr int V = 16 / sizeof(T);
+  __shared__ T As[BM * P];  // rows = (kh,kw,ci), k = m
+  __shared__ T Bs[BN * P];  // rows = co, k = m
+
+  const int M = cs.N * cs.OH * cs.OW;         // reduction dim
+  const int Kout = cs.KH * cs.KW * cs.Cin;    // output rows
+  const int r0 = blockIdx.x * BM;
+  const int n0 = blockIdx.y * BN;
+  const int64_t m_begin = (int64_t)M * blockIdx.z / gridDim.z;
+  const int64_t m_end = (int64_t)M * (blockIdx.z + 1) / gridDim.z;
+  const WaveCoord wc;
+  f32x4 acc[FM][FN] = {};
+  using VecT = Pack16<T>;
+
+  for (int64_t k0 = m_begin; k0 < m_end; k0 += BK) {
+    // ---- stage A^T: x rows ci-contiguous, scatter into As[ko][m] ----
+#pragma unroll
+    for (int c = threadIdx.x; c < BK * (BM / V); c += THREADS) {
+      int mm = c / (BM / V);
+      int rr = (c % (BM / V)) * V;
+      int64_t gm = k0 + mm;
+      int gr = r0 + rr;
+      VecT v = {};
+      if (gm < m_end && gr < Kout) {
+        int n = (int)(gm / (cs.OH * cs.OW));
+        int rem = (int)(gm % (cs.OH * cs.OW));
+        int oh = rem / cs.OW, ow = rem % cs.OW;
+        int ci = gr % cs.Cin;
+        int kidx = gr / cs.Cin;
+        int kw = kidx % cs.KW, kh = kidx / cs.KW;
+        if (ci + V <= cs.Cin && gr + V <= Kout) {
+          int ih = oh * cs.SH - cs.PH + kh;
+          int iw = ow * cs.SW - cs.PW + kw;
+          if (ih >= 0 && ih < cs.H && iw >= 0 && iw < cs.W)
+            v = *(const VecT*)&X[(((int64_t)n * cs.H + ih) * cs.W + iw) *
+                                     cs.Cin + ci];
+        } else {
+#pragma unroll
+          for (int j = 0; j < V; ++j) {
+            int r = gr + j;
+            if (r < Kout) {
+              int cij = r % cs.Cin;
+              int kj = r / cs.Cin;
+              int kwj = kj % cs.KW, khj = kj / cs.KW;
+              int ih = oh * cs.SH - cs.PH + khj;
+              int iw = ow * cs.SW - cs.PW + kwj;
+              if (ih >= 0 && ih < cs.H && iw >= 0 && iw < cs.W)
+                v.e[j] = X[(((int64_t)n * cs.H + ih) * cs.W + iw) * cs.Cin + cij];
+            }
+          }
+        }
+      }
+#pragma unroll
+      for (int j = 0; j < V; ++j) As[(rr + j) * P + mm] = v.e[j];
+    }
+    // ---- stage B: dy[m][co] -> Bs[co][m] ----
+#pragma unroll
+    for (int c = threadIdx.x; c < BK * (BN / V); c += THREADS) {
+      int mm = c / (BN / V);
+      int nn = (c % (BN / V)) * V;
+      int64_t gm = k0 + mm;
+      int gn = n0 + nn;
+      VecT v = {};
+      if (gm < m_end) {
+        if (gn + V <= cs.Cout) {
+          v = *(const VecT*)&DY[gm * cs.Cout + gn];
+        } else {
+#pragma unroll
+          for (int j = 0; j < V; ++j)
+            if (gn + j < cs.Cout) v.e[j] = DY[gm * cs.Cout + gn + j];
+        }
+      }
+#pragma unroll
+      for (int j = 0; j < V; ++j) Bs[(nn + j) * P + mm] = v.e[j];
+    }
+    __syncthreads();
+    mfma_compute_tile(As, Bs, wc, acc);
+    __syncthreads();
+  }
+
+  const bool single = gridDim.z == 1;
+  epilogue_visit(wc, acc, r0, n0, [&](int row, int col, float v) {
+    if (row < Kout && col < cs.Cout) {
+      if (single)
+        DW[(int64_t)row * cs.Cout + col] = v;
+      else
+        atomicAdd(&DW[(int64_t)row * cs.Cout + col], v);
+    }
+  });
+}
+
+// ---------------------------------------------------------------------------
+template <typename T>
+__global__ void k_transpose_w(const T* __restrict__ W, T* __restrict__ WT,
+                              int KHW, int Cin, int Cout) {
+  // [khw, ci, co] -> [khw, co, ci]
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t n = (int64_t)KHW * Cin * Cout;
+  if (i >= n) return;
+  int co = i % Cout;
+  int ci = (i / Cout) % Cin;
+  int khw = i / ((int64_t)Cin * Cout);
+  WT[((int64_t)khw * Cout + co) * Cin + ci] = W[i];
+}
+
+// ---------------------------------------------------------------------------
+void conv2d_fwd_launch(DT dt, const void* x, const void* w, const void* bias,
+                       void* y, const ConvShape& cs, bool relu, hipStream_t s) {
+  int M = cs.N * cs.OH * cs.OW;
+  dim3 grid(ceil_div(M, BM), ceil_div(cs.Cout, BN));
+  int act = relu ? ACT_RELU : ACT_LINEAR;
+  if (dt == DT::F32)
+    hipLaunchKernelGGL(k_conv_fwd<float>, grid, dim3(THREADS), 0, s,
+                       (const float*)x, (const float*)w, (const float*)bias,
+                       (const float*)nullptr, (float*)y, cs, act);
+  else
+    hipLaunchKernelGGL(k_conv_fwd<bf16>, grid, dim3(THREADS), 0, s,
+                       (const bf16*)x, (const bf16*)w, (const float*)nullptr,
+                       (const bf16*)bias, (bf16*)y, cs, act);
+}
+
+void conv2d_dgrad_launch(DT dt, const void* dy, const void* w_t, void* dx,
+                         const ConvShape& cs, hipStream_t s) {
+  int M = cs.N * cs.H * cs.W;
+  dim3 grid(ceil_div(M, BM), ceil_div(cs.Cin, BN));
+  if (dt == DT::F32)
+    hipLaunchKernelGGL(k_conv_dgrad<float>, grid, dim3(THREADS), 0, s,
+                       (const float*)dy, (const float*)w_t, (float*)dx, cs);
+  else
+    hipLaunchKernelGGL(k_conv_dgrad<bf16>, grid, dim3(THREADS), 0, s,
+                       (const bf16*)dy, (const bf16*)w_t, (bf16*)dx, cs);
+}
+
+void conv2d_wgrad_launch(DT dt, const void* x, const void* dy, float* dw_f32,
+                         const ConvShape& cs, hipStream_t s) {
+  int M = cs.N * cs.OH * cs.OW;
+  int Kout = cs.KH * cs.KW * cs.Cin;
+  int base = ceil_div(Kout, BM) * ceil_div(cs.Cout, BN);
+  int want = 512;
+  int z = base >= want ? 1
+                       : std::min(ceil_div(M, 4 * BK), ceil_div(want, base));
+  z = std::max(z, 1);
+  dim3 grid(ceil_div(Kout, BM), ceil_div(cs.Cout, BN), z);
+  if (dt == DT::F32)
+    hipLaunchKernelGGL(k_conv_wgrad<float>, grid, dim3(THREADS), 0, s,
+                       (const float*)x, (const float*)dy, dw_f32, cs);
+  else
+    hipLaunchKernelGGL(k_conv_wgrad<bf16>, grid, dim3(THREADS), 0, s,
+                       (const bf16*)x, (const bf16*)dy, dw_f32, cs);
+}
+
+void transpose_w_launch(DT dt, const void* w, void* w_t, int KH, int KW,
+                        int Cin, int Cout, hipStream_t s) {
+  int64_t n = (int64_t)KH * KW * Cin * Cout;
+  int blocks = (int)((n + 255) / 256);
+  if (dt == DT::F32)
+    hipLaunchKernelGGL(k_transpose_w<float>, dim3(blocks), dim3(256), 0, s,
+                       (const float*)w, (float*)w_t, KH * KW, Cin, Cout);
+  else
+    hipLaunchKernelGGL(k_transpose_w<bf16>, dim3(blocks), dim3(256), 0, s,
+                       (const bf16*)w, (bf16*)w_t, KH * KW, Cin, Cout);
+}
+
+}  // namespace tnn

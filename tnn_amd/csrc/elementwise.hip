@@ -1,0 +1,200 @@
+// Elementwise kernels: activations, ReLU backward mask, dropout (Philox),
+// column sums (replaces reference src/ops/cuda/kernels.cu element-wise set,
+// src/nn/activations_impl/cuda/*, dropout.cu, add_bias/bgrad column reduce).
+//
+// All grid-stride, 256-thread workgroups (wave64 x 4), vectorized 4-wide.
+
+#include "common.h"
+#include "kernels.h"
+
+namespace tnn {
+
+template <typename T, typename F>
+__global__ void k_ewise1(const T* __restrict__ x, T* __restrict__ y,
+                         int64_t n, F f) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) y[i] = VecIO<T>::from_f32(f(VecIO<T>::to_f32(x[i])));
+}
+
+static inline int ew_blocks(int64_t n) {
+  int64_t b = (n + 255) / 256;
+  return (int)(b < 2048 ? b : 2048);
+}
+
+// ---- activations -----------------------------------------------------------
+template <typename T>
+__global__ void k_act_fwd(const T* x, T* y, int64_t n, int kind) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride)
+    y[i] = VecIO<T>::from_f32(act_apply(VecIO<T>::to_f32(x[i]), kind));
+}
+
+template <typename T>
+__global__ void k_act_bwd(const T* dy, const T* x, const T* y, T* dx,
+                          int64_t n, int kind) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride)
+    dx[i] = VecIO<T>::from_f32(act_grad(VecIO<T>::to_f32(dy[i]),
+                                        VecIO<T>::to_f32(x[i]),
+                                        VecIO<T>::to_f32(y[i]), kind));
+}
+
+void act_fwd_launch(DT dt, const void* x, void* y, int64_t n, int kind,
+                    hipStream_t s) {
+  if (dt == DT::F32)
+    hipLaunchKernelGGL(k_act_fwd<float>, dim3(ew_blocks(n)), dim3(256), 0, s,
+                       (const float*)x, (float*)y, n, kind);
+  else
+    hipLaunchKernelGGL(k_act_fwd<bf16>, dim3(ew_blocks(n)), dim3(256), 0, s,
+                       (const bf16*)x, (bf16*)y, n, kind);
+}
+
+void act_bwd_launch(DT dt, const void* dy, const void* x, const void* y,
+                    void* dx, int64_t n, int kind, hipStream_t s) {
+  if (dt == DT::F32)
+    hipLaunchKernelGGL(k_act_bwd<float>, dim3(ew_blocks(n)), dim3(256), 0, s,
+                       (const float*)dy, (const float*)x, (const float*)y,
+                       (float*)dx, n, kind);
+  else
+    hipLaunchKernelGGL(k_act_bwd<bf16>, dim3(ew_blocks(n)), dim3(256), 0, s,
+                       (const bf16*)dy, (const bf16*)x, (const bf16*)y,
+                       (bf16*)dx, n, kind);
+}
+
+// ---- relu backward through the saved output (mask = y > 0) -----------------
+template <typename T>
+__global__ void k_relu_bwd_mask(const T* dy, const T* y, T* dx, int64_t n) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride)
+    dx[i] = VecIO<T>::to_f32(y[i]) > 0.0f ? dy[i] : VecIO<T>::from_f32(0.0f);
+}
+
+void relu_bwd_mask_launch(DT dt, const void* dy, const void* y, void* dx,
+                          int64_t n, hipStream_t s) {
+  if (dt == DT::F32)
+    hipLaunchKernelGGL(k_relu_bwd_mask<float>, dim3(ew_blocks(n)), dim3(256), 0,
+                       s, (const float*)dy, (const float*)y, (float*)dx, n);
+  else
+    hipLaunchKernelGGL(k_relu_bwd_mask<bf16>, dim3(ew_blocks(n)), dim3(256), 0,
+                       s, (const bf16*)dy, (const bf16*)y, (bf16*)dx, n);
+}
+
+// ---- dropout (Philox 4x32; reference dropout.cu:33 vectorized form) --------
+template <typename T>
+__global__ void k_dropout_fwd(const T* x, T* y, uint8_t* mask, int64_t n,
+                              float p, uint64_t seed) {
+  const float scale = 1.0f / (1.0f - p);
+  Philox rng(seed);
+  int64_t i4 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i4 * 4 < n; i4 += stride) {
+    uint4 r = rng(i4);
+    unsigned int rs[4] = {r.x, r.y, r.z, r.w};
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int64_t i = i4 * 4 + j;
+      if (i < n) {
+        bool keep = u32_to_uniform(rs[j]) > p;
+        mask[i] = keep;
+        y[i] = keep ? VecIO<T>::from_f32(VecIO<T>::to_f32(x[i]) * scale)
+                    : VecIO<T>::from_f32(0.0f);
+      }
+    }
+  }
+}
+
+template <typename T>
+__global__ void k_dropout_bwd(const T* dy, const uint8_t* mask, T* dx,
+                              int64_t n, float p) {
+  const float scale = 1.0f / (1.0f - p);
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride)
+    dx[i] = mask[i] ? VecIO<T>::from_f32(VecIO<T>::to_f32(dy[i]) * scale)
+                    : VecIO<T>::from_f32(0.0f);
+}
+
+void dropout_fwd_launch(DT dt, const void* x, void* y, uint8_t* mask,
+                        int64_t n, float p, uint64_t seed, hipStream_t s) {
+  int blocks = ew_blocks((n + 3) / 4);
+  if (dt == DT::F32)
+    hipLaunchKernelGGL(k_dropout_fwd<float>, dim3(blocks), dim3(256), 0, s,
+                       (const float*)x, (float*)y, mask, n, p, seed);
+  else
+    hipLaunchKernelGGL(k_dropout_fwd<bf16>, dim3(blocks), dim3(256), 0, s,
+                       (const bf16*)x, (bf16*)y, mask, n, p, seed);
+}
+
+void dropout_bwd_launch(DT dt, const void* dy, const uint8_t* mask, void* dx,
+                        int64_t n, float p, hipStream_t s) {
+  if (dt == DT::F32)
+    hipLaunchKernelGGL(k_dropout_bwd<float>, dim3(ew_blocks(n)), dim3(256), 0,
+                       s, (const float*)dy, mask, (float*)dx, n, p);
+  else
+    hipLaunchKernelGGL(k_dropout_bwd<bf16>, dim3(ew_blocks(n)), dim3(256), 0,
+                       s, (const bf16*)dy, mask, (bf16*)dx, n, p);
+}
+
+// ---- column sum (bias gradient; reference run_bgrad_kernel_ex) -------------
+// x: [rows, cols] -> out_f32[cols]; each block owns a col-chunk x row-slice,
+// partials combined with one atomic per (block, col).
+template <typename T>
+__global__ void k_colsum(const T* __restrict__ x, float* __restrict__ out,
+                         int64_t rows, int cols) {
+  const int col = blockIdx.x * 64 + (threadIdx.x & 63);
+  const int rslice = blockIdx.y;
+  const int nrs = gridDim.y;
+  if (col >= cols) return;
+  const int64_t r0 = rows * rslice / nrs, r1 = rows * (rslice + 1) / nrs;
+  float acc = 0.0f;
+  // 4 waves stride over rows
+  for (int64_t r = r0 + (threadIdx.x >> 6); r < r1; r += 4)
+    acc += VecIO<T>::to_f32(x[r * cols + col]);
+  __shared__ float sh[4][64];
+  sh[threadIdx.x >> 6][threadIdx.x & 63] = acc;
+  __syncthreads();
+  if (threadIdx.x < 64) {
+    float v = sh[0][threadIdx.x] + sh[1][threadIdx.x] + sh[2][threadIdx.x] +
+              sh[3][threadIdx.x];
+    if (nrs == 1)
+      out[col] = v;
+    else
+      atomicAdd(&out[col], v);
+  }
+}
+
+void colsum_launch(DT dt, const void* x, void* out_f32, int64_t rows,
+                   int64_t cols, hipStream_t s) {
+  int rslices = rows > 16384 ? 8 : 1;
+  dim3 grid((cols + 63) / 64, rslices);
+  if (dt == DT::F32)
+    hipLaunchKernelGGL(k_colsum<float>, grid, dim3(256), 0, s, (const float*)x,
+                       (float*)out_f32, rows, (int)cols);
+  else
+    hipLaunchKernelGGL(k_colsum<bf16>, grid, dim3(256), 0, s, (const bf16*)x,
+                       (float*)out_f32, rows, (int)cols);
+}
+
+// ---- f32 -> T cast ---------------------------------------------------------
+template <typename T>
+__global__ void k_cast(const float* x, T* y, int64_t n) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) y[i] = VecIO<T>::from_f32(x[i]);
+}
+
+void cast_f32_launch(DT dt_out, const float* x, void* y, int64_t n,
+                     hipStream_t s) {
+  if (dt_out == DT::F32)
+    hipLaunchKernelGGL(k_cast<float>, dim3(ew_blocks(n)), dim3(256), 0, s, x,
+                       (float*)y, n);
+  else
+    hipLaunchKernelGGL(k_cast<bf16>, dim3(ew_blocks(n)), dim3(256), 0, s, x,
+                       (bf16*)y, n);
+}
+
+}  // namespace tnn

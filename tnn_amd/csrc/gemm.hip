@@ -1,0 +1,276 @@
+// MFMA GEMM kernels: NN (fused bias+activation epilogue), NT (dgrad),
+// TN (wgrad, split-M atomic fp32 accumulate).
+// Replaces the reference's cublasGemmEx / cuDNN-frontend matmul call sites
+// (src/math/cuda/gemm.cu:64, src/math/cuda/cudnn_gemm.cu:285,
+// src/nn/layers_impl/cuda/dense_ops.cu:17-117).
+
+#include "common.h"
+#include "kernels.h"
+#include "tile_gemm.h"
+
+namespace tnn {
+
+using namespace tile;
+
+// ---------------------------------------------------------------------------
+// NN / NT kernel: C[M,N] = act(A[M,K] @ B + bias)
+//   TRANS_B = false: B is [K, N] row-major (staged transposed into LDS)
+//   TRANS_B = true:  B is [N, K] row-major (staged directly; C = A @ B^T)
+// ---------------------------------------------------------------------------
+
+template <typename T, bool TRANS_B>
+__launch_bounds__(THREADS)
+__global__ void k_gemm(const T* __restrict__ A, const T* __restrict__ B,
+                       const float* __restrict__ bias_f32,
+                       const T* __restrict__ bias_t, T* __restrict__ C,
+                       int M, int N, int K, int act_kind) {
+  constexpr int P = BKP<T>();
+  constexpr int V = 16 / sizeof(T);  // elems per 16B vector
+  __shared__ T As[BM * P];
+  __shared__ T Bs[BN * P];
+
+  const int m0 = blockIdx.x * BM;
+  const int n0 = blockIdx.y * BN;
+  const WaveCoord wc;
+  f32x4 acc[FM][FN] = {};
+
+  using VecT = Pack16<T>;
+
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    // ---- stage A: [BM][BK], vectorized, zero-filled at edges ----
+#pragma unroll
+    for (int c = threadIdx.x; c < BM * (BK / V); c += THREADS) {
+      int row = c / (BK / V);
+      int kk = (c % (BK / V)) * V;
+      int gm = m0 + row, gk = k0 + kk;
+      VecT v = {};
+      if (gm < M) {
+        if (gk + V <= K) {
+          v = *(const VecT*)&A[(int64_t)gm * K + gk];
+        } else {
+#pragma unroll
+          for (int j = 0; j < V; ++j)
+            if (gk + j < K) v.e[j] = A[(int64_t)gm * K + gk + j];
+        }
+      }
+      *(VecT*)&As[row * P + kk] = v;
+    }
+    // ---- stage B into Bs[n][k] ----
+    if constexpr (TRANS_B) {
+      // B[N,K]: rows are output columns; contiguous copy
+#pragma unroll
+      for (int c = threadIdx.x; c < BN * (BK / V); c += THREADS) {
+        int col = c / (BK / V);
+        int kk = (c % (BK / V)) * V;
+        int gn = n0 + col, gk = k0 + kk;
+        VecT v = {};
+        if (gn < N) {
+          if (gk + V <= K) {
+            v = *(const VecT*)&B[(int64_t)gn * K + gk];
+          } else {
+#pragma unroll
+            for (int j = 0; j < V; ++j)
+              if (gk + j < K) v.e[j] = B[(int64_t)gn * K + gk + j];
+          }
+        }
+        *(VecT*)&Bs[col * P + kk] = v;
+      }
+    } else {
+      // B[K,N]: load n-contiguous vectors, scatter-transpose into LDS
+#pragma unroll
+      for (int c = threadIdx.x; c < BK * (BN / V); c += THREADS) {
+        int kk = c / (BN / V);
+        int nn = (c % (BN / V)) * V;
+        int gk = k0 + kk, gn = n0 + nn;
+        VecT v = {};
+        if (gk < K) {
+          if (gn + V <= N) {
+            v = *(const VecT*)&B[(int64_t)gk * N + gn];
+          } else {
+#pragma unroll
+            for (int j = 0; j < V; ++j)
+              if (gn + j < N) v.e[j] = B[(int64_t)gk * N + gn + j];
+          }
+        }
+#pragma unroll
+        for (int j = 0; j < V; ++j) Bs[(nn + j) * P + kk] = v.e[j];
+      }
+    }
+    __syncthreads();
+    mfma_compute_tile(As, Bs, wc, acc);
+    __syncthreads();
+  }
+
+  // ---- epilogue: bias + activation + cast store ----
+  epilogue_visit(wc, acc, m0, n0, [&](int row, int col, float v) {
+    if (row < M && col < N) {
+      if (bias_f32) v += bias_f32[col];
+      if (bias_t) v += VecIO<T>::to_f32(bias_t[col]);
+      if (act_kind != ACT_LINEAR) v = act_apply(v, act_kind);
+      C[(int64_t)row * N + col] = VecIO<T>::from_f32(v);
+    }
+  });
+}
+
+// ---------------------------------------------------------------------------
+// TN kernel: C[K,N] += A[M,K]^T @ B[M,N], fp32 atomic accumulate.
+// Grid.z slices the M reduction (split-K in the GEMM sense) so small-output
+// wgrads still fill 256 CUs (SURVEY §7 hard part 1).
+// ---------------------------------------------------------------------------
+
+template <typename T>
+__launch_bounds__(THREADS)
+__global__ void k_gemm_tn(const T* __restrict__ A, const T* __restrict__ B,
+                          float* __restrict__ C, int M, int N, int K) {
+  constexpr int P = BKP<T>();
+  constexpr int V = 16 / sizeof(T);
+  __shared__ T As[BM * P];  // rows = K-dim, k = m-chunk
+  __shared__ T Bs[BN * P];
+
+  const int r0 = blockIdx.x * BM;  // output row = A column
+  const int n0 = blockIdx.y * BN;
+  const int64_t m_begin = (int64_t)M * blockIdx.z / gridDim.z;
+  const int64_t m_end = (int64_t)M * (blockIdx.z + 1) / gridDim.z;
+  const WaveCoord wc;
+  f32x4 acc[FM][FN] = {};
+
+  using VecT = Pack16<T>;
+
+  for (int64_t k0 = m_begin; k0 < m_end; k0 += BK) {
+    // ---- stage A^T: load A[m][r..r+V] (contiguous), scatter to As[r][m] ----
+#pragma unroll
+    for (int c = threadIdx.x; c < BK * (BM / V); c += THREADS) {
+      int mm = c / (BM / V);
+      int rr = (c % (BM / V)) * V;
+      int64_t gm = k0 + mm;
+      int gr = r0 + rr;
+      VecT v = {};
+      if (gm < m_end) {
+        if (gr + V <= K) {
+          v = *(const VecT*)&A[gm * K + gr];
+        } else {
+#pragma unroll
+          for (int j = 0; j < V; ++j)
+            if (gr + j < K) v.e[j] = A[gm * K + gr + j];
+        }
+      }
+#pragma unroll
+      for (int j = 0; j < V; ++j) As[(rr + j) * P + mm] = v.e[j];
+    }
+    // ---- stage B[m][n] -> Bs[n][m] ----
+#pragma unroll
+    for (int c = threadIdx.x; c < BK * (BN / V); c += THREADS) {
+      int mm = c / (BN / V);
+      int nn = (c % (BN / V)) * V;
+      int64_t gm = k0 + mm;
+      int gn = n0 + nn;
+      VecT v = {};
+      if (gm < m_end) {
+        if (gn + V <= N) {
+          v = *(const VecT*)&B[gm * N + gn];
+        } else {
+#pragma unroll
+          for (int j = 0; j < V; ++j)
+            if (gn + j < N) v.e[j] = B[gm * N + gn + j];
+        }
+      }
+#pragma unroll
+      for (int j = 0; j < V; ++j) Bs[(nn + j) * P + mm] = v.e[j];
+    }
+    __syncthreads();
+    mfma_compute_tile(As, Bs, wc, acc);
+    __syncthreads();
+  }
+
+  const bool single = gridDim.z == 1;
+  epilogue_visit(wc, acc, r0, n0, [&](int row, int col, float v) {
+    if (row < K && col < N) {
+      if (single)
+        C[(int64_t)row * N + col] = v;
+      else
+        atomicAdd(&C[(int64_t)row * N + col], v);
+    }
+  });
+}
+
+// ---------------------------------------------------------------------------
+// launchers
+// ---------------------------------------------------------------------------
+
+void gemm_launch(DT dt, const void* a, const void* b, const void* bias,
+                 void* c, int M, int N, int K, bool trans_b, int act_kind,
+                 hipStream_t s) {
+  dim3 grid(ceil_div(M, BM), ceil_div(N, BN));
+  dim3 blk(THREADS);
+  if (dt == DT::F32) {
+    auto kern = trans_b ? k_gemm<float, true> : k_gemm<float, false>;
+    hipLaunchKernelGGL(kern, grid, blk, 0, s, (const float*)a, (const float*)b,
+                       (const float*)bias, (const float*)nullptr, (float*)c, M,
+                       N, K, act_kind);
+  } else {
+    auto kern = trans_b ? k_gemm<bf16, true> : k_gemm<bf16, false>;
+    hipLaunchKernelGGL(kern, grid, blk, 0, s, (const bf16*)a, (const bf16*)b,
+                       (const float*)nullptr, (const bf16*)bias, (bf16*)c, M, N,
+                       K, act_kind);
+  }
+}
+
+void gemm_tn_launch(DT dt, const void* a, const void* b, float* c_f32, int M,
+                    int N, int K, hipStream_t s) {
+  // pick the M split so total blocks ~ 2x CUs minimum
+  int base = ceil_div(K, BM) * ceil_div(N, BN);
+  int want = 512;
+  int z = base >= want ? 1 : std::min(ceil_div(M, BK), ceil_div(want, base));
+  z = std::max(z, 1);
+  dim3 grid(ceil_div(K, BM), ceil_div(N, BN), z);
+  if (dt == DT::F32)
+    hipLaunchKernelGGL(k_gemm_tn<float>, grid, dim3(THREADS), 0, s,
+                       (const float*)a, (const float*)b, c_f32, M, N, K);
+  else
+    hipLaunchKernelGGL(k_gemm_tn<bf16>, grid, dim3(THREADS), 0, s,
+                       (const bf16*)a, (const bf16*)b, c_f32, M, N, K);
+}
+
+// ---------------------------------------------------------------------------
+// MFMA layout self-test: single-wave D = A@B for the exact fragment maps
+// the tile core assumes (run once on real hardware; asymmetric inputs).
+// ---------------------------------------------------------------------------
+
+__global__ void k_mfma_selftest(const bf16* A, const bf16* B, float* D) {
+  // A [16][32] row-major, B [32][16] row-major, D [16][16]
+  int l = threadIdx.x;
+  bf16x8 a, b;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    a[j] = (__bf16)A[(l & 15) * 32 + (l >> 4) * 8 + j];
+    b[j] = (__bf16)B[((l >> 4) * 8 + j) * 16 + (l & 15)];
+  }
+  f32x4 acc = {};
+  acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+#pragma unroll
+  for (int j = 0; j < 4; ++j) D[((l >> 4) * 4 + j) * 16 + (l & 15)] = acc[j];
+}
+
+__global__ void k_mfma_selftest_f32(const float* A, const float* B, float* D) {
+  // A [16][4], B [4][16], D [16][16]
+  int l = threadIdx.x;
+  float a = A[(l & 15) * 4 + (l >> 4)];
+  float b = B[(l >> 4) * 16 + (l & 15)];
+  f32x4 acc = {};
+  acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
+#pragma unroll
+  for (int j = 0; j < 4; ++j) D[((l >> 4) * 4 + j) * 16 + (l & 15)] = acc[j];
+}
+
+void mfma_selftest_launch(const void* a, const void* b, float* d,
+                          hipStream_t s) {
+  hipLaunchKernelGGL(k_mfma_selftest, dim3(1), dim3(64), 0, s, (const bf16*)a,
+                     (const bf16*)b, d);
+}
+
+void mfma_selftest_f32_launch(const float* a, const float* b, float* d,
+                              hipStream_t s) {
+  hipLaunchKernelGGL(k_mfma_selftest_f32, dim3(1), dim3(64), 0, s, a, b, d);
+}
+
+}  // namespace tnn

@@ -1,0 +1,121 @@
+// Launcher declarations shared between .hip translation units and the
+// torch bindings. All functions are synchronous-on-stream (enqueue only).
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <algorithm>
+#include <cstdint>
+
+namespace tnn {
+
+enum class DT { F32, BF16 };
+
+// ---- elementwise.hip -------------------------------------------------------
+void act_fwd_launch(DT dt, const void* x, void* y, int64_t n, int kind,
+                    hipStream_t s);
+void act_bwd_launch(DT dt, const void* dy, const void* x, const void* y,
+                    void* dx, int64_t n, int kind, hipStream_t s);
+void relu_bwd_mask_launch(DT dt, const void* dy, const void* y, void* dx,
+                          int64_t n, hipStream_t s);
+void dropout_fwd_launch(DT dt, const void* x, void* y, uint8_t* mask,
+                        int64_t n, float p, uint64_t seed, hipStream_t s);
+void dropout_bwd_launch(DT dt, const void* dy, const uint8_t* mask, void* dx,
+                        int64_t n, float p, hipStream_t s);
+void colsum_launch(DT dt, const void* x, void* out_f32, int64_t rows,
+                   int64_t cols, hipStream_t s);
+void cast_f32_launch(DT dt_out, const float* x, void* y, int64_t n,
+                     hipStream_t s);
+
+// ---- gemm.hip --------------------------------------------------------------
+void gemm_launch(DT dt, const void* a, const void* b, const void* bias,
+                 void* c, int M, int N, int K, bool trans_b, int act_kind,
+                 hipStream_t s);
+void gemm_tn_launch(DT dt, const void* a, const void* b, float* c_f32,
+                    int M, int N, int K, hipStream_t s);
+void mfma_selftest_launch(const void* a_bf16, const void* b_bf16, float* d,
+                          hipStream_t s);
+void mfma_selftest_f32_launch(const float* a, const float* b, float* d,
+                              hipStream_t s);
+
+// ---- conv2d.hip ------------------------------------------------------------
+struct ConvShape {
+  int N, H, W, Cin, Cout, KH, KW, SH, SW, PH, PW, OH, OW;
+};
+void conv2d_fwd_launch(DT dt, const void* x, const void* w, const void* bias,
+                       void* y, const ConvShape& cs, bool relu, hipStream_t s);
+void conv2d_dgrad_launch(DT dt, const void* dy, const void* w_t, void* dx,
+                         const ConvShape& cs, hipStream_t s);
+void conv2d_wgrad_launch(DT dt, const void* x, const void* dy, float* dw_f32,
+                         const ConvShape& cs, hipStream_t s);
+void transpose_w_launch(DT dt, const void* w, void* w_t, int KH, int KW,
+                        int Cin, int Cout, hipStream_t s);
+
+// ---- batchnorm.hip ---------------------------------------------------------
+void bn_stats_launch(DT dt, const void* x, float* mean, float* invstd,
+                     int64_t rows, int cols, float eps, hipStream_t s);
+void bn_apply_launch(DT dt, const void* x, const float* mean,
+                     const float* invstd, const float* gamma, const float* beta,
+                     void* y, int64_t rows, int cols, bool relu, hipStream_t s);
+void bn_infer_launch(DT dt, const void* x, const float* rmean,
+                     const float* rvar, const float* gamma, const float* beta,
+                     void* y, int64_t rows, int cols, float eps, bool relu,
+                     hipStream_t s);
+void bn_bwd_reduce_launch(DT dt, const void* x, const void* dy, const void* y_relu,
+                          const float* mean, const float* invstd,
+                          float* sum_dy, float* sum_dy_xhat,
+                          int64_t rows, int cols, hipStream_t s);
+void bn_bwd_apply_launch(DT dt, const void* x, const void* dy, const void* y_relu,
+                         const float* mean, const float* invstd,
+                         const float* gamma, const float* sum_dy,
+                         const float* sum_dy_xhat, void* dx,
+                         int64_t rows, int cols, hipStream_t s);
+
+// ---- pool.hip --------------------------------------------------------------
+struct PoolShape {
+  int N, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW;
+};
+void maxpool_fwd_launch(DT dt, const void* x, void* y, int32_t* idx,
+                        const PoolShape& ps, hipStream_t s);
+void maxpool_bwd_launch(DT dt, const void* dy, const int32_t* idx, float* dx_f32,
+                        const PoolShape& ps, hipStream_t s);
+void avgpool_fwd_launch(DT dt, const void* x, void* y, const PoolShape& ps,
+                        hipStream_t s);
+void avgpool_bwd_launch(DT dt, const void* dy, void* dx, const PoolShape& ps,
+                        hipStream_t s);
+
+// ---- loss.hip --------------------------------------------------------------
+void ce_fwd_launch(DT dt, const void* logits, const int64_t* targets,
+                   float* loss, float* lse, int64_t rows, int cols,
+                   hipStream_t s);
+void ce_bwd_launch(DT dt, const void* logits, const int64_t* targets,
+                   const float* lse, const float* dloss, void* dlogits,
+                   int64_t rows, int cols, hipStream_t s);
+
+// ---- layernorm.hip ---------------------------------------------------------
+void ln_fwd_launch(DT dt, const void* x, const float* gamma, const float* beta,
+                   void* y, float* mean, float* invstd, int64_t rows, int cols,
+                   float eps, hipStream_t s);
+void ln_bwd_launch(DT dt, const void* x, const void* dy, const float* gamma,
+                   const float* mean, const float* invstd, void* dx,
+                   float* dgamma, float* dbeta, int64_t rows, int cols,
+                   hipStream_t s);
+
+// ---- embedding.hip ---------------------------------------------------------
+void embedding_fwd_launch(DT dt, const int64_t* ids, const void* table,
+                          void* y, int64_t n_ids, int dim, hipStream_t s);
+void embedding_bwd_launch(DT dt, const int64_t* ids, const void* dy,
+                          float* dtable_f32, int64_t n_ids, int dim,
+                          hipStream_t s);
+
+// ---- optim.hip -------------------------------------------------------------
+void sgd_step_launch(DT dt_p, const void* grad, DT dt_g, void* param,
+                     float* master, float* momentum_buf, int64_t n, float lr,
+                     float momentum, float weight_decay, bool nesterov,
+                     bool has_master, hipStream_t s);
+void adam_step_launch(DT dt_p, const void* grad, DT dt_g, void* param,
+                      float* master, float* m, float* v, int64_t n, int step,
+                      float lr, float beta1, float beta2, float eps,
+                      float weight_decay, bool adamw, bool has_master,
+                      hipStream_t s);
+
+}  // namespace tnn

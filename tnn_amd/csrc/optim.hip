@@ -1,0 +1,91 @@
+// Fused optimizer update kernels (reference sgd_kernels.cu:17,26,
+// adam_kernels.cu:19,56). One pass: read grad (any dtype), update fp32
+// master + moments, write the low-precision parameter — the mixed-precision
+// upgrade over the reference's fp32-only updates.
+
+#include "common.h"
+#include "kernels.h"
+
+namespace tnn {
+
+template <typename TP, typename TG>
+__global__ void k_sgd(TP* __restrict__ p, float* __restrict__ master,
+                      const TG* __restrict__ g, float* __restrict__ buf,
+                      int64_t n, float lr, float momentum, float wd,
+                      bool nesterov, bool has_master) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    float w = has_master ? master[i] : VecIO<TP>::to_f32(p[i]);
+    float gr = VecIO<TG>::to_f32(g[i]) + wd * w;
+    if (buf) {
+      float b = buf[i] * momentum + gr;
+      buf[i] = b;
+      gr = nesterov ? gr + momentum * b : b;
+    }
+    w -= lr * gr;
+    if (has_master) master[i] = w;
+    p[i] = VecIO<TP>::from_f32(w);
+  }
+}
+
+template <typename TP, typename TG>
+__global__ void k_adam(TP* __restrict__ p, float* __restrict__ master,
+                       const TG* __restrict__ g, float* __restrict__ m,
+                       float* __restrict__ v, int64_t n, float lr, float beta1,
+                       float beta2, float eps, float wd, float bc1, float bc2,
+                       bool adamw, bool has_master) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    float w = has_master ? master[i] : VecIO<TP>::to_f32(p[i]);
+    float gr = VecIO<TG>::to_f32(g[i]);
+    if (!adamw) gr += wd * w;
+    float mi = m[i] = beta1 * m[i] + (1.0f - beta1) * gr;
+    float vi = v[i] = beta2 * v[i] + (1.0f - beta2) * gr * gr;
+    if (adamw) w *= (1.0f - lr * wd);
+    w -= lr / bc1 * mi / (sqrtf(vi / bc2) + eps);
+    if (has_master) master[i] = w;
+    p[i] = VecIO<TP>::from_f32(w);
+  }
+}
+
+static inline int ob(int64_t n) {
+  int64_t b = (n + 255) / 256;
+  return (int)(b < 2048 ? b : 2048);
+}
+
+void sgd_step_launch(DT dt_p, const void* grad, DT dt_g, void* param,
+                     float* master, float* momentum_buf, int64_t n, float lr,
+                     float momentum, float weight_decay, bool nesterov,
+                     bool has_master, hipStream_t s) {
+#define CASE(TP, TG)                                                          \
+  hipLaunchKernelGGL((k_sgd<TP, TG>), dim3(ob(n)), dim3(256), 0, s,           \
+                     (TP*)param, master, (const TG*)grad, momentum_buf, n,    \
+                     lr, momentum, weight_decay, nesterov, has_master)
+  if (dt_p == DT::F32 && dt_g == DT::F32) CASE(float, float);
+  else if (dt_p == DT::BF16 && dt_g == DT::BF16) CASE(bf16, bf16);
+  else if (dt_p == DT::BF16 && dt_g == DT::F32) CASE(bf16, float);
+  else CASE(float, bf16);
+#undef CASE
+}
+
+void adam_step_launch(DT dt_p, const void* grad, DT dt_g, void* param,
+                      float* master, float* m, float* v, int64_t n, int step,
+                      float lr, float beta1, float beta2, float eps,
+                      float weight_decay, bool adamw, bool has_master,
+                      hipStream_t s) {
+  float bc1 = 1.0f - powf(beta1, (float)step);
+  float bc2 = 1.0f - powf(beta2, (float)step);
+#define CASE(TP, TG)                                                          \
+  hipLaunchKernelGGL((k_adam<TP, TG>), dim3(ob(n)), dim3(256), 0, s,          \
+                     (TP*)param, master, (const TG*)grad, m, v, n, lr, beta1, \
+                     beta2, eps, weight_decay, bc1, bc2, adamw, has_master)
+  if (dt_p == DT::F32 && dt_g == DT::F32) CASE(float, float);
+  else if (dt_p == DT::BF16 && dt_g == DT::BF16) CASE(bf16, bf16);
+  else if (dt_p == DT::BF16 && dt_g == DT::F32) CASE(bf16, float);
+  else CASE(float, bf16);
+#undef CASE
+}
+
+}  // namespace tnn

@@ -1,0 +1,122 @@
+// MFMA block-tile GEMM core shared by the dense GEMM and the implicit-GEMM
+// convolution kernels (replaces reference cublasGemmEx call sites,
+// src/math/cuda/gemm.cu:64, and the cuDNN-frontend conv graphs).
+//
+// Structure (cdna_hip_programming.md §5 canonical GEMM):
+//   - 256 threads = 4 waves in a 2x2 grid; block tile BM=128 x BN=64, BK=32.
+//   - A staged in LDS as [BM][BKP] row-major (+16B row pad against bank
+//     conflicts); B staged TRANSPOSED as [BN][BKP] so both operands read
+//     contiguous k-vectors per lane.
+//   - bf16: one v_mfma_f32_16x16x32_bf16 per (fm, fn) per tile;
+//     fp32: v_mfma_f32_16x16x4_f32 (exact f32 at the f32 vector rate —
+//     there is no xf32 on gfx950), 8 k-steps per tile.
+//   - Each wave owns a 64x32 sub-tile: 4x2 fragments, f32x4 accumulators.
+//
+// Kernels using this header implement their own global->LDS staging
+// (dense/NT/implicit-gemm gather) and epilogue (bias+act store / atomic
+// accumulate), then call mfma_compute_tile() between barriers.
+#pragma once
+
+#include "common.h"
+
+namespace tile {
+
+constexpr int BM = 128, BN = 64, BK = 32, THREADS = 256;
+constexpr int WAVES_M = 2, WAVES_N = 2;      // wave grid
+constexpr int WM = BM / WAVES_M;             // 64 rows per wave
+constexpr int WN = BN / WAVES_N;             // 32 cols per wave
+constexpr int FM = WM / 16;                  // 4 row fragments
+constexpr int FN = WN / 16;                  // 2 col fragments
+
+// 16-byte staging pack (dwordx4 load/store); ext_vector_type cannot hold
+// the __hip_bfloat16 struct, an aligned POD array can.
+template <typename T> struct alignas(16) Pack16 {
+  T e[16 / sizeof(T)];
+};
+
+template <typename T> struct Pad;            // 16B row padding in elements
+template <> struct Pad<float> { static constexpr int E = 4; };
+template <> struct Pad<bf16> { static constexpr int E = 8; };
+
+template <typename T> constexpr int BKP() { return BK + Pad<T>::E; }
+
+template <typename T> struct LDSBytes {
+  static constexpr int A = BM * BKP<T>() * sizeof(T);
+  static constexpr int B = BN * BKP<T>() * sizeof(T);
+  static constexpr int total = A + B;
+};
+
+struct WaveCoord {
+  int wid, lane, wrow0, wcol0;
+  DEV WaveCoord() {
+    wid = threadIdx.x >> 6;
+    lane = threadIdx.x & 63;
+    wrow0 = (wid / WAVES_N) * WM;
+    wcol0 = (wid % WAVES_N) * WN;
+  }
+};
+
+// ---- MFMA tile compute: acc[FM][FN] += A_tile * B_tile^T-stored -----------
+DEV void mfma_compute_tile(const bf16* As, const bf16* Bs, const WaveCoord& w,
+                           f32x4 acc[FM][FN]) {
+  constexpr int P = BK + Pad<bf16>::E;
+  const int r = w.lane & 15;          // fragment row/col within 16
+  const int kb = (w.lane >> 4) * 8;   // 8 bf16 per lane
+  bf16x8 a[FM], b[FN];
+#pragma unroll
+  for (int fm = 0; fm < FM; ++fm)
+    a[fm] = *(const bf16x8*)&As[(w.wrow0 + fm * 16 + r) * P + kb];
+#pragma unroll
+  for (int fn = 0; fn < FN; ++fn)
+    b[fn] = *(const bf16x8*)&Bs[(w.wcol0 + fn * 16 + r) * P + kb];
+#pragma unroll
+  for (int fm = 0; fm < FM; ++fm)
+#pragma unroll
+    for (int fn = 0; fn < FN; ++fn)
+      acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          a[fm], b[fn], acc[fm][fn], 0, 0, 0);
+}
+
+DEV void mfma_compute_tile(const float* As, const float* Bs, const WaveCoord& w,
+                           f32x4 acc[FM][FN]) {
+  constexpr int P = BK + Pad<float>::E;
+  const int r = w.lane & 15;
+  const int kq = w.lane >> 4;         // one f32 per lane per k-step of 4
+#pragma unroll
+  for (int ks = 0; ks < BK / 4; ++ks) {
+    float a[FM], b[FN];
+#pragma unroll
+    for (int fm = 0; fm < FM; ++fm)
+      a[fm] = As[(w.wrow0 + fm * 16 + r) * P + ks * 4 + kq];
+#pragma unroll
+    for (int fn = 0; fn < FN; ++fn)
+      b[fn] = Bs[(w.wcol0 + fn * 16 + r) * P + ks * 4 + kq];
+#pragma unroll
+    for (int fm = 0; fm < FM; ++fm)
+#pragma unroll
+      for (int fn = 0; fn < FN; ++fn)
+        acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+            a[fm], b[fn], acc[fm][fn], 0, 0, 0);
+  }
+}
+
+// ---- epilogue: C/D fragment map col=lane&15, row=(lane>>4)*4+j ------------
+// Visits every accumulator element with its global (row, col).
+template <typename F>
+DEV void epilogue_visit(const WaveCoord& w, f32x4 acc[FM][FN],
+                        int row0, int col0, F&& emit) {
+  const int cr = (w.lane >> 4) * 4;
+  const int cc = w.lane & 15;
+#pragma unroll
+  for (int fm = 0; fm < FM; ++fm)
+#pragma unroll
+    for (int fn = 0; fn < FN; ++fn)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int row = row0 + w.wrow0 + fm * 16 + cr + j;
+        int col = col0 + w.wcol0 + fn * 16 + cc;
+        emit(row, col, acc[fm][fn][j]);
+      }
+}
+
+}  // namespace tile

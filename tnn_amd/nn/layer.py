@@ -56,6 +56,20 @@ def layer_from_config(cfg: Dict[str, Any]) -> "Layer":
     return cls.from_config(cfg)
 
 
+def cast_compute_dtype(model: nn.Module, dtype: torch.dtype) -> nn.Module:
+    """Cast compute params (conv/dense weights...) to ``dtype`` while norm
+    affine params (named gamma/beta) and buffers stay fp32 — the mixed
+    precision policy (reference keeps BN params fp32,
+    batchnorm_layer.cpp:140-148)."""
+    for mod in model.modules():
+        for name, p in mod.named_parameters(recurse=False):
+            if p.dtype == torch.float32 and name not in ("gamma", "beta"):
+                p.data = p.data.to(dtype)
+        if hasattr(mod, "io_dtype"):
+            mod.io_dtype = dtype
+    return model
+
+
 class Layer(nn.Module):
     """Base layer: named, dtype-aware, config round-trippable."""
 
