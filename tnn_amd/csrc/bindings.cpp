@@ -93,6 +93,8 @@ at::Tensor gemm(const at::Tensor& a, const at::Tensor& b,
   CHECK_IN(b);
   TORCH_CHECK(a.dim() == 2 && b.dim() == 2 && a.size(1) == b.size(0),
               "gemm shapes ", a.sizes(), " @ ", b.sizes());
+  TORCH_CHECK(a.scalar_type() == b.scalar_type(), "gemm dtype mismatch ",
+              a.scalar_type(), " vs ", b.scalar_type());
   int M = a.size(0), K = a.size(1), N = b.size(1);
   auto c = at::empty({M, N}, a.options());
   const void* bp = bias.has_value() ? bias->data_ptr() : nullptr;
@@ -172,6 +174,10 @@ at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
   CHECK_IN(w);
   TORCH_CHECK(x.dim() == 4 && w.dim() == 4 && x.size(3) == w.size(2),
               "conv2d_fwd x ", x.sizes(), " w ", w.sizes());
+  TORCH_CHECK(x.scalar_type() == w.scalar_type(), "conv2d dtype mismatch ",
+              x.scalar_type(), " vs ", w.scalar_type());
+  TORCH_CHECK(!bias.has_value() || bias->scalar_type() == x.scalar_type(),
+              "conv2d bias dtype mismatch");
   auto cs = conv_shape(x, w.size(2), w.size(3), w.size(0), w.size(1), sh, sw,
                        ph, pw);
   auto y = at::empty({cs.N, cs.OH, cs.OW, cs.Cout}, x.options());

@@ -33,8 +33,8 @@ __global__ void k_conv_fwd(const T* __restrict__ X, const T* __restrict__ Wt,
                            ConvShape cs, int act_kind) {
   constexpr int P = BKP<T>();
   constexpr int V = 16 / sizeof(T);
-  __shared__ T As[BM * P];
-  __shared__ T Bs[BN * P];
+  __shared__ alignas(16) T As[BM * P];
+  __shared__ alignas(16) T Bs[BN * P];
 
   const int M = cs.N * cs.OH * cs.OW;
   const int K = cs.KH * cs.KW * cs.Cin;
@@ -59,7 +59,8 @@ __global__ void k_conv_fwd(const T* __restrict__ X, const T* __restrict__ Wt,
         int ci = gk % cs.Cin;
         int kidx = gk / cs.Cin;
         int kw = kidx % cs.KW, kh = kidx / cs.KW;
-        if (ci + V <= cs.Cin && gk + V <= K) {
+        if (ci + V <= cs.Cin && gk + V <= K && (cs.Cin % V) == 0 &&
+            aligned16(X)) {
           int ih = oh * cs.SH - cs.PH + kh;
           int iw = ow * cs.SW - cs.PW + kw;
           if (ih >= 0 && ih < cs.H && iw >= 0 && iw < cs.W)
@@ -91,8 +92,9 @@ __global__ void k_conv_fwd(const T* __restrict__ X, const T* __restrict__ Wt,
       int gk = k0 + kk, gn = n0 + nn;
       VecT v = {};
       if (gk < K) {
-        if (gn + V <= cs.Cout) {
-          v = *(const VecT*)&Wt[(int64_t)gk * cs.Cout + gn];
+        const T* src = &Wt[(int64_t)gk * cs.Cout + gn];
+        if (gn + V <= cs.Cout && aligned16(src)) {
+          v = *(const VecT*)src;
         } else {
 #pragma unroll
           for (int j = 0; j < V; ++j)
@@ -128,8 +130,8 @@ __global__ void k_conv_dgrad(const T* __restrict__ DY, const T* __restrict__ WT,
                              T* __restrict__ DX, ConvShape cs) {
   constexpr int P = BKP<T>();
   constexpr int V = 16 / sizeof(T);
-  __shared__ T As[BM * P];
-  __shared__ T Bs[BN * P];
+  __shared__ alignas(16) T As[BM * P];
+  __shared__ alignas(16) T Bs[BN * P];
 
   const int M = cs.N * cs.H * cs.W;
   const int K = cs.KH * cs.KW * cs.Cout;
@@ -161,7 +163,8 @@ __global__ void k_conv_dgrad(const T* __restrict__ DY, const T* __restrict__ WT,
           if (oh >= cs.OH || ow >= cs.OW) return T(0.0f);
           return DY[(((int64_t)n * cs.OH + oh) * cs.OW + ow) * cs.Cout + coj];
         };
-        if (co + V <= cs.Cout && gk + V <= K) {
+        if (co + V <= cs.Cout && gk + V <= K && (cs.Cout % V) == 0 &&
+            aligned16(DY)) {
           int th = ih + cs.PH - kh, tw = iw + cs.PW - kw;
           if (th >= 0 && tw >= 0 && th % cs.SH == 0 && tw % cs.SW == 0) {
             int oh = th / cs.SH, ow = tw / cs.SW;
@@ -191,8 +194,9 @@ __global__ void k_conv_dgrad(const T* __restrict__ DY, const T* __restrict__ WT,
       int gk = k0 + kk, gn = n0 + nn;
       VecT v = {};
       if (gk < K) {
-        if (gn + V <= cs.Cin) {
-          v = *(const VecT*)&WT[(int64_t)gk * cs.Cin + gn];
+        const T* src = &WT[(int64_t)gk * cs.Cin + gn];
+        if (gn + V <= cs.Cin && aligned16(src)) {
+          v = *(const VecT*)src;
         } else {
 #pragma unroll
           for (int j = 0; j < V; ++j)
@@ -223,8 +227,8 @@ __global__ void k_conv_wgrad(const T* __restrict__ X, const T* __restrict__ DY,
                              float* __restrict__ DW, ConvShape cs) {
   constexpr int P = BKP<T>();
   constexpr int V = 16 / sizeof(T);
-  __shared__ T As[BM * P];  // rows = (kh,kw,ci), k = m
-  __shared__ T Bs[BN * P];  // rows = co, k = m
+  __shared__ alignas(16) T As[BM * P];  // rows = (kh,kw,ci), k = m
+  __shared__ alignas(16) T Bs[BN * P];  // rows = co, k = m
 
   const int M = cs.N * cs.OH * cs.OW;         // reduction dim
   const int Kout = cs.KH * cs.KW * cs.Cin;    // output rows
@@ -252,7 +256,8 @@ __global__ void k_conv_wgrad(const T* __restrict__ X, const T* __restrict__ DY,
         int ci = gr % cs.Cin;
         int kidx = gr / cs.Cin;
         int kw = kidx % cs.KW, kh = kidx / cs.KW;
-        if (ci + V <= cs.Cin && gr + V <= Kout) {
+        if (ci + V <= cs.Cin && gr + V <= Kout && (cs.Cin % V) == 0 &&
+            aligned16(X)) {
           int ih = oh * cs.SH - cs.PH + kh;
           int iw = ow * cs.SW - cs.PW + kw;
           if (ih >= 0 && ih < cs.H && iw >= 0 && iw < cs.W)
@@ -286,8 +291,9 @@ __global__ void k_conv_wgrad(const T* __restrict__ X, const T* __restrict__ DY,
       int gn = n0 + nn;
       VecT v = {};
       if (gm < m_end) {
-        if (gn + V <= cs.Cout) {
-          v = *(const VecT*)&DY[gm * cs.Cout + gn];
+        const T* src = &DY[gm * cs.Cout + gn];
+        if (gn + V <= cs.Cout && aligned16(src)) {
+          v = *(const VecT*)src;
         } else {
 #pragma unroll
           for (int j = 0; j < V; ++j)

@@ -26,8 +26,8 @@ __global__ void k_gemm(const T* __restrict__ A, const T* __restrict__ B,
                        int M, int N, int K, int act_kind) {
   constexpr int P = BKP<T>();
   constexpr int V = 16 / sizeof(T);  // elems per 16B vector
-  __shared__ T As[BM * P];
-  __shared__ T Bs[BN * P];
+  __shared__ alignas(16) T As[BM * P];
+  __shared__ alignas(16) T Bs[BN * P];
 
   const int m0 = blockIdx.x * BM;
   const int n0 = blockIdx.y * BN;
@@ -45,8 +45,9 @@ __global__ void k_gemm(const T* __restrict__ A, const T* __restrict__ B,
       int gm = m0 + row, gk = k0 + kk;
       VecT v = {};
       if (gm < M) {
-        if (gk + V <= K) {
-          v = *(const VecT*)&A[(int64_t)gm * K + gk];
+        const T* src = &A[(int64_t)gm * K + gk];
+        if (gk + V <= K && aligned16(src)) {
+          v = *(const VecT*)src;
         } else {
 #pragma unroll
           for (int j = 0; j < V; ++j)
@@ -65,8 +66,9 @@ __global__ void k_gemm(const T* __restrict__ A, const T* __restrict__ B,
         int gn = n0 + col, gk = k0 + kk;
         VecT v = {};
         if (gn < N) {
-          if (gk + V <= K) {
-            v = *(const VecT*)&B[(int64_t)gn * K + gk];
+          const T* src = &B[(int64_t)gn * K + gk];
+          if (gk + V <= K && aligned16(src)) {
+            v = *(const VecT*)src;
           } else {
 #pragma unroll
             for (int j = 0; j < V; ++j)
@@ -84,8 +86,9 @@ __global__ void k_gemm(const T* __restrict__ A, const T* __restrict__ B,
         int gk = k0 + kk, gn = n0 + nn;
         VecT v = {};
         if (gk < K) {
-          if (gn + V <= N) {
-            v = *(const VecT*)&B[(int64_t)gk * N + gn];
+          const T* src = &B[(int64_t)gk * N + gn];
+          if (gn + V <= N && aligned16(src)) {
+            v = *(const VecT*)src;
           } else {
 #pragma unroll
             for (int j = 0; j < V; ++j)
@@ -124,8 +127,8 @@ __global__ void k_gemm_tn(const T* __restrict__ A, const T* __restrict__ B,
                           float* __restrict__ C, int M, int N, int K) {
   constexpr int P = BKP<T>();
   constexpr int V = 16 / sizeof(T);
-  __shared__ T As[BM * P];  // rows = K-dim, k = m-chunk
-  __shared__ T Bs[BN * P];
+  __shared__ alignas(16) T As[BM * P];  // rows = K-dim, k = m-chunk
+  __shared__ alignas(16) T Bs[BN * P];
 
   const int r0 = blockIdx.x * BM;  // output row = A column
   const int n0 = blockIdx.y * BN;
@@ -146,8 +149,9 @@ __global__ void k_gemm_tn(const T* __restrict__ A, const T* __restrict__ B,
       int gr = r0 + rr;
       VecT v = {};
       if (gm < m_end) {
-        if (gr + V <= K) {
-          v = *(const VecT*)&A[gm * K + gr];
+        const T* src = &A[gm * K + gr];
+        if (gr + V <= K && aligned16(src)) {
+          v = *(const VecT*)src;
         } else {
 #pragma unroll
           for (int j = 0; j < V; ++j)
@@ -166,8 +170,9 @@ __global__ void k_gemm_tn(const T* __restrict__ A, const T* __restrict__ B,
       int gn = n0 + nn;
       VecT v = {};
       if (gm < m_end) {
-        if (gn + V <= N) {
-          v = *(const VecT*)&B[gm * N + gn];
+        const T* src = &B[gm * N + gn];
+        if (gn + V <= N && aligned16(src)) {
+          v = *(const VecT*)src;
         } else {
 #pragma unroll
           for (int j = 0; j < V; ++j)

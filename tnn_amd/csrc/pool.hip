@@ -102,12 +102,14 @@ __global__ void k_avgpool_bwd(const T* __restrict__ dy, T* __restrict__ dx,
     int ih = rest % ps.H;
     int n = rest / ps.H;
     float acc = 0.0f;
-    int oh_lo = (ih + ps.PH - ps.KH) / ps.SH + 1;
-    if (oh_lo < 0) oh_lo = 0;
+    // smallest oh with window covering ih: ceil((ih+PH-KH+1)/SH), clamped;
+    // C division truncates toward zero so guard the negative numerator
+    int a1 = ih + ps.PH - ps.KH + 1;
+    int oh_lo = a1 > 0 ? (a1 + ps.SH - 1) / ps.SH : 0;
     int oh_hi = (ih + ps.PH) / ps.SH;
     if (oh_hi >= ps.OH) oh_hi = ps.OH - 1;
-    int ow_lo = (iw + ps.PW - ps.KW) / ps.SW + 1;
-    if (ow_lo < 0) ow_lo = 0;
+    int b1 = iw + ps.PW - ps.KW + 1;
+    int ow_lo = b1 > 0 ? (b1 + ps.SW - 1) / ps.SW : 0;
     int ow_hi = (iw + ps.PW) / ps.SW;
     if (ow_hi >= ps.OW) ow_hi = ps.OW - 1;
     for (int oh = oh_lo; oh <= oh_hi; ++oh) {

@@ -34,6 +34,11 @@ template <typename T> struct alignas(16) Pack16 {
   T e[16 / sizeof(T)];
 };
 
+template <typename T>
+DEV bool aligned16(const T* p) {
+  return ((unsigned long long)(const void*)p & 15ull) == 0;
+}
+
 template <typename T> struct Pad;            // 16B row padding in elements
 template <> struct Pad<float> { static constexpr int E = 4; };
 template <> struct Pad<bf16> { static constexpr int E = 8; };

@@ -105,12 +105,10 @@ class PipelineEngine:
 
     # ------------------------------------------------------------------
     def _cast_stage(self, dtype):
-        """Cast compute params to bf16; BN/LN affine + buffers stay fp32
-        (they are created fp32 and are never touched here)."""
-        for layer in self.stage.modules():
-            for name, p in layer.named_parameters(recurse=False):
-                if p.dtype == torch.float32 and name not in ("gamma", "beta"):
-                    p.data = p.data.to(dtype)
+        """Cast compute params + layer io dtypes to bf16; BN/LN affine and
+        running stats stay fp32."""
+        from ..nn.layer import cast_compute_dtype
+        cast_compute_dtype(self.stage, dtype)
 
     def _sync_weights(self):
         """Ship rank-0's partitioned weights to every stage owner so all
