@@ -152,10 +152,11 @@ static inline dim3 bn_reduce_grid(int64_t rows, int cols) {
   int cpb = cols < 256 ? cols : 256;
   int cblocks = (cols + cpb - 1) / cpb;
   int rows_per_iter = 256 / cpb;
-  // aim for >=512 blocks to fill 256 CUs
+  // fill the chip: >=2048 blocks where the row count allows, >=16
+  // iterations per block so atomics stay a rounding error
   int64_t iters = (rows + rows_per_iter - 1) / rows_per_iter;
-  int rslices = (int)std::min<int64_t>(std::max<int64_t>(512 / cblocks, 1),
-                                  std::max<int64_t>(iters / 16, 1));
+  int rslices = (int)std::min<int64_t>(std::max<int64_t>(2048 / cblocks, 1),
+                                       std::max<int64_t>(iters / 16, 1));
   return dim3(cblocks, rslices);
 }
 

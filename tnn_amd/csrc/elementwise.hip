@@ -148,16 +148,17 @@ __global__ void k_colsum(const T* __restrict__ x, float* __restrict__ out,
   const int col = blockIdx.x * 64 + (threadIdx.x & 63);
   const int rslice = blockIdx.y;
   const int nrs = gridDim.y;
-  if (col >= cols) return;
   const int64_t r0 = rows * rslice / nrs, r1 = rows * (rslice + 1) / nrs;
   float acc = 0.0f;
-  // 4 waves stride over rows
-  for (int64_t r = r0 + (threadIdx.x >> 6); r < r1; r += 4)
-    acc += VecIO<T>::to_f32(x[r * cols + col]);
+  if (col < cols) {
+    // 4 waves stride over the block's row slice; lanes = consecutive cols
+    for (int64_t r = r0 + (threadIdx.x >> 6); r < r1; r += 4)
+      acc += VecIO<T>::to_f32(x[r * cols + col]);
+  }
   __shared__ float sh[4][64];
   sh[threadIdx.x >> 6][threadIdx.x & 63] = acc;
   __syncthreads();
-  if (threadIdx.x < 64) {
+  if (threadIdx.x < 64 && col < cols) {
     float v = sh[0][threadIdx.x] + sh[1][threadIdx.x] + sh[2][threadIdx.x] +
               sh[3][threadIdx.x];
     if (nrs == 1)
@@ -169,8 +170,11 @@ __global__ void k_colsum(const T* __restrict__ x, float* __restrict__ out,
 
 void colsum_launch(DT dt, const void* x, void* out_f32, int64_t rows,
                    int64_t cols, hipStream_t s) {
-  int rslices = rows > 16384 ? 8 : 1;
-  dim3 grid((cols + 63) / 64, rslices);
+  // fill the chip: one block per ~64-row slice, capped so atomics stay cheap
+  int cblocks = (int)((cols + 63) / 64);
+  int cap = std::max(1, 2048 / cblocks);
+  int rslices = (int)std::min<int64_t>((rows + 63) / 64, (int64_t)cap);
+  dim3 grid(cblocks, rslices);
   if (dt == DT::F32)
     hipLaunchKernelGGL(k_colsum<float>, grid, dim3(256), 0, s, (const float*)x,
                        (float*)out_f32, rows, (int)cols);
