@@ -123,9 +123,16 @@ at::Tensor gemm_tn(const at::Tensor& a, const at::Tensor& b) {
   TORCH_CHECK(a.dim() == 2 && b.dim() == 2 && a.size(0) == b.size(0),
               "gemm_tn shapes ", a.sizes(), "^T @ ", b.sizes());
   int M = a.size(0), K = a.size(1), N = b.size(1);
-  auto c = at::zeros({K, N}, a.options().dtype(at::kFloat));
-  gemm_tn_launch(dt_of(a), a.data_ptr(), b.data_ptr(), c.data_ptr<float>(), M,
-                 N, K, cur_stream());
+  auto c = at::empty({K, N}, a.options().dtype(at::kFloat));
+  int z = gemm_tn_zsplits(M, N, K);
+  at::Tensor ws;
+  float* wsp = nullptr;
+  if (z > 1) {
+    ws = at::empty({(int64_t)z * K * N}, a.options().dtype(at::kFloat));
+    wsp = ws.data_ptr<float>();
+  }
+  gemm_tn_launch(dt_of(a), a.data_ptr(), b.data_ptr(), c.data_ptr<float>(),
+                 wsp, z, M, N, K, cur_stream());
   return c;
 }
 
@@ -224,10 +231,18 @@ at::Tensor conv2d_wgrad(const at::Tensor& x, const at::Tensor& dy, int64_t KH,
   CHECK_IN(dy);
   auto cs = conv_shape(x, x.size(3), dy.size(3), KH, KW, sh, sw, ph, pw);
   TORCH_CHECK(dy.size(1) == cs.OH && dy.size(2) == cs.OW, "wgrad shape");
-  auto dw = at::zeros({KH, KW, cs.Cin, cs.Cout},
+  auto dw = at::empty({KH, KW, cs.Cin, cs.Cout},
                       x.options().dtype(at::kFloat));
+  int z = conv2d_wgrad_zsplits(cs);
+  at::Tensor ws;
+  float* wsp = nullptr;
+  if (z > 1) {
+    ws = at::empty({(int64_t)z * KH * KW * cs.Cin * cs.Cout},
+                   x.options().dtype(at::kFloat));
+    wsp = ws.data_ptr<float>();
+  }
   conv2d_wgrad_launch(dt_of(x), x.data_ptr(), dy.data_ptr(),
-                      dw.data_ptr<float>(), cs, cur_stream());
+                      dw.data_ptr<float>(), wsp, z, cs, cur_stream());
   return dw;
 }
 

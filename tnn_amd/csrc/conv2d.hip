@@ -31,10 +31,9 @@ __global__ void k_conv_fwd(const T* __restrict__ X, const T* __restrict__ Wt,
                            const float* __restrict__ bias_f32,
                            const T* __restrict__ bias_t, T* __restrict__ Y,
                            ConvShape cs, int act_kind) {
-  constexpr int P = BKP<T>();
   constexpr int V = 16 / sizeof(T);
-  __shared__ alignas(16) T As[BM * P];
-  __shared__ alignas(16) T Bs[BN * P];
+  __shared__ alignas(16) T As[BM * BK];
+  __shared__ alignas(16) T Bs[BN * BK];
 
   const int M = cs.N * cs.OH * cs.OW;
   const int K = cs.KH * cs.KW * cs.Cin;
@@ -82,7 +81,7 @@ __global__ void k_conv_fwd(const T* __restrict__ X, const T* __restrict__ Wt,
           }
         }
       }
-      *(VecT*)&As[row * P + kk] = v;
+      *(VecT*)&As[lds_off<T>(row, kk)] = v;
     }
     // ---- stage B (weights [K, Cout]) -> Bs[n][k] scatter-transpose ----
 #pragma unroll
@@ -102,7 +101,8 @@ __global__ void k_conv_fwd(const T* __restrict__ X, const T* __restrict__ Wt,
         }
       }
 #pragma unroll
-      for (int j = 0; j < V; ++j) Bs[(nn + j) * P + kk] = v.e[j];
+      for (int j = 0; j < V; ++j)
+        Bs[lds_off<T>(nn + j, kk)] = v.e[j];
     }
     __syncthreads();
     mfma_compute_tile(As, Bs, wc, acc);
@@ -128,10 +128,9 @@ template <typename T>
 __launch_bounds__(THREADS)
 __global__ void k_conv_dgrad(const T* __restrict__ DY, const T* __restrict__ WT,
                              T* __restrict__ DX, ConvShape cs) {
-  constexpr int P = BKP<T>();
   constexpr int V = 16 / sizeof(T);
-  __shared__ alignas(16) T As[BM * P];
-  __shared__ alignas(16) T Bs[BN * P];
+  __shared__ alignas(16) T As[BM * BK];
+  __shared__ alignas(16) T Bs[BN * BK];
 
   const int M = cs.N * cs.H * cs.W;
   const int K = cs.KH * cs.KW * cs.Cout;
@@ -184,7 +183,7 @@ __global__ void k_conv_dgrad(const T* __restrict__ DY, const T* __restrict__ WT,
           }
         }
       }
-      *(VecT*)&As[row * P + kk] = v;
+      *(VecT*)&As[lds_off<T>(row, kk)] = v;
     }
     // ---- stage B (w_t [K, Cin]) -> Bs[ci][k] ----
 #pragma unroll
@@ -204,7 +203,8 @@ __global__ void k_conv_dgrad(const T* __restrict__ DY, const T* __restrict__ WT,
         }
       }
 #pragma unroll
-      for (int j = 0; j < V; ++j) Bs[(nn + j) * P + kk] = v.e[j];
+      for (int j = 0; j < V; ++j)
+        Bs[lds_off<T>(nn + j, kk)] = v.e[j];
     }
     __syncthreads();
     mfma_compute_tile(As, Bs, wc, acc);
@@ -225,33 +225,32 @@ template <typename T>
 __launch_bounds__(THREADS)
 __global__ void k_conv_wgrad(const T* __restrict__ X, const T* __restrict__ DY,
                              float* __restrict__ DW, ConvShape cs) {
-  constexpr int P = BKP<T>();
   constexpr int V = 16 / sizeof(T);
-  __shared__ alignas(16) T As[BM * P];  // rows = (kh,kw,ci), k = m
-  __shared__ alignas(16) T Bs[BN * P];  // rows = co, k = m
+  __shared__ alignas(16) T As[BM * BK];  // rows = (kh,kw,ci), k = m
+  __shared__ alignas(16) T Bs[BN * BK];  // rows = co, k = m
 
   const int M = cs.N * cs.OH * cs.OW;         // reduction dim
   const int Kout = cs.KH * cs.KW * cs.Cin;    // output rows
   const int r0 = blockIdx.x * BM;
   const int n0 = blockIdx.y * BN;
-  const int64_t m_begin = (int64_t)M * blockIdx.z / gridDim.z;
-  const int64_t m_end = (int64_t)M * (blockIdx.z + 1) / gridDim.z;
+  const int m_begin = (int)((int64_t)M * blockIdx.z / gridDim.z);
+  const int m_end = (int)((int64_t)M * (blockIdx.z + 1) / gridDim.z);
   const WaveCoord wc;
   f32x4 acc[FM][FN] = {};
   using VecT = Pack16<T>;
 
-  for (int64_t k0 = m_begin; k0 < m_end; k0 += BK) {
+  for (int k0 = m_begin; k0 < m_end; k0 += BK) {
     // ---- stage A^T: x rows ci-contiguous, scatter into As[ko][m] ----
 #pragma unroll
     for (int c = threadIdx.x; c < BK * (BM / V); c += THREADS) {
       int mm = c / (BM / V);
       int rr = (c % (BM / V)) * V;
-      int64_t gm = k0 + mm;
+      int gm = k0 + mm;
       int gr = r0 + rr;
       VecT v = {};
       if (gm < m_end && gr < Kout) {
-        int n = (int)(gm / (cs.OH * cs.OW));
-        int rem = (int)(gm % (cs.OH * cs.OW));
+        int n = gm / (cs.OH * cs.OW);
+        int rem = gm % (cs.OH * cs.OW);
         int oh = rem / cs.OW, ow = rem % cs.OW;
         int ci = gr % cs.Cin;
         int kidx = gr / cs.Cin;
@@ -280,42 +279,39 @@ __global__ void k_conv_wgrad(const T* __restrict__ X, const T* __restrict__ DY,
         }
       }
 #pragma unroll
-      for (int j = 0; j < V; ++j) As[(rr + j) * P + mm] = v.e[j];
+      for (int j = 0; j < V; ++j)
+        As[lds_off<T>(rr + j, mm)] = v.e[j];
     }
     // ---- stage B: dy[m][co] -> Bs[co][m] ----
 #pragma unroll
     for (int c = threadIdx.x; c < BK * (BN / V); c += THREADS) {
       int mm = c / (BN / V);
       int nn = (c % (BN / V)) * V;
-      int64_t gm = k0 + mm;
+      int gm = k0 + mm;
       int gn = n0 + nn;
       VecT v = {};
       if (gm < m_end) {
-        const T* src = &DY[gm * cs.Cout + gn];
+        const T* src = &DY[(int64_t)gm * cs.Cout + gn];
         if (gn + V <= cs.Cout && aligned16(src)) {
           v = *(const VecT*)src;
         } else {
 #pragma unroll
           for (int j = 0; j < V; ++j)
-            if (gn + j < cs.Cout) v.e[j] = DY[gm * cs.Cout + gn + j];
+            if (gn + j < cs.Cout) v.e[j] = DY[(int64_t)gm * cs.Cout + gn + j];
         }
       }
 #pragma unroll
-      for (int j = 0; j < V; ++j) Bs[(nn + j) * P + mm] = v.e[j];
+      for (int j = 0; j < V; ++j)
+        Bs[lds_off<T>(nn + j, mm)] = v.e[j];
     }
     __syncthreads();
     mfma_compute_tile(As, Bs, wc, acc);
     __syncthreads();
   }
 
-  const bool single = gridDim.z == 1;
+  float* out = DW + (int64_t)blockIdx.z * Kout * cs.Cout;
   epilogue_visit(wc, acc, r0, n0, [&](int row, int col, float v) {
-    if (row < Kout && col < cs.Cout) {
-      if (single)
-        DW[(int64_t)row * cs.Cout + col] = v;
-      else
-        atomicAdd(&DW[(int64_t)row * cs.Cout + col], v);
-    }
+    if (row < Kout && col < cs.Cout) out[(int64_t)row * cs.Cout + col] = v;
   });
 }
 
@@ -361,22 +357,30 @@ void conv2d_dgrad_launch(DT dt, const void* dy, const void* w_t, void* dx,
                        (const bf16*)dy, (const bf16*)w_t, (bf16*)dx, cs);
 }
 
-void conv2d_wgrad_launch(DT dt, const void* x, const void* dy, float* dw_f32,
-                         const ConvShape& cs, hipStream_t s) {
+int conv2d_wgrad_zsplits(const ConvShape& cs) {
   int M = cs.N * cs.OH * cs.OW;
   int Kout = cs.KH * cs.KW * cs.Cin;
   int base = ceil_div(Kout, BM) * ceil_div(cs.Cout, BN);
-  int want = 512;
+  int want = 2048;
   int z = base >= want ? 1
                        : std::min(ceil_div(M, 4 * BK), ceil_div(want, base));
-  z = std::max(z, 1);
+  return std::max(z, 1);
+}
+
+void conv2d_wgrad_launch(DT dt, const void* x, const void* dy, float* dw_f32,
+                         float* ws, int z, const ConvShape& cs,
+                         hipStream_t s) {
+  int Kout = cs.KH * cs.KW * cs.Cin;
+  float* target = z == 1 ? dw_f32 : ws;
   dim3 grid(ceil_div(Kout, BM), ceil_div(cs.Cout, BN), z);
   if (dt == DT::F32)
     hipLaunchKernelGGL(k_conv_wgrad<float>, grid, dim3(THREADS), 0, s,
-                       (const float*)x, (const float*)dy, dw_f32, cs);
+                       (const float*)x, (const float*)dy, target, cs);
   else
     hipLaunchKernelGGL(k_conv_wgrad<bf16>, grid, dim3(THREADS), 0, s,
-                       (const bf16*)x, (const bf16*)dy, dw_f32, cs);
+                       (const bf16*)x, (const bf16*)dy, target, cs);
+  if (z > 1)
+    splitk_reduce_launch(ws, dw_f32, z, (int64_t)Kout * cs.Cout, s);
 }
 
 void transpose_w_launch(DT dt, const void* w, void* w_t, int KH, int KW,

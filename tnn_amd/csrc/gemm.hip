@@ -24,10 +24,9 @@ __global__ void k_gemm(const T* __restrict__ A, const T* __restrict__ B,
                        const float* __restrict__ bias_f32,
                        const T* __restrict__ bias_t, T* __restrict__ C,
                        int M, int N, int K, int act_kind) {
-  constexpr int P = BKP<T>();
   constexpr int V = 16 / sizeof(T);  // elems per 16B vector
-  __shared__ alignas(16) T As[BM * P];
-  __shared__ alignas(16) T Bs[BN * P];
+  __shared__ alignas(16) T As[BM * BK];
+  __shared__ alignas(16) T Bs[BN * BK];
 
   const int m0 = blockIdx.x * BM;
   const int n0 = blockIdx.y * BN;
@@ -54,7 +53,7 @@ __global__ void k_gemm(const T* __restrict__ A, const T* __restrict__ B,
             if (gk + j < K) v.e[j] = A[(int64_t)gm * K + gk + j];
         }
       }
-      *(VecT*)&As[row * P + kk] = v;
+      *(VecT*)&As[lds_off<T>(row, kk)] = v;
     }
     // ---- stage B into Bs[n][k] ----
     if constexpr (TRANS_B) {
@@ -75,7 +74,7 @@ __global__ void k_gemm(const T* __restrict__ A, const T* __restrict__ B,
               if (gk + j < K) v.e[j] = B[(int64_t)gn * K + gk + j];
           }
         }
-        *(VecT*)&Bs[col * P + kk] = v;
+        *(VecT*)&Bs[lds_off<T>(col, kk)] = v;
       }
     } else {
       // B[K,N]: load n-contiguous vectors, scatter-transpose into LDS
@@ -96,7 +95,8 @@ __global__ void k_gemm(const T* __restrict__ A, const T* __restrict__ B,
           }
         }
 #pragma unroll
-        for (int j = 0; j < V; ++j) Bs[(nn + j) * P + kk] = v.e[j];
+        for (int j = 0; j < V; ++j)
+          Bs[lds_off<T>(nn + j, kk)] = v.e[j];
       }
     }
     __syncthreads();
@@ -125,77 +125,95 @@ template <typename T>
 __launch_bounds__(THREADS)
 __global__ void k_gemm_tn(const T* __restrict__ A, const T* __restrict__ B,
                           float* __restrict__ C, int M, int N, int K) {
-  constexpr int P = BKP<T>();
   constexpr int V = 16 / sizeof(T);
-  __shared__ alignas(16) T As[BM * P];  // rows = K-dim, k = m-chunk
-  __shared__ alignas(16) T Bs[BN * P];
+  __shared__ alignas(16) T As[BM * BK];  // rows = K-dim, k = m-chunk
+  __shared__ alignas(16) T Bs[BN * BK];
 
   const int r0 = blockIdx.x * BM;  // output row = A column
   const int n0 = blockIdx.y * BN;
-  const int64_t m_begin = (int64_t)M * blockIdx.z / gridDim.z;
-  const int64_t m_end = (int64_t)M * (blockIdx.z + 1) / gridDim.z;
+  // 32-bit m index: int64 div/mod in the staging loop costs ~100 VALU
+  // cycles per chunk (measured: wgrad at 1/4 the MFMA residency of fwd)
+  const int m_begin = (int)((int64_t)M * blockIdx.z / gridDim.z);
+  const int m_end = (int)((int64_t)M * (blockIdx.z + 1) / gridDim.z);
   const WaveCoord wc;
   f32x4 acc[FM][FN] = {};
 
   using VecT = Pack16<T>;
 
-  for (int64_t k0 = m_begin; k0 < m_end; k0 += BK) {
+  for (int k0 = m_begin; k0 < m_end; k0 += BK) {
     // ---- stage A^T: load A[m][r..r+V] (contiguous), scatter to As[r][m] ----
 #pragma unroll
     for (int c = threadIdx.x; c < BK * (BM / V); c += THREADS) {
       int mm = c / (BM / V);
       int rr = (c % (BM / V)) * V;
-      int64_t gm = k0 + mm;
+      int gm = k0 + mm;
       int gr = r0 + rr;
       VecT v = {};
       if (gm < m_end) {
-        const T* src = &A[gm * K + gr];
+        const T* src = &A[(int64_t)gm * K + gr];
         if (gr + V <= K && aligned16(src)) {
           v = *(const VecT*)src;
         } else {
 #pragma unroll
           for (int j = 0; j < V; ++j)
-            if (gr + j < K) v.e[j] = A[gm * K + gr + j];
+            if (gr + j < K) v.e[j] = A[(int64_t)gm * K + gr + j];
         }
       }
 #pragma unroll
-      for (int j = 0; j < V; ++j) As[(rr + j) * P + mm] = v.e[j];
+      for (int j = 0; j < V; ++j)
+        As[lds_off<T>(rr + j, mm)] = v.e[j];
     }
     // ---- stage B[m][n] -> Bs[n][m] ----
 #pragma unroll
     for (int c = threadIdx.x; c < BK * (BN / V); c += THREADS) {
       int mm = c / (BN / V);
       int nn = (c % (BN / V)) * V;
-      int64_t gm = k0 + mm;
+      int gm = k0 + mm;
       int gn = n0 + nn;
       VecT v = {};
       if (gm < m_end) {
-        const T* src = &B[gm * N + gn];
+        const T* src = &B[(int64_t)gm * N + gn];
         if (gn + V <= N && aligned16(src)) {
           v = *(const VecT*)src;
         } else {
 #pragma unroll
           for (int j = 0; j < V; ++j)
-            if (gn + j < N) v.e[j] = B[gm * N + gn + j];
+            if (gn + j < N) v.e[j] = B[(int64_t)gm * N + gn + j];
         }
       }
 #pragma unroll
-      for (int j = 0; j < V; ++j) Bs[(nn + j) * P + mm] = v.e[j];
+      for (int j = 0; j < V; ++j)
+        Bs[lds_off<T>(nn + j, mm)] = v.e[j];
     }
     __syncthreads();
     mfma_compute_tile(As, Bs, wc, acc);
     __syncthreads();
   }
 
-  const bool single = gridDim.z == 1;
+  // slab write: block z owns slab z of the fp32 workspace (C when z==1);
+  // a separate reduce pass sums slabs — deterministic, no atomic contention
+  float* out = C + (int64_t)blockIdx.z * K * N;
   epilogue_visit(wc, acc, r0, n0, [&](int row, int col, float v) {
-    if (row < K && col < N) {
-      if (single)
-        C[(int64_t)row * N + col] = v;
-      else
-        atomicAdd(&C[(int64_t)row * N + col], v);
-    }
+    if (row < K && col < N) out[(int64_t)row * N + col] = v;
   });
+}
+
+__global__ void k_splitk_reduce(const float* __restrict__ ws,
+                                float* __restrict__ out, int z, int64_t n) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    float acc = 0.0f;
+    for (int s = 0; s < z; ++s) acc += ws[(int64_t)s * n + i];
+    out[i] = acc;
+  }
+}
+
+void splitk_reduce_launch(const float* ws, float* out, int z, int64_t n,
+                          hipStream_t s) {
+  int blocks = (int)std::min<int64_t>((n + 255) / 256, (int64_t)2048);
+  hipLaunchKernelGGL(k_splitk_reduce, dim3(blocks), dim3(256), 0, s, ws, out,
+                     z, n);
 }
 
 // ---------------------------------------------------------------------------
@@ -220,20 +238,24 @@ void gemm_launch(DT dt, const void* a, const void* b, const void* bias,
   }
 }
 
-void gemm_tn_launch(DT dt, const void* a, const void* b, float* c_f32, int M,
-                    int N, int K, hipStream_t s) {
-  // pick the M split so total blocks ~ 2x CUs minimum
+int gemm_tn_zsplits(int M, int N, int K) {
   int base = ceil_div(K, BM) * ceil_div(N, BN);
-  int want = 512;
+  int want = 2048;
   int z = base >= want ? 1 : std::min(ceil_div(M, BK), ceil_div(want, base));
-  z = std::max(z, 1);
+  return std::max(z, 1);
+}
+
+void gemm_tn_launch(DT dt, const void* a, const void* b, float* c_f32,
+                    float* ws, int z, int M, int N, int K, hipStream_t s) {
   dim3 grid(ceil_div(K, BM), ceil_div(N, BN), z);
+  float* target = z == 1 ? c_f32 : ws;
   if (dt == DT::F32)
     hipLaunchKernelGGL(k_gemm_tn<float>, grid, dim3(THREADS), 0, s,
-                       (const float*)a, (const float*)b, c_f32, M, N, K);
+                       (const float*)a, (const float*)b, target, M, N, K);
   else
     hipLaunchKernelGGL(k_gemm_tn<bf16>, grid, dim3(THREADS), 0, s,
-                       (const bf16*)a, (const bf16*)b, c_f32, M, N, K);
+                       (const bf16*)a, (const bf16*)b, target, M, N, K);
+  if (z > 1) splitk_reduce_launch(ws, c_f32, z, (int64_t)K * N, s);
 }
 
 // ---------------------------------------------------------------------------

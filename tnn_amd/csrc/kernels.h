@@ -30,8 +30,11 @@ void cast_f32_launch(DT dt_out, const float* x, void* y, int64_t n,
 void gemm_launch(DT dt, const void* a, const void* b, const void* bias,
                  void* c, int M, int N, int K, bool trans_b, int act_kind,
                  hipStream_t s);
+int gemm_tn_zsplits(int M, int N, int K);
 void gemm_tn_launch(DT dt, const void* a, const void* b, float* c_f32,
-                    int M, int N, int K, hipStream_t s);
+                    float* ws, int z, int M, int N, int K, hipStream_t s);
+void splitk_reduce_launch(const float* ws, float* out, int z, int64_t n,
+                          hipStream_t s);
 void mfma_selftest_launch(const void* a_bf16, const void* b_bf16, float* d,
                           hipStream_t s);
 void mfma_selftest_f32_launch(const float* a, const float* b, float* d,
@@ -45,8 +48,10 @@ void conv2d_fwd_launch(DT dt, const void* x, const void* w, const void* bias,
                        void* y, const ConvShape& cs, bool relu, hipStream_t s);
 void conv2d_dgrad_launch(DT dt, const void* dy, const void* w_t, void* dx,
                          const ConvShape& cs, hipStream_t s);
+int conv2d_wgrad_zsplits(const ConvShape& cs);
 void conv2d_wgrad_launch(DT dt, const void* x, const void* dy, float* dw_f32,
-                         const ConvShape& cs, hipStream_t s);
+                         float* ws, int z, const ConvShape& cs,
+                         hipStream_t s);
 void transpose_w_launch(DT dt, const void* w, void* w_t, int KH, int KW,
                         int Cin, int Cout, hipStream_t s);
 

@@ -3,25 +3,27 @@
 // src/math/cuda/gemm.cu:64, and the cuDNN-frontend conv graphs).
 //
 // Structure (cdna_hip_programming.md §5 canonical GEMM):
-//   - 256 threads = 4 waves in a 2x2 grid; block tile BM=128 x BN=64, BK=32.
-//   - A staged in LDS as [BM][BKP] row-major (+16B row pad against bank
-//     conflicts); B staged TRANSPOSED as [BN][BKP] so both operands read
-//     contiguous k-vectors per lane.
-//   - bf16: one v_mfma_f32_16x16x32_bf16 per (fm, fn) per tile;
-//     fp32: v_mfma_f32_16x16x4_f32 (exact f32 at the f32 vector rate —
-//     there is no xf32 on gfx950), 8 k-steps per tile.
+//   - 256 threads = 4 waves in a 2x2 grid; block tile BM=128 x BN=64, BK=64.
+//   - A staged in LDS as [BM][BK] row-major; B staged TRANSPOSED as
+//     [BN][BK] so both operands read contiguous k-vectors per lane.
+//   - LDS banking: element (row, col) lives at byte
+//     (col*sizeof(T)) ^ ((((row>>3)^row)&7)<<4) within its 128B+ row.
+//     The XOR key mixes low AND mid row bits so BOTH access patterns are
+//     spread across banks: fragment reads touch 16 consecutive rows
+//     (key varies via row&7) and transpose-scatter writes touch rows at
+//     stride 8 across lanes (key varies via row>>3). Plain +pad schemes
+//     can't fix the scatter side: any 16B-aligned row stride puts an
+//     8-row lane stride on one bank (measured 3e8 conflict cycles).
+//   - bf16: v_mfma_f32_16x16x32_bf16; fp32: v_mfma_f32_16x16x4_f32
+//     (exact f32 at the f32 vector rate — no xf32 on gfx950).
 //   - Each wave owns a 64x32 sub-tile: 4x2 fragments, f32x4 accumulators.
-//
-// Kernels using this header implement their own global->LDS staging
-// (dense/NT/implicit-gemm gather) and epilogue (bias+act store / atomic
-// accumulate), then call mfma_compute_tile() between barriers.
 #pragma once
 
 #include "common.h"
 
 namespace tile {
 
-constexpr int BM = 128, BN = 64, BK = 32, THREADS = 256;
+constexpr int BM = 128, BN = 64, BK = 64, THREADS = 256;
 constexpr int WAVES_M = 2, WAVES_N = 2;      // wave grid
 constexpr int WM = BM / WAVES_M;             // 64 rows per wave
 constexpr int WN = BN / WAVES_N;             // 32 cols per wave
@@ -39,15 +41,19 @@ DEV bool aligned16(const T* p) {
   return ((unsigned long long)(const void*)p & 15ull) == 0;
 }
 
-template <typename T> struct Pad;            // 16B row padding in elements
-template <> struct Pad<float> { static constexpr int E = 4; };
-template <> struct Pad<bf16> { static constexpr int E = 8; };
-
-template <typename T> constexpr int BKP() { return BK + Pad<T>::E; }
+// element offset of tile element (row, col) in the swizzled [rows][BK]
+// image; the XOR only touches byte bits 4-6, so any 16B-aligned col keeps
+// its vector contiguous and the swizzle stays inside the row.
+template <typename T>
+DEV int lds_off(int row, int col) {
+  int byte = col * (int)sizeof(T);
+  byte ^= ((((row >> 3) ^ row) & 7) << 4);
+  return row * BK + byte / (int)sizeof(T);
+}
 
 template <typename T> struct LDSBytes {
-  static constexpr int A = BM * BKP<T>() * sizeof(T);
-  static constexpr int B = BN * BKP<T>() * sizeof(T);
+  static constexpr int A = BM * BK * sizeof(T);
+  static constexpr int B = BN * BK * sizeof(T);
   static constexpr int total = A + B;
 };
 
@@ -64,27 +70,28 @@ struct WaveCoord {
 // ---- MFMA tile compute: acc[FM][FN] += A_tile * B_tile^T-stored -----------
 DEV void mfma_compute_tile(const bf16* As, const bf16* Bs, const WaveCoord& w,
                            f32x4 acc[FM][FN]) {
-  constexpr int P = BK + Pad<bf16>::E;
-  const int r = w.lane & 15;          // fragment row/col within 16
-  const int kb = (w.lane >> 4) * 8;   // 8 bf16 per lane
-  bf16x8 a[FM], b[FN];
+  const int r = w.lane & 15;            // fragment row/col within 16
 #pragma unroll
-  for (int fm = 0; fm < FM; ++fm)
-    a[fm] = *(const bf16x8*)&As[(w.wrow0 + fm * 16 + r) * P + kb];
+  for (int ks = 0; ks < BK / 32; ++ks) {
+    const int kb = ks * 32 + (w.lane >> 4) * 8;  // 8 bf16 per lane
+    bf16x8 a[FM], b[FN];
 #pragma unroll
-  for (int fn = 0; fn < FN; ++fn)
-    b[fn] = *(const bf16x8*)&Bs[(w.wcol0 + fn * 16 + r) * P + kb];
-#pragma unroll
-  for (int fm = 0; fm < FM; ++fm)
+    for (int fm = 0; fm < FM; ++fm)
+      a[fm] = *(const bf16x8*)&As[lds_off<bf16>(w.wrow0 + fm * 16 + r, kb)];
 #pragma unroll
     for (int fn = 0; fn < FN; ++fn)
-      acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-          a[fm], b[fn], acc[fm][fn], 0, 0, 0);
+      b[fn] = *(const bf16x8*)&Bs[lds_off<bf16>(w.wcol0 + fn * 16 + r, kb)];
+#pragma unroll
+    for (int fm = 0; fm < FM; ++fm)
+#pragma unroll
+      for (int fn = 0; fn < FN; ++fn)
+        acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a[fm], b[fn], acc[fm][fn], 0, 0, 0);
+  }
 }
 
 DEV void mfma_compute_tile(const float* As, const float* Bs, const WaveCoord& w,
                            f32x4 acc[FM][FN]) {
-  constexpr int P = BK + Pad<float>::E;
   const int r = w.lane & 15;
   const int kq = w.lane >> 4;         // one f32 per lane per k-step of 4
 #pragma unroll
@@ -92,10 +99,10 @@ DEV void mfma_compute_tile(const float* As, const float* Bs, const WaveCoord& w,
     float a[FM], b[FN];
 #pragma unroll
     for (int fm = 0; fm < FM; ++fm)
-      a[fm] = As[(w.wrow0 + fm * 16 + r) * P + ks * 4 + kq];
+      a[fm] = As[lds_off<float>(w.wrow0 + fm * 16 + r, ks * 4 + kq)];
 #pragma unroll
     for (int fn = 0; fn < FN; ++fn)
-      b[fn] = Bs[(w.wcol0 + fn * 16 + r) * P + ks * 4 + kq];
+      b[fn] = Bs[lds_off<float>(w.wcol0 + fn * 16 + r, ks * 4 + kq)];
 #pragma unroll
     for (int fm = 0; fm < FM; ++fm)
 #pragma unroll
