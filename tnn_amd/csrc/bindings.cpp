@@ -171,6 +171,7 @@ static ConvShape conv_shape(const at::Tensor& x_like, int Cin, int Cout, int KH,
   cs.PW = PW;
   cs.OH = (cs.H + 2 * PH - KH) / SH + 1;
   cs.OW = (cs.W + 2 * PW - KW) / SW + 1;
+  cs.init_fdiv();
   return cs;
 }
 
@@ -214,6 +215,7 @@ at::Tensor conv2d_dgrad(const at::Tensor& dy, const at::Tensor& w, int64_t H,
   cs.PW = pw;
   cs.OH = dy.size(1);
   cs.OW = dy.size(2);
+  cs.init_fdiv();
   // dgrad consumes the weight as B[(kh,kw,co)][ci]: transpose once per call
   auto w_t = at::empty({KH, KW, Cout, Cin}, w.options());
   transpose_w_launch(dt_of(w), w.data_ptr(), w_t.data_ptr(), KH, KW, Cin, Cout,

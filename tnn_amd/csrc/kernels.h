@@ -41,8 +41,29 @@ void mfma_selftest_f32_launch(const float* a, const float* b, float* d,
                               hipStream_t s);
 
 // ---- conv2d.hip ------------------------------------------------------------
+// shift-based fast division for the im2col address math (every hot divisor
+// — OW, OH*OW, Cin, Cout — is a power of two in the CNN zoo; others fall
+// back to hardware division)
+struct IDiv {
+  int d = 1;
+  int lg = 0;  // log2(d) if power of two, else -1
+  void set(int dd) {
+    d = dd;
+    lg = (dd & (dd - 1)) == 0 ? __builtin_ctz(dd) : -1;
+  }
+};
+
 struct ConvShape {
   int N, H, W, Cin, Cout, KH, KW, SH, SW, PH, PW, OH, OW;
+  IDiv d_ohow, d_ow, d_cin, d_hw, d_w, d_cout;
+  void init_fdiv() {
+    d_ohow.set(OH * OW);
+    d_ow.set(OW);
+    d_cin.set(Cin);
+    d_hw.set(H * W);
+    d_w.set(W);
+    d_cout.set(Cout);
+  }
 };
 void conv2d_fwd_launch(DT dt, const void* x, const void* w, const void* bias,
                        void* y, const ConvShape& cs, bool relu, hipStream_t s);

@@ -171,9 +171,19 @@ class PipelineEngine:
                 inp = self._recv_act(mb_size)
             out = self.stage(inp)
             if self.is_last:
-                loss = self.criterion(out, micro_y[fwd_idx]) / M
+                my = micro_y[fwd_idx]
+                loss = self.criterion(out, my) / M
                 losses.append(loss)
-                accs.append(accuracy(out.detach(), micro_y[fwd_idx]))
+                # keep accuracy on-device: a .item() here would stall the
+                # pipeline once per micro-batch
+                with torch.no_grad():
+                    pred = out.detach().reshape(-1, out.shape[-1]).argmax(-1)
+                    t = my.reshape(-1)
+                    if t.dim() < pred.dim() or t.dtype not in (torch.int64,
+                                                               torch.int32):
+                        t = my.reshape(-1, my.shape[-1]).argmax(-1) \
+                            if my.dim() == out.dim() else t.long()
+                    accs.append((pred == t).float().mean())
                 fifo.append((inp, loss))
             else:
                 send_works.append(self.comm.isend(out, self.rank + 1))
@@ -206,8 +216,12 @@ class PipelineEngine:
             self.optimizer.zero_grad()
             if self.scheduler is not None:
                 self.scheduler.step()
-        stats = {"loss": float(sum(l.item() for l in losses)) if losses else 0.0,
-                 "accuracy": float(sum(accs) / len(accs)) if accs else 0.0}
+        if losses:  # one host sync for the whole batch
+            packed = torch.stack([torch.stack(losses).sum(),
+                                  torch.stack(accs).mean()])
+            stats = {"loss": float(packed[0]), "accuracy": float(packed[1])}
+        else:
+            stats = {"loss": 0.0, "accuracy": 0.0}
         return stats
 
     @torch.no_grad()
