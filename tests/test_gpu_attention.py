@@ -1,0 +1,117 @@
+"""Flash attention kernels vs fp32 composed reference (full-tensor check,
+asymmetric random operands — cdna guide §5.4 rules 16/25)."""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from tnn_amd import _C
+    ext = _C.ext()
+
+DEV = "cuda"
+
+
+def ref_attention(q, k, v, causal):
+    scale = q.shape[-1] ** -0.5
+    s = (q.float() @ k.float().transpose(-1, -2)) * scale
+    if causal:
+        S = q.shape[-2]
+        mask = torch.ones(S, S, dtype=torch.bool, device=q.device).triu(1)
+        s = s.masked_fill(mask, float("-inf"))
+    p = torch.softmax(s, dim=-1)
+    return p @ v.float()
+
+
+def maxerr(a, b):
+    return (a.float() - b.float()).abs().max().item()
+
+
+@pytest.mark.parametrize("causal", [False, True])
+@pytest.mark.parametrize("S", [64, 128, 192, 1000])
+@pytest.mark.parametrize("D", [64, 128])
+def test_attn_fwd(S, D, causal):
+    torch.manual_seed(0)
+    B, H = 2, 3
+    q = torch.randn(B, H, S, D, dtype=torch.bfloat16, device=DEV)
+    k = torch.randn(B, H, S, D, dtype=torch.bfloat16, device=DEV)
+    v = torch.randn(B, H, S, D, dtype=torch.bfloat16, device=DEV)
+    o, lse = ext.attn_fwd(q, k, v, causal)
+    ref = ref_attention(q, k, v, causal)
+    assert maxerr(o, ref) < 3e-2, (S, D, causal, maxerr(o, ref))
+    # lse check
+    scale = D ** -0.5
+    s = (q.float() @ k.float().transpose(-1, -2)) * scale
+    if causal:
+        mask = torch.ones(S, S, dtype=torch.bool, device=DEV).triu(1)
+        s = s.masked_fill(mask, float("-inf"))
+    ref_lse = torch.logsumexp(s, dim=-1)
+    assert maxerr(lse, ref_lse) < 2e-2
+
+
+@pytest.mark.parametrize("causal", [False, True])
+@pytest.mark.parametrize("S,D", [(128, 64), (256, 64), (192, 128)])
+def test_attn_bwd(S, D, causal):
+    torch.manual_seed(1)
+    B, H = 2, 2
+    q = torch.randn(B, H, S, D, dtype=torch.bfloat16, device=DEV)
+    k = torch.randn(B, H, S, D, dtype=torch.bfloat16, device=DEV)
+    v = torch.randn(B, H, S, D, dtype=torch.bfloat16, device=DEV)
+    do = torch.randn(B, H, S, D, dtype=torch.bfloat16, device=DEV)
+
+    o, lse = ext.attn_fwd(q, k, v, causal)
+    dq, dk, dv = ext.attn_bwd(q, k, v, o, do, lse, causal)
+
+    qf = q.float().requires_grad_(True)
+    kf = k.float().requires_grad_(True)
+    vf = v.float().requires_grad_(True)
+    ref_attention(qf, kf, vf, causal).backward(do.float())
+    tol = 6e-2
+    assert maxerr(dq, qf.grad) < tol, ("dq", maxerr(dq, qf.grad))
+    assert maxerr(dk, kf.grad) < tol, ("dk", maxerr(dk, kf.grad))
+    assert maxerr(dv, vf.grad) < tol, ("dv", maxerr(dv, vf.grad))
+
+
+def test_attn_spiked_rescale_path():
+    """Force large per-tile max jumps (guide §5.4 rule 26): one K row
+    spiked so the running max moves far past earlier tiles."""
+    torch.manual_seed(2)
+    S, D = 512, 64
+    q = torch.randn(1, 1, S, D, dtype=torch.bfloat16, device=DEV)
+    k = torch.randn(1, 1, S, D, dtype=torch.bfloat16, device=DEV)
+    k[0, 0, 400] *= 30  # spike in a late tile
+    v = torch.randn(1, 1, S, D, dtype=torch.bfloat16, device=DEV)
+    o, _ = ext.attn_fwd(q, k, v, False)
+    ref = ref_attention(q, k, v, False)
+    assert maxerr(o, ref) < 5e-2
+
+
+def test_flash_autograd_function():
+    from tnn_amd.ops.flash import flash_attention
+    torch.manual_seed(3)
+    q = torch.randn(1, 2, 128, 64, dtype=torch.bfloat16, device=DEV,
+                    requires_grad=True)
+    k = torch.randn_like(q, requires_grad=True)
+    v = torch.randn_like(q, requires_grad=True)
+    o = flash_attention(q, k, v, causal=True)
+    o.sum().backward()
+    assert q.grad is not None and torch.isfinite(q.grad.float()).all()
+
+
+def test_gpt_block_gpu():
+    from tnn_amd.nn.blocks import GPTBlock
+    from tnn_amd.nn.layer import cast_compute_dtype
+    torch.manual_seed(4)
+    blk = GPTBlock(256, 4, flash=True, dtype=torch.float32)
+    blk_ref = GPTBlock(256, 4, flash=False, dtype=torch.float32)
+    blk_ref.load_state_dict(blk.state_dict())
+    x = torch.randn(2, 128, 256)
+    ref = blk_ref(x)  # CPU fp32 naive attention
+    cast_compute_dtype(blk, torch.bfloat16)
+    blk.to(DEV)
+    y = blk(x.to(DEV).bfloat16())
+    err = (y.float().cpu() - ref).abs().max().item()
+    assert err < 0.15, err

@@ -469,6 +469,44 @@ at::Tensor embedding_bwd(const at::Tensor& ids, const at::Tensor& dy,
   return dt;
 }
 
+// ---- flash attention -------------------------------------------------------
+std::vector<at::Tensor> attn_fwd(const at::Tensor& q, const at::Tensor& k,
+                                 const at::Tensor& v, bool causal) {
+  CHECK_IN(q); CHECK_IN(k); CHECK_IN(v);
+  TORCH_CHECK(q.scalar_type() == at::kBFloat16, "flash attention is bf16");
+  TORCH_CHECK(q.dim() == 4, "q must be [B,H,S,D]");
+  int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
+  TORCH_CHECK(D == 64 || D == 128, "head dim must be 64 or 128, got ", D);
+  auto o = at::empty_like(q);
+  auto lse = at::empty({B, H, S}, q.options().dtype(at::kFloat));
+  float scale = 1.0f / std::sqrt((float)D);
+  attn_fwd_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                  lse.data_ptr<float>(), B * H, S, D, causal, scale,
+                  cur_stream());
+  return {o, lse};
+}
+
+std::vector<at::Tensor> attn_bwd(const at::Tensor& q, const at::Tensor& k,
+                                 const at::Tensor& v, const at::Tensor& o,
+                                 const at::Tensor& dout, const at::Tensor& lse,
+                                 bool causal) {
+  CHECK_IN(q); CHECK_IN(dout);
+  int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
+  auto dqw = at::zeros({B, H, S, D}, q.options().dtype(at::kFloat));
+  auto di = at::empty({B, H, S}, q.options().dtype(at::kFloat));
+  auto dk = at::empty_like(k);
+  auto dv = at::empty_like(v);
+  float scale = 1.0f / std::sqrt((float)D);
+  attn_bwd_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                  dout.contiguous().data_ptr(), lse.data_ptr<float>(),
+                  di.data_ptr<float>(), dqw.data_ptr<float>(), dk.data_ptr(),
+                  dv.data_ptr(), B * H, S, D, causal, scale, cur_stream());
+  auto dq = at::empty_like(q);
+  cast_f32_launch(DT::BF16, dqw.data_ptr<float>(), dq.data_ptr(), dq.numel(),
+                  cur_stream());
+  return {dq, dk, dv};
+}
+
 // ---- optimizers ------------------------------------------------------------
 void sgd_step(at::Tensor param, at::Tensor master, const at::Tensor& grad,
               const c10::optional<at::Tensor>& momentum_buf, double lr,
@@ -526,6 +564,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ln_bwd", &tnn::ln_bwd);
   m.def("embedding_fwd", &tnn::embedding_fwd);
   m.def("embedding_bwd", &tnn::embedding_bwd);
+  m.def("attn_fwd", &tnn::attn_fwd);
+  m.def("attn_bwd", &tnn::attn_bwd);
   m.def("sgd_step", &tnn::sgd_step);
   m.def("adam_step", &tnn::adam_step);
 }

@@ -475,7 +475,7 @@ def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
               causal: bool = True) -> torch.Tensor:
     """q/k/v: [B, H, S, D] -> [B, H, S, D]."""
     if _use_hip(q):
-        from .attention import flash_attention
+        from .flash import flash_attention
         return flash_attention(q, k, v, causal=causal)
     scale = q.shape[-1] ** -0.5
     s = (q.float() @ k.float().transpose(-1, -2)) * scale
