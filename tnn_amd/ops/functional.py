@@ -215,8 +215,13 @@ def linear(x: torch.Tensor, w: torch.Tensor, bias: Optional[torch.Tensor] = None
     lead = x.shape[:-1]
     x2 = x.reshape(-1, x.shape[-1])
     if _use_hip(x):
+        # relu fuses into the GEMM epilogue (mask-replay backward); other
+        # activations need the pre-activation saved, so run them unfused
+        fused = act if act in ("linear", "relu") else "linear"
         y = _Linear.apply(x2.contiguous(), w.contiguous(),
-                          None if bias is None else bias.contiguous(), act)
+                          None if bias is None else bias.contiguous(), fused)
+        if fused != act:
+            y = _Activation.apply(y, act)
     else:
         y = x2 @ w
         if bias is not None:
