@@ -64,13 +64,14 @@ def _reference_losses(M=4):
     return losses, model.state_dict()
 
 
-def _pipeline_worker(rank, world, port, result_q):
+def _pipeline_worker(rank, world, tmpdir):
     os.environ.update({
-        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
         "RANK": str(rank), "WORLD_SIZE": str(world), "LOCAL_RANK": str(rank),
     })
     import torch.distributed as dist
-    dist.init_process_group("gloo", rank=rank, world_size=world)
+    dist.init_process_group(
+        "gloo", init_method=f"file://{tmpdir}/pg_init", rank=rank,
+        world_size=world)
     from tnn_amd.parallel import Communicator, PipelineEngine
     comm = Communicator()
     model = _build_model() if rank == 0 else None
@@ -85,28 +86,27 @@ def _pipeline_worker(rank, world, port, result_q):
         stats = engine.broadcast_stats(stats)
         losses.append(stats["loss"])
     sd = {k: v.clone() for k, v in engine.state_dict().items()}
-    result_q.put((rank, losses, sd))
+    torch.save((rank, losses, sd), os.path.join(tmpdir, f"result_{rank}.pt"))
     dist.barrier()
     dist.destroy_process_group()
 
 
 @pytest.mark.timeout(300)
-def test_pipeline_matches_single_process():
+def test_pipeline_matches_single_process(tmp_path):
     ref_losses, ref_sd = _reference_losses()
-    port = _free_port()
     ctx = mp.get_context("spawn")
-    q = ctx.SimpleQueue()
-    procs = [ctx.Process(target=_pipeline_worker, args=(r, 2, port, q))
+    procs = [ctx.Process(target=_pipeline_worker, args=(r, 2, str(tmp_path)))
              for r in range(2)]
     for p in procs:
         p.start()
-    results = {}
-    for _ in range(2):
-        rank, losses, sd = q.get()
-        results[rank] = (losses, sd)
     for p in procs:
-        p.join(timeout=120)
+        p.join(timeout=180)
         assert p.exitcode == 0
+    results = {}
+    for r in range(2):
+        rank, losses, sd = torch.load(tmp_path / f"result_{r}.pt",
+                                      weights_only=False)
+        results[rank] = (losses, sd)
 
     pipe_losses = results[0][0]
     assert pipe_losses == pytest.approx(ref_losses, rel=1e-4), \

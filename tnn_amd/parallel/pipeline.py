@@ -219,7 +219,8 @@ class PipelineEngine:
         if losses:  # one host sync for the whole batch
             packed = torch.stack([torch.stack(losses).sum(),
                                   torch.stack(accs).mean()])
-            stats = {"loss": float(packed[0]), "accuracy": float(packed[1])}
+            stats = {"loss": float(packed[0].detach()),
+                     "accuracy": float(packed[1])}
         else:
             stats = {"loss": 0.0, "accuracy": 0.0}
         return stats
