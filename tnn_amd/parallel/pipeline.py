@@ -33,6 +33,7 @@ from ..nn.losses import Loss, CrossEntropyLoss
 from ..nn.optim import Optimizer, optimizer_from_config
 from ..nn.schedulers import scheduler_from_config
 from ..utils.logging import get_logger
+from ..utils.profiler import Profiler, EventType
 from .comm import Communicator
 from .partitioner import partition_model, Partitioner
 
@@ -92,6 +93,10 @@ class PipelineEngine:
         self.out_shape = self.boundary[self.rank + 1]
         self.is_first = self.rank == 0
         self.is_last = self.rank == self.world - 1
+
+        # per-rank profiler (reference worker.hpp:146-204 hook points);
+        # enable with engine.profiler.start(), merge via gather_profiles()
+        self.profiler = Profiler(source=f"rank{self.rank}")
 
         opt_cfg = optimizer_config or {"type": "adamw", "lr": 1e-3}
         self.optimizer: Optimizer = optimizer_from_config(
@@ -168,8 +173,11 @@ class PipelineEngine:
             if self.is_first:
                 inp = micro_x[fwd_idx].to(self.io_dtype)
             else:
-                inp = self._recv_act(mb_size)
-            out = self.stage(inp)
+                with self.profiler.span(EventType.COMMUNICATION,
+                                        f"recv_act mb{fwd_idx}"):
+                    inp = self._recv_act(mb_size)
+            with self.profiler.span(EventType.COMPUTE, f"forward mb{fwd_idx}"):
+                out = self.stage(inp)
             if self.is_last:
                 my = micro_y[fwd_idx]
                 loss = self.criterion(out, my) / M
@@ -190,15 +198,23 @@ class PipelineEngine:
                 fifo.append((inp, out))
             fwd_idx += 1
 
+        bwd_idx = 0
+
         def do_backward():
+            nonlocal bwd_idx
             inp, out = fifo.pop(0)
             if self.is_last:
-                out.backward()  # out is the micro-loss
+                with self.profiler.span(EventType.COMPUTE, f"backward mb{bwd_idx}"):
+                    out.backward()  # out is the micro-loss
             else:
-                grad = self._recv_grad(out)
-                torch.autograd.backward(out, grad)
+                with self.profiler.span(EventType.COMMUNICATION,
+                                        f"recv_grad mb{bwd_idx}"):
+                    grad = self._recv_grad(out)
+                with self.profiler.span(EventType.COMPUTE, f"backward mb{bwd_idx}"):
+                    torch.autograd.backward(out, grad)
             if not self.is_first:
                 send_works.append(self.comm.isend(inp.grad, self.rank - 1))
+            bwd_idx += 1
 
         warmup = min(self.num_stages - 1 - self.rank, M)
         for _ in range(warmup):
@@ -212,12 +228,13 @@ class PipelineEngine:
         for w in send_works:
             w.wait()
         if step:
-            self.optimizer.step()
-            self.optimizer.zero_grad()
-            if self.scheduler is not None:
-                self.scheduler.step()
+            with self.profiler.span(EventType.COMPUTE, "update_parameters"):
+                self.optimizer.step()
+                self.optimizer.zero_grad()
+                if self.scheduler is not None:
+                    self.scheduler.step()
         if losses:  # one host sync for the whole batch
-            packed = torch.stack([torch.stack(losses).sum(),
+            packed = torch.stack([torch.stack([l.detach() for l in losses]).sum(),
                                   torch.stack(accs).mean()])
             stats = {"loss": float(packed[0].detach()),
                      "accuracy": float(packed[1])}
@@ -262,3 +279,25 @@ class PipelineEngine:
 
     def state_dict(self):
         return self.stage.state_dict()
+
+    # -- profiling control plane (reference coordinator.hpp:277-362:
+    #    START/REPORT/CLEAR fan-out + merge) --------------------------------
+    def start_profiling(self):
+        self.profiler.start()
+
+    def stop_profiling(self):
+        self.profiler.stop()
+
+    def gather_profiles(self) -> Optional[Profiler]:
+        """Collect every rank's events on rank 0 (reference merges returned
+        Profiler payloads, coordinator.hpp:291-362)."""
+        dumps = self.comm.gather_objects(self.profiler.to_dict(), dst=0)
+        if self.rank != 0:
+            return None
+        merged = Profiler("pipeline")
+        for d in dumps:
+            merged.merge(Profiler.from_dict(d))
+        return merged
+
+    def clear_profiling(self):
+        self.profiler.clear()
