@@ -25,16 +25,23 @@ sources = sorted(p for p in glob.glob(os.path.join(CSRC, "*.hip"))
     os.path.join(CSRC, "bindings.cpp")
 ]
 
+cxx_flags = ["-O3", "-std=c++17"]
+hip_flags = ["-O3", "-std=c++17", "--offload-arch=gfx950"]
+if os.environ.get("TNN_DEBUG"):
+    # debug builds get host sanitizers (reference cmake/CompilerFlags.cmake
+    # -fsanitize=address,undefined analog); device-side: run under
+    # AMD_SERIALIZE_KERNEL=3 HIP_LAUNCH_BLOCKING=1
+    san = ["-g", "-fsanitize=address,undefined", "-fno-omit-frame-pointer"]
+    cxx_flags += san
+    hip_flags += ["-g"]
+
 setup(
     name="tnn_amd_hip",
     ext_modules=[
         CUDAExtension(
             name="tnn_amd._hip",
             sources=sources,
-            extra_compile_args={
-                "cxx": ["-O3", "-std=c++17"],
-                "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
-            },
+            extra_compile_args={"cxx": cxx_flags, "nvcc": hip_flags},
         )
     ],
     cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},
