@@ -58,9 +58,19 @@ def main():
     num_micro = max(1, global_batch // args.microbatch) if world > 1 else 1
 
     model = models.create_model(args.model) if rank == 0 else None
-    in_shape = (32, 32, 3)
-    num_classes = {"cifar100_wrn16_8": 100, "cifar10_resnet9": 10}.get(
-        args.model, 100)
+    shapes = {
+        "cifar100_wrn16_8": ((32, 32, 3), 100),
+        "cifar10_resnet9": ((32, 32, 3), 10),
+        "cifar10_vgg": ((32, 32, 3), 10),
+        "cifar100_resnet18": ((32, 32, 3), 100),
+        "tiny_imagenet_wrn16_8": ((64, 64, 3), 200),
+        "tiny_imagenet_resnet18": ((64, 64, 3), 200),
+        "tiny_imagenet_resnet50": ((64, 64, 3), 200),
+        "tiny_imagenet_vit": ((64, 64, 3), 200),
+        "tiny_imagenet_flash_vit": ((64, 64, 3), 200),
+        "imagenet_resnet50": ((224, 224, 3), 1000),
+    }
+    in_shape, num_classes = shapes.get(args.model, ((32, 32, 3), 100))
 
     engine = PipelineEngine(
         model, comm, input_shape=in_shape, num_microbatches=num_micro,
@@ -114,7 +124,7 @@ def main():
             "config": {
                 "model": args.model,
                 "global_batch": global_batch,
-                "input": "32x32x3",
+                "input": "x".join(map(str, in_shape)),
                 "microbatches": num_micro,
                 "parallelism": f"pp{world}" if world > 1 else "single",
             },
