@@ -118,3 +118,46 @@ def test_pipeline_matches_single_process(tmp_path):
     n_ref = sum(v.numel() for v in ref_sd.values())
     n_pipe = sum(sum(v.numel() for v in results[r][1].values()) for r in (0, 1))
     assert n_ref == n_pipe
+
+
+def _pipeline_worker_n(rank, world, tmpdir, M):
+    os.environ.update({
+        "RANK": str(rank), "WORLD_SIZE": str(world), "LOCAL_RANK": str(rank),
+    })
+    import torch.distributed as dist
+    dist.init_process_group(
+        "gloo", init_method=f"file://{tmpdir}/pg_init", rank=rank,
+        world_size=world)
+    from tnn_amd.parallel import Communicator, PipelineEngine
+    comm = Communicator()
+    model = _build_model() if rank == 0 else None
+    engine = PipelineEngine(model, comm, input_shape=(8, 8, 3),
+                            num_microbatches=M,
+                            optimizer_config={"type": "adamw", "lr": 1e-3},
+                            device=torch.device("cpu"),
+                            sync_weights=True)
+    losses = []
+    for x, y in _data():
+        stats = engine.broadcast_stats(engine.train_batch(x, y))
+        losses.append(stats["loss"])
+    if rank == 0:
+        torch.save(losses, os.path.join(tmpdir, "losses.pt"))
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_pipeline_4stage_matches_single_process(tmp_path):
+    """4 ranks x 8 micro-batches: middle ranks run the batched
+    send-forward-recv-backward steady state (the RCCL deadlock-free path)."""
+    ref_losses, _ = _reference_losses(M=8)
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_pipeline_worker_n,
+                         args=(r, 4, str(tmp_path), 8)) for r in range(4)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=240)
+        assert p.exitcode == 0
+    losses = torch.load(tmp_path / "losses.pt", weights_only=False)
+    assert losses == pytest.approx(ref_losses, rel=1e-4), (losses, ref_losses)

@@ -162,68 +162,118 @@ class PipelineEngine:
         if self.is_last:
             micro_y = list(y.to(self.device).chunk(M, dim=0))
 
-        send_works: List[Any] = []
+        # -- comm pattern -----------------------------------------------------
+        # Warmup is forward-only traffic and cooldown backward-only (both
+        # acyclic), but in the steady 1F1B phase every middle rank both
+        # sends an activation to r+1 and receives a gradient from r+1 (and
+        # symmetrically with r-1). Issued as separate ops those two
+        # rendezvous transfers deadlock on RCCL — the send kernel occupies
+        # the comm stream until the peer posts its recv, and the peer is
+        # symmetrically blocked (gloo's buffering hides this on CPU). So
+        # bidirectional pairs always travel as ONE batched P2P group
+        # (rcclGroupStart/End), the Megatron send_forward_recv_backward /
+        # send_backward_recv_forward pattern.
+        import torch.distributed as dist
+
         fifo: List[Tuple[Optional[torch.Tensor], torch.Tensor]] = []
         losses: List[torch.Tensor] = []
-        accs: List[float] = []
-        fwd_idx = 0
+        accs: List[torch.Tensor] = []
+        send_works: List[Any] = []
 
-        def do_forward():
-            nonlocal fwd_idx
-            if self.is_first:
-                inp = micro_x[fwd_idx].to(self.io_dtype)
-            else:
-                with self.profiler.span(EventType.COMMUNICATION,
-                                        f"recv_act mb{fwd_idx}"):
-                    inp = self._recv_act(mb_size)
-            with self.profiler.span(EventType.COMPUTE, f"forward mb{fwd_idx}"):
+        def forward_mb(inp, idx):
+            with self.profiler.span(EventType.COMPUTE, f"forward mb{idx}"):
                 out = self.stage(inp)
             if self.is_last:
-                my = micro_y[fwd_idx]
+                my = micro_y[idx]
                 loss = self.criterion(out, my) / M
                 losses.append(loss)
-                # keep accuracy on-device: a .item() here would stall the
-                # pipeline once per micro-batch
-                with torch.no_grad():
+                with torch.no_grad():  # on-device accuracy (no .item() stall)
                     pred = out.detach().reshape(-1, out.shape[-1]).argmax(-1)
-                    t = my.reshape(-1)
-                    if t.dim() < pred.dim() or t.dtype not in (torch.int64,
-                                                               torch.int32):
-                        t = my.reshape(-1, my.shape[-1]).argmax(-1) \
-                            if my.dim() == out.dim() else t.long()
+                    t = (my.reshape(-1, my.shape[-1]).argmax(-1)
+                         if my.dim() == out.dim() else my.reshape(-1).long())
                     accs.append((pred == t).float().mean())
                 fifo.append((inp, loss))
-            else:
-                send_works.append(self.comm.isend(out, self.rank + 1))
-                fifo.append((inp, out))
-            fwd_idx += 1
+                return None
+            fifo.append((inp, out))
+            return out
 
-        bwd_idx = 0
-
-        def do_backward():
-            nonlocal bwd_idx
+        def backward_mb(grad_out, idx):
             inp, out = fifo.pop(0)
-            if self.is_last:
-                with self.profiler.span(EventType.COMPUTE, f"backward mb{bwd_idx}"):
+            with self.profiler.span(EventType.COMPUTE, f"backward mb{idx}"):
+                if self.is_last:
                     out.backward()  # out is the micro-loss
-            else:
-                with self.profiler.span(EventType.COMMUNICATION,
-                                        f"recv_grad mb{bwd_idx}"):
-                    grad = self._recv_grad(out)
-                with self.profiler.span(EventType.COMPUTE, f"backward mb{bwd_idx}"):
-                    torch.autograd.backward(out, grad)
-            if not self.is_first:
-                send_works.append(self.comm.isend(inp.grad, self.rank - 1))
-            bwd_idx += 1
+                else:
+                    torch.autograd.backward(out, grad_out)
+            return None if self.is_first else inp.grad
+
+        def recv_forward(idx):
+            if self.is_first:
+                return micro_x[idx].to(self.io_dtype)
+            with self.profiler.span(EventType.COMMUNICATION, f"recv_act {idx}"):
+                return self._recv_act(mb_size)
+
+        def send_forward(out):
+            if out is not None:
+                send_works.append(self.comm.isend(out, self.rank + 1))
+
+        def send_fwd_recv_bwd(out):
+            """Batched: ship the newest activation to r+1 and receive the
+            gradient for the oldest in-flight one (same peer, one group)."""
+            if self.is_last:
+                return None
+            grad = torch.empty_like(fifo[0][1])
+            with self.profiler.span(EventType.COMMUNICATION, "send_fwd_recv_bwd"):
+                reqs = self.comm.batch_p2p([
+                    dist.P2POp(dist.isend, out.contiguous(), self.rank + 1),
+                    dist.P2POp(dist.irecv, grad, self.rank + 1)])
+                for r in reqs:
+                    r.wait()
+            return grad
+
+        def send_bwd_recv_fwd(ingrad):
+            """Batched: ship the input-gradient to r-1 and receive the next
+            micro-batch activation from it."""
+            act = torch.empty(mb_size, *self.in_shape, device=self.device,
+                              dtype=self.io_dtype)
+            with self.profiler.span(EventType.COMMUNICATION, "send_bwd_recv_fwd"):
+                reqs = self.comm.batch_p2p([
+                    dist.P2POp(dist.isend, ingrad.contiguous(), self.rank - 1),
+                    dist.P2POp(dist.irecv, act, self.rank - 1)])
+                for r in reqs:
+                    r.wait()
+            return act.requires_grad_(True)
+
+        def send_backward(ingrad):
+            if ingrad is not None:
+                send_works.append(self.comm.isend(ingrad, self.rank - 1))
 
         warmup = min(self.num_stages - 1 - self.rank, M)
+        steady = M - warmup
+        fwd_idx = bwd_idx = 0
+
         for _ in range(warmup):
-            do_forward()
-        for _ in range(M - warmup):
-            do_forward()
-            do_backward()
-        for _ in range(warmup):
-            do_backward()
+            send_forward(forward_mb(recv_forward(fwd_idx), fwd_idx))
+            fwd_idx += 1
+        inp = recv_forward(fwd_idx) if steady > 0 else None
+        for i in range(steady):
+            out = forward_mb(inp, fwd_idx)
+            fwd_idx += 1
+            grad_out = send_fwd_recv_bwd(out)
+            ingrad = backward_mb(grad_out, bwd_idx)
+            bwd_idx += 1
+            last_steady = i == steady - 1
+            if not last_steady:
+                if self.is_first:
+                    inp = recv_forward(fwd_idx)
+                elif ingrad is not None:
+                    inp = send_bwd_recv_fwd(ingrad)
+            else:
+                send_backward(ingrad)
+        for _ in range(warmup):  # cooldown: backward-only traffic, acyclic
+            grad_out = (None if self.is_last
+                        else self._recv_grad(fifo[0][1]))
+            send_backward(backward_mb(grad_out, bwd_idx))
+            bwd_idx += 1
 
         for w in send_works:
             w.wait()
