@@ -16,7 +16,7 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from tnn_amd import models
 from tnn_amd.nn import CrossEntropyLoss, AdamW
 from tnn_amd.nn.layer import cast_compute_dtype
-from tnn_amd.models.generate import generate
+from tnn_amd.models.generate import generate, generate_cached
 
 
 def main():
@@ -66,13 +66,14 @@ def main():
 
     # ---- greedy decode (reference gpt2_inference loop) ----
     model.eval()
-    t0 = time.perf_counter()
-    out = generate(model, list(range(16)), max_new_tokens=args.decode_tokens,
-                   seq_len=args.seq_len, device=dev, eot_token=None)
-    dt = time.perf_counter() - t0
-    n = len(out) - 16
-    print(f"decode: {n} tokens in {dt:.2f}s = {n / dt:.2f} tok/s "
-          f"(full-sequence recompute, reference parity)")
+    for name, fn in [("recompute (reference parity)", generate),
+                     ("kv-cache", generate_cached)]:
+        t0 = time.perf_counter()
+        out = fn(model, list(range(16)), max_new_tokens=args.decode_tokens,
+                 seq_len=args.seq_len, device=dev, eot_token=None)
+        dt = time.perf_counter() - t0
+        n = len(out) - 16
+        print(f"decode [{name}]: {n} tokens in {dt:.2f}s = {n / dt:.2f} tok/s")
 
 
 if __name__ == "__main__":

@@ -20,7 +20,7 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 from tnn_amd import models
 from tnn_amd.data import Tokenizer
-from tnn_amd.models.generate import generate
+from tnn_amd.models.generate import generate, generate_cached
 from tnn_amd.nn.layer import cast_compute_dtype
 from tnn_amd.utils.checkpoint import load_model
 
@@ -34,6 +34,8 @@ def main():
     p.add_argument("--tokens", type=int, default=50)
     p.add_argument("--seq-len", type=int, default=512)
     p.add_argument("--bf16", action="store_true")
+    p.add_argument("--kv-cache", action="store_true",
+                   help="cached decode (the reference recomputes)")
     args = p.parse_args()
 
     model = (load_model(args.snapshot) if args.snapshot
@@ -46,14 +48,15 @@ def main():
     tok = Tokenizer().load(args.vocab) if args.vocab else None
     prompt_ids = tok.encode(args.prompt) if tok else list(range(10))
 
+    gen = generate_cached if args.kv_cache else generate
     t0 = time.perf_counter()
-    out = generate(model, prompt_ids, max_new_tokens=args.tokens,
-                   seq_len=args.seq_len, device=dev,
-                   eot_token=50256 if tok else None)
+    out = gen(model, prompt_ids, max_new_tokens=args.tokens,
+              seq_len=args.seq_len, device=dev,
+              eot_token=50256 if tok else None)
     dt = time.perf_counter() - t0
     n_new = len(out) - len(prompt_ids)
-    print(f"{n_new} tokens in {dt:.2f}s ({n_new / dt:.2f} tok/s, "
-          f"full-sequence recompute)")
+    mode = "kv-cache" if args.kv_cache else "full-sequence recompute"
+    print(f"{n_new} tokens in {dt:.2f}s ({n_new / dt:.2f} tok/s, {mode})")
     print(tok.decode(out) if tok else out)
 
 

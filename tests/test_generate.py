@@ -4,7 +4,7 @@ randomly initialized GPT."""
 import torch
 
 from tnn_amd.nn import LayerBuilder
-from tnn_amd.models.generate import generate
+from tnn_amd.models.generate import generate, generate_cached
 
 
 def _tiny_gpt(vocab=64, dim=32, seq=16):
@@ -33,3 +33,31 @@ def test_generate_window_clamps():
     out = generate(m, list(range(6)), max_new_tokens=6, seq_len=8,
                    eot_token=None)
     assert len(out) == 12  # windowed recompute keeps going past seq_len
+
+
+def test_generate_cached_matches_recompute():
+    torch.manual_seed(0)
+    m = _tiny_gpt(seq=32)
+    ref = generate(m, [1, 2, 3], max_new_tokens=10, seq_len=32, eot_token=None)
+    fast = generate_cached(m, [1, 2, 3], max_new_tokens=10, seq_len=32,
+                           eot_token=None)
+    assert fast == ref
+    # cache cleared afterwards: training-mode forward still works
+    y = m(torch.randint(0, 64, (2, 16)))
+    assert y.shape == (2, 16, 64)
+
+
+def test_generate_cached_flash_block_cpu():
+    torch.manual_seed(1)
+    from tnn_amd.nn import LayerBuilder
+    m = (LayerBuilder((16,))
+         .embedding(64, 32, "tok")
+         .positional_embedding(16, "pos")
+         .gpt_block(4, 2, flash=True, name="b0")
+         .layernorm(name="ln_f")
+         .dense(64, True, "head")
+         .build("tiny_flash_gpt"))
+    ref = generate(m, [5, 6], max_new_tokens=6, seq_len=16, eot_token=None)
+    fast = generate_cached(m, [5, 6], max_new_tokens=6, seq_len=16,
+                           eot_token=None)
+    assert fast == ref

@@ -180,13 +180,42 @@ class _MHABase(Layer):
         self.bv = nn.Parameter(torch.zeros(dim, dtype=dtype))
         self.bo = nn.Parameter(torch.zeros(dim, dtype=dtype))
 
+    _kv_cache = None  # plain attr: (k, v) [B,H,T,D]; None = training mode
+
+    def reset_cache(self):
+        self._kv_cache = None
+
+    def enable_cache(self):
+        self._kv_cache = (None, None)
+
     def _project(self, x):
         b, s, d = x.shape
         h, hd = self.num_heads, self.head_dim
         q = ops.linear(x, self.wq, self.bq).view(b, s, h, hd).transpose(1, 2)
         k = ops.linear(x, self.wk, self.bk).view(b, s, h, hd).transpose(1, 2)
         v = ops.linear(x, self.wv, self.bv).view(b, s, h, hd).transpose(1, 2)
+        if self._kv_cache is not None:
+            ck, cv = self._kv_cache
+            if ck is not None:
+                k = torch.cat([ck, k], dim=2)
+                v = torch.cat([cv, v], dim=2)
+            self._kv_cache = (k.detach(), v.detach())
         return q, k, v
+
+    def _cached_attention(self, q, k, v):
+        """Decode-path attention: q covers the s_new newest positions of
+        the k/v sequence; causal within the suffix, full over the prefix
+        (the KV-cache fast path the reference lacks —
+        examples/gpt2_inference.cpp recomputes the full sequence)."""
+        t, s_new = k.shape[-2], q.shape[-2]
+        scale = self.head_dim ** -0.5
+        scores = torch.matmul(q.float(), k.float().transpose(-1, -2)) * scale
+        if self.causal and s_new > 1:
+            pos_q = torch.arange(t - s_new, t, device=q.device).unsqueeze(-1)
+            pos_k = torch.arange(t, device=q.device)
+            scores = scores.masked_fill(pos_k > pos_q, float("-inf"))
+        p = torch.softmax(scores, dim=-1).to(v.dtype)
+        return torch.matmul(p, v)
 
     def _merge(self, o, b, s):
         o = o.transpose(1, 2).reshape(b, s, self.dim)
@@ -210,6 +239,9 @@ class AttentionBlock(_MHABase):
     def forward(self, x):
         b, s, _ = x.shape
         q, k, v = self._project(x)
+        if self._kv_cache is not None:
+            o = self._cached_attention(q, k, v)
+            return self._merge(o, b, s)
         scale = self.head_dim ** -0.5
         scores = torch.matmul(q.float(), k.float().transpose(-1, -2)) * scale
         if self.causal:
@@ -228,6 +260,9 @@ class FlashAttentionBlock(_MHABase):
     def forward(self, x):
         b, s, _ = x.shape
         q, k, v = self._project(x)
+        if self._kv_cache is not None:
+            o = self._cached_attention(q, k, v)
+            return self._merge(o, b, s)
         o = ops.attention(q.contiguous(), k.contiguous(), v.contiguous(),
                           causal=self.causal)
         return self._merge(o, b, s)

@@ -329,9 +329,12 @@ class PositionalEmbedding(Layer):
         self.max_len, self.dim = max_len, dim
         self.weight = nn.Parameter(torch.randn(max_len, dim, dtype=dtype) * 0.02)
 
+    _pos_offset = 0  # set by cached decode (models/generate.py)
+
     def forward(self, x):
         s = x.shape[-2]
-        return x + self.weight[:s]
+        off = self._pos_offset
+        return x + self.weight[off:off + s]
 
     def extra_config(self):
         return {"max_len": self.max_len, "dim": self.dim}
