@@ -54,7 +54,7 @@ def generate_cached(model: Sequential, prompt_ids: List[int],
     attns = [m for m in model.modules() if isinstance(m, _MHABase)]
     poss = [m for m in model.modules() if isinstance(m, PositionalEmbedding)]
     for a in attns:
-        a.enable_cache()
+        a.enable_cache(max_len=seq_len)
     try:
         ids = list(prompt_ids)
         x = torch.tensor([ids], dtype=torch.int64, device=device)

@@ -180,13 +180,15 @@ class _MHABase(Layer):
         self.bv = nn.Parameter(torch.zeros(dim, dtype=dtype))
         self.bo = nn.Parameter(torch.zeros(dim, dtype=dtype))
 
-    _kv_cache = None  # plain attr: (k, v) [B,H,T,D]; None = training mode
+    _kv_cache = None  # dict cache state; None = training mode
 
     def reset_cache(self):
         self._kv_cache = None
 
-    def enable_cache(self):
-        self._kv_cache = (None, None)
+    def enable_cache(self, max_len: int = 1024):
+        # buffers allocated lazily at prefill (batch size unknown here);
+        # appends are in-place slice writes, never torch.cat reallocation
+        self._kv_cache = {"k": None, "v": None, "len": 0, "cap": max_len}
 
     def _project(self, x):
         b, s, d = x.shape
@@ -194,12 +196,18 @@ class _MHABase(Layer):
         q = ops.linear(x, self.wq, self.bq).view(b, s, h, hd).transpose(1, 2)
         k = ops.linear(x, self.wk, self.bk).view(b, s, h, hd).transpose(1, 2)
         v = ops.linear(x, self.wv, self.bv).view(b, s, h, hd).transpose(1, 2)
-        if self._kv_cache is not None:
-            ck, cv = self._kv_cache
-            if ck is not None:
-                k = torch.cat([ck, k], dim=2)
-                v = torch.cat([cv, v], dim=2)
-            self._kv_cache = (k.detach(), v.detach())
+        c = self._kv_cache
+        if c is not None:
+            if c["k"] is None:
+                c["k"] = torch.empty(b, h, c["cap"], hd, device=x.device,
+                                     dtype=k.dtype)
+                c["v"] = torch.empty_like(c["k"])
+            t0, t1 = c["len"], c["len"] + s
+            c["k"][:, :, t0:t1] = k.detach()
+            c["v"][:, :, t0:t1] = v.detach()
+            c["len"] = t1
+            k = c["k"][:, :, :t1]
+            v = c["v"][:, :, :t1]
         return q, k, v
 
     def _cached_attention(self, q, k, v):
