@@ -65,6 +65,10 @@ def main():
               f"loss {loss.item():.3f})")
 
     # ---- greedy decode (reference gpt2_inference loop) ----
+    # note: at batch 1 and short sequences decode is kernel-launch-bound on
+    # a 2.5PF GPU, so the full-recompute loop (one big batch of launches)
+    # can beat the KV cache (many tiny launches); the cache wins as the
+    # sequence grows. hipGraph capture of the decode step is future work.
     model.eval()
     for name, fn in [("recompute (reference parity)", generate),
                      ("kv-cache", generate_cached)]:
