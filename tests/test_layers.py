@@ -121,3 +121,15 @@ def test_residual_block_shapes():
     b.eval()
     assert b(x).shape == (2, 4, 4, 32)
     assert b.output_shape((8, 8, 16)) == (4, 4, 32)
+
+
+def test_nary_and_mbroadcast():
+    from tnn_amd.nn import NAry, MBroadcast
+    a, b = torch.randn(3, 4), torch.randn(3, 4)
+    assert torch.allclose(NAry("add")(a, b), a + b)
+    assert torch.allclose(NAry("mul")([a, b]), a * b)
+    x = torch.randn(3, 4, requires_grad=True)
+    outs = MBroadcast(3)(x)
+    assert len(outs) == 3
+    sum(o.sum() for o in outs).backward()
+    assert torch.allclose(x.grad, torch.full_like(x, 3.0))

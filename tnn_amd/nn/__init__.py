@@ -13,6 +13,7 @@ from .layers import (
     Dense, Conv2D, BatchNorm, LayerNorm, GroupNorm,
     MaxPool2D, AvgPool2D, Dropout, Activation, Flatten,
     Embedding, PositionalEmbedding, ClassToken, Identity, Transpose, Slice,
+    NAry, MBroadcast,
 )
 from .blocks import Sequential, ResidualBlock, MSequential, AttentionBlock, FlashAttentionBlock, GPTBlock
 from .builder import LayerBuilder
@@ -29,6 +30,7 @@ __all__ = [
     "Dense", "Conv2D", "BatchNorm", "LayerNorm", "GroupNorm",
     "MaxPool2D", "AvgPool2D", "Dropout", "Activation", "Flatten",
     "Embedding", "PositionalEmbedding", "ClassToken", "Identity", "Transpose", "Slice",
+    "NAry", "MBroadcast",
     "Sequential", "ResidualBlock", "MSequential", "AttentionBlock",
     "FlashAttentionBlock", "GPTBlock", "LayerBuilder",
     "Loss", "CrossEntropyLoss", "MSELoss", "MAELoss", "HuberLoss", "loss_from_config",

@@ -360,3 +360,48 @@ class ClassToken(Layer):
 
     def extra_config(self):
         return {"dim": self.dim}
+
+
+@register_layer("nary")
+class NAry(Layer):
+    """Elementwise join of multiple inputs (reference n_ary_ops.cu:32-163:
+    add/sub/mul/div joins). Takes a list/tuple of tensors."""
+
+    OPS = {"add": torch.add, "sub": torch.sub, "mul": torch.mul,
+           "div": torch.div}
+
+    def __init__(self, op: str = "add", name: str = "nary",
+                 dtype: torch.dtype = torch.float32):
+        super().__init__(name, dtype)
+        self.op = op
+
+    def forward(self, *xs):
+        if len(xs) == 1 and isinstance(xs[0], (list, tuple)):
+            xs = xs[0]
+        fn = self.OPS[self.op]
+        y = xs[0]
+        for x in xs[1:]:
+            y = fn(y, x)
+        return y
+
+    def extra_config(self):
+        return {"op": self.op}
+
+
+@register_layer("mbroadcast")
+class MBroadcast(Layer):
+    """Fan-out one input to N consumers; backward sums the incoming grads
+    (reference include/nn/layers_impl/mbroadcast_layer.hpp:12). With
+    autograd the fan-out is just returning the tensor N times — grads sum
+    automatically."""
+
+    def __init__(self, n: int = 2, name: str = "mbroadcast",
+                 dtype: torch.dtype = torch.float32):
+        super().__init__(name, dtype)
+        self.n = n
+
+    def forward(self, x):
+        return tuple(x for _ in range(self.n))
+
+    def extra_config(self):
+        return {"n": self.n}
