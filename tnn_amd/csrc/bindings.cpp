@@ -15,6 +15,15 @@ static hipStream_t cur_stream() {
   return at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
 }
 
+// 16-byte zero page for glds source redirection (invalid lanes load
+// zeros instead of being exec-masked, which would leave stale LDS)
+static void* zero_page(const at::Tensor& like) {
+  static at::Tensor z;
+  if (!z.defined() || z.device() != like.device())
+    z = at::zeros({16}, like.options().dtype(at::kFloat));
+  return z.data_ptr();
+}
+
 static DT dt_of(const at::Tensor& t) {
   if (t.scalar_type() == at::kFloat) return DT::F32;
   if (t.scalar_type() == at::kBFloat16) return DT::BF16;
@@ -190,8 +199,8 @@ at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
                        ph, pw);
   auto y = at::empty({cs.N, cs.OH, cs.OW, cs.Cout}, x.options());
   const void* bp = bias.has_value() ? bias->data_ptr() : nullptr;
-  conv2d_fwd_launch(dt_of(x), x.data_ptr(), w.data_ptr(), bp, y.data_ptr(), cs,
-                    relu, cur_stream());
+  conv2d_fwd_launch(dt_of(x), x.data_ptr(), w.data_ptr(), bp, y.data_ptr(),
+                    zero_page(x), cs, relu, cur_stream());
   return y;
 }
 
@@ -222,7 +231,7 @@ at::Tensor conv2d_dgrad(const at::Tensor& dy, const at::Tensor& w, int64_t H,
                      cur_stream());
   auto dx = at::empty({cs.N, H, W, Cin}, dy.options());
   conv2d_dgrad_launch(dt_of(dy), dy.data_ptr(), w_t.data_ptr(), dx.data_ptr(),
-                      cs, cur_stream());
+                      zero_page(dy), cs, cur_stream());
   return dx;
 }
 

@@ -38,12 +38,13 @@ DEV int imod(int x, const IDiv& f) {
 }
 
 // ---------------------------------------------------------------------------
-template <typename T, bool POW2>
+template <typename T, bool POW2, bool GLDS>
 __launch_bounds__(THREADS)
 __global__ void k_conv_fwd(const T* __restrict__ X, const T* __restrict__ Wt,
                            const float* __restrict__ bias_f32,
                            const T* __restrict__ bias_t, T* __restrict__ Y,
-                           ConvShape cs, int act_kind) {
+                           const T* __restrict__ zero16, ConvShape cs,
+                           int act_kind) {
   constexpr int V = 16 / sizeof(T);
   __shared__ alignas(16) T As[BM * BK];
   __shared__ alignas(16) T Bs[BN * BK];
@@ -56,7 +57,29 @@ __global__ void k_conv_fwd(const T* __restrict__ X, const T* __restrict__ Wt,
   f32x4 acc[FM][FN] = {};
   using VecT = Pack16<T>;
 
+  // fused-im2col source address for a 16B chunk (row, kk elem), or the
+  // zero page when the chunk is out of image/padding/tail range; only
+  // called on the GLDS path where Cin % V == 0 guarantees one (kh,kw)
+  // slice per chunk.
+  auto a_src = [&](int k0, int rl, int kk) -> const T* {
+    int gm = m0 + rl, gk = k0 + kk;
+    if (gm >= M || gk >= K) return zero16;
+    int n = idiv<POW2>(gm, cs.d_ohow);
+    int rem = gm - n * (cs.OH * cs.OW);
+    int oh = idiv<POW2>(rem, cs.d_ow), ow = rem - oh * cs.OW;
+    int ci = imod<POW2>(gk, cs.d_cin);
+    int kidx = idiv<POW2>(gk, cs.d_cin);
+    int kw = kidx % cs.KW, kh = kidx / cs.KW;
+    int ih = oh * cs.SH - cs.PH + kh;
+    int iw = ow * cs.SW - cs.PW + kw;
+    if (ih < 0 || ih >= cs.H || iw < 0 || iw >= cs.W) return zero16;
+    return &X[(((int64_t)n * cs.H + ih) * cs.W + iw) * cs.Cin + ci];
+  };
+
   for (int k0 = 0; k0 < K; k0 += BK) {
+    if constexpr (GLDS) {
+      glds_stage_a<T>(As, wc, [&](int rl, int kk) { return a_src(k0, rl, kk); });
+    } else {
     // ---- stage A: im2col gather ----
 #pragma unroll
     for (int c = threadIdx.x; c < BM * (BK / V); c += THREADS) {
@@ -95,6 +118,7 @@ __global__ void k_conv_fwd(const T* __restrict__ X, const T* __restrict__ Wt,
         }
       }
       *(VecT*)&As[lds_off<T>(row, kk)] = v;
+    }
     }
     // ---- stage B (weights [K, Cout]) -> Bs[n][k] scatter-transpose ----
 #pragma unroll
@@ -137,10 +161,11 @@ __global__ void k_conv_fwd(const T* __restrict__ X, const T* __restrict__ Wt,
 // with oh = (ih+PH-kh)/SH when divisible. Wt here is the transposed weight
 // [KH,KW,Cout,Cin] so B rows are k=(kh,kw,co) with ci contiguous.
 // ---------------------------------------------------------------------------
-template <typename T, bool POW2>
+template <typename T, bool POW2, bool GLDS>
 __launch_bounds__(THREADS)
 __global__ void k_conv_dgrad(const T* __restrict__ DY, const T* __restrict__ WT,
-                             T* __restrict__ DX, ConvShape cs) {
+                             T* __restrict__ DX, const T* __restrict__ zero16,
+                             ConvShape cs) {
   constexpr int V = 16 / sizeof(T);
   __shared__ alignas(16) T As[BM * BK];
   __shared__ alignas(16) T Bs[BN * BK];
@@ -153,7 +178,26 @@ __global__ void k_conv_dgrad(const T* __restrict__ DY, const T* __restrict__ WT,
   f32x4 acc[FM][FN] = {};
   using VecT = Pack16<T>;
 
+  auto a_src = [&](int k0, int rl, int kk) -> const T* {
+    int gm = m0 + rl, gk = k0 + kk;
+    if (gm >= M || gk >= K) return zero16;
+    int n = idiv<POW2>(gm, cs.d_hw);
+    int rem = gm - n * (cs.H * cs.W);
+    int ih = idiv<POW2>(rem, cs.d_w), iw = rem - ih * cs.W;
+    int co = imod<POW2>(gk, cs.d_cout);
+    int kidx = idiv<POW2>(gk, cs.d_cout);
+    int kw = kidx % cs.KW, kh = kidx / cs.KW;
+    int th = ih + cs.PH - kh, tw = iw + cs.PW - kw;
+    if (th < 0 || tw < 0 || th % cs.SH || tw % cs.SW) return zero16;
+    int oh = th / cs.SH, ow = tw / cs.SW;
+    if (oh >= cs.OH || ow >= cs.OW) return zero16;
+    return &DY[(((int64_t)n * cs.OH + oh) * cs.OW + ow) * cs.Cout + co];
+  };
+
   for (int k0 = 0; k0 < K; k0 += BK) {
+    if constexpr (GLDS) {
+      glds_stage_a<T>(As, wc, [&](int rl, int kk) { return a_src(k0, rl, kk); });
+    } else {
     // ---- stage A: gather dy with stride/padding inversion ----
 #pragma unroll
     for (int c = threadIdx.x; c < BM * (BK / V); c += THREADS) {
@@ -197,6 +241,7 @@ __global__ void k_conv_dgrad(const T* __restrict__ DY, const T* __restrict__ WT,
         }
       }
       *(VecT*)&As[lds_off<T>(row, kk)] = v;
+    }
     }
     // ---- stage B (w_t [K, Cin]) -> Bs[ci][k] ----
 #pragma unroll
@@ -349,37 +394,55 @@ static bool all_pow2(const ConvShape& cs) {
 }
 
 void conv2d_fwd_launch(DT dt, const void* x, const void* w, const void* bias,
-                       void* y, const ConvShape& cs, bool relu, hipStream_t s) {
+                       void* y, const void* zero16, const ConvShape& cs,
+                       bool relu, hipStream_t s) {
   int M = cs.N * cs.OH * cs.OW;
   dim3 grid(ceil_div(M, BM), ceil_div(cs.Cout, BN));
   int act = relu ? ACT_RELU : ACT_LINEAR;
   bool p2 = all_pow2(cs);
   if (dt == DT::F32) {
-    auto kern = p2 ? k_conv_fwd<float, true> : k_conv_fwd<float, false>;
+    bool g = p2 && cs.Cin % 4 == 0 && (((uintptr_t)x & 15) == 0);
+    auto kern = g ? k_conv_fwd<float, true, true>
+                  : (p2 ? k_conv_fwd<float, true, false>
+                        : k_conv_fwd<float, false, false>);
     hipLaunchKernelGGL(kern, grid, dim3(THREADS), 0, s,
                        (const float*)x, (const float*)w, (const float*)bias,
-                       (const float*)nullptr, (float*)y, cs, act);
+                       (const float*)nullptr, (float*)y, (const float*)zero16,
+                       cs, act);
   } else {
-    auto kern = p2 ? k_conv_fwd<bf16, true> : k_conv_fwd<bf16, false>;
+    bool g = p2 && cs.Cin % 8 == 0 && (((uintptr_t)x & 15) == 0);
+    auto kern = g ? k_conv_fwd<bf16, true, true>
+                  : (p2 ? k_conv_fwd<bf16, true, false>
+                        : k_conv_fwd<bf16, false, false>);
     hipLaunchKernelGGL(kern, grid, dim3(THREADS), 0, s,
                        (const bf16*)x, (const bf16*)w, (const float*)nullptr,
-                       (const bf16*)bias, (bf16*)y, cs, act);
+                       (const bf16*)bias, (bf16*)y, (const bf16*)zero16, cs,
+                       act);
   }
 }
 
 void conv2d_dgrad_launch(DT dt, const void* dy, const void* w_t, void* dx,
-                         const ConvShape& cs, hipStream_t s) {
+                         const void* zero16, const ConvShape& cs,
+                         hipStream_t s) {
   int M = cs.N * cs.H * cs.W;
   dim3 grid(ceil_div(M, BM), ceil_div(cs.Cin, BN));
   bool p2 = all_pow2(cs);
   if (dt == DT::F32) {
-    auto kern = p2 ? k_conv_dgrad<float, true> : k_conv_dgrad<float, false>;
+    bool g = p2 && cs.Cout % 4 == 0 && (((uintptr_t)dy & 15) == 0);
+    auto kern = g ? k_conv_dgrad<float, true, true>
+                  : (p2 ? k_conv_dgrad<float, true, false>
+                        : k_conv_dgrad<float, false, false>);
     hipLaunchKernelGGL(kern, grid, dim3(THREADS), 0, s,
-                       (const float*)dy, (const float*)w_t, (float*)dx, cs);
+                       (const float*)dy, (const float*)w_t, (float*)dx,
+                       (const float*)zero16, cs);
   } else {
-    auto kern = p2 ? k_conv_dgrad<bf16, true> : k_conv_dgrad<bf16, false>;
+    bool g = p2 && cs.Cout % 8 == 0 && (((uintptr_t)dy & 15) == 0);
+    auto kern = g ? k_conv_dgrad<bf16, true, true>
+                  : (p2 ? k_conv_dgrad<bf16, true, false>
+                        : k_conv_dgrad<bf16, false, false>);
     hipLaunchKernelGGL(kern, grid, dim3(THREADS), 0, s,
-                       (const bf16*)dy, (const bf16*)w_t, (bf16*)dx, cs);
+                       (const bf16*)dy, (const bf16*)w_t, (bf16*)dx,
+                       (const bf16*)zero16, cs);
   }
 }
 

@@ -66,9 +66,11 @@ struct ConvShape {
   }
 };
 void conv2d_fwd_launch(DT dt, const void* x, const void* w, const void* bias,
-                       void* y, const ConvShape& cs, bool relu, hipStream_t s);
+                       void* y, const void* zero16, const ConvShape& cs,
+                       bool relu, hipStream_t s);
 void conv2d_dgrad_launch(DT dt, const void* dy, const void* w_t, void* dx,
-                         const ConvShape& cs, hipStream_t s);
+                         const void* zero16, const ConvShape& cs,
+                         hipStream_t s);
 int conv2d_wgrad_zsplits(const ConvShape& cs);
 void conv2d_wgrad_launch(DT dt, const void* x, const void* dy, float* dw_f32,
                          float* ws, int z, const ConvShape& cs,

@@ -51,6 +51,7 @@ DEV int lds_off(int row, int col) {
   return row * BK + byte / (int)sizeof(T);
 }
 
+
 template <typename T> struct LDSBytes {
   static constexpr int A = BM * BK * sizeof(T);
   static constexpr int B = BN * BK * sizeof(T);
@@ -66,6 +67,39 @@ struct WaveCoord {
     wcol0 = (wid % WAVES_N) * WN;
   }
 };
+
+// ---- async global->LDS staging for the A tile (gfx950 global_load_lds) ---
+// Writes the whole [BM][BK] swizzled image with 16B-per-lane DMA: each
+// wave issues (BM*BK*sizeof(T))/4096 instructions, each filling a 1 KiB
+// lane-linear LDS region. The swizzle therefore moves to the SOURCE
+// address (guide rule 21): each lane asks AddrFn for the element that
+// belongs at its linear LDS slot. Invalid lanes (image padding, M/K
+// tails) point their source at a 16-byte zero page — no exec masking, no
+// pre-zeroing pass (a masked glds lane would leave stale LDS bytes).
+// AddrFn: (row_in_tile, col_elem) -> const T* (16B-aligned) or zero16.
+template <typename T, typename AddrFn>
+DEV void glds_stage_a(T* As, const WaveCoord& w, AddrFn&& addr) {
+  constexpr int ROWB = BK * (int)sizeof(T);      // bytes per image row
+  constexpr int RPK = 1024 / ROWB;               // rows per 1 KiB region
+  constexpr int LPR = ROWB / 16;                 // lanes per row
+  constexpr int NREG = BM * ROWB / 1024;         // 1 KiB regions in the tile
+  constexpr int NPW = NREG / 4;                  // regions per wave
+#pragma unroll
+  for (int i = 0; i < NPW; ++i) {
+    const int j = w.wid * NPW + i;
+    const int rl = RPK * j + w.lane / LPR;
+    const int byte_in_row = (w.lane % LPR) * 16;
+    const int kk = (byte_in_row ^ ((((rl >> 3) ^ rl) & 7) << 4)) /
+                   (int)sizeof(T);
+    const T* src = addr(rl, kk);
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)src,
+        (__attribute__((address_space(3))) unsigned int*)
+            &As[j * (1024 / (int)sizeof(T))],
+        16, 0, 0);
+  }
+}
+
 
 // ---- MFMA tile compute: acc[FM][FN] += A_tile * B_tile^T-stored -----------
 DEV void mfma_compute_tile(const bf16* As, const bf16* Bs, const WaveCoord& w,
