@@ -107,8 +107,9 @@ at::Tensor gemm(const at::Tensor& a, const at::Tensor& b,
   int M = a.size(0), K = a.size(1), N = b.size(1);
   auto c = at::empty({M, N}, a.options());
   const void* bp = bias.has_value() ? bias->data_ptr() : nullptr;
-  gemm_launch(dt_of(a), a.data_ptr(), b.data_ptr(), bp, c.data_ptr(), M, N, K,
-              /*trans_b=*/false, (int)act_kind, cur_stream());
+  gemm_launch(dt_of(a), a.data_ptr(), b.data_ptr(), bp, c.data_ptr(),
+              zero_page(a), M, N, K, /*trans_b=*/false, (int)act_kind,
+              cur_stream());
   return c;
 }
 
@@ -120,8 +121,8 @@ at::Tensor gemm_nt(const at::Tensor& a, const at::Tensor& b) {
               "gemm_nt shapes ", a.sizes(), " @ ", b.sizes(), "^T");
   int M = a.size(0), K = a.size(1), N = b.size(0);
   auto c = at::empty({M, N}, a.options());
-  gemm_launch(dt_of(a), a.data_ptr(), b.data_ptr(), nullptr, c.data_ptr(), M, N,
-              K, /*trans_b=*/true, 0, cur_stream());
+  gemm_launch(dt_of(a), a.data_ptr(), b.data_ptr(), nullptr, c.data_ptr(),
+              zero_page(a), M, N, K, /*trans_b=*/true, 0, cur_stream());
   return c;
 }
 

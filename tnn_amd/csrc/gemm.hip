@@ -18,12 +18,13 @@ using namespace tile;
 //   TRANS_B = true:  B is [N, K] row-major (staged directly; C = A @ B^T)
 // ---------------------------------------------------------------------------
 
-template <typename T, bool TRANS_B>
+template <typename T, bool TRANS_B, bool GLDS>
 __launch_bounds__(THREADS)
 __global__ void k_gemm(const T* __restrict__ A, const T* __restrict__ B,
                        const float* __restrict__ bias_f32,
                        const T* __restrict__ bias_t, T* __restrict__ C,
-                       int M, int N, int K, int act_kind) {
+                       const T* __restrict__ zero16, int M, int N, int K,
+                       int act_kind) {
   constexpr int V = 16 / sizeof(T);  // elems per 16B vector
   __shared__ alignas(16) T As[BM * BK];
   __shared__ alignas(16) T Bs[BN * BK];
@@ -36,6 +37,14 @@ __global__ void k_gemm(const T* __restrict__ A, const T* __restrict__ B,
   using VecT = Pack16<T>;
 
   for (int k0 = 0; k0 < K; k0 += BK) {
+    if constexpr (GLDS) {
+      // K % V == 0 and A 16B-aligned guaranteed by the launcher
+      glds_stage_a<T>(As, wc, [&](int rl, int kk) -> const T* {
+        int gm = m0 + rl, gk = k0 + kk;
+        if (gm >= M || gk >= K) return zero16;
+        return &A[(int64_t)gm * K + gk];
+      });
+    } else {
     // ---- stage A: [BM][BK], vectorized, zero-filled at edges ----
 #pragma unroll
     for (int c = threadIdx.x; c < BM * (BK / V); c += THREADS) {
@@ -55,8 +64,16 @@ __global__ void k_gemm(const T* __restrict__ A, const T* __restrict__ B,
       }
       *(VecT*)&As[lds_off<T>(row, kk)] = v;
     }
+    }
     // ---- stage B into Bs[n][k] ----
-    if constexpr (TRANS_B) {
+    if constexpr (TRANS_B && GLDS) {
+      // B[N,K] rows are already the image rows: same DMA form as A
+      glds_stage<T, BN>(Bs, wc, [&](int rl, int kk) -> const T* {
+        int gn = n0 + rl, gk = k0 + kk;
+        if (gn >= N || gk >= K) return zero16;
+        return &B[(int64_t)gn * K + gk];
+      });
+    } else if constexpr (TRANS_B) {
       // B[N,K]: rows are output columns; contiguous copy
 #pragma unroll
       for (int c = threadIdx.x; c < BN * (BK / V); c += THREADS) {
@@ -221,20 +238,28 @@ void splitk_reduce_launch(const float* ws, float* out, int z, int64_t n,
 // ---------------------------------------------------------------------------
 
 void gemm_launch(DT dt, const void* a, const void* b, const void* bias,
-                 void* c, int M, int N, int K, bool trans_b, int act_kind,
-                 hipStream_t s) {
+                 void* c, const void* zero16, int M, int N, int K,
+                 bool trans_b, int act_kind, hipStream_t s) {
   dim3 grid(ceil_div(M, BM), ceil_div(N, BN));
   dim3 blk(THREADS);
   if (dt == DT::F32) {
-    auto kern = trans_b ? k_gemm<float, true> : k_gemm<float, false>;
+    bool g = K % 4 == 0 && (((uintptr_t)a & 15) == 0);
+    auto kern = trans_b ? (g ? k_gemm<float, true, true>
+                             : k_gemm<float, true, false>)
+                        : (g ? k_gemm<float, false, true>
+                             : k_gemm<float, false, false>);
     hipLaunchKernelGGL(kern, grid, blk, 0, s, (const float*)a, (const float*)b,
-                       (const float*)bias, (const float*)nullptr, (float*)c, M,
-                       N, K, act_kind);
+                       (const float*)bias, (const float*)nullptr, (float*)c,
+                       (const float*)zero16, M, N, K, act_kind);
   } else {
-    auto kern = trans_b ? k_gemm<bf16, true> : k_gemm<bf16, false>;
+    bool g = K % 8 == 0 && (((uintptr_t)a & 15) == 0);
+    auto kern = trans_b ? (g ? k_gemm<bf16, true, true>
+                             : k_gemm<bf16, true, false>)
+                        : (g ? k_gemm<bf16, false, true>
+                             : k_gemm<bf16, false, false>);
     hipLaunchKernelGGL(kern, grid, blk, 0, s, (const bf16*)a, (const bf16*)b,
-                       (const float*)nullptr, (const bf16*)bias, (bf16*)c, M, N,
-                       K, act_kind);
+                       (const float*)nullptr, (const bf16*)bias, (bf16*)c,
+                       (const bf16*)zero16, M, N, K, act_kind);
   }
 }
 

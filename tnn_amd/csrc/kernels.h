@@ -28,8 +28,8 @@ void cast_f32_launch(DT dt_out, const float* x, void* y, int64_t n,
 
 // ---- gemm.hip --------------------------------------------------------------
 void gemm_launch(DT dt, const void* a, const void* b, const void* bias,
-                 void* c, int M, int N, int K, bool trans_b, int act_kind,
-                 hipStream_t s);
+                 void* c, const void* zero16, int M, int N, int K,
+                 bool trans_b, int act_kind, hipStream_t s);
 int gemm_tn_zsplits(int M, int N, int K);
 void gemm_tn_launch(DT dt, const void* a, const void* b, float* c_f32,
                     float* ws, int z, int M, int N, int K, hipStream_t s);

@@ -77,12 +77,12 @@ struct WaveCoord {
 // tails) point their source at a 16-byte zero page — no exec masking, no
 // pre-zeroing pass (a masked glds lane would leave stale LDS bytes).
 // AddrFn: (row_in_tile, col_elem) -> const T* (16B-aligned) or zero16.
-template <typename T, typename AddrFn>
-DEV void glds_stage_a(T* As, const WaveCoord& w, AddrFn&& addr) {
+template <typename T, int ROWS, typename AddrFn>
+DEV void glds_stage(T* As, const WaveCoord& w, AddrFn&& addr) {
   constexpr int ROWB = BK * (int)sizeof(T);      // bytes per image row
   constexpr int RPK = 1024 / ROWB;               // rows per 1 KiB region
   constexpr int LPR = ROWB / 16;                 // lanes per row
-  constexpr int NREG = BM * ROWB / 1024;         // 1 KiB regions in the tile
+  constexpr int NREG = ROWS * ROWB / 1024;       // 1 KiB regions in the tile
   constexpr int NPW = NREG / 4;                  // regions per wave
 #pragma unroll
   for (int i = 0; i < NPW; ++i) {
@@ -98,6 +98,11 @@ DEV void glds_stage_a(T* As, const WaveCoord& w, AddrFn&& addr) {
             &As[j * (1024 / (int)sizeof(T))],
         16, 0, 0);
   }
+}
+
+template <typename T, typename AddrFn>
+DEV void glds_stage_a(T* As, const WaveCoord& w, AddrFn&& addr) {
+  glds_stage<T, BM>(As, w, addr);
 }
 
 
