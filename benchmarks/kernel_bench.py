@@ -96,8 +96,25 @@ def bench_ce(dtype=torch.bfloat16):
            bytes_=logits.numel() * logits.element_size())
 
 
+def bench_attn():
+    for B, H, S, D, causal in [(8, 12, 512, 64, True), (8, 12, 1024, 64, True),
+                               (2, 16, 2048, 128, False)]:
+        q = torch.randn(B, H, S, D, dtype=torch.bfloat16, device=DEV)
+        k = torch.randn_like(q)
+        v = torch.randn_like(q)
+        fl = 4.0 * B * H * S * S * D * (0.5 if causal else 1.0)
+        secs = timeit(lambda: ext.attn_fwd(q, k, v, causal))
+        report("attn_fwd", f"B{B} H{H} S{S} D{D} causal={causal}", secs, fl)
+        o, lse = ext.attn_fwd(q, k, v, causal)
+        do = torch.randn_like(q)
+        secs = timeit(lambda: ext.attn_bwd(q, k, v, o, do, lse, causal),
+                      iters=20)
+        report("attn_bwd", f"B{B} H{H} S{S} D{D} causal={causal}", secs,
+               2.5 * fl)
+
+
 ALL = {"gemm": bench_gemm, "conv": bench_conv, "bn": bench_bn,
-       "colsum": bench_colsum, "ce": bench_ce}
+       "colsum": bench_colsum, "ce": bench_ce, "attn": bench_attn}
 
 if __name__ == "__main__":
     which = sys.argv[1:] or list(ALL)
