@@ -280,7 +280,7 @@ __global__ void k_conv_dgrad(const T* __restrict__ DY, const T* __restrict__ WT,
 // grid.z m-slices.
 // ---------------------------------------------------------------------------
 template <typename T, bool POW2>
-__launch_bounds__(THREADS)
+__launch_bounds__(THREADS, 3)  // cap VGPRs: unconstrained allocation hit 221-256 VGPR = 1-2 waves/SIMD
 __global__ void k_conv_wgrad(const T* __restrict__ X, const T* __restrict__ DY,
                              float* __restrict__ DW, ConvShape cs) {
   constexpr int V = 16 / sizeof(T);

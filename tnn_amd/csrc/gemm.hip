@@ -139,7 +139,7 @@ __global__ void k_gemm(const T* __restrict__ A, const T* __restrict__ B,
 // ---------------------------------------------------------------------------
 
 template <typename T>
-__launch_bounds__(THREADS)
+__launch_bounds__(THREADS, 3)  // cap VGPRs: unconstrained allocation hit 221-256 VGPR = 1-2 waves/SIMD
 __global__ void k_gemm_tn(const T* __restrict__ A, const T* __restrict__ B,
                           float* __restrict__ C, int M, int N, int K) {
   constexpr int V = 16 / sizeof(T);
