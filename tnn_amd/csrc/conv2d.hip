@@ -299,7 +299,7 @@ __global__ void k_conv_wgrad(const T* __restrict__ X, const T* __restrict__ DY,
 
   for (int k0 = m_begin; k0 < m_end; k0 += BK) {
     // ---- stage A^T: x rows ci-contiguous, scatter into As[ko][m] ----
-#pragma unroll
+#pragma unroll 1  // full unroll quadruples live address state (spills)
     for (int c = threadIdx.x; c < BK * (BM / V); c += THREADS) {
       int mm = c / (BM / V);
       int rr = (c % (BM / V)) * V;
@@ -341,7 +341,7 @@ __global__ void k_conv_wgrad(const T* __restrict__ X, const T* __restrict__ DY,
         As[lds_off<T>(rr + j, mm)] = v.e[j];
     }
     // ---- stage B: dy[m][co] -> Bs[co][m] ----
-#pragma unroll
+#pragma unroll 1
     for (int c = threadIdx.x; c < BK * (BN / V); c += THREADS) {
       int mm = c / (BN / V);
       int nn = (c % (BN / V)) * V;

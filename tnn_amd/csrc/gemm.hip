@@ -159,7 +159,7 @@ __global__ void k_gemm_tn(const T* __restrict__ A, const T* __restrict__ B,
 
   for (int k0 = m_begin; k0 < m_end; k0 += BK) {
     // ---- stage A^T: load A[m][r..r+V] (contiguous), scatter to As[r][m] ----
-#pragma unroll
+#pragma unroll 1  // full unroll quadruples live address state (spills)
     for (int c = threadIdx.x; c < BK * (BM / V); c += THREADS) {
       int mm = c / (BM / V);
       int rr = (c % (BM / V)) * V;
@@ -181,7 +181,7 @@ __global__ void k_gemm_tn(const T* __restrict__ A, const T* __restrict__ B,
         As[lds_off<T>(rr + j, mm)] = v.e[j];
     }
     // ---- stage B[m][n] -> Bs[n][m] ----
-#pragma unroll
+#pragma unroll 1
     for (int c = threadIdx.x; c < BK * (BN / V); c += THREADS) {
       int mm = c / (BN / V);
       int nn = (c % (BN / V)) * V;
