@@ -341,7 +341,7 @@ __global__ void k_conv_wgrad(const T* __restrict__ X, const T* __restrict__ DY,
         As[lds_off<T>(rr + j, mm)] = v.e[j];
     }
     // ---- stage B: dy[m][co] -> Bs[co][m] ----
-#pragma unroll 1
+#pragma unroll
     for (int c = threadIdx.x; c < BK * (BN / V); c += THREADS) {
       int mm = c / (BN / V);
       int nn = (c % (BN / V)) * V;

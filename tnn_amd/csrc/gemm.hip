@@ -95,7 +95,7 @@ __global__ void k_gemm(const T* __restrict__ A, const T* __restrict__ B,
       }
     } else {
       // B[K,N]: load n-contiguous vectors, scatter-transpose into LDS
-#pragma unroll
+#pragma unroll 1
       for (int c = threadIdx.x; c < BK * (BN / V); c += THREADS) {
         int kk = c / (BN / V);
         int nn = (c % (BN / V)) * V;
