@@ -19,7 +19,8 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from tnn_amd import models
 from tnn_amd.data import DataLoaderFactory
 from tnn_amd.nn import CrossEntropyLoss
-from tnn_amd.parallel import init_distributed, PipelineEngine
+from tnn_amd.parallel import (init_distributed, PipelineEngine,
+                              train_pipeline_model)
 from tnn_amd.utils import get_logger
 
 log = get_logger("pipeline_trainer")
@@ -35,6 +36,9 @@ def main():
     p.add_argument("--lr", type=float, default=1e-3)
     p.add_argument("--dtype", default="bf16" if torch.cuda.is_available()
                    else "fp32")
+    p.add_argument("--profile", default=None,
+                   help="write a merged chrome trace here (rank 0)")
+    p.add_argument("--snapshot-dir", default=None)
     args = p.parse_args()
 
     comm = init_distributed()
@@ -52,15 +56,18 @@ def main():
     loader = DataLoaderFactory.create(
         "synthetic_image", shape=(32, 32, 3), num_classes=100,
         num_samples=50 * args.batch_size, batch_size=args.batch_size, seed=7)
-    for epoch in range(args.epochs):
-        for i, (x, y) in enumerate(loader):
-            stats = engine.train_batch(x, y)
-            if engine.is_last and i % 10 == 0:
-                log.info("epoch %d batch %d loss %.4f acc %.3f", epoch, i,
-                         stats["loss"], stats["accuracy"])
-        vstats = engine.broadcast_stats(engine.eval_batch(x, y))
-        if comm.rank == 0:
-            log.info("epoch %d done, val-on-last-batch %s", epoch, vstats)
+    val_loader = DataLoaderFactory.create(
+        "synthetic_image", shape=(32, 32, 3), num_classes=100,
+        num_samples=10 * args.batch_size, batch_size=args.batch_size, seed=8)
+    if args.profile:
+        engine.start_profiling()
+    train_pipeline_model(engine, loader, val_loader, epochs=args.epochs,
+                         snapshot_dir=args.snapshot_dir)
+    if args.profile:
+        merged = engine.gather_profiles()
+        if merged is not None:
+            merged.export_chrome_trace(args.profile)
+            log.info("chrome trace written to %s", args.profile)
     comm.barrier()
     comm.destroy()
 

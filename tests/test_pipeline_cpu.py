@@ -87,6 +87,13 @@ def _pipeline_worker(rank, world, tmpdir):
         losses.append(stats["loss"])
     sd = {k: v.clone() for k, v in engine.state_dict().items()}
     torch.save((rank, losses, sd), os.path.join(tmpdir, f"result_{rank}.pt"))
+    # per-stage checkpoint round trip (reference Worker SAVE_TO_FILE)
+    from tnn_amd.parallel import save_stage_checkpoint, load_stage_checkpoint
+    save_stage_checkpoint(engine, os.path.join(tmpdir, "snap"))
+    header = load_stage_checkpoint(engine, os.path.join(tmpdir, "snap"))
+    assert header["rank"] == rank and header["world"] == world
+    for k, v in engine.state_dict().items():
+        assert torch.equal(v, sd[k]), k
     dist.barrier()
     dist.destroy_process_group()
 

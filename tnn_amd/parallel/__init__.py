@@ -13,9 +13,15 @@ from .comm import (init_distributed, is_initialized, rank, world_size,
 from .partitioner import (NaivePipelinePartitioner, WeightedPipelinePartitioner,
                           NaiveDataPartitioner, partition_model)
 from .pipeline import PipelineEngine
+from .train import (train_pipeline_epoch, validate_pipeline_epoch,
+                    train_pipeline_model, save_stage_checkpoint,
+                    load_stage_checkpoint)
 from .ddp import DataParallelEngine
 
 __all__ = ["init_distributed", "is_initialized", "rank", "world_size",
            "Communicator", "NaivePipelinePartitioner",
            "WeightedPipelinePartitioner", "NaiveDataPartitioner",
-           "partition_model", "PipelineEngine", "DataParallelEngine"]
+           "partition_model", "PipelineEngine", "DataParallelEngine",
+           "train_pipeline_epoch", "validate_pipeline_epoch",
+           "train_pipeline_model", "save_stage_checkpoint",
+           "load_stage_checkpoint"]
