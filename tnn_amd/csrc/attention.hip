@@ -44,17 +44,31 @@ DEV int aoff(int row, int col) {
 }
 
 // cooperative stage of a [ROWS][D] bf16 tile from global (row-major,
-// row stride D) into the swizzled LDS image; zero-fills rows >= nrows.
+// row stride D) into the swizzled LDS image via global_load_lds: lane-
+// linear 1KiB regions, swizzle applied to the per-lane source address,
+// rows >= nrows redirected to the zero page (same scheme as
+// tile::glds_stage; see tile_gemm.h).
 template <int ROWS, int D>
-DEV void stage_tile(const bf16* __restrict__ g, bf16* lds, int nrows) {
-  constexpr int V = 8;
+DEV void stage_tile(const bf16* __restrict__ g, bf16* lds, int nrows,
+                    const bf16* __restrict__ zero16) {
+  constexpr int RB = D * 2;            // bytes per image row
+  constexpr int RPK = 1024 / RB;       // rows per 1 KiB region
+  constexpr int LPR = RB / 16;         // lanes per row
+  constexpr int NREG = ROWS * RB / 1024;
+  constexpr int NPW = NREG / 4;
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
 #pragma unroll
-  for (int c = threadIdx.x; c < ROWS * (D / V); c += THREADS) {
-    int row = c / (D / V);
-    int col = (c % (D / V)) * V;
-    Pack16<bf16> v = {};
-    if (row < nrows) v = *(const Pack16<bf16>*)&g[row * D + col];
-    *(Pack16<bf16>*)&lds[aoff<bf16, D>(row, col)] = v;
+  for (int i = 0; i < NPW; ++i) {
+    const int j = wid * NPW + i;
+    const int rl = RPK * j + lane / LPR;
+    const int byte_in_row = (lane % LPR) * 16;
+    const int col = (byte_in_row ^ ((((rl >> 3) ^ rl) & 7) << 4)) / 2;
+    const bf16* src = rl < nrows ? &g[rl * D + col] : zero16;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)src,
+        (__attribute__((address_space(3))) unsigned int*)&lds[j * 512],
+        16, 0, 0);
   }
 }
 
@@ -139,7 +153,9 @@ template <int D, bool CAUSAL>
 __launch_bounds__(THREADS)
 __global__ void k_attn_fwd(const bf16* __restrict__ Q, const bf16* __restrict__ K,
                            const bf16* __restrict__ V, bf16* __restrict__ O,
-                           float* __restrict__ LSE, int S, float scale) {
+                           float* __restrict__ LSE,
+                           const bf16* __restrict__ zero16, int S,
+                           float scale) {
   constexpr int FN = BKV / 16;   // 4 score col fragments
   constexpr int FO = D / 16;     // output col fragments
   __shared__ alignas(16) bf16 q_lds[BQ * D];
@@ -156,7 +172,7 @@ __global__ void k_attn_fwd(const bf16* __restrict__ Q, const bf16* __restrict__ 
   const int lane = threadIdx.x & 63;
   const int wrow = wid * 16;                 // wave's q-row offset in tile
 
-  stage_tile<BQ, D>(q + (int64_t)q0 * D, q_lds, S - q0);
+  stage_tile<BQ, D>(q + (int64_t)q0 * D, q_lds, S - q0, zero16);
   // per-lane row stats for the 4 rows this lane's fragments touch share
   // one (m, l) per row; every lane keeps its row's copy (cr group)
   float m_run[4], l_run[4];
@@ -170,7 +186,7 @@ __global__ void k_attn_fwd(const bf16* __restrict__ Q, const bf16* __restrict__ 
   const int kv_end = CAUSAL ? min(S, q0 + BQ) : S;
   for (int kv0 = 0; kv0 < kv_end; kv0 += BKV) {
     __syncthreads();
-    stage_tile<BKV, D>(k + (int64_t)kv0 * D, k_lds, S - kv0);
+    stage_tile<BKV, D>(k + (int64_t)kv0 * D, k_lds, S - kv0, zero16);
     stage_tile_t<BKV, D>(v + (int64_t)kv0 * D, vt_lds, S - kv0);
     __syncthreads();
 
@@ -273,7 +289,9 @@ __global__ void k_attn_bwd(const bf16* __restrict__ Q, const bf16* __restrict__ 
                            const float* __restrict__ LSE,
                            const float* __restrict__ Di,
                            float* __restrict__ dQw, bf16* __restrict__ dK,
-                           bf16* __restrict__ dV, int S, float scale) {
+                           bf16* __restrict__ dV,
+                           const bf16* __restrict__ zero16, int S,
+                           float scale) {
   constexpr int FN = BQ / 16;    // 4 q-col fragments
   constexpr int FO = D / 16;
   __shared__ alignas(16) bf16 k_lds[BKV * D];
@@ -298,8 +316,8 @@ __global__ void k_attn_bwd(const bf16* __restrict__ Q, const bf16* __restrict__ 
   const int lane = threadIdx.x & 63;
   const int wrow = wid * 16;                 // wave's kv-row offset
 
-  stage_tile<BKV, D>(k + (int64_t)kv0 * D, k_lds, S - kv0);
-  stage_tile<BKV, D>(v + (int64_t)kv0 * D, v_lds, S - kv0);
+  stage_tile<BKV, D>(k + (int64_t)kv0 * D, k_lds, S - kv0, zero16);
+  stage_tile<BKV, D>(v + (int64_t)kv0 * D, v_lds, S - kv0, zero16);
   stage_tile_t<BKV, D>(k + (int64_t)kv0 * D, kt_lds, S - kv0);
 
   f32x4 dk_acc[FO] = {}, dv_acc[FO] = {};
@@ -307,9 +325,9 @@ __global__ void k_attn_bwd(const bf16* __restrict__ Q, const bf16* __restrict__ 
   const int q_start = CAUSAL ? (kv0 / BQ) * BQ : 0;
   for (int qt = q_start; qt < S; qt += BQ) {
     __syncthreads();
-    stage_tile<BQ, D>(q + (int64_t)qt * D, q_lds, S - qt);
+    stage_tile<BQ, D>(q + (int64_t)qt * D, q_lds, S - qt, zero16);
     stage_tile_t<BQ, D>(q + (int64_t)qt * D, qt_lds, S - qt);
-    stage_tile<BQ, D>(dout + (int64_t)qt * D, dot_lds, S - qt);
+    stage_tile<BQ, D>(dout + (int64_t)qt * D, dot_lds, S - qt, zero16);
     stage_tile_t<BQ, D>(dout + (int64_t)qt * D, dott_lds, S - qt);
     for (int i = threadIdx.x; i < BQ; i += THREADS) {
       int g = qt + i;
@@ -417,13 +435,13 @@ namespace tnn {
 using namespace attn;
 
 void attn_fwd_launch(const void* q, const void* k, const void* v, void* o,
-                     float* lse, int BH, int S, int D, bool causal,
-                     float scale, hipStream_t s) {
+                     float* lse, const void* zero16, int BH, int S, int D,
+                     bool causal, float scale, hipStream_t s) {
   dim3 grid((S + BQ - 1) / BQ, BH);
 #define LAUNCH(DD, CC)                                                     \
   hipLaunchKernelGGL((k_attn_fwd<DD, CC>), grid, dim3(THREADS), 0, s,      \
                      (const bf16*)q, (const bf16*)k, (const bf16*)v,       \
-                     (bf16*)o, lse, S, scale)
+                     (bf16*)o, lse, (const bf16*)zero16, S, scale)
   if (D == 64) { if (causal) LAUNCH(64, true); else LAUNCH(64, false); }
   else if (D == 128) { if (causal) LAUNCH(128, true); else LAUNCH(128, false); }
 #undef LAUNCH
@@ -431,8 +449,9 @@ void attn_fwd_launch(const void* q, const void* k, const void* v, void* o,
 
 void attn_bwd_launch(const void* q, const void* k, const void* v,
                      const void* o, const void* dout, const float* lse,
-                     float* di, float* dq_ws, void* dk, void* dv, int BH,
-                     int S, int D, bool causal, float scale, hipStream_t s) {
+                     float* di, float* dq_ws, void* dk, void* dv,
+                     const void* zero16, int BH, int S, int D, bool causal,
+                     float scale, hipStream_t s) {
   int64_t rows = (int64_t)BH * S;
   hipLaunchKernelGGL(k_attn_dot, dim3((rows * 64 + 255) / 256), dim3(256), 0,
                      s, (const bf16*)dout, (const bf16*)o, di, rows, D);
@@ -441,7 +460,7 @@ void attn_bwd_launch(const void* q, const void* k, const void* v,
   hipLaunchKernelGGL((k_attn_bwd<DD, CC>), grid, dim3(THREADS), 0, s,      \
                      (const bf16*)q, (const bf16*)k, (const bf16*)v,       \
                      (const bf16*)dout, lse, di, dq_ws, (bf16*)dk,         \
-                     (bf16*)dv, S, scale)
+                     (bf16*)dv, (const bf16*)zero16, S, scale)
   if (D == 64) { if (causal) LAUNCH(64, true); else LAUNCH(64, false); }
   else if (D == 128) { if (causal) LAUNCH(128, true); else LAUNCH(128, false); }
 #undef LAUNCH

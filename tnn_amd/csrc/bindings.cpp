@@ -491,8 +491,8 @@ std::vector<at::Tensor> attn_fwd(const at::Tensor& q, const at::Tensor& k,
   auto lse = at::empty({B, H, S}, q.options().dtype(at::kFloat));
   float scale = 1.0f / std::sqrt((float)D);
   attn_fwd_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
-                  lse.data_ptr<float>(), B * H, S, D, causal, scale,
-                  cur_stream());
+                  lse.data_ptr<float>(), zero_page(q), B * H, S, D, causal,
+                  scale, cur_stream());
   return {o, lse};
 }
 
@@ -510,7 +510,8 @@ std::vector<at::Tensor> attn_bwd(const at::Tensor& q, const at::Tensor& k,
   attn_bwd_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
                   dout.contiguous().data_ptr(), lse.data_ptr<float>(),
                   di.data_ptr<float>(), dqw.data_ptr<float>(), dk.data_ptr(),
-                  dv.data_ptr(), B * H, S, D, causal, scale, cur_stream());
+                  dv.data_ptr(), zero_page(q), B * H, S, D, causal, scale,
+                  cur_stream());
   auto dq = at::empty_like(q);
   cast_f32_launch(DT::BF16, dqw.data_ptr<float>(), dq.data_ptr(), dq.numel(),
                   cur_stream());

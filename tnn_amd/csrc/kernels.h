@@ -137,12 +137,13 @@ void embedding_bwd_launch(DT dt, const int64_t* ids, const void* dy,
 
 // ---- attention.hip (bf16 only; D in {64,128}) ------------------------------
 void attn_fwd_launch(const void* q, const void* k, const void* v, void* o,
-                     float* lse, int BH, int S, int D, bool causal,
-                     float scale, hipStream_t s);
+                     float* lse, const void* zero16, int BH, int S, int D,
+                     bool causal, float scale, hipStream_t s);
 void attn_bwd_launch(const void* q, const void* k, const void* v,
                      const void* o, const void* dout, const float* lse,
-                     float* di, float* dq_ws, void* dk, void* dv, int BH,
-                     int S, int D, bool causal, float scale, hipStream_t s);
+                     float* di, float* dq_ws, void* dk, void* dv,
+                     const void* zero16, int BH, int S, int D, bool causal,
+                     float scale, hipStream_t s);
 
 // ---- optim.hip -------------------------------------------------------------
 void sgd_step_launch(DT dt_p, const void* grad, DT dt_g, void* param,
