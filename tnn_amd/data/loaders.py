@@ -174,6 +174,12 @@ class TinyImageNetLoader(BaseDataLoader):
             np.load(os.path.join(self.path, f"{split}_y.npy")).astype(np.int64))
 
 
+class ImageNet100Loader(TinyImageNetLoader):
+    """ImageNet-100 from pre-decoded .npy splits (reference decodes JPEG
+    via stb_image; this image has no decoder, so images are converted
+    offline to {train,val}_{x,y}.npy exactly like TinyImageNetLoader)."""
+
+
 class OpenWebTextLoader(BaseDataLoader):
     """mmap'd uint16 GPT-2 token file, random windows
     (reference include/data_loading/open_webtext_data_loader.hpp:11-95).
@@ -221,6 +227,7 @@ class DataLoaderFactory:
         "cifar10": CIFAR10Loader,
         "cifar100": CIFAR100Loader,
         "tiny_imagenet": TinyImageNetLoader,
+        "imagenet_100": ImageNet100Loader,
         "openwebtext": OpenWebTextLoader,
     }
 
