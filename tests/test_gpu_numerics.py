@@ -503,3 +503,22 @@ def test_wrn_converges_on_learnable_task():
         pred = model(xg).argmax(-1)
         acc = (pred == yg).float().mean().item()
     assert acc > 0.9, f"did not converge: acc={acc}"
+
+
+def test_graphed_inference_matches_eager():
+    """hipGraph capture of an eval forward (utils/graphstep.py): the
+    replayed graph must match eager outputs for fresh inputs."""
+    from tnn_amd import models
+    from tnn_amd.nn.layer import cast_compute_dtype
+    from tnn_amd.utils.graphstep import GraphedInference
+    torch.manual_seed(21)
+    m = models.create_model("cifar10_resnet9")
+    cast_compute_dtype(m, torch.bfloat16)
+    m.to(DEV).eval()
+    x0 = torch.randn(16, 32, 32, 3, dtype=torch.bfloat16, device=DEV)
+    g = GraphedInference(m, x0)
+    x1 = torch.randn_like(x0)
+    with torch.no_grad():
+        ref = m(x1)
+    out = g(x1)
+    assert torch.equal(out, ref)
