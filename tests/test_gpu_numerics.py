@@ -186,7 +186,7 @@ def test_bn_fwd_train(dtype, C):
     x = torch.randn(8, 6, 6, C, dtype=dtype, device=DEV) * 2 + 0.5
     gamma = torch.rand(C, device=DEV) + 0.5
     beta = torch.randn(C, device=DEV)
-    y, mean, invstd = ext.bn_fwd_train(x, gamma, beta, 1e-5, False)
+    y, mean, invstd = ext.bn_fwd_train(x, gamma, beta, None, None, 0.1, 1e-5, False)
     xf = x.float().reshape(-1, C)
     rmean = xf.mean(0)
     rvar = xf.var(0, unbiased=False)
@@ -201,7 +201,7 @@ def test_bn_relu_and_infer(dtype):
     x = torch.randn(4, 5, 5, C, dtype=dtype, device=DEV)
     gamma = torch.ones(C, device=DEV)
     beta = torch.zeros(C, device=DEV)
-    y, _, _ = ext.bn_fwd_train(x, gamma, beta, 1e-5, True)
+    y, _, _ = ext.bn_fwd_train(x, gamma, beta, None, None, 0.1, 1e-5, True)
     assert (y.float() >= 0).all()
     rm = torch.randn(C, device=DEV) * 0.1
     rv = torch.rand(C, device=DEV) + 0.5
@@ -226,7 +226,7 @@ def test_bn_bwd_matches_autograd(dtype):
     y.backward(dy)
     xdt = x.to(dtype)
     y_k, mean_k, invstd_k = ext.bn_fwd_train(xdt, gamma.detach(), beta.detach(),
-                                             1e-5, False)
+                                             None, None, 0.1, 1e-5, False)
     dx, dgamma, dbeta = ext.bn_bwd(xdt, dy.to(dtype), gamma.detach(), mean_k,
                                    invstd_k, None)
     tol = TOL[dtype] * 4

@@ -47,7 +47,8 @@ __global__ void k_bn_partial(const T* __restrict__ x, float* __restrict__ sum,
 }
 
 __global__ void k_bn_finalize(float* mean, float* invstd, const float* sum,
-                              const float* sumsq, int64_t rows, int cols,
+                              const float* sumsq, float* rmean, float* rvar,
+                              float momentum, int64_t rows, int cols,
                               float eps) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= cols) return;
@@ -55,6 +56,11 @@ __global__ void k_bn_finalize(float* mean, float* invstd, const float* sum,
   float var = fmaxf(sumsq[c] / rows - m * m, 0.0f);
   mean[c] = m;
   invstd[c] = rsqrtf(var + eps);
+  if (rmean) {  // fused running-stat update (unbiased var, torch semantics)
+    float unbiased = var * ((float)rows / (float)(rows > 1 ? rows - 1 : 1));
+    rmean[c] = rmean[c] * (1.0f - momentum) + m * momentum;
+    rvar[c] = rvar[c] * (1.0f - momentum) + unbiased * momentum;
+  }
 }
 
 template <typename T>
@@ -161,6 +167,7 @@ static inline dim3 bn_reduce_grid(int64_t rows, int cols) {
 }
 
 void bn_stats_launch(DT dt, const void* x, float* mean, float* invstd,
+                     float* rmean, float* rvar, float momentum,
                      int64_t rows, int cols, float eps, hipStream_t s) {
   // mean/invstd double as the scratch sum/sumsq buffers (finalized in place);
   // they must be zeroed first.
@@ -174,7 +181,8 @@ void bn_stats_launch(DT dt, const void* x, float* mean, float* invstd,
     hipLaunchKernelGGL(k_bn_partial<bf16>, grid, dim3(256), 0, s,
                        (const bf16*)x, mean, invstd, rows, cols);
   hipLaunchKernelGGL(k_bn_finalize, dim3((cols + 255) / 256), dim3(256), 0, s,
-                     mean, invstd, mean, invstd, rows, cols, eps);
+                     mean, invstd, mean, invstd, rmean, rvar, momentum, rows,
+                     cols, eps);
 }
 
 void bn_apply_launch(DT dt, const void* x, const float* mean,

@@ -261,16 +261,21 @@ at::Tensor conv2d_wgrad(const at::Tensor& x, const at::Tensor& dy, int64_t KH,
 // ---- batch norm ------------------------------------------------------------
 std::vector<at::Tensor> bn_fwd_train(const at::Tensor& x,
                                      const at::Tensor& gamma,
-                                     const at::Tensor& beta, double eps,
-                                     bool relu) {
+                                     const at::Tensor& beta,
+                                     const c10::optional<at::Tensor>& rmean,
+                                     const c10::optional<at::Tensor>& rvar,
+                                     double momentum, double eps, bool relu) {
   CHECK_IN(x);
   int C = x.size(-1);
   int64_t rows = x.numel() / C;
   auto mean = at::empty({C}, x.options().dtype(at::kFloat));
   auto invstd = at::empty({C}, x.options().dtype(at::kFloat));
   auto y = at::empty_like(x);
+  float* rm = rmean.has_value() ? rmean->data_ptr<float>() : nullptr;
+  float* rv = rvar.has_value() ? rvar->data_ptr<float>() : nullptr;
   bn_stats_launch(dt_of(x), x.data_ptr(), mean.data_ptr<float>(),
-                  invstd.data_ptr<float>(), rows, C, (float)eps, cur_stream());
+                  invstd.data_ptr<float>(), rm, rv, (float)momentum, rows, C,
+                  (float)eps, cur_stream());
   bn_apply_launch(dt_of(x), x.data_ptr(), mean.data_ptr<float>(),
                   invstd.data_ptr<float>(), gamma.data_ptr<float>(),
                   beta.data_ptr<float>(), y.data_ptr(), rows, C, relu,

@@ -80,6 +80,7 @@ void transpose_w_launch(DT dt, const void* w, void* w_t, int KH, int KW,
 
 // ---- batchnorm.hip ---------------------------------------------------------
 void bn_stats_launch(DT dt, const void* x, float* mean, float* invstd,
+                     float* rmean, float* rvar, float momentum,
                      int64_t rows, int cols, float eps, hipStream_t s);
 void bn_apply_launch(DT dt, const void* x, const float* mean,
                      const float* invstd, const float* gamma, const float* beta,

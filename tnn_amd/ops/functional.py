@@ -109,9 +109,12 @@ def conv2d_nhwc(
 
 class _BatchNormAct(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, gamma, beta, eps, relu):
+    def forward(ctx, x, gamma, beta, running_mean, running_var, momentum,
+                eps, relu):
         ext = _C.ext()
-        y, mean, invstd = ext.bn_fwd_train(x, gamma, beta, eps, relu)
+        # running stats update fused into the finalize kernel
+        y, mean, invstd = ext.bn_fwd_train(x, gamma, beta, running_mean,
+                                           running_var, momentum, eps, relu)
         ctx.save_for_backward(x, gamma, mean, invstd, y if relu else None)
         ctx.relu = relu
         return y, mean, invstd
@@ -122,7 +125,7 @@ class _BatchNormAct(torch.autograd.Function):
         ext = _C.ext()
         dx, dgamma, dbeta = ext.bn_bwd(x, dy.contiguous(), gamma, mean, invstd,
                                        y if ctx.relu else None)
-        return dx, dgamma, dbeta, None, None
+        return dx, dgamma, dbeta, None, None, None, None, None
 
 
 def batch_norm_act(
@@ -146,12 +149,9 @@ def batch_norm_act(
     n = x.numel() // C
     if _use_hip(x):
         if training:
-            y, mean, invstd = _BatchNormAct.apply(x.contiguous(), gamma, beta, eps, relu)
-            with torch.no_grad():
-                var = (1.0 / (invstd * invstd)) - eps
-                unbiased = var * (n / max(n - 1, 1))
-                running_mean.mul_(1 - momentum).add_(mean.float(), alpha=momentum)
-                running_var.mul_(1 - momentum).add_(unbiased.float(), alpha=momentum)
+            y, _, _ = _BatchNormAct.apply(x.contiguous(), gamma, beta,
+                                          running_mean, running_var, momentum,
+                                          eps, relu)
             return y
         ext = _C.ext()
         return ext.bn_fwd_infer(x.contiguous(), gamma, beta, running_mean, running_var,
