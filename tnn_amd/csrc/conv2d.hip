@@ -161,7 +161,7 @@ __global__ void k_conv_fwd(const T* __restrict__ X, const T* __restrict__ Wt,
 // with oh = (ih+PH-kh)/SH when divisible. Wt here is the transposed weight
 // [KH,KW,Cout,Cin] so B rows are k=(kh,kw,co) with ci contiguous.
 // ---------------------------------------------------------------------------
-template <typename T, bool POW2, bool GLDS>
+template <typename T, bool POW2, bool GLDS, bool S2 = false>
 __launch_bounds__(THREADS)
 __global__ void k_conv_dgrad(const T* __restrict__ DY, const T* __restrict__ WT,
                              T* __restrict__ DX, const T* __restrict__ zero16,
@@ -170,8 +170,21 @@ __global__ void k_conv_dgrad(const T* __restrict__ DY, const T* __restrict__ WT,
   __shared__ alignas(16) T As[BM * BK];
   __shared__ alignas(16) T Bs[BN * BK];
 
-  const int M = cs.N * cs.H * cs.W;
-  const int K = cs.KH * cs.KW * cs.Cout;
+  // S2: stride-2 parity decomposition. blockIdx.z selects the
+  // (ih%2, iw%2) class so every (kh,kw) visited satisfies the stride
+  // divisibility test by construction -- the generic kernel at stride 2
+  // burns 3/4 of its MFMAs multiplying gathered zeros. The launcher
+  // rewrites cs.d_hw/d_w to the per-class (Hc*Wc, Wc) divisors.
+  const int cls_a = S2 ? ((int)blockIdx.z >> 1) : 0;
+  const int cls_b = S2 ? ((int)blockIdx.z & 1) : 0;
+  const int start_h = S2 ? ((cls_a + cs.PH) & 1) : 0;
+  const int start_w = S2 ? ((cls_b + cs.PW) & 1) : 0;
+  const int nkh = S2 ? ((cs.KH - start_h + 1) >> 1) : cs.KH;
+  const int nkw = S2 ? ((cs.KW - start_w + 1) >> 1) : cs.KW;
+  const int Hc = S2 ? (cs.H >> 1) : cs.H;
+  const int Wc = S2 ? (cs.W >> 1) : cs.W;
+  const int M = cs.N * Hc * Wc;
+  const int K = nkh * nkw * cs.Cout;
   const int m0 = blockIdx.x * BM;
   const int n0 = blockIdx.y * BN;
   const WaveCoord wc;
@@ -182,16 +195,29 @@ __global__ void k_conv_dgrad(const T* __restrict__ DY, const T* __restrict__ WT,
     int gm = m0 + rl, gk = k0 + kk;
     if (gm >= M || gk >= K) return zero16;
     int n = idiv<POW2>(gm, cs.d_hw);
-    int rem = gm - n * (cs.H * cs.W);
-    int ih = idiv<POW2>(rem, cs.d_w), iw = rem - ih * cs.W;
+    int rem = gm - n * (Hc * Wc);
+    int ih = idiv<POW2>(rem, cs.d_w), iw = rem - ih * Wc;
     int co = imod<POW2>(gk, cs.d_cout);
     int kidx = idiv<POW2>(gk, cs.d_cout);
-    int kw = kidx % cs.KW, kh = kidx / cs.KW;
-    int th = ih + cs.PH - kh, tw = iw + cs.PW - kw;
-    if (th < 0 || tw < 0 || th % cs.SH || tw % cs.SW) return zero16;
-    int oh = th / cs.SH, ow = tw / cs.SW;
-    if (oh >= cs.OH || ow >= cs.OW) return zero16;
-    return &DY[(((int64_t)n * cs.OH + oh) * cs.OW + ow) * cs.Cout + co];
+    if constexpr (S2) {
+      ih = 2 * ih + cls_a;
+      iw = 2 * iw + cls_b;
+      int kwi = kidx & (nkw - 1);              // nkw is 1 or 2
+      int khi = nkw == 2 ? (kidx >> 1) : kidx;
+      int kh = start_h + 2 * khi, kw = start_w + 2 * kwi;
+      int th = ih + cs.PH - kh, tw = iw + cs.PW - kw;
+      if (th < 0 || tw < 0) return zero16;
+      int oh = th >> 1, ow = tw >> 1;          // divisible by construction
+      if (oh >= cs.OH || ow >= cs.OW) return zero16;
+      return &DY[(((int64_t)n * cs.OH + oh) * cs.OW + ow) * cs.Cout + co];
+    } else {
+      int kw = kidx % cs.KW, kh = kidx / cs.KW;
+      int th = ih + cs.PH - kh, tw = iw + cs.PW - kw;
+      if (th < 0 || tw < 0 || th % cs.SH || tw % cs.SW) return zero16;
+      int oh = th / cs.SH, ow = tw / cs.SW;
+      if (oh >= cs.OH || ow >= cs.OW) return zero16;
+      return &DY[(((int64_t)n * cs.OH + oh) * cs.OW + ow) * cs.Cout + co];
+    }
   };
 
   for (int k0 = 0; k0 < K; k0 += BK) {
@@ -243,7 +269,7 @@ __global__ void k_conv_dgrad(const T* __restrict__ DY, const T* __restrict__ WT,
       *(VecT*)&As[lds_off<T>(row, kk)] = v;
     }
     }
-    // ---- stage B (w_t [K, Cin]) -> Bs[ci][k] ----
+    // ---- stage B (w_t [KH*KW*Cout, Cin]) -> Bs[ci][k] ----
 #pragma unroll
     for (int c = threadIdx.x; c < BK * (BN / V); c += THREADS) {
       int kk = c / (BN / V);
@@ -251,13 +277,23 @@ __global__ void k_conv_dgrad(const T* __restrict__ DY, const T* __restrict__ WT,
       int gk = k0 + kk, gn = n0 + nn;
       VecT v = {};
       if (gk < K) {
-        const T* src = &WT[(int64_t)gk * cs.Cin + gn];
+        int64_t wrow = gk;
+        if constexpr (S2) {
+          // compacted (khi,kwi,co) -> full (kh*KW+kw)*Cout+co row of WT
+          int co = imod<POW2>(gk, cs.d_cout);
+          int kidx = idiv<POW2>(gk, cs.d_cout);
+          int kwi = kidx & (nkw - 1);
+          int khi = nkw == 2 ? (kidx >> 1) : kidx;
+          wrow = (int64_t)((start_h + 2 * khi) * cs.KW + start_w + 2 * kwi) *
+                     cs.Cout + co;
+        }
+        const T* src = &WT[wrow * cs.Cin + gn];
         if (gn + V <= cs.Cin && aligned16(src)) {
           v = *(const VecT*)src;
         } else {
 #pragma unroll
           for (int j = 0; j < V; ++j)
-            if (gn + j < cs.Cin) v.e[j] = WT[(int64_t)gk * cs.Cin + gn + j];
+            if (gn + j < cs.Cin) v.e[j] = src[j];
         }
       }
 #pragma unroll
@@ -270,8 +306,17 @@ __global__ void k_conv_dgrad(const T* __restrict__ DY, const T* __restrict__ WT,
   }
 
   epilogue_visit(wc, acc, m0, n0, [&](int row, int col, float v) {
-    if (row < M && col < cs.Cin)
-      DX[(int64_t)row * cs.Cin + col] = VecIO<T>::from_f32(v);
+    if (row < M && col < cs.Cin) {
+      int64_t pix = row;
+      if constexpr (S2) {
+        int n = idiv<POW2>(row, cs.d_hw);
+        int rem = row - n * (Hc * Wc);
+        int ihp = idiv<POW2>(rem, cs.d_w);
+        int ih = 2 * ihp + cls_a, iw = 2 * (rem - ihp * Wc) + cls_b;
+        pix = ((int64_t)n * cs.H + ih) * cs.W + iw;
+      }
+      DX[pix * cs.Cin + col] = VecIO<T>::from_f32(v);
+    }
   });
 }
 
@@ -427,8 +472,27 @@ void conv2d_dgrad_launch(DT dt, const void* dy, const void* w_t, void* dx,
   int M = cs.N * cs.H * cs.W;
   dim3 grid(ceil_div(M, BM), ceil_div(cs.Cin, BN));
   bool p2 = all_pow2(cs);
+  // stride-2 parity decomposition (see k_conv_dgrad<..., S2>): 4 classes in
+  // grid.z over an M/4-row problem with only the stride-compatible (kh,kw)
+  // taps in K. Requires even spatial dims; routed through the glds variant.
+  bool s2 = p2 && cs.SH == 2 && cs.SW == 2 && cs.H % 2 == 0 && cs.W % 2 == 0;
+  ConvShape cs2 = cs;
+  if (s2) {
+    cs2.d_hw.set((cs.H / 2) * (cs.W / 2));
+    cs2.d_w.set(cs.W / 2);
+    s2 = cs2.d_hw.lg >= 0 && cs2.d_w.lg >= 0;
+  }
+  dim3 grid_s2(ceil_div(cs.N * (cs.H / 2) * (cs.W / 2), BM),
+               ceil_div(cs.Cin, BN), 4);
   if (dt == DT::F32) {
     bool g = p2 && cs.Cout % 4 == 0 && (((uintptr_t)dy & 15) == 0);
+    if (g && s2) {
+      hipLaunchKernelGGL((k_conv_dgrad<float, true, true, true>), grid_s2,
+                         dim3(THREADS), 0, s, (const float*)dy,
+                         (const float*)w_t, (float*)dx, (const float*)zero16,
+                         cs2);
+      return;
+    }
     auto kern = g ? k_conv_dgrad<float, true, true>
                   : (p2 ? k_conv_dgrad<float, true, false>
                         : k_conv_dgrad<float, false, false>);
@@ -437,6 +501,13 @@ void conv2d_dgrad_launch(DT dt, const void* dy, const void* w_t, void* dx,
                        (const float*)zero16, cs);
   } else {
     bool g = p2 && cs.Cout % 8 == 0 && (((uintptr_t)dy & 15) == 0);
+    if (g && s2) {
+      hipLaunchKernelGGL((k_conv_dgrad<bf16, true, true, true>), grid_s2,
+                         dim3(THREADS), 0, s, (const bf16*)dy,
+                         (const bf16*)w_t, (bf16*)dx, (const bf16*)zero16,
+                         cs2);
+      return;
+    }
     auto kern = g ? k_conv_dgrad<bf16, true, true>
                   : (p2 ? k_conv_dgrad<bf16, true, false>
                         : k_conv_dgrad<bf16, false, false>);
