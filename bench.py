@@ -41,8 +41,7 @@ def main():
     args = parse_args()
     from tnn_amd import models
     from tnn_amd.nn import CrossEntropyLoss
-    from tnn_amd.nn.layer import cast_compute_dtype
-    from tnn_amd.parallel import init_distributed, Communicator, PipelineEngine
+    from tnn_amd.parallel import init_distributed, PipelineEngine
 
     have_gpu = torch.cuda.is_available()
     device = torch.device("cuda" if have_gpu else "cpu")

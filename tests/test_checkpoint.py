@@ -1,7 +1,6 @@
 """Checkpoint archive format round trips (reference archiver_test.cpp +
 Graph::save_state/load_state)."""
 
-import os
 
 import torch
 

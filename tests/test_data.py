@@ -3,8 +3,7 @@
 import numpy as np
 import torch
 
-from tnn_amd.data import (SyntheticImageLoader, SyntheticTokenLoader,
-                          CIFAR100Loader, OpenWebTextLoader, DataLoaderFactory,
+from tnn_amd.data import (CIFAR100Loader, OpenWebTextLoader, DataLoaderFactory,
                           AugmentationStrategy, HorizontalFlip, RandomCrop,
                           Cutout, Normalize, Tokenizer)
 

@@ -1,7 +1,6 @@
 """Flash attention kernels vs fp32 composed reference (full-tensor check,
 asymmetric random operands — cdna guide §5.4 rules 16/25)."""
 
-import math
 
 import pytest
 import torch

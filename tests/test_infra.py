@@ -4,7 +4,6 @@
 import json
 import time
 
-import numpy as np
 import torch
 
 from tnn_amd import nn as tnn

@@ -1,6 +1,5 @@
 """Optimizer/scheduler behavior (reference optimizers.hpp / schedulers.hpp)."""
 
-import math
 
 import pytest
 import torch

@@ -3,7 +3,6 @@
 import torch
 
 from tnn_amd import models
-from tnn_amd.nn import LayerBuilder
 from tnn_amd.parallel.partitioner import (NaivePipelinePartitioner,
                                           WeightedPipelinePartitioner,
                                           NaiveDataPartitioner, Partitioner)

@@ -5,8 +5,7 @@ import torch
 
 from tnn_amd import nn as tnn
 from tnn_amd.data import SyntheticImageLoader
-from tnn_amd.nn import (TrainingConfig, train_model, CrossEntropyLoss,
-                        AdamW, accuracy)
+from tnn_amd.nn import TrainingConfig, train_model, CrossEntropyLoss, AdamW
 
 
 def _tiny_cnn():

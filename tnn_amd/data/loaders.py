@@ -9,7 +9,6 @@ benchmark path (no network ⇒ no real datasets in this environment).
 from __future__ import annotations
 
 import os
-import struct
 from typing import Iterator, Optional, Tuple
 
 import numpy as np

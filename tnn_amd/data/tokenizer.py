@@ -9,7 +9,7 @@ work without tiktoken.
 from __future__ import annotations
 
 import struct
-from typing import List, Optional
+from typing import List
 
 
 class Tokenizer:

@@ -8,7 +8,7 @@ a Sequential into per-rank stage Sequentials, which round-trip through config.
 from __future__ import annotations
 
 import math
-from typing import Any, Dict, List, Optional, Tuple
+from typing import List, Optional
 
 import torch
 import torch.nn as nn

@@ -13,8 +13,7 @@ allocator + a flat ``zero_grads`` over parameters here.
 
 from __future__ import annotations
 
-import json
-from typing import Any, Dict, List, Optional, Tuple
+from typing import Any, Dict, List, Optional
 
 import torch
 

@@ -14,7 +14,7 @@ in-flight micro-batch.
 
 from __future__ import annotations
 
-from typing import Any, Dict, Optional, Sequence, Tuple
+from typing import Any, Dict, Tuple
 
 import torch
 import torch.nn as nn

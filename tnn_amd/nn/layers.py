@@ -8,13 +8,13 @@ directly by the CDNA4 MFMA conv kernels.
 from __future__ import annotations
 
 import math
-from typing import Any, Dict, Optional, Tuple
+from typing import Optional
 
 import torch
 import torch.nn as nn
 
 from .. import ops
-from .layer import Layer, register_layer, dtype_name
+from .layer import Layer, register_layer
 
 
 def _pair(v):

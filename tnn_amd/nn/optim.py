@@ -10,8 +10,7 @@ MI355X-native mixed-precision upgrade).
 
 from __future__ import annotations
 
-import math
-from typing import Any, Dict, Iterable, List, Optional
+from typing import Any, Dict, Iterable, List
 
 import torch
 

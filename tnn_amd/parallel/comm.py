@@ -15,7 +15,7 @@ from __future__ import annotations
 
 import datetime
 import os
-from typing import Any, List, Optional, Sequence
+from typing import Any, List, Optional
 
 import torch
 import torch.distributed as dist

@@ -10,7 +10,7 @@ collectives (per-link-bound ⇒ fewer, larger buckets than on NVSwitch).
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 import torch
 import torch.distributed as dist

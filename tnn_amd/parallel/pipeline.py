@@ -21,7 +21,6 @@ reference, on purpose:
 
 from __future__ import annotations
 
-import time
 from typing import Any, Dict, List, Optional, Tuple
 
 import torch

@@ -11,7 +11,6 @@ counts that a captured graph would freeze (documented round-2 work).
 
 from __future__ import annotations
 
-from typing import Optional
 
 import torch
 
