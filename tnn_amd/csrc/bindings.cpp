@@ -305,8 +305,10 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& x, const at::Tensor& dy,
   CHECK_IN(dy);
   int C = x.size(-1);
   int64_t rows = x.numel() / C;
-  auto sum_dy = at::zeros({C}, x.options().dtype(at::kFloat));
-  auto sum_dy_xhat = at::zeros({C}, x.options().dtype(at::kFloat));
+  // one zeroed slab for both channel sums: halves the per-BN fill launches
+  auto sums = at::zeros({2, C}, x.options().dtype(at::kFloat));
+  auto sum_dy = sums[0];
+  auto sum_dy_xhat = sums[1];
   auto dx = at::empty_like(x);
   const void* yr = y_relu.has_value() ? y_relu->data_ptr() : nullptr;
   bn_bwd_reduce_launch(dt_of(x), x.data_ptr(), dy.data_ptr(), yr,

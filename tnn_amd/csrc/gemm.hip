@@ -226,8 +226,48 @@ __global__ void k_splitk_reduce(const float* __restrict__ ws,
   }
 }
 
+// float4 variant with 4 independent slab accumulators: the scalar kernel is
+// latency-bound (one dependent 4B load per z iteration measured ~1.25 TB/s);
+// 16B loads x 4-deep MLP reach the HBM roofline.
+__global__ void k_splitk_reduce4(const float4* __restrict__ ws,
+                                 float4* __restrict__ out, int z, int64_t n4) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n4; i += stride) {
+    float4 a0{0, 0, 0, 0}, a1{0, 0, 0, 0}, a2{0, 0, 0, 0}, a3{0, 0, 0, 0};
+    int s = 0;
+    for (; s + 4 <= z; s += 4) {
+      float4 v0 = ws[(int64_t)s * n4 + i];
+      float4 v1 = ws[(int64_t)(s + 1) * n4 + i];
+      float4 v2 = ws[(int64_t)(s + 2) * n4 + i];
+      float4 v3 = ws[(int64_t)(s + 3) * n4 + i];
+      a0.x += v0.x; a0.y += v0.y; a0.z += v0.z; a0.w += v0.w;
+      a1.x += v1.x; a1.y += v1.y; a1.z += v1.z; a1.w += v1.w;
+      a2.x += v2.x; a2.y += v2.y; a2.z += v2.z; a2.w += v2.w;
+      a3.x += v3.x; a3.y += v3.y; a3.z += v3.z; a3.w += v3.w;
+    }
+    for (; s < z; ++s) {
+      float4 v = ws[(int64_t)s * n4 + i];
+      a0.x += v.x; a0.y += v.y; a0.z += v.z; a0.w += v.w;
+    }
+    float4 r;
+    r.x = (a0.x + a1.x) + (a2.x + a3.x);
+    r.y = (a0.y + a1.y) + (a2.y + a3.y);
+    r.z = (a0.z + a1.z) + (a2.z + a3.z);
+    r.w = (a0.w + a1.w) + (a2.w + a3.w);
+    out[i] = r;
+  }
+}
+
 void splitk_reduce_launch(const float* ws, float* out, int z, int64_t n,
                           hipStream_t s) {
+  if ((n & 3) == 0) {
+    int64_t n4 = n >> 2;
+    int blocks = (int)std::min<int64_t>((n4 + 255) / 256, (int64_t)2048);
+    hipLaunchKernelGGL(k_splitk_reduce4, dim3(blocks), dim3(256), 0, s,
+                       (const float4*)ws, (float4*)out, z, n4);
+    return;
+  }
   int blocks = (int)std::min<int64_t>((n + 255) / 256, (int64_t)2048);
   hipLaunchKernelGGL(k_splitk_reduce, dim3(blocks), dim3(256), 0, s, ws, out,
                      z, n);
