@@ -13,6 +13,11 @@
 
 namespace tnn {
 
+// 16-byte load pack (ext_vector_type cannot hold the __hip_bfloat16 struct)
+template <typename T> struct alignas(16) BnPack {
+  T e[16 / sizeof(T)];
+};
+
 // pass 1a: partial sum / sumsq per channel, atomics into f32 buffers
 template <typename T>
 __global__ void k_bn_partial(const T* __restrict__ x, float* __restrict__ sum,
@@ -43,6 +48,132 @@ __global__ void k_bn_partial(const T* __restrict__ x, float* __restrict__ sum,
     }
     atomicAdd(&sum[c], s);
     atomicAdd(&sumsq[c], ss);
+  }
+}
+
+// Vectorized reduce variants (cols % V == 0, 16B-aligned): the scalar
+// kernels issue one 2B/4B load per iteration feeding two short dependent
+// FMA chains -- measured ~1.4 TB/s. 16B loads + V independent accumulator
+// lanes per thread give the MLP to reach the HBM roofline. Each thread owns
+// V consecutive channels; per-block partials combine through LDS and emit
+// one atomicAdd per channel.
+template <typename T>
+__global__ void k_bn_partial_vec(const T* __restrict__ x,
+                                 float* __restrict__ sum,
+                                 float* __restrict__ sumsq, int64_t rows,
+                                 int cols) {
+  constexpr int V = 16 / sizeof(T);
+  using VecT = BnPack<T>;
+  const int groups = cols / V;
+  const int gpb = min(groups, 256);
+  const int rows_per_iter = 256 / gpb;
+  const int g = threadIdx.x % gpb + blockIdx.x * gpb;
+  const int r_off = threadIdx.x / gpb;
+  __shared__ float sh[2 * 256 * V];
+  float* sh_s = sh;
+  float* sh_ss = sh + 256 * V;
+  float s[V] = {}, ss[V] = {};
+  if (g < groups && r_off < rows_per_iter) {
+    const int64_t r0 = rows * blockIdx.y / gridDim.y;
+    const int64_t r1 = rows * (blockIdx.y + 1) / gridDim.y;
+    for (int64_t r = r0 + r_off; r < r1; r += rows_per_iter) {
+      VecT v = *(const VecT*)&x[r * cols + (int64_t)g * V];
+#pragma unroll
+      for (int j = 0; j < V; ++j) {
+        float f = VecIO<T>::to_f32(v.e[j]);
+        s[j] += f;
+        ss[j] += f * f;
+      }
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < V; ++j) {
+    sh_s[threadIdx.x * V + j] = s[j];
+    sh_ss[threadIdx.x * V + j] = ss[j];
+  }
+  __syncthreads();
+  const int cpb_total = gpb * V;
+  for (int cl = threadIdx.x; cl < cpb_total; cl += 256) {
+    int gl = cl / V, j = cl % V;
+    float ts = 0.0f, tss = 0.0f;
+    for (int r = 0; r < rows_per_iter; ++r) {
+      ts += sh_s[(r * gpb + gl) * V + j];
+      tss += sh_ss[(r * gpb + gl) * V + j];
+    }
+    int c = blockIdx.x * cpb_total + cl;
+    atomicAdd(&sum[c], ts);
+    atomicAdd(&sumsq[c], tss);
+  }
+}
+
+template <typename T>
+__global__ void k_bn_bwd_reduce_vec(const T* __restrict__ x,
+                                    const T* __restrict__ dy,
+                                    const T* __restrict__ y_relu,
+                                    const float* mean, const float* invstd,
+                                    float* __restrict__ sum_dy,
+                                    float* __restrict__ sum_dy_xhat,
+                                    int64_t rows, int cols) {
+  constexpr int V = 16 / sizeof(T);
+  using VecT = BnPack<T>;
+  const int groups = cols / V;
+  const int gpb = min(groups, 256);
+  const int rows_per_iter = 256 / gpb;
+  const int g = threadIdx.x % gpb + blockIdx.x * gpb;
+  const int r_off = threadIdx.x / gpb;
+  __shared__ float sh[2 * 256 * V];
+  float* sh_s = sh;
+  float* sh_sx = sh + 256 * V;
+  float s[V] = {}, sx[V] = {};
+  if (g < groups && r_off < rows_per_iter) {
+    const int64_t r0 = rows * blockIdx.y / gridDim.y;
+    const int64_t r1 = rows * (blockIdx.y + 1) / gridDim.y;
+    float m[V], is[V];
+#pragma unroll
+    for (int j = 0; j < V; ++j) {
+      m[j] = mean[g * V + j];
+      is[j] = invstd[g * V + j];
+    }
+    for (int64_t r = r0 + r_off; r < r1; r += rows_per_iter) {
+      int64_t i = r * cols + (int64_t)g * V;
+      VecT vg = *(const VecT*)&dy[i];
+      VecT vx = *(const VecT*)&x[i];
+      if (y_relu) {
+        VecT vy = *(const VecT*)&y_relu[i];
+#pragma unroll
+        for (int j = 0; j < V; ++j) {
+          float gr = VecIO<T>::to_f32(vy.e[j]) > 0.0f
+                         ? VecIO<T>::to_f32(vg.e[j]) : 0.0f;
+          s[j] += gr;
+          sx[j] += gr * (VecIO<T>::to_f32(vx.e[j]) - m[j]) * is[j];
+        }
+      } else {
+#pragma unroll
+        for (int j = 0; j < V; ++j) {
+          float gr = VecIO<T>::to_f32(vg.e[j]);
+          s[j] += gr;
+          sx[j] += gr * (VecIO<T>::to_f32(vx.e[j]) - m[j]) * is[j];
+        }
+      }
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < V; ++j) {
+    sh_s[threadIdx.x * V + j] = s[j];
+    sh_sx[threadIdx.x * V + j] = sx[j];
+  }
+  __syncthreads();
+  const int cpb_total = gpb * V;
+  for (int cl = threadIdx.x; cl < cpb_total; cl += 256) {
+    int gl = cl / V, j = cl % V;
+    float ts = 0.0f, tsx = 0.0f;
+    for (int r = 0; r < rows_per_iter; ++r) {
+      ts += sh_s[(r * gpb + gl) * V + j];
+      tsx += sh_sx[(r * gpb + gl) * V + j];
+    }
+    int c = blockIdx.x * cpb_total + cl;
+    atomicAdd(&sum_dy[c], ts);
+    atomicAdd(&sum_dy_xhat[c], tsx);
   }
 }
 
@@ -154,6 +285,25 @@ __global__ void k_bn_bwd_apply(const T* __restrict__ x, const T* __restrict__ dy
 }
 
 // ---------------------------------------------------------------------------
+template <typename T>
+static inline bool bn_vec_ok(const void* x, int cols) {
+  constexpr int V = 16 / sizeof(T);
+  return cols % V == 0 && (((uintptr_t)x & 15) == 0);
+}
+
+template <typename T>
+static inline dim3 bn_reduce_grid_vec(int64_t rows, int cols) {
+  constexpr int V = 16 / sizeof(T);
+  int groups = cols / V;
+  int gpb = groups < 256 ? groups : 256;
+  int cblocks = (groups + gpb - 1) / gpb;
+  int rows_per_iter = 256 / gpb;
+  int64_t iters = (rows + rows_per_iter - 1) / rows_per_iter;
+  int rslices = (int)std::min<int64_t>(std::max<int64_t>(2048 / cblocks, 1),
+                                       std::max<int64_t>(iters / 8, 1));
+  return dim3(cblocks, rslices);
+}
+
 static inline dim3 bn_reduce_grid(int64_t rows, int cols) {
   int cpb = cols < 256 ? cols : 256;
   int cblocks = (cols + cpb - 1) / cpb;
@@ -173,13 +323,25 @@ void bn_stats_launch(DT dt, const void* x, float* mean, float* invstd,
   // they must be zeroed first.
   hipMemsetAsync(mean, 0, cols * sizeof(float), s);
   hipMemsetAsync(invstd, 0, cols * sizeof(float), s);
-  dim3 grid = bn_reduce_grid(rows, cols);
-  if (dt == DT::F32)
-    hipLaunchKernelGGL(k_bn_partial<float>, grid, dim3(256), 0, s,
-                       (const float*)x, mean, invstd, rows, cols);
-  else
-    hipLaunchKernelGGL(k_bn_partial<bf16>, grid, dim3(256), 0, s,
-                       (const bf16*)x, mean, invstd, rows, cols);
+  if (dt == DT::F32) {
+    if (bn_vec_ok<float>(x, cols))
+      hipLaunchKernelGGL(k_bn_partial_vec<float>,
+                         bn_reduce_grid_vec<float>(rows, cols), dim3(256), 0, s,
+                         (const float*)x, mean, invstd, rows, cols);
+    else
+      hipLaunchKernelGGL(k_bn_partial<float>, bn_reduce_grid(rows, cols),
+                         dim3(256), 0, s, (const float*)x, mean, invstd, rows,
+                         cols);
+  } else {
+    if (bn_vec_ok<bf16>(x, cols))
+      hipLaunchKernelGGL(k_bn_partial_vec<bf16>,
+                         bn_reduce_grid_vec<bf16>(rows, cols), dim3(256), 0, s,
+                         (const bf16*)x, mean, invstd, rows, cols);
+    else
+      hipLaunchKernelGGL(k_bn_partial<bf16>, bn_reduce_grid(rows, cols),
+                         dim3(256), 0, s, (const bf16*)x, mean, invstd, rows,
+                         cols);
+  }
   hipLaunchKernelGGL(k_bn_finalize, dim3((cols + 255) / 256), dim3(256), 0, s,
                      mean, invstd, mean, invstd, rmean, rvar, momentum, rows,
                      cols, eps);
@@ -222,15 +384,30 @@ void bn_bwd_reduce_launch(DT dt, const void* x, const void* dy,
                           const float* invstd, float* sum_dy,
                           float* sum_dy_xhat, int64_t rows, int cols,
                           hipStream_t s) {
-  dim3 grid = bn_reduce_grid(rows, cols);
-  if (dt == DT::F32)
-    hipLaunchKernelGGL(k_bn_bwd_reduce<float>, grid, dim3(256), 0, s,
-                       (const float*)x, (const float*)dy, (const float*)y_relu,
-                       mean, invstd, sum_dy, sum_dy_xhat, rows, cols);
-  else
-    hipLaunchKernelGGL(k_bn_bwd_reduce<bf16>, grid, dim3(256), 0, s,
-                       (const bf16*)x, (const bf16*)dy, (const bf16*)y_relu,
-                       mean, invstd, sum_dy, sum_dy_xhat, rows, cols);
+  if (dt == DT::F32) {
+    if (bn_vec_ok<float>(x, cols) && bn_vec_ok<float>(dy, cols))
+      hipLaunchKernelGGL(k_bn_bwd_reduce_vec<float>,
+                         bn_reduce_grid_vec<float>(rows, cols), dim3(256), 0, s,
+                         (const float*)x, (const float*)dy,
+                         (const float*)y_relu, mean, invstd, sum_dy,
+                         sum_dy_xhat, rows, cols);
+    else
+      hipLaunchKernelGGL(k_bn_bwd_reduce<float>, bn_reduce_grid(rows, cols),
+                         dim3(256), 0, s, (const float*)x, (const float*)dy,
+                         (const float*)y_relu, mean, invstd, sum_dy,
+                         sum_dy_xhat, rows, cols);
+  } else {
+    if (bn_vec_ok<bf16>(x, cols) && bn_vec_ok<bf16>(dy, cols))
+      hipLaunchKernelGGL(k_bn_bwd_reduce_vec<bf16>,
+                         bn_reduce_grid_vec<bf16>(rows, cols), dim3(256), 0, s,
+                         (const bf16*)x, (const bf16*)dy, (const bf16*)y_relu,
+                         mean, invstd, sum_dy, sum_dy_xhat, rows, cols);
+    else
+      hipLaunchKernelGGL(k_bn_bwd_reduce<bf16>, bn_reduce_grid(rows, cols),
+                         dim3(256), 0, s, (const bf16*)x, (const bf16*)dy,
+                         (const bf16*)y_relu, mean, invstd, sum_dy,
+                         sum_dy_xhat, rows, cols);
+  }
 }
 
 void bn_bwd_apply_launch(DT dt, const void* x, const void* dy,
