@@ -254,7 +254,8 @@ at::Tensor conv2d_wgrad(const at::Tensor& x, const at::Tensor& dy, int64_t KH,
     wsp = ws.data_ptr<float>();
   }
   conv2d_wgrad_launch(dt_of(x), x.data_ptr(), dy.data_ptr(),
-                      dw.data_ptr<float>(), wsp, z, cs, cur_stream());
+                      dw.data_ptr<float>(), wsp, z, zero_page(x), cs,
+                      cur_stream());
   return dw;
 }
 

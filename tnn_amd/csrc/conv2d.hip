@@ -419,6 +419,102 @@ __global__ void k_conv_wgrad(const T* __restrict__ X, const T* __restrict__ DY,
 }
 
 // ---------------------------------------------------------------------------
+// vector-guaranteed wgrad (Cin%V==0, Cout%V==0, 16B-aligned tensors): stages
+// in three phases -- addresses (zero-page for OOB/padding), then all 16B
+// loads, then the LDS scatter -- so RA+RB global loads stay in flight.
+// The generic kernel's `#pragma unroll 1` staging (needed there to avoid
+// 221-256 VGPR spills from per-iteration gather state) serializes to one
+// load in flight per wave; holding only pointers between phases keeps the
+// live state at ~24 VGPR.
+// ---------------------------------------------------------------------------
+template <typename T, bool POW2>
+__launch_bounds__(THREADS, 3)
+__global__ void k_conv_wgrad_vec(const T* __restrict__ X,
+                                 const T* __restrict__ DY,
+                                 float* __restrict__ DW,
+                                 const T* __restrict__ zero16, ConvShape cs) {
+  constexpr int V = 16 / sizeof(T);
+  constexpr int RA = BK * (BM / V) / THREADS;
+  constexpr int RB = BK * (BN / V) / THREADS;
+  __shared__ alignas(16) T As[BM * BK];  // rows = (kh,kw,ci), k = m
+  __shared__ alignas(16) T Bs[BN * BK];  // rows = co, k = m
+  const int M = cs.N * cs.OH * cs.OW;
+  const int Kout = cs.KH * cs.KW * cs.Cin;
+  const int r0 = blockIdx.x * BM;
+  const int n0 = blockIdx.y * BN;
+  const int m_begin = (int)((int64_t)M * blockIdx.z / gridDim.z);
+  const int m_end = (int)((int64_t)M * (blockIdx.z + 1) / gridDim.z);
+  const WaveCoord wc;
+  f32x4 acc[FM][FN] = {};
+  using VecT = Pack16<T>;
+
+  for (int k0 = m_begin; k0 < m_end; k0 += BK) {
+    const T* asrc[RA];
+#pragma unroll
+    for (int it = 0; it < RA; ++it) {
+      int c = threadIdx.x + it * THREADS;
+      int mm = c / (BM / V);
+      int rr = (c % (BM / V)) * V;
+      int gm = k0 + mm, gr = r0 + rr;
+      const T* p = zero16;
+      if (gm < m_end && gr < Kout) {
+        int n = idiv<POW2>(gm, cs.d_ohow);
+        int rem = gm - n * (cs.OH * cs.OW);
+        int oh = idiv<POW2>(rem, cs.d_ow), ow = rem - oh * cs.OW;
+        int ci = imod<POW2>(gr, cs.d_cin);
+        int kidx = idiv<POW2>(gr, cs.d_cin);
+        int kw = kidx % cs.KW, kh = kidx / cs.KW;
+        int ih = oh * cs.SH - cs.PH + kh;
+        int iw = ow * cs.SW - cs.PW + kw;
+        if (ih >= 0 && ih < cs.H && iw >= 0 && iw < cs.W)
+          p = &X[(((int64_t)n * cs.H + ih) * cs.W + iw) * cs.Cin + ci];
+      }
+      asrc[it] = p;
+    }
+    const T* bsrc[RB];
+#pragma unroll
+    for (int it = 0; it < RB; ++it) {
+      int c = threadIdx.x + it * THREADS;
+      int mm = c / (BN / V);
+      int nn = (c % (BN / V)) * V;
+      int gm = k0 + mm;
+      bsrc[it] = (gm < m_end && n0 + nn + V <= cs.Cout)
+                     ? &DY[(int64_t)gm * cs.Cout + n0 + nn]
+                     : zero16;
+    }
+    VecT va[RA], vb[RB];
+#pragma unroll
+    for (int it = 0; it < RA; ++it) va[it] = *(const VecT*)asrc[it];
+#pragma unroll
+    for (int it = 0; it < RB; ++it) vb[it] = *(const VecT*)bsrc[it];
+#pragma unroll
+    for (int it = 0; it < RA; ++it) {
+      int c = threadIdx.x + it * THREADS;
+      int mm = c / (BM / V);
+      int rr = (c % (BM / V)) * V;
+#pragma unroll
+      for (int j = 0; j < V; ++j) As[lds_off<T>(rr + j, mm)] = va[it].e[j];
+    }
+#pragma unroll
+    for (int it = 0; it < RB; ++it) {
+      int c = threadIdx.x + it * THREADS;
+      int mm = c / (BN / V);
+      int nn = (c % (BN / V)) * V;
+#pragma unroll
+      for (int j = 0; j < V; ++j) Bs[lds_off<T>(nn + j, mm)] = vb[it].e[j];
+    }
+    __syncthreads();
+    mfma_compute_tile(As, Bs, wc, acc);
+    __syncthreads();
+  }
+
+  float* out = DW + (int64_t)blockIdx.z * Kout * cs.Cout;
+  epilogue_visit(wc, acc, r0, n0, [&](int row, int col, float v) {
+    if (row < Kout && col < cs.Cout) out[(int64_t)row * cs.Cout + col] = v;
+  });
+}
+
+// ---------------------------------------------------------------------------
 template <typename T>
 __global__ void k_transpose_w(const T* __restrict__ W, T* __restrict__ WT,
                               int KHW, int Cin, int Cout) {
@@ -528,20 +624,36 @@ int conv2d_wgrad_zsplits(const ConvShape& cs) {
 }
 
 void conv2d_wgrad_launch(DT dt, const void* x, const void* dy, float* dw_f32,
-                         float* ws, int z, const ConvShape& cs,
-                         hipStream_t s) {
+                         float* ws, int z, const void* zero16,
+                         const ConvShape& cs, hipStream_t s) {
   int Kout = cs.KH * cs.KW * cs.Cin;
   float* target = z == 1 ? dw_f32 : ws;
   dim3 grid(ceil_div(Kout, BM), ceil_div(cs.Cout, BN), z);
   bool p2 = all_pow2(cs);
   if (dt == DT::F32) {
-    auto kern = p2 ? k_conv_wgrad<float, true> : k_conv_wgrad<float, false>;
-    hipLaunchKernelGGL(kern, grid, dim3(THREADS), 0, s,
-                       (const float*)x, (const float*)dy, target, cs);
+    bool vec = p2 && cs.Cin % 4 == 0 && cs.Cout % 4 == 0 &&
+               (((uintptr_t)x & 15) == 0) && (((uintptr_t)dy & 15) == 0);
+    if (vec) {
+      hipLaunchKernelGGL((k_conv_wgrad_vec<float, true>), grid, dim3(THREADS),
+                         0, s, (const float*)x, (const float*)dy, target,
+                         (const float*)zero16, cs);
+    } else {
+      auto kern = p2 ? k_conv_wgrad<float, true> : k_conv_wgrad<float, false>;
+      hipLaunchKernelGGL(kern, grid, dim3(THREADS), 0, s, (const float*)x,
+                         (const float*)dy, target, cs);
+    }
   } else {
-    auto kern = p2 ? k_conv_wgrad<bf16, true> : k_conv_wgrad<bf16, false>;
-    hipLaunchKernelGGL(kern, grid, dim3(THREADS), 0, s,
-                       (const bf16*)x, (const bf16*)dy, target, cs);
+    bool vec = p2 && cs.Cin % 8 == 0 && cs.Cout % 8 == 0 &&
+               (((uintptr_t)x & 15) == 0) && (((uintptr_t)dy & 15) == 0);
+    if (vec) {
+      hipLaunchKernelGGL((k_conv_wgrad_vec<bf16, true>), grid, dim3(THREADS),
+                         0, s, (const bf16*)x, (const bf16*)dy, target,
+                         (const bf16*)zero16, cs);
+    } else {
+      auto kern = p2 ? k_conv_wgrad<bf16, true> : k_conv_wgrad<bf16, false>;
+      hipLaunchKernelGGL(kern, grid, dim3(THREADS), 0, s, (const bf16*)x,
+                         (const bf16*)dy, target, cs);
+    }
   }
   if (z > 1)
     splitk_reduce_launch(ws, dw_f32, z, (int64_t)Kout * cs.Cout, s);

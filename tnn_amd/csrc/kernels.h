@@ -73,8 +73,8 @@ void conv2d_dgrad_launch(DT dt, const void* dy, const void* w_t, void* dx,
                          hipStream_t s);
 int conv2d_wgrad_zsplits(const ConvShape& cs);
 void conv2d_wgrad_launch(DT dt, const void* x, const void* dy, float* dw_f32,
-                         float* ws, int z, const ConvShape& cs,
-                         hipStream_t s);
+                         float* ws, int z, const void* zero16,
+                         const ConvShape& cs, hipStream_t s);
 void transpose_w_launch(DT dt, const void* w, void* w_t, int KH, int KW,
                         int Cin, int Cout, hipStream_t s);
 
