@@ -55,6 +55,18 @@ def test_mfma_selftest_f32():
 # ---------------------------------------------------------------------------
 
 @pytest.mark.parametrize("dtype", DTYPES)
+@pytest.mark.parametrize("cols", [16, 100, 128, 2304, 50264])
+def test_colsum(dtype, cols):
+    # 2304 spans >1 channel-block in the vectorized kernel (regression:
+    # the final per-channel loop wrote past cols without a bound check)
+    torch.manual_seed(11)
+    x = torch.randn(777, cols, dtype=dtype, device=DEV)
+    out = ext.colsum(x)
+    ref = x.float().sum(0)
+    assert relerr(out, ref) < 1e-2, cols
+
+
+@pytest.mark.parametrize("dtype", DTYPES)
 @pytest.mark.parametrize("M,N,K", [(128, 64, 32), (256, 128, 512),
                                    (100, 10, 27), (64, 100, 512),
                                    (130, 70, 33)])

@@ -101,6 +101,7 @@ __global__ void k_bn_partial_vec(const T* __restrict__ x,
       tss += sh_ss[(r * gpb + gl) * V + j];
     }
     int c = blockIdx.x * cpb_total + cl;
+    if (c >= cols) break;  // last block may cover fewer than gpb groups
     atomicAdd(&sum[c], ts);
     atomicAdd(&sumsq[c], tss);
   }
@@ -172,6 +173,7 @@ __global__ void k_bn_bwd_reduce_vec(const T* __restrict__ x,
       tsx += sh_sx[(r * gpb + gl) * V + j];
     }
     int c = blockIdx.x * cpb_total + cl;
+    if (c >= cols) break;  // last block may cover fewer than gpb groups
     atomicAdd(&sum_dy[c], ts);
     atomicAdd(&sum_dy_xhat[c], tsx);
   }

@@ -142,7 +142,7 @@ at::Tensor gemm_tn(const at::Tensor& a, const at::Tensor& b) {
     wsp = ws.data_ptr<float>();
   }
   gemm_tn_launch(dt_of(a), a.data_ptr(), b.data_ptr(), c.data_ptr<float>(),
-                 wsp, z, M, N, K, cur_stream());
+                 wsp, z, zero_page(a), M, N, K, cur_stream());
   return c;
 }
 

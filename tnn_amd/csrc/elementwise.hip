@@ -201,6 +201,7 @@ __global__ void k_colsum_vec(const T* __restrict__ x, float* __restrict__ out,
     float t = 0.0f;
     for (int r = 0; r < rows_per_iter; ++r) t += sh[(r * gpb + gl) * V + j];
     int c = blockIdx.x * cpb_total + cl;
+    if (c >= cols) break;  // last block may cover fewer than gpb groups
     if (gridDim.y == 1)
       out[c] = t;
     else

@@ -215,6 +215,80 @@ __global__ void k_gemm_tn(const T* __restrict__ A, const T* __restrict__ B,
   });
 }
 
+// vector-guaranteed TN gemm (K%V==0, N%V==0, aligned): 3-phase staging --
+// addresses (zero-page OOB), all 16B loads, LDS scatter -- keeps RA+RB
+// loads in flight where `#pragma unroll 1` serializes to one (see
+// k_conv_wgrad_vec).
+template <typename T>
+__launch_bounds__(THREADS, 3)
+__global__ void k_gemm_tn_vec(const T* __restrict__ A, const T* __restrict__ B,
+                              float* __restrict__ C,
+                              const T* __restrict__ zero16, int M, int N,
+                              int K) {
+  constexpr int V = 16 / sizeof(T);
+  constexpr int RA = BK * (BM / V) / THREADS;
+  constexpr int RB = BK * (BN / V) / THREADS;
+  __shared__ alignas(16) T As[BM * BK];
+  __shared__ alignas(16) T Bs[BN * BK];
+  const int r0 = blockIdx.x * BM;
+  const int n0 = blockIdx.y * BN;
+  const int m_begin = (int)((int64_t)M * blockIdx.z / gridDim.z);
+  const int m_end = (int)((int64_t)M * (blockIdx.z + 1) / gridDim.z);
+  const WaveCoord wc;
+  f32x4 acc[FM][FN] = {};
+  using VecT = Pack16<T>;
+
+  for (int k0 = m_begin; k0 < m_end; k0 += BK) {
+    const T* asrc[RA];
+#pragma unroll
+    for (int it = 0; it < RA; ++it) {
+      int c = threadIdx.x + it * THREADS;
+      int mm = c / (BM / V);
+      int rr = (c % (BM / V)) * V;
+      int gm = k0 + mm, gr = r0 + rr;
+      asrc[it] = (gm < m_end && gr < K) ? &A[(int64_t)gm * K + gr] : zero16;
+    }
+    const T* bsrc[RB];
+#pragma unroll
+    for (int it = 0; it < RB; ++it) {
+      int c = threadIdx.x + it * THREADS;
+      int mm = c / (BN / V);
+      int nn = (c % (BN / V)) * V;
+      int gm = k0 + mm, gn = n0 + nn;
+      bsrc[it] = (gm < m_end && gn < N) ? &B[(int64_t)gm * N + gn] : zero16;
+    }
+    VecT va[RA], vb[RB];
+#pragma unroll
+    for (int it = 0; it < RA; ++it) va[it] = *(const VecT*)asrc[it];
+#pragma unroll
+    for (int it = 0; it < RB; ++it) vb[it] = *(const VecT*)bsrc[it];
+#pragma unroll
+    for (int it = 0; it < RA; ++it) {
+      int c = threadIdx.x + it * THREADS;
+      int mm = c / (BM / V);
+      int rr = (c % (BM / V)) * V;
+#pragma unroll
+      for (int j = 0; j < V; ++j) As[lds_off<T>(rr + j, mm)] = va[it].e[j];
+    }
+#pragma unroll
+    for (int it = 0; it < RB; ++it) {
+      int c = threadIdx.x + it * THREADS;
+      int mm = c / (BN / V);
+      int nn = (c % (BN / V)) * V;
+#pragma unroll
+      for (int j = 0; j < V; ++j) Bs[lds_off<T>(nn + j, mm)] = vb[it].e[j];
+    }
+    __syncthreads();
+    mfma_compute_tile(As, Bs, wc, acc);
+    __syncthreads();
+  }
+
+  float* out = C + (int64_t)blockIdx.z * K * N;
+  epilogue_visit(wc, acc, r0, n0, [&](int row, int col, float v) {
+    if (row < K && col < N) out[(int64_t)row * N + col] = v;
+  });
+}
+
 __global__ void k_splitk_reduce(const float* __restrict__ ws,
                                 float* __restrict__ out, int z, int64_t n) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -311,15 +385,29 @@ int gemm_tn_zsplits(int M, int N, int K) {
 }
 
 void gemm_tn_launch(DT dt, const void* a, const void* b, float* c_f32,
-                    float* ws, int z, int M, int N, int K, hipStream_t s) {
+                    float* ws, int z, const void* zero16, int M, int N, int K,
+                    hipStream_t s) {
   dim3 grid(ceil_div(K, BM), ceil_div(N, BN), z);
   float* target = z == 1 ? c_f32 : ws;
-  if (dt == DT::F32)
-    hipLaunchKernelGGL(k_gemm_tn<float>, grid, dim3(THREADS), 0, s,
-                       (const float*)a, (const float*)b, target, M, N, K);
-  else
-    hipLaunchKernelGGL(k_gemm_tn<bf16>, grid, dim3(THREADS), 0, s,
-                       (const bf16*)a, (const bf16*)b, target, M, N, K);
+  if (dt == DT::F32) {
+    if (K % 4 == 0 && N % 4 == 0 && (((uintptr_t)a & 15) == 0) &&
+        (((uintptr_t)b & 15) == 0))
+      hipLaunchKernelGGL(k_gemm_tn_vec<float>, grid, dim3(THREADS), 0, s,
+                         (const float*)a, (const float*)b, target,
+                         (const float*)zero16, M, N, K);
+    else
+      hipLaunchKernelGGL(k_gemm_tn<float>, grid, dim3(THREADS), 0, s,
+                         (const float*)a, (const float*)b, target, M, N, K);
+  } else {
+    if (K % 8 == 0 && N % 8 == 0 && (((uintptr_t)a & 15) == 0) &&
+        (((uintptr_t)b & 15) == 0))
+      hipLaunchKernelGGL(k_gemm_tn_vec<bf16>, grid, dim3(THREADS), 0, s,
+                         (const bf16*)a, (const bf16*)b, target,
+                         (const bf16*)zero16, M, N, K);
+    else
+      hipLaunchKernelGGL(k_gemm_tn<bf16>, grid, dim3(THREADS), 0, s,
+                         (const bf16*)a, (const bf16*)b, target, M, N, K);
+  }
   if (z > 1) splitk_reduce_launch(ws, c_f32, z, (int64_t)K * N, s);
 }
 

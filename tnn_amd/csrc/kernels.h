@@ -32,7 +32,8 @@ void gemm_launch(DT dt, const void* a, const void* b, const void* bias,
                  bool trans_b, int act_kind, hipStream_t s);
 int gemm_tn_zsplits(int M, int N, int K);
 void gemm_tn_launch(DT dt, const void* a, const void* b, float* c_f32,
-                    float* ws, int z, int M, int N, int K, hipStream_t s);
+                    float* ws, int z, const void* zero16, int M, int N, int K,
+                    hipStream_t s);
 void splitk_reduce_launch(const float* ws, float* out, int z, int64_t n,
                           hipStream_t s);
 void mfma_selftest_launch(const void* a_bf16, const void* b_bf16, float* d,
