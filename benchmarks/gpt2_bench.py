@@ -16,7 +16,7 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from tnn_amd import models
 from tnn_amd.nn import CrossEntropyLoss, AdamW
 from tnn_amd.nn.layer import cast_compute_dtype
-from tnn_amd.models.generate import generate, generate_cached
+from tnn_amd.models.generate import generate, generate_cached, generate_graphed
 
 
 def main():
@@ -67,11 +67,12 @@ def main():
     # ---- greedy decode (reference gpt2_inference loop) ----
     # note: at batch 1 and short sequences decode is kernel-launch-bound on
     # a 2.5PF GPU, so the full-recompute loop (one big batch of launches)
-    # can beat the KV cache (many tiny launches); the cache wins as the
-    # sequence grows. hipGraph capture of the decode step is future work.
+    # can beat the eager KV cache (many tiny launches); the hipGraph path
+    # captures the whole per-token step into one graph launch.
     model.eval()
     for name, fn in [("recompute (reference parity)", generate),
-                     ("kv-cache", generate_cached)]:
+                     ("kv-cache", generate_cached),
+                     ("hipGraph", generate_graphed)]:
         t0 = time.perf_counter()
         out = fn(model, list(range(16)), max_new_tokens=args.decode_tokens,
                  seq_len=args.seq_len, device=dev, eot_token=None)

@@ -69,7 +69,8 @@ def test_colsum(dtype, cols):
 @pytest.mark.parametrize("dtype", DTYPES)
 @pytest.mark.parametrize("M,N,K", [(128, 64, 32), (256, 128, 512),
                                    (100, 10, 27), (64, 100, 512),
-                                   (130, 70, 33)])
+                                   (130, 70, 33),
+                                   (1, 3072, 768), (1, 97, 33)])
 def test_gemm_nn(dtype, M, N, K):
     torch.manual_seed(1)
     a = torch.randn(M, K, dtype=dtype, device=DEV)
@@ -91,7 +92,8 @@ def test_gemm_relu_epilogue(dtype):
 
 
 @pytest.mark.parametrize("dtype", DTYPES)
-@pytest.mark.parametrize("M,N,K", [(128, 64, 32), (100, 512, 100), (64, 27, 130)])
+@pytest.mark.parametrize("M,N,K", [(128, 64, 32), (100, 512, 100), (64, 27, 130),
+                                   (1, 3072, 768), (1, 100, 33)])
 def test_gemm_nt(dtype, M, N, K):
     torch.manual_seed(2)
     a = torch.randn(M, K, dtype=dtype, device=DEV)
@@ -534,3 +536,23 @@ def test_graphed_inference_matches_eager():
         ref = m(x1)
     out = g(x1)
     assert torch.equal(out, ref)
+
+
+def test_generate_graphed_gpu_matches_recompute():
+    # real hipGraph capture of the decode step vs full recompute
+    from tnn_amd.nn import LayerBuilder
+    from tnn_amd.models.generate import generate, generate_graphed
+    torch.manual_seed(3)
+    m = (LayerBuilder((32,))
+         .embedding(96, 64, "tok")
+         .positional_embedding(32, "pos")
+         .gpt_block(8, 2, flash=False, name="b0")
+         .gpt_block(8, 2, flash=False, name="b1")
+         .layernorm(name="ln_f")
+         .dense(96, True, "head")
+         .build("tiny_gpt")).to(DEV)
+    ref = generate(m, [1, 2, 3], max_new_tokens=12, seq_len=32,
+                   eot_token=None, device=torch.device(DEV))
+    fast = generate_graphed(m, [1, 2, 3], max_new_tokens=12, seq_len=32,
+                            eot_token=None, device=torch.device(DEV))
+    assert fast == ref

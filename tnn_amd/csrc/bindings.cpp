@@ -107,6 +107,14 @@ at::Tensor gemm(const at::Tensor& a, const at::Tensor& b,
   int M = a.size(0), K = a.size(1), N = b.size(1);
   auto c = at::empty({M, N}, a.options());
   const void* bp = bias.has_value() ? bias->data_ptr() : nullptr;
+  if (M == 1) {  // decode-path matvec: K-split partials + finalize
+    int ks = gemv_nn_ksplits(N, K);
+    auto ws = at::empty({(int64_t)ks * N}, a.options().dtype(at::kFloat));
+    gemv_nn_launch(dt_of(a), a.data_ptr(), b.data_ptr(), bp, c.data_ptr(),
+                   ws.data_ptr<float>(), ks, N, K, (int)act_kind,
+                   cur_stream());
+    return c;
+  }
   gemm_launch(dt_of(a), a.data_ptr(), b.data_ptr(), bp, c.data_ptr(),
               zero_page(a), M, N, K, /*trans_b=*/false, (int)act_kind,
               cur_stream());

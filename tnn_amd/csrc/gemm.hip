@@ -215,6 +215,85 @@ __global__ void k_gemm_tn(const T* __restrict__ A, const T* __restrict__ B,
   });
 }
 
+// M=1 NT matvec (decode-path linears): one wave per output row, 16B loads
+// over K, cross-lane reduce. The MFMA tile kernel wastes 127/128 of its
+// A-tile rows at M=1 and is ~30x slower on gpt2 decode shapes.
+template <typename T>
+__launch_bounds__(256)
+__global__ void k_gemv_nt(const T* __restrict__ x, const T* __restrict__ B,
+                          const float* __restrict__ bias_f32,
+                          const T* __restrict__ bias_t, T* __restrict__ y,
+                          int N, int K, int act_kind) {
+  constexpr int V = 16 / sizeof(T);
+  using VecT = Pack16<T>;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int n = blockIdx.x * 4 + wave;
+  if (n >= N) return;
+  const T* row = &B[(int64_t)n * K];
+  float acc = 0.0f;
+  if ((K % V) == 0 && aligned16(row) && aligned16(x)) {
+    for (int k = lane * V; k < K; k += 64 * V) {
+      VecT vb = *(const VecT*)&row[k];
+      VecT vx = *(const VecT*)&x[k];
+#pragma unroll
+      for (int j = 0; j < V; ++j)
+        acc += VecIO<T>::to_f32(vx.e[j]) * VecIO<T>::to_f32(vb.e[j]);
+    }
+  } else {
+    for (int k = lane; k < K; k += 64)
+      acc += VecIO<T>::to_f32(x[k]) * VecIO<T>::to_f32(row[k]);
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    acc += __shfl_xor(acc, off, 64);
+  if (lane == 0) {
+    if (bias_f32) acc += bias_f32[n];
+    if (bias_t) acc += VecIO<T>::to_f32(bias_t[n]);
+    if (act_kind != ACT_LINEAR) acc = act_apply(acc, act_kind);
+    y[n] = VecIO<T>::from_f32(acc);
+  }
+}
+
+// M=1 NN matvec: out[N] = x[K] @ B[K,N] with B row-major over N (the
+// layout every linear layer stores). Split K over grid.y into fp32
+// partials, then a finalize pass adds bias/activation. Two tiny launches
+// beat one M=1 MFMA tile kernel ~6x on gpt2 decode shapes.
+template <typename T>
+__launch_bounds__(256)
+__global__ void k_gemv_nn_part(const T* __restrict__ x, const T* __restrict__ B,
+                               float* __restrict__ ws, int N, int K) {
+  int n = blockIdx.x * 256 + threadIdx.x;
+  int ks = blockIdx.y, KS = gridDim.y;
+  if (n >= N) return;
+  int k0 = (int)((int64_t)K * ks / KS), k1 = (int)((int64_t)K * (ks + 1) / KS);
+  float a0 = 0.0f, a1 = 0.0f;
+  int k = k0;
+  for (; k + 2 <= k1; k += 2) {
+    a0 += VecIO<T>::to_f32(x[k]) * VecIO<T>::to_f32(B[(int64_t)k * N + n]);
+    a1 += VecIO<T>::to_f32(x[k + 1]) *
+          VecIO<T>::to_f32(B[(int64_t)(k + 1) * N + n]);
+  }
+  if (k < k1) a0 += VecIO<T>::to_f32(x[k]) * VecIO<T>::to_f32(B[(int64_t)k * N + n]);
+  ws[(int64_t)ks * N + n] = a0 + a1;
+}
+
+template <typename T>
+__launch_bounds__(256)
+__global__ void k_gemv_nn_fin(const float* __restrict__ ws,
+                              const float* __restrict__ bias_f32,
+                              const T* __restrict__ bias_t, T* __restrict__ y,
+                              int N, int KS, int act_kind) {
+  int n = blockIdx.x * 256 + threadIdx.x;
+  if (n >= N) return;
+  float a = 0.0f;
+  for (int s = 0; s < KS; ++s) a += ws[(int64_t)s * N + n];
+  if (bias_f32) a += bias_f32[n];
+  if (bias_t) a += VecIO<T>::to_f32(bias_t[n]);
+  if (act_kind != ACT_LINEAR) a = act_apply(a, act_kind);
+  y[n] = VecIO<T>::from_f32(a);
+}
+
 // vector-guaranteed TN gemm (K%V==0, N%V==0, aligned): 3-phase staging --
 // addresses (zero-page OOB), all 16B loads, LDS scatter -- keeps RA+RB
 // loads in flight where `#pragma unroll 1` serializes to one (see
@@ -354,6 +433,18 @@ void splitk_reduce_launch(const float* ws, float* out, int z, int64_t n,
 void gemm_launch(DT dt, const void* a, const void* b, const void* bias,
                  void* c, const void* zero16, int M, int N, int K,
                  bool trans_b, int act_kind, hipStream_t s) {
+  if (M == 1 && trans_b) {  // decode-path matvec
+    dim3 gg(ceil_div(N, 4));
+    if (dt == DT::F32)
+      hipLaunchKernelGGL(k_gemv_nt<float>, gg, dim3(256), 0, s,
+                         (const float*)a, (const float*)b, (const float*)bias,
+                         (const float*)nullptr, (float*)c, N, K, act_kind);
+    else
+      hipLaunchKernelGGL(k_gemv_nt<bf16>, gg, dim3(256), 0, s, (const bf16*)a,
+                         (const bf16*)b, (const float*)nullptr,
+                         (const bf16*)bias, (bf16*)c, N, K, act_kind);
+    return;
+  }
   dim3 grid(ceil_div(M, BM), ceil_div(N, BN));
   dim3 blk(THREADS);
   if (dt == DT::F32) {
@@ -374,6 +465,33 @@ void gemm_launch(DT dt, const void* a, const void* b, const void* bias,
     hipLaunchKernelGGL(kern, grid, blk, 0, s, (const bf16*)a, (const bf16*)b,
                        (const float*)nullptr, (const bf16*)bias, (bf16*)c,
                        (const bf16*)zero16, M, N, K, act_kind);
+  }
+}
+
+int gemv_nn_ksplits(int N, int K) {
+  int ncb = ceil_div(N, 256);
+  int ks = std::max(1, 512 / ncb);
+  ks = std::min(ks, std::max(1, K / 16));
+  return ks;
+}
+
+void gemv_nn_launch(DT dt, const void* x, const void* b, const void* bias,
+                    void* y, float* ws, int ks, int N, int K, int act_kind,
+                    hipStream_t s) {
+  dim3 pg(ceil_div(N, 256), ks);
+  dim3 fg(ceil_div(N, 256));
+  if (dt == DT::F32) {
+    hipLaunchKernelGGL(k_gemv_nn_part<float>, pg, dim3(256), 0, s,
+                       (const float*)x, (const float*)b, ws, N, K);
+    hipLaunchKernelGGL(k_gemv_nn_fin<float>, fg, dim3(256), 0, s, ws,
+                       (const float*)bias, (const float*)nullptr, (float*)y, N,
+                       ks, act_kind);
+  } else {
+    hipLaunchKernelGGL(k_gemv_nn_part<bf16>, pg, dim3(256), 0, s,
+                       (const bf16*)x, (const bf16*)b, ws, N, K);
+    hipLaunchKernelGGL(k_gemv_nn_fin<bf16>, fg, dim3(256), 0, s, ws,
+                       (const float*)nullptr, (const bf16*)bias, (bf16*)y, N,
+                       ks, act_kind);
   }
 }
 

@@ -30,6 +30,10 @@ void cast_f32_launch(DT dt_out, const float* x, void* y, int64_t n,
 void gemm_launch(DT dt, const void* a, const void* b, const void* bias,
                  void* c, const void* zero16, int M, int N, int K,
                  bool trans_b, int act_kind, hipStream_t s);
+int gemv_nn_ksplits(int N, int K);
+void gemv_nn_launch(DT dt, const void* x, const void* b, const void* bias,
+                    void* y, float* ws, int ks, int N, int K, int act_kind,
+                    hipStream_t s);
 int gemm_tn_zsplits(int M, int N, int K);
 void gemm_tn_launch(DT dt, const void* a, const void* b, float* c_f32,
                     float* ws, int z, const void* zero16, int M, int N, int K,

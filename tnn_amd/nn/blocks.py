@@ -190,6 +190,16 @@ class _MHABase(Layer):
         # appends are in-place slice writes, never torch.cat reallocation
         self._kv_cache = {"k": None, "v": None, "len": 0, "cap": max_len}
 
+    def enable_static_cache(self, max_len: int, pos_t: torch.Tensor):
+        """hipGraph-capturable decode cache: every tensor shape is fixed at
+        ``max_len`` and the current position lives in the shared int64
+        device tensor ``pos_t`` ([1]), so a captured single-token step can
+        be replayed with no host round-trips. Prefill (s>1) still runs the
+        eager path; single-token steps write k/v at pos_t via index_copy_
+        and attend over the full capacity with a pos_t-derived mask."""
+        self._kv_cache = {"k": None, "v": None, "len": 0, "cap": max_len,
+                          "pos_t": pos_t}
+
     def _project(self, x):
         b, s, d = x.shape
         h, hd = self.num_heads, self.head_dim
@@ -199,9 +209,14 @@ class _MHABase(Layer):
         c = self._kv_cache
         if c is not None:
             if c["k"] is None:
-                c["k"] = torch.empty(b, h, c["cap"], hd, device=x.device,
+                c["k"] = torch.zeros(b, h, c["cap"], hd, device=x.device,
                                      dtype=k.dtype)
-                c["v"] = torch.empty_like(c["k"])
+                c["v"] = torch.zeros_like(c["k"])
+            if "pos_t" in c and s == 1:
+                # graph-capturable single-token append at pos_t
+                c["k"].index_copy_(2, c["pos_t"], k.detach())
+                c["v"].index_copy_(2, c["pos_t"], v.detach())
+                return q, c["k"], c["v"]
             t0, t1 = c["len"], c["len"] + s
             c["k"][:, :, t0:t1] = k.detach()
             c["v"][:, :, t0:t1] = v.detach()
@@ -214,11 +229,21 @@ class _MHABase(Layer):
         """Decode-path attention: q covers the s_new newest positions of
         the k/v sequence; causal within the suffix, full over the prefix
         (the KV-cache fast path the reference lacks —
-        examples/gpt2_inference.cpp recomputes the full sequence)."""
+        examples/gpt2_inference.cpp recomputes the full sequence).
+
+        Static mode (s_new == 1 with a pos_t cache): k/v span the full
+        capacity and positions beyond pos_t are masked from pos_t itself,
+        keeping every shape fixed for hipGraph capture."""
+        c = self._kv_cache
         t, s_new = k.shape[-2], q.shape[-2]
         scale = self.head_dim ** -0.5
         scores = torch.matmul(q.float(), k.float().transpose(-1, -2)) * scale
-        if self.causal and s_new > 1:
+        if c is not None and "pos_t" in c and s_new == 1:
+            if "arange" not in c:
+                c["arange"] = torch.arange(c["cap"], device=q.device)
+            scores = scores.masked_fill(c["arange"] > c["pos_t"],
+                                        float("-inf"))
+        elif self.causal and s_new > 1:
             pos_q = torch.arange(t - s_new, t, device=q.device).unsqueeze(-1)
             pos_k = torch.arange(t, device=q.device)
             scores = scores.masked_fill(pos_k > pos_q, float("-inf"))

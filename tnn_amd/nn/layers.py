@@ -329,10 +329,13 @@ class PositionalEmbedding(Layer):
         self.max_len, self.dim = max_len, dim
         self.weight = nn.Parameter(torch.randn(max_len, dim, dtype=dtype) * 0.02)
 
-    _pos_offset = 0  # set by cached decode (models/generate.py)
+    _pos_offset = 0      # set by cached decode (models/generate.py)
+    _pos_tensor = None   # device int64 [1]: graph-capturable position
 
     def forward(self, x):
         s = x.shape[-2]
+        if self._pos_tensor is not None and s == 1:
+            return x + self.weight.index_select(0, self._pos_tensor)
         off = self._pos_offset
         return x + self.weight[off:off + s]
 
