@@ -208,8 +208,17 @@ at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
                        ph, pw);
   auto y = at::empty({cs.N, cs.OH, cs.OW, cs.Cout}, x.options());
   const void* bp = bias.has_value() ? bias->data_ptr() : nullptr;
-  conv2d_fwd_launch(dt_of(x), x.data_ptr(), w.data_ptr(), bp, y.data_ptr(),
-                    zero_page(x), cs, relu, cur_stream());
+  at::Tensor wt2;
+  const void* wt2p = nullptr;
+  if (conv2d_fwd_wants_db(dt_of(x), x.data_ptr(), cs)) {
+    wt2 = at::empty({(int64_t)cs.Cout, (int64_t)cs.KH * cs.KW * cs.Cin},
+                    w.options());
+    transpose_w_fwd_launch(dt_of(x), w.data_ptr(), wt2.data_ptr(),
+                           cs.KH * cs.KW, cs.Cin, cs.Cout, cur_stream());
+    wt2p = wt2.data_ptr();
+  }
+  conv2d_fwd_launch(dt_of(x), x.data_ptr(), w.data_ptr(), wt2p, bp,
+                    y.data_ptr(), zero_page(x), cs, relu, cur_stream());
   return y;
 }
 
@@ -234,13 +243,22 @@ at::Tensor conv2d_dgrad(const at::Tensor& dy, const at::Tensor& w, int64_t H,
   cs.OH = dy.size(1);
   cs.OW = dy.size(2);
   cs.init_fdiv();
+  auto dx = at::empty({cs.N, H, W, Cin}, dy.options());
+  if (conv2d_dgrad_wants_db(dt_of(dy), dy.data_ptr(), cs)) {
+    // double-buffered path: weight as [Cin][(kh,kw,co)] k-contiguous rows
+    auto w_t2d = at::empty({Cin, (int64_t)KH * KW * Cout}, w.options());
+    transpose_w_dgrad_launch(dt_of(w), w.data_ptr(), w_t2d.data_ptr(), KH * KW,
+                             Cin, Cout, cur_stream());
+    conv2d_dgrad_launch(dt_of(dy), dy.data_ptr(), nullptr, w_t2d.data_ptr(),
+                        dx.data_ptr(), zero_page(dy), cs, cur_stream());
+    return dx;
+  }
   // dgrad consumes the weight as B[(kh,kw,co)][ci]: transpose once per call
   auto w_t = at::empty({KH, KW, Cout, Cin}, w.options());
   transpose_w_launch(dt_of(w), w.data_ptr(), w_t.data_ptr(), KH, KW, Cin, Cout,
                      cur_stream());
-  auto dx = at::empty({cs.N, H, W, Cin}, dy.options());
-  conv2d_dgrad_launch(dt_of(dy), dy.data_ptr(), w_t.data_ptr(), dx.data_ptr(),
-                      zero_page(dy), cs, cur_stream());
+  conv2d_dgrad_launch(dt_of(dy), dy.data_ptr(), w_t.data_ptr(), nullptr,
+                      dx.data_ptr(), zero_page(dy), cs, cur_stream());
   return dx;
 }
 

@@ -38,6 +38,83 @@ DEV int imod(int x, const IDiv& f) {
 }
 
 // ---------------------------------------------------------------------------
+// double-buffered pure-glds forward: both operands DMA straight to LDS
+// (weights pre-transposed to [Cout, K] so B rows are k-contiguous), the
+// next chunk's DMAs overlap the current chunk's MFMAs. Counted
+// s_waitcnt vmcnt + raw s_barrier per the CDNA4 glds idiom -- a
+// __syncthreads() here would drain the in-flight next-chunk DMAs.
+template <typename T, bool POW2>
+__launch_bounds__(THREADS)
+__global__ void k_conv_fwd_db(const T* __restrict__ X,
+                              const T* __restrict__ WT2,
+                              const float* __restrict__ bias_f32,
+                              const T* __restrict__ bias_t, T* __restrict__ Y,
+                              const T* __restrict__ zero16, ConvShape cs,
+                              int act_kind) {
+  __shared__ alignas(16) T As[2][BM * BK];
+  __shared__ alignas(16) T Bs[2][BN * BK];
+
+  const int M = cs.N * cs.OH * cs.OW;
+  const int K = cs.KH * cs.KW * cs.Cin;
+  const int m0 = blockIdx.x * BM;
+  const int n0 = blockIdx.y * BN;
+  const WaveCoord wc;
+  f32x4 acc[FM][FN] = {};
+
+  auto a_src = [&](int kk0, int rl, int kk) -> const T* {
+    int gm = m0 + rl, gk = kk0 + kk;
+    if (gm >= M || gk >= K) return zero16;
+    int n = idiv<POW2>(gm, cs.d_ohow);
+    int rem = gm - n * (cs.OH * cs.OW);
+    int oh = idiv<POW2>(rem, cs.d_ow), ow = rem - oh * cs.OW;
+    int ci = imod<POW2>(gk, cs.d_cin);
+    int kidx = idiv<POW2>(gk, cs.d_cin);
+    int kw = kidx % cs.KW, kh = kidx / cs.KW;
+    int ih = oh * cs.SH - cs.PH + kh;
+    int iw = ow * cs.SW - cs.PW + kw;
+    if (ih < 0 || ih >= cs.H || iw < 0 || iw >= cs.W) return zero16;
+    return &X[(((int64_t)n * cs.H + ih) * cs.W + iw) * cs.Cin + ci];
+  };
+  auto b_src = [&](int kk0, int rl, int kk) -> const T* {
+    int gn = n0 + rl, gk = kk0 + kk;
+    if (gn >= cs.Cout || gk >= K) return zero16;
+    return &WT2[(int64_t)gn * K + gk];
+  };
+  auto stage = [&](int t, int which) {
+    int kk0 = t * BK;
+    glds_stage<T, BM>(As[which], wc,
+                      [&](int rl, int kk) { return a_src(kk0, rl, kk); });
+    glds_stage<T, BN>(Bs[which], wc,
+                      [&](int rl, int kk) { return b_src(kk0, rl, kk); });
+  };
+  constexpr int NPER = glds_count<T, BM>() + glds_count<T, BN>();
+
+  const int nch = (K + BK - 1) / BK;
+  stage(0, 0);
+  for (int t = 0; t < nch; ++t) {
+    const int cur = t & 1;
+    if (t + 1 < nch) {
+      stage(t + 1, cur ^ 1);
+      wait_vmcnt<NPER>();   // chunk t landed; t+1 stays in flight
+    } else {
+      wait_vmcnt<0>();
+    }
+    __builtin_amdgcn_s_barrier();
+    mfma_compute_tile(As[cur], Bs[cur], wc, acc);
+    __builtin_amdgcn_s_barrier();   // cur free for the t+2 stage
+  }
+
+  epilogue_visit(wc, acc, m0, n0, [&](int row, int col, float v) {
+    if (row < M && col < cs.Cout) {
+      if (bias_f32) v += bias_f32[col];
+      if (bias_t) v += VecIO<T>::to_f32(bias_t[col]);
+      if (act_kind != ACT_LINEAR) v = act_apply(v, act_kind);
+      Y[(int64_t)row * cs.Cout + col] = VecIO<T>::from_f32(v);
+    }
+  });
+}
+
+// ---------------------------------------------------------------------------
 template <typename T, bool POW2, bool GLDS>
 __launch_bounds__(THREADS)
 __global__ void k_conv_fwd(const T* __restrict__ X, const T* __restrict__ Wt,
@@ -321,6 +398,116 @@ __global__ void k_conv_dgrad(const T* __restrict__ DY, const T* __restrict__ WT,
 }
 
 // ---------------------------------------------------------------------------
+// double-buffered pure-glds dgrad (see k_conv_fwd_db): the weight comes
+// pre-transposed to [Cin, KH*KW*Cout] so B rows are k-contiguous; with S2
+// the compacted parity-class k index remaps into the full column space.
+template <typename T, bool POW2, bool S2 = false>
+__launch_bounds__(THREADS)
+__global__ void k_conv_dgrad_db(const T* __restrict__ DY,
+                                const T* __restrict__ WT2D,
+                                T* __restrict__ DX,
+                                const T* __restrict__ zero16, ConvShape cs) {
+  __shared__ alignas(16) T As[2][BM * BK];
+  __shared__ alignas(16) T Bs[2][BN * BK];
+
+  const int cls_a = S2 ? ((int)blockIdx.z >> 1) : 0;
+  const int cls_b = S2 ? ((int)blockIdx.z & 1) : 0;
+  const int start_h = S2 ? ((cls_a + cs.PH) & 1) : 0;
+  const int start_w = S2 ? ((cls_b + cs.PW) & 1) : 0;
+  const int nkh = S2 ? ((cs.KH - start_h + 1) >> 1) : cs.KH;
+  const int nkw = S2 ? ((cs.KW - start_w + 1) >> 1) : cs.KW;
+  const int Hc = S2 ? (cs.H >> 1) : cs.H;
+  const int Wc = S2 ? (cs.W >> 1) : cs.W;
+  const int M = cs.N * Hc * Wc;
+  const int K = nkh * nkw * cs.Cout;
+  const int KFULL = cs.KH * cs.KW * cs.Cout;
+  const int m0 = blockIdx.x * BM;
+  const int n0 = blockIdx.y * BN;
+  const WaveCoord wc;
+  f32x4 acc[FM][FN] = {};
+
+  auto a_src = [&](int kk0, int rl, int kk) -> const T* {
+    int gm = m0 + rl, gk = kk0 + kk;
+    if (gm >= M || gk >= K) return zero16;
+    int n = idiv<POW2>(gm, cs.d_hw);
+    int rem = gm - n * (Hc * Wc);
+    int ih = idiv<POW2>(rem, cs.d_w), iw = rem - ih * Wc;
+    int co = imod<POW2>(gk, cs.d_cout);
+    int kidx = idiv<POW2>(gk, cs.d_cout);
+    if constexpr (S2) {
+      ih = 2 * ih + cls_a;
+      iw = 2 * iw + cls_b;
+      int kwi = kidx & (nkw - 1);
+      int khi = nkw == 2 ? (kidx >> 1) : kidx;
+      int kh = start_h + 2 * khi, kw = start_w + 2 * kwi;
+      int th = ih + cs.PH - kh, tw = iw + cs.PW - kw;
+      if (th < 0 || tw < 0) return zero16;
+      int oh = th >> 1, ow = tw >> 1;
+      if (oh >= cs.OH || ow >= cs.OW) return zero16;
+      return &DY[(((int64_t)n * cs.OH + oh) * cs.OW + ow) * cs.Cout + co];
+    } else {
+      int kw = kidx % cs.KW, kh = kidx / cs.KW;
+      int th = ih + cs.PH - kh, tw = iw + cs.PW - kw;
+      if (th < 0 || tw < 0 || th % cs.SH || tw % cs.SW) return zero16;
+      int oh = th / cs.SH, ow = tw / cs.SW;
+      if (oh >= cs.OH || ow >= cs.OW) return zero16;
+      return &DY[(((int64_t)n * cs.OH + oh) * cs.OW + ow) * cs.Cout + co];
+    }
+  };
+  auto b_src = [&](int kk0, int rl, int kk) -> const T* {
+    int gn = n0 + rl, gk = kk0 + kk;
+    if (gn >= cs.Cin || gk >= K) return zero16;
+    int64_t col = gk;
+    if constexpr (S2) {
+      int co = imod<POW2>(gk, cs.d_cout);
+      int kidx = idiv<POW2>(gk, cs.d_cout);
+      int kwi = kidx & (nkw - 1);
+      int khi = nkw == 2 ? (kidx >> 1) : kidx;
+      col = (int64_t)((start_h + 2 * khi) * cs.KW + start_w + 2 * kwi) *
+                cs.Cout + co;
+    }
+    return &WT2D[(int64_t)gn * KFULL + col];
+  };
+  auto stage = [&](int t, int which) {
+    int kk0 = t * BK;
+    glds_stage<T, BM>(As[which], wc,
+                      [&](int rl, int kk) { return a_src(kk0, rl, kk); });
+    glds_stage<T, BN>(Bs[which], wc,
+                      [&](int rl, int kk) { return b_src(kk0, rl, kk); });
+  };
+  constexpr int NPER = glds_count<T, BM>() + glds_count<T, BN>();
+
+  const int nch = (K + BK - 1) / BK;
+  stage(0, 0);
+  for (int t = 0; t < nch; ++t) {
+    const int cur = t & 1;
+    if (t + 1 < nch) {
+      stage(t + 1, cur ^ 1);
+      wait_vmcnt<NPER>();
+    } else {
+      wait_vmcnt<0>();
+    }
+    __builtin_amdgcn_s_barrier();
+    mfma_compute_tile(As[cur], Bs[cur], wc, acc);
+    __builtin_amdgcn_s_barrier();
+  }
+
+  epilogue_visit(wc, acc, m0, n0, [&](int row, int col, float v) {
+    if (row < M && col < cs.Cin) {
+      int64_t pix = row;
+      if constexpr (S2) {
+        int n = idiv<POW2>(row, cs.d_hw);
+        int rem = row - n * (Hc * Wc);
+        int ihp = idiv<POW2>(rem, cs.d_w);
+        int ih = 2 * ihp + cls_a, iw = 2 * (rem - ihp * Wc) + cls_b;
+        pix = ((int64_t)n * cs.H + ih) * cs.W + iw;
+      }
+      DX[pix * cs.Cin + col] = VecIO<T>::from_f32(v);
+    }
+  });
+}
+
+// ---------------------------------------------------------------------------
 // wgrad: dw[(kh,kw,ci), co] += sum_m x_gather * dy ; fp32 atomics over
 // grid.z m-slices.
 // ---------------------------------------------------------------------------
@@ -529,19 +716,80 @@ __global__ void k_transpose_w(const T* __restrict__ W, T* __restrict__ WT,
 }
 
 // ---------------------------------------------------------------------------
+// [KH,KW,Ci,Co] -> [Ci, KH*KW*Co] (k-contiguous rows for the dgrad B glds)
+template <typename T>
+__global__ void k_transpose_w_dgrad(const T* __restrict__ W,
+                                    T* __restrict__ WT2D, int KHW, int Cin,
+                                    int Cout) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t n = (int64_t)KHW * Cin * Cout;
+  if (i >= n) return;
+  int co = i % Cout;
+  int ci = (i / Cout) % Cin;
+  int64_t khw = i / ((int64_t)Cin * Cout);
+  WT2D[((int64_t)ci * KHW + khw) * Cout + co] = W[i];
+}
+
+// ---------------------------------------------------------------------------
+// [KH,KW,Ci,Co] -> [Co, KH*KW*Ci] (k-contiguous rows for the fwd B glds)
+template <typename T>
+__global__ void k_transpose_w_fwd(const T* __restrict__ W, T* __restrict__ WT2,
+                                  int KHW, int Cin, int Cout) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t n = (int64_t)KHW * Cin * Cout;
+  if (i >= n) return;
+  int co = i % Cout;
+  int64_t k = i / Cout;  // khw*Cin + ci
+  WT2[(int64_t)co * (KHW * Cin) + k] = W[i];
+}
+
+// ---------------------------------------------------------------------------
 static bool all_pow2(const ConvShape& cs) {
   return cs.d_ohow.lg >= 0 && cs.d_ow.lg >= 0 && cs.d_cin.lg >= 0 &&
          cs.d_hw.lg >= 0 && cs.d_w.lg >= 0 && cs.d_cout.lg >= 0;
 }
 
-void conv2d_fwd_launch(DT dt, const void* x, const void* w, const void* bias,
-                       void* y, const void* zero16, const ConvShape& cs,
-                       bool relu, hipStream_t s) {
+bool conv2d_fwd_wants_db(DT dt, const void* x, const ConvShape& cs) {
+  static const bool db_on = []() {
+    const char* e = getenv("TNN_FWD_DB");
+    return !(e && atoi(e) == 0);
+  }();
+  if (!db_on) return false;
+  int v = dt == DT::F32 ? 4 : 8;
+  // 2x LDS buffers cap occupancy at 3 blocks/CU: a win while the k-loop is
+  // short (measured +8% at K<=1152, -15% at K=4608 where the single-buffer
+  // kernel's 6 resident blocks hide latency better)
+  return cs.KH * cs.KW * cs.Cin <= 2304 && all_pow2(cs) &&
+         cs.Cin % v == 0 && (((uintptr_t)x & 15) == 0);
+}
+
+void transpose_w_fwd_launch(DT dt, const void* w, void* w_t2, int KHW, int Cin,
+                            int Cout, hipStream_t s) {
+  int64_t n = (int64_t)KHW * Cin * Cout;
+  int blocks = (int)((n + 255) / 256);
+  if (dt == DT::F32)
+    hipLaunchKernelGGL(k_transpose_w_fwd<float>, dim3(blocks), dim3(256), 0, s,
+                       (const float*)w, (float*)w_t2, KHW, Cin, Cout);
+  else
+    hipLaunchKernelGGL(k_transpose_w_fwd<bf16>, dim3(blocks), dim3(256), 0, s,
+                       (const bf16*)w, (bf16*)w_t2, KHW, Cin, Cout);
+}
+
+void conv2d_fwd_launch(DT dt, const void* x, const void* w, const void* w_t2,
+                       const void* bias, void* y, const void* zero16,
+                       const ConvShape& cs, bool relu, hipStream_t s) {
   int M = cs.N * cs.OH * cs.OW;
   dim3 grid(ceil_div(M, BM), ceil_div(cs.Cout, BN));
   int act = relu ? ACT_RELU : ACT_LINEAR;
   bool p2 = all_pow2(cs);
   if (dt == DT::F32) {
+    if (w_t2) {
+      hipLaunchKernelGGL((k_conv_fwd_db<float, true>), grid, dim3(THREADS), 0,
+                         s, (const float*)x, (const float*)w_t2,
+                         (const float*)bias, (const float*)nullptr, (float*)y,
+                         (const float*)zero16, cs, act);
+      return;
+    }
     bool g = p2 && cs.Cin % 4 == 0 && (((uintptr_t)x & 15) == 0);
     auto kern = g ? k_conv_fwd<float, true, true>
                   : (p2 ? k_conv_fwd<float, true, false>
@@ -551,6 +799,13 @@ void conv2d_fwd_launch(DT dt, const void* x, const void* w, const void* bias,
                        (const float*)nullptr, (float*)y, (const float*)zero16,
                        cs, act);
   } else {
+    if (w_t2) {
+      hipLaunchKernelGGL((k_conv_fwd_db<bf16, true>), grid, dim3(THREADS), 0,
+                         s, (const bf16*)x, (const bf16*)w_t2,
+                         (const float*)nullptr, (const bf16*)bias, (bf16*)y,
+                         (const bf16*)zero16, cs, act);
+      return;
+    }
     bool g = p2 && cs.Cin % 8 == 0 && (((uintptr_t)x & 15) == 0);
     auto kern = g ? k_conv_fwd<bf16, true, true>
                   : (p2 ? k_conv_fwd<bf16, true, false>
@@ -562,9 +817,32 @@ void conv2d_fwd_launch(DT dt, const void* x, const void* w, const void* bias,
   }
 }
 
-void conv2d_dgrad_launch(DT dt, const void* dy, const void* w_t, void* dx,
-                         const void* zero16, const ConvShape& cs,
-                         hipStream_t s) {
+bool conv2d_dgrad_wants_db(DT dt, const void* dy, const ConvShape& cs) {
+  static const bool db_on = []() {
+    const char* e = getenv("TNN_FWD_DB");
+    return !(e && atoi(e) == 0);
+  }();
+  if (!db_on) return false;
+  int v = dt == DT::F32 ? 4 : 8;
+  return cs.KH * cs.KW * cs.Cout <= 2304 && all_pow2(cs) &&
+         cs.Cout % v == 0 && (((uintptr_t)dy & 15) == 0);
+}
+
+void transpose_w_dgrad_launch(DT dt, const void* w, void* w_t2d, int KHW,
+                              int Cin, int Cout, hipStream_t s) {
+  int64_t n = (int64_t)KHW * Cin * Cout;
+  int blocks = (int)((n + 255) / 256);
+  if (dt == DT::F32)
+    hipLaunchKernelGGL(k_transpose_w_dgrad<float>, dim3(blocks), dim3(256), 0,
+                       s, (const float*)w, (float*)w_t2d, KHW, Cin, Cout);
+  else
+    hipLaunchKernelGGL(k_transpose_w_dgrad<bf16>, dim3(blocks), dim3(256), 0,
+                       s, (const bf16*)w, (bf16*)w_t2d, KHW, Cin, Cout);
+}
+
+void conv2d_dgrad_launch(DT dt, const void* dy, const void* w_t,
+                         const void* w_t2d, void* dx, const void* zero16,
+                         const ConvShape& cs, hipStream_t s) {
   int M = cs.N * cs.H * cs.W;
   dim3 grid(ceil_div(M, BM), ceil_div(cs.Cin, BN));
   bool p2 = all_pow2(cs);
@@ -582,6 +860,19 @@ void conv2d_dgrad_launch(DT dt, const void* dy, const void* w_t, void* dx,
                ceil_div(cs.Cin, BN), 4);
   if (dt == DT::F32) {
     bool g = p2 && cs.Cout % 4 == 0 && (((uintptr_t)dy & 15) == 0);
+    if (w_t2d && s2) {
+      hipLaunchKernelGGL((k_conv_dgrad_db<float, true, true>), grid_s2,
+                         dim3(THREADS), 0, s, (const float*)dy,
+                         (const float*)w_t2d, (float*)dx, (const float*)zero16,
+                         cs2);
+      return;
+    }
+    if (w_t2d) {
+      hipLaunchKernelGGL((k_conv_dgrad_db<float, true>), grid, dim3(THREADS),
+                         0, s, (const float*)dy, (const float*)w_t2d,
+                         (float*)dx, (const float*)zero16, cs);
+      return;
+    }
     if (g && s2) {
       hipLaunchKernelGGL((k_conv_dgrad<float, true, true, true>), grid_s2,
                          dim3(THREADS), 0, s, (const float*)dy,
@@ -597,6 +888,19 @@ void conv2d_dgrad_launch(DT dt, const void* dy, const void* w_t, void* dx,
                        (const float*)zero16, cs);
   } else {
     bool g = p2 && cs.Cout % 8 == 0 && (((uintptr_t)dy & 15) == 0);
+    if (w_t2d && s2) {
+      hipLaunchKernelGGL((k_conv_dgrad_db<bf16, true, true>), grid_s2,
+                         dim3(THREADS), 0, s, (const bf16*)dy,
+                         (const bf16*)w_t2d, (bf16*)dx, (const bf16*)zero16,
+                         cs2);
+      return;
+    }
+    if (w_t2d) {
+      hipLaunchKernelGGL((k_conv_dgrad_db<bf16, true>), grid, dim3(THREADS),
+                         0, s, (const bf16*)dy, (const bf16*)w_t2d, (bf16*)dx,
+                         (const bf16*)zero16, cs);
+      return;
+    }
     if (g && s2) {
       hipLaunchKernelGGL((k_conv_dgrad<bf16, true, true, true>), grid_s2,
                          dim3(THREADS), 0, s, (const bf16*)dy,

@@ -70,12 +70,18 @@ struct ConvShape {
     d_cout.set(Cout);
   }
 };
-void conv2d_fwd_launch(DT dt, const void* x, const void* w, const void* bias,
-                       void* y, const void* zero16, const ConvShape& cs,
-                       bool relu, hipStream_t s);
-void conv2d_dgrad_launch(DT dt, const void* dy, const void* w_t, void* dx,
-                         const void* zero16, const ConvShape& cs,
-                         hipStream_t s);
+bool conv2d_fwd_wants_db(DT dt, const void* x, const ConvShape& cs);
+void transpose_w_fwd_launch(DT dt, const void* w, void* w_t2, int KHW, int Cin,
+                            int Cout, hipStream_t s);
+void conv2d_fwd_launch(DT dt, const void* x, const void* w, const void* w_t2,
+                       const void* bias, void* y, const void* zero16,
+                       const ConvShape& cs, bool relu, hipStream_t s);
+bool conv2d_dgrad_wants_db(DT dt, const void* dy, const ConvShape& cs);
+void transpose_w_dgrad_launch(DT dt, const void* w, void* w_t2d, int KHW,
+                              int Cin, int Cout, hipStream_t s);
+void conv2d_dgrad_launch(DT dt, const void* dy, const void* w_t,
+                         const void* w_t2d, void* dx, const void* zero16,
+                         const ConvShape& cs, hipStream_t s);
 int conv2d_wgrad_zsplits(const ConvShape& cs);
 void conv2d_wgrad_launch(DT dt, const void* x, const void* dy, float* dw_f32,
                          float* ws, int z, const void* zero16,

@@ -105,6 +105,27 @@ DEV void glds_stage_a(T* As, const WaveCoord& w, AddrFn&& addr) {
   glds_stage<T, BM>(As, w, addr);
 }
 
+// glds instructions per wave for one ROWS-row tile (vmcnt bookkeeping)
+template <typename T, int ROWS>
+constexpr int glds_count() {
+  return ROWS * (BK * (int)sizeof(T)) / 1024 / 4;
+}
+
+// counted s_waitcnt vmcnt(N): __syncthreads() with a glds in flight drains
+// vmcnt(0) (guide: LDS-DMA is a pending LDS write on the VM counter), so
+// double-buffered loops wait an explicit partial count + raw s_barrier.
+template <int N>
+DEV void wait_vmcnt() {
+  static_assert(N == 0 || N == 4 || N == 6 || N == 8 || N == 12 || N == 24,
+                "add an asm literal for this count");
+  if constexpr (N == 0) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  else if constexpr (N == 4) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  else if constexpr (N == 6) asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+  else if constexpr (N == 8) asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+  else if constexpr (N == 12) asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
+  else if constexpr (N == 24) asm volatile("s_waitcnt vmcnt(24)" ::: "memory");
+}
+
 
 // ---- MFMA tile compute: acc[FM][FN] += A_tile * B_tile^T-stored -----------
 DEV void mfma_compute_tile(const bf16* As, const bf16* Bs, const WaveCoord& w,
