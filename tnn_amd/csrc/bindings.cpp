@@ -115,6 +115,19 @@ at::Tensor gemm(const at::Tensor& a, const at::Tensor& b,
                    cur_stream());
     return c;
   }
+  bool vec_ok = (dt_of(a) == DT::F32 ? K % 4 == 0 : K % 8 == 0) &&
+                (((uintptr_t)a.data_ptr() & 15) == 0);
+  if (M >= 64 && vec_ok) {
+    // transpose the (weight-sized) B once and take the NT glds path: B rows
+    // become k-contiguous LDS-DMA targets instead of a VGPR scatter
+    auto bt = at::empty({(int64_t)N, (int64_t)K}, b.options());
+    transpose_w_launch(dt_of(b), b.data_ptr(), bt.data_ptr(), 1, 1, K, N,
+                       cur_stream());
+    gemm_launch(dt_of(a), a.data_ptr(), bt.data_ptr(), bp, c.data_ptr(),
+                zero_page(a), M, N, K, /*trans_b=*/true, (int)act_kind,
+                cur_stream());
+    return c;
+  }
   gemm_launch(dt_of(a), a.data_ptr(), b.data_ptr(), bp, c.data_ptr(),
               zero_page(a), M, N, K, /*trans_b=*/false, (int)act_kind,
               cur_stream());

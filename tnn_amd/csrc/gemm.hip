@@ -215,6 +215,65 @@ __global__ void k_gemm_tn(const T* __restrict__ A, const T* __restrict__ B,
   });
 }
 
+// double-buffered pure-glds NT gemm (see k_conv_fwd_db): A[M,K] and
+// B[N,K] rows are both k-contiguous, so the next chunk's LDS-DMAs overlap
+// the current chunk's MFMAs. Dispatched for K <= 3072 (2x LDS buffers cap
+// occupancy at 3 blocks/CU; long k-loops prefer the 6-block single-buffer
+// kernel).
+template <typename T>
+__launch_bounds__(THREADS)
+__global__ void k_gemm_nt_db(const T* __restrict__ A, const T* __restrict__ B,
+                             const float* __restrict__ bias_f32,
+                             const T* __restrict__ bias_t, T* __restrict__ C,
+                             const T* __restrict__ zero16, int M, int N, int K,
+                             int act_kind) {
+  __shared__ alignas(16) T As[2][BM * BK];
+  __shared__ alignas(16) T Bs[2][BN * BK];
+  const int m0 = blockIdx.x * BM;
+  const int n0 = blockIdx.y * BN;
+  const WaveCoord wc;
+  f32x4 acc[FM][FN] = {};
+
+  auto stage = [&](int t, int which) {
+    int kk0 = t * BK;
+    glds_stage<T, BM>(As[which], wc, [&](int rl, int kk) -> const T* {
+      int gm = m0 + rl, gk = kk0 + kk;
+      if (gm >= M || gk >= K) return zero16;
+      return &A[(int64_t)gm * K + gk];
+    });
+    glds_stage<T, BN>(Bs[which], wc, [&](int rl, int kk) -> const T* {
+      int gn = n0 + rl, gk = kk0 + kk;
+      if (gn >= N || gk >= K) return zero16;
+      return &B[(int64_t)gn * K + gk];
+    });
+  };
+  constexpr int NPER = glds_count<T, BM>() + glds_count<T, BN>();
+
+  const int nch = (K + BK - 1) / BK;
+  stage(0, 0);
+  for (int t = 0; t < nch; ++t) {
+    const int cur = t & 1;
+    if (t + 1 < nch) {
+      stage(t + 1, cur ^ 1);
+      wait_vmcnt<NPER>();
+    } else {
+      wait_vmcnt<0>();
+    }
+    __builtin_amdgcn_s_barrier();
+    mfma_compute_tile(As[cur], Bs[cur], wc, acc);
+    __builtin_amdgcn_s_barrier();
+  }
+
+  epilogue_visit(wc, acc, m0, n0, [&](int row, int col, float v) {
+    if (row < M && col < N) {
+      if (bias_f32) v += bias_f32[col];
+      if (bias_t) v += VecIO<T>::to_f32(bias_t[col]);
+      if (act_kind != ACT_LINEAR) v = act_apply(v, act_kind);
+      C[(int64_t)row * N + col] = VecIO<T>::from_f32(v);
+    }
+  });
+}
+
 // M=1 NT matvec (decode-path linears): one wave per output row, 16B loads
 // over K, cross-lane reduce. The MFMA tile kernel wastes 127/128 of its
 // A-tile rows at M=1 and is ~30x slower on gpt2 decode shapes.
@@ -449,6 +508,13 @@ void gemm_launch(DT dt, const void* a, const void* b, const void* bias,
   dim3 blk(THREADS);
   if (dt == DT::F32) {
     bool g = K % 4 == 0 && (((uintptr_t)a & 15) == 0);
+    if (trans_b && g && K <= 3072 && (((uintptr_t)b & 15) == 0)) {
+      hipLaunchKernelGGL(k_gemm_nt_db<float>, grid, blk, 0, s,
+                         (const float*)a, (const float*)b, (const float*)bias,
+                         (const float*)nullptr, (float*)c,
+                         (const float*)zero16, M, N, K, act_kind);
+      return;
+    }
     auto kern = trans_b ? (g ? k_gemm<float, true, true>
                              : k_gemm<float, true, false>)
                         : (g ? k_gemm<float, false, true>
@@ -458,6 +524,13 @@ void gemm_launch(DT dt, const void* a, const void* b, const void* bias,
                        (const float*)zero16, M, N, K, act_kind);
   } else {
     bool g = K % 8 == 0 && (((uintptr_t)a & 15) == 0);
+    if (trans_b && g && K <= 3072 && (((uintptr_t)b & 15) == 0)) {
+      hipLaunchKernelGGL(k_gemm_nt_db<bf16>, grid, blk, 0, s, (const bf16*)a,
+                         (const bf16*)b, (const float*)nullptr,
+                         (const bf16*)bias, (bf16*)c, (const bf16*)zero16, M,
+                         N, K, act_kind);
+      return;
+    }
     auto kern = trans_b ? (g ? k_gemm<bf16, true, true>
                              : k_gemm<bf16, true, false>)
                         : (g ? k_gemm<bf16, false, true>
