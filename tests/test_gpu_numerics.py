@@ -200,7 +200,7 @@ def test_bn_fwd_train(dtype, C):
     x = torch.randn(8, 6, 6, C, dtype=dtype, device=DEV) * 2 + 0.5
     gamma = torch.rand(C, device=DEV) + 0.5
     beta = torch.randn(C, device=DEV)
-    y, mean, invstd = ext.bn_fwd_train(x, gamma, beta, None, None, 0.1, 1e-5, False)
+    y, mean, invstd = ext.bn_fwd_train(x, gamma, beta, None, None, 0.1, 1e-5, False, 0.0, 0)
     xf = x.float().reshape(-1, C)
     rmean = xf.mean(0)
     rvar = xf.var(0, unbiased=False)
@@ -215,7 +215,7 @@ def test_bn_relu_and_infer(dtype):
     x = torch.randn(4, 5, 5, C, dtype=dtype, device=DEV)
     gamma = torch.ones(C, device=DEV)
     beta = torch.zeros(C, device=DEV)
-    y, _, _ = ext.bn_fwd_train(x, gamma, beta, None, None, 0.1, 1e-5, True)
+    y, _, _ = ext.bn_fwd_train(x, gamma, beta, None, None, 0.1, 1e-5, True, 0.0, 0)
     assert (y.float() >= 0).all()
     rm = torch.randn(C, device=DEV) * 0.1
     rv = torch.rand(C, device=DEV) + 0.5
@@ -240,9 +240,10 @@ def test_bn_bwd_matches_autograd(dtype):
     y.backward(dy)
     xdt = x.to(dtype)
     y_k, mean_k, invstd_k = ext.bn_fwd_train(xdt, gamma.detach(), beta.detach(),
-                                             None, None, 0.1, 1e-5, False)
+                                             None, None, 0.1, 1e-5, False, 0.0,
+                                             0)
     dx, dgamma, dbeta = ext.bn_bwd(xdt, dy.to(dtype), gamma.detach(), mean_k,
-                                   invstd_k, None)
+                                   invstd_k, None, 1.0)
     tol = TOL[dtype] * 4
     assert relerr(dx, xg.grad) < tol
     assert relerr(dgamma, gamma.grad) < tol
@@ -556,3 +557,49 @@ def test_generate_graphed_gpu_matches_recompute():
     fast = generate_graphed(m, [1, 2, 3], max_new_tokens=12, seq_len=32,
                             eot_token=None, device=torch.device(DEV))
     assert fast == ref
+
+
+@pytest.mark.parametrize("dtype", DTYPES)
+def test_bn_relu_dropout_fused(dtype):
+    # fused BN+ReLU+dropout: dropped positions are exactly 0, kept positions
+    # are relu(bn(x))/keep; backward = autograd with the realized mask
+    torch.manual_seed(9)
+    p = 0.4
+    N, C = 4096, 128
+    x = torch.randn(N, C, dtype=dtype, device=DEV)
+    gamma = torch.rand(C, device=DEV, dtype=torch.float32) + 0.5
+    beta = torch.randn(C, device=DEV, dtype=torch.float32)
+    dy = torch.randn(N, C, dtype=dtype, device=DEV)
+
+    xr = x.detach().clone().requires_grad_(True)
+    gr = gamma.detach().clone().requires_grad_(True)
+    br = beta.detach().clone().requires_grad_(True)
+    from tnn_amd.ops.functional import _BatchNormAct
+    y, _, _ = _BatchNormAct.apply(xr, gr, br, None, None, 0.1, 1e-5, True,
+                                  p, 1234)
+    # reference bn+relu
+    xf = x.float()
+    mean = xf.mean(0)
+    var = xf.var(0, unbiased=False)
+    ref = ((xf - mean) / torch.sqrt(var + 1e-5)) * gamma + beta
+    ref = torch.relu(ref)
+    kept = y.float() != 0.0
+    frac = kept.float().mean().item()
+    pos_frac = (ref > 0).float().mean().item()
+    assert abs(frac - pos_frac * (1 - p)) < 0.02
+    assert relerr(y.float()[kept], (ref / (1 - p))[kept]) < TOL[dtype] * 2
+
+    # backward vs autograd using the realized mask
+    y.backward(dy)
+    xa = x.detach().clone().float().requires_grad_(True)
+    ga = gamma.detach().clone().requires_grad_(True)
+    ba = beta.detach().clone().requires_grad_(True)
+    m2 = xa.mean(0)
+    v2 = xa.var(0, unbiased=False)
+    ya = torch.relu(((xa - m2) / torch.sqrt(v2 + 1e-5)) * ga + ba)
+    ya = ya * kept.float() / (1 - p)
+    ya.backward(dy.float())
+    tol = TOL[dtype] * 8
+    assert relerr(xr.grad, xa.grad) < tol
+    assert relerr(gr.grad, ga.grad) < tol
+    assert relerr(br.grad, ba.grad) < tol

@@ -114,7 +114,7 @@ __global__ void k_bn_bwd_reduce_vec(const T* __restrict__ x,
                                     const float* mean, const float* invstd,
                                     float* __restrict__ sum_dy,
                                     float* __restrict__ sum_dy_xhat,
-                                    int64_t rows, int cols) {
+                                    int64_t rows, int cols, float dy_scale) {
   constexpr int V = 16 / sizeof(T);
   using VecT = BnPack<T>;
   const int groups = cols / V;
@@ -144,14 +144,14 @@ __global__ void k_bn_bwd_reduce_vec(const T* __restrict__ x,
 #pragma unroll
         for (int j = 0; j < V; ++j) {
           float gr = VecIO<T>::to_f32(vy.e[j]) > 0.0f
-                         ? VecIO<T>::to_f32(vg.e[j]) : 0.0f;
+                         ? VecIO<T>::to_f32(vg.e[j]) * dy_scale : 0.0f;
           s[j] += gr;
           sx[j] += gr * (VecIO<T>::to_f32(vx.e[j]) - m[j]) * is[j];
         }
       } else {
 #pragma unroll
         for (int j = 0; j < V; ++j) {
-          float gr = VecIO<T>::to_f32(vg.e[j]);
+          float gr = VecIO<T>::to_f32(vg.e[j]) * dy_scale;
           s[j] += gr;
           sx[j] += gr * (VecIO<T>::to_f32(vx.e[j]) - m[j]) * is[j];
         }
@@ -211,6 +211,38 @@ __global__ void k_bn_apply(const T* __restrict__ x, const float* mean,
   }
 }
 
+// fused BN+ReLU+dropout apply (train, relu only): dropped positions write
+// 0 so the saved output doubles as both the relu mask and the dropout
+// mask for backward -- no mask tensor, no separate dropout pass. Kept
+// positions are pre-scaled by 1/(1-p); backward just scales dy by the
+// same constant wherever the saved output is > 0.
+template <typename T>
+__global__ void k_bn_apply_drop(const T* __restrict__ x, const float* mean,
+                                const float* invstd, const float* gamma,
+                                const float* beta, T* __restrict__ y,
+                                int64_t n, int cols, float p, uint64_t seed) {
+  const float scale = 1.0f / (1.0f - p);
+  Philox rng(seed);
+  int64_t i4 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i4 * 4 < n; i4 += stride) {
+    uint4 r = rng(i4);
+    unsigned int rs[4] = {r.x, r.y, r.z, r.w};
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int64_t i = i4 * 4 + j;
+      if (i < n) {
+        int c = i % cols;
+        float v =
+            (VecIO<T>::to_f32(x[i]) - mean[c]) * invstd[c] * gamma[c] + beta[c];
+        v = fmaxf(v, 0.0f);
+        bool keep = u32_to_uniform(rs[j]) > p;
+        y[i] = VecIO<T>::from_f32(keep ? v * scale : 0.0f);
+      }
+    }
+  }
+}
+
 template <typename T>
 __global__ void k_bn_infer(const T* __restrict__ x, const float* rmean,
                            const float* rvar, const float* gamma,
@@ -233,7 +265,7 @@ __global__ void k_bn_bwd_reduce(const T* __restrict__ x, const T* __restrict__ d
                                 const T* __restrict__ y_relu, const float* mean,
                                 const float* invstd, float* __restrict__ sum_dy,
                                 float* __restrict__ sum_dy_xhat, int64_t rows,
-                                int cols) {
+                                int cols, float dy_scale) {
   const int cpb = min(cols, 256);
   const int rows_per_iter = 256 / cpb;
   const int c = threadIdx.x % cpb + blockIdx.x * cpb;
@@ -245,7 +277,7 @@ __global__ void k_bn_bwd_reduce(const T* __restrict__ x, const T* __restrict__ d
   float s = 0.0f, sx = 0.0f;
   for (int64_t r = r0 + r_off; r < r1; r += rows_per_iter) {
     int64_t i = r * cols + c;
-    float g = VecIO<T>::to_f32(dy[i]);
+    float g = VecIO<T>::to_f32(dy[i]) * dy_scale;
     if (y_relu && VecIO<T>::to_f32(y_relu[i]) <= 0.0f) g = 0.0f;
     float xhat = (VecIO<T>::to_f32(x[i]) - m) * is;
     s += g;
@@ -271,13 +303,13 @@ __global__ void k_bn_bwd_apply(const T* __restrict__ x, const T* __restrict__ dy
                                const float* invstd, const float* gamma,
                                const float* sum_dy, const float* sum_dy_xhat,
                                T* __restrict__ dx, int64_t n, int64_t rows,
-                               int cols) {
+                               int cols, float dy_scale) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   const float inv_n = 1.0f / (float)rows;
   for (; i < n; i += stride) {
     int c = i % cols;
-    float g = VecIO<T>::to_f32(dy[i]);
+    float g = VecIO<T>::to_f32(dy[i]) * dy_scale;
     if (y_relu && VecIO<T>::to_f32(y_relu[i]) <= 0.0f) g = 0.0f;
     float xhat = (VecIO<T>::to_f32(x[i]) - mean[c]) * invstd[c];
     float v = gamma[c] * invstd[c] *
@@ -365,6 +397,22 @@ void bn_apply_launch(DT dt, const void* x, const float* mean,
                        cols, relu);
 }
 
+void bn_apply_drop_launch(DT dt, const void* x, const float* mean,
+                          const float* invstd, const float* gamma,
+                          const float* beta, void* y, int64_t rows, int cols,
+                          float p, uint64_t seed, hipStream_t s) {
+  int64_t n = rows * cols;
+  int blocks = (int)std::min<int64_t>(((n + 3) / 4 + 255) / 256, (int64_t)2048);
+  if (dt == DT::F32)
+    hipLaunchKernelGGL(k_bn_apply_drop<float>, dim3(blocks), dim3(256), 0, s,
+                       (const float*)x, mean, invstd, gamma, beta, (float*)y, n,
+                       cols, p, seed);
+  else
+    hipLaunchKernelGGL(k_bn_apply_drop<bf16>, dim3(blocks), dim3(256), 0, s,
+                       (const bf16*)x, mean, invstd, gamma, beta, (bf16*)y, n,
+                       cols, p, seed);
+}
+
 void bn_infer_launch(DT dt, const void* x, const float* rmean,
                      const float* rvar, const float* gamma, const float* beta,
                      void* y, int64_t rows, int cols, float eps, bool relu,
@@ -385,30 +433,30 @@ void bn_bwd_reduce_launch(DT dt, const void* x, const void* dy,
                           const void* y_relu, const float* mean,
                           const float* invstd, float* sum_dy,
                           float* sum_dy_xhat, int64_t rows, int cols,
-                          hipStream_t s) {
+                          float dy_scale, hipStream_t s) {
   if (dt == DT::F32) {
     if (bn_vec_ok<float>(x, cols) && bn_vec_ok<float>(dy, cols))
       hipLaunchKernelGGL(k_bn_bwd_reduce_vec<float>,
                          bn_reduce_grid_vec<float>(rows, cols), dim3(256), 0, s,
                          (const float*)x, (const float*)dy,
                          (const float*)y_relu, mean, invstd, sum_dy,
-                         sum_dy_xhat, rows, cols);
+                         sum_dy_xhat, rows, cols, dy_scale);
     else
       hipLaunchKernelGGL(k_bn_bwd_reduce<float>, bn_reduce_grid(rows, cols),
                          dim3(256), 0, s, (const float*)x, (const float*)dy,
                          (const float*)y_relu, mean, invstd, sum_dy,
-                         sum_dy_xhat, rows, cols);
+                         sum_dy_xhat, rows, cols, dy_scale);
   } else {
     if (bn_vec_ok<bf16>(x, cols) && bn_vec_ok<bf16>(dy, cols))
       hipLaunchKernelGGL(k_bn_bwd_reduce_vec<bf16>,
                          bn_reduce_grid_vec<bf16>(rows, cols), dim3(256), 0, s,
                          (const bf16*)x, (const bf16*)dy, (const bf16*)y_relu,
-                         mean, invstd, sum_dy, sum_dy_xhat, rows, cols);
+                         mean, invstd, sum_dy, sum_dy_xhat, rows, cols, dy_scale);
     else
       hipLaunchKernelGGL(k_bn_bwd_reduce<bf16>, bn_reduce_grid(rows, cols),
                          dim3(256), 0, s, (const bf16*)x, (const bf16*)dy,
                          (const bf16*)y_relu, mean, invstd, sum_dy,
-                         sum_dy_xhat, rows, cols);
+                         sum_dy_xhat, rows, cols, dy_scale);
   }
 }
 
@@ -416,19 +464,20 @@ void bn_bwd_apply_launch(DT dt, const void* x, const void* dy,
                          const void* y_relu, const float* mean,
                          const float* invstd, const float* gamma,
                          const float* sum_dy, const float* sum_dy_xhat,
-                         void* dx, int64_t rows, int cols, hipStream_t s) {
+                         void* dx, int64_t rows, int cols, float dy_scale,
+                         hipStream_t s) {
   int64_t n = rows * cols;
   int blocks = (int)std::min<int64_t>((n + 255) / 256, (int64_t)2048);
   if (dt == DT::F32)
     hipLaunchKernelGGL(k_bn_bwd_apply<float>, dim3(blocks), dim3(256), 0, s,
                        (const float*)x, (const float*)dy, (const float*)y_relu,
                        mean, invstd, gamma, sum_dy, sum_dy_xhat, (float*)dx, n,
-                       rows, cols);
+                       rows, cols, dy_scale);
   else
     hipLaunchKernelGGL(k_bn_bwd_apply<bf16>, dim3(blocks), dim3(256), 0, s,
                        (const bf16*)x, (const bf16*)dy, (const bf16*)y_relu,
                        mean, invstd, gamma, sum_dy, sum_dy_xhat, (bf16*)dx, n,
-                       rows, cols);
+                       rows, cols, dy_scale);
 }
 
 }  // namespace tnn

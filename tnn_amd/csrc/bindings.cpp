@@ -304,8 +304,10 @@ std::vector<at::Tensor> bn_fwd_train(const at::Tensor& x,
                                      const at::Tensor& beta,
                                      const c10::optional<at::Tensor>& rmean,
                                      const c10::optional<at::Tensor>& rvar,
-                                     double momentum, double eps, bool relu) {
+                                     double momentum, double eps, bool relu,
+                                     double dropout_p, int64_t seed) {
   CHECK_IN(x);
+  TORCH_CHECK(dropout_p == 0.0 || relu, "fused BN dropout requires relu");
   int C = x.size(-1);
   int64_t rows = x.numel() / C;
   auto mean = at::empty({C}, x.options().dtype(at::kFloat));
@@ -316,10 +318,16 @@ std::vector<at::Tensor> bn_fwd_train(const at::Tensor& x,
   bn_stats_launch(dt_of(x), x.data_ptr(), mean.data_ptr<float>(),
                   invstd.data_ptr<float>(), rm, rv, (float)momentum, rows, C,
                   (float)eps, cur_stream());
-  bn_apply_launch(dt_of(x), x.data_ptr(), mean.data_ptr<float>(),
-                  invstd.data_ptr<float>(), gamma.data_ptr<float>(),
-                  beta.data_ptr<float>(), y.data_ptr(), rows, C, relu,
-                  cur_stream());
+  if (dropout_p > 0.0)
+    bn_apply_drop_launch(dt_of(x), x.data_ptr(), mean.data_ptr<float>(),
+                         invstd.data_ptr<float>(), gamma.data_ptr<float>(),
+                         beta.data_ptr<float>(), y.data_ptr(), rows, C,
+                         (float)dropout_p, (uint64_t)seed, cur_stream());
+  else
+    bn_apply_launch(dt_of(x), x.data_ptr(), mean.data_ptr<float>(),
+                    invstd.data_ptr<float>(), gamma.data_ptr<float>(),
+                    beta.data_ptr<float>(), y.data_ptr(), rows, C, relu,
+                    cur_stream());
   return {y, mean, invstd};
 }
 
@@ -340,7 +348,8 @@ at::Tensor bn_fwd_infer(const at::Tensor& x, const at::Tensor& gamma,
 std::vector<at::Tensor> bn_bwd(const at::Tensor& x, const at::Tensor& dy,
                                const at::Tensor& gamma, const at::Tensor& mean,
                                const at::Tensor& invstd,
-                               const c10::optional<at::Tensor>& y_relu) {
+                               const c10::optional<at::Tensor>& y_relu,
+                               double dy_scale) {
   CHECK_IN(x);
   CHECK_IN(dy);
   int C = x.size(-1);
@@ -354,12 +363,12 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& x, const at::Tensor& dy,
   bn_bwd_reduce_launch(dt_of(x), x.data_ptr(), dy.data_ptr(), yr,
                        mean.data_ptr<float>(), invstd.data_ptr<float>(),
                        sum_dy.data_ptr<float>(), sum_dy_xhat.data_ptr<float>(),
-                       rows, C, cur_stream());
+                       rows, C, (float)dy_scale, cur_stream());
   bn_bwd_apply_launch(dt_of(x), x.data_ptr(), dy.data_ptr(), yr,
                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
                       gamma.data_ptr<float>(), sum_dy.data_ptr<float>(),
                       sum_dy_xhat.data_ptr<float>(), dx.data_ptr(), rows, C,
-                      cur_stream());
+                      (float)dy_scale, cur_stream());
   // dgamma = sum_dy_xhat, dbeta = sum_dy (fp32, matching fp32 gamma/beta)
   return {dx, sum_dy_xhat, sum_dy};
 }

@@ -96,19 +96,23 @@ void bn_stats_launch(DT dt, const void* x, float* mean, float* invstd,
 void bn_apply_launch(DT dt, const void* x, const float* mean,
                      const float* invstd, const float* gamma, const float* beta,
                      void* y, int64_t rows, int cols, bool relu, hipStream_t s);
+void bn_apply_drop_launch(DT dt, const void* x, const float* mean,
+                          const float* invstd, const float* gamma,
+                          const float* beta, void* y, int64_t rows, int cols,
+                          float p, uint64_t seed, hipStream_t s);
 void bn_infer_launch(DT dt, const void* x, const float* rmean,
                      const float* rvar, const float* gamma, const float* beta,
                      void* y, int64_t rows, int cols, float eps, bool relu,
                      hipStream_t s);
 void bn_bwd_reduce_launch(DT dt, const void* x, const void* dy, const void* y_relu,
-                          const float* mean, const float* invstd,
-                          float* sum_dy, float* sum_dy_xhat,
-                          int64_t rows, int cols, hipStream_t s);
+                          const float* mean, const float* invstd, float* sum_dy,
+                          float* sum_dy_xhat, int64_t rows, int cols,
+                          float dy_scale, hipStream_t s);
 void bn_bwd_apply_launch(DT dt, const void* x, const void* dy, const void* y_relu,
                          const float* mean, const float* invstd,
                          const float* gamma, const float* sum_dy,
-                         const float* sum_dy_xhat, void* dx,
-                         int64_t rows, int cols, hipStream_t s);
+                         const float* sum_dy_xhat, void* dx, int64_t rows,
+                         int cols, float dy_scale, hipStream_t s);
 
 // ---- pool.hip --------------------------------------------------------------
 struct PoolShape {

@@ -25,8 +25,26 @@ class Sequential(Layer):
         self.layers = nn.ModuleList(layers or [])
 
     def forward(self, x):
-        for layer in self.layers:
+        from .layers import BatchNorm, Dropout
+        n = len(self.layers)
+        i = 0
+        while i < n:
+            layer = self.layers[i]
+            # peephole: BN(relu) directly followed by Dropout runs as one
+            # fused kernel pass on GPU during training (ops.batch_norm_act
+            # dropout_p) -- saves the dropout read/write passes entirely
+            if (i + 1 < n and self.training and x.is_cuda
+                    and isinstance(layer, BatchNorm) and layer.relu
+                    and isinstance(self.layers[i + 1], Dropout)
+                    and self.layers[i + 1].p > 0):
+                x = ops.batch_norm_act(x, layer.gamma, layer.beta,
+                                       layer.running_mean, layer.running_var,
+                                       True, layer.momentum, layer.eps,
+                                       True, self.layers[i + 1].p)
+                i += 2
+                continue
             x = layer(x)
+            i += 1
         return x
 
     def output_shape(self, in_shape):

@@ -110,13 +110,19 @@ def conv2d_nhwc(
 class _BatchNormAct(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, gamma, beta, running_mean, running_var, momentum,
-                eps, relu):
+                eps, relu, dropout_p=0.0, seed=0):
         ext = _C.ext()
-        # running stats update fused into the finalize kernel
+        # running stats update fused into the finalize kernel; dropout (if
+        # any) fused into the apply kernel -- dropped positions write 0, so
+        # the saved output doubles as relu mask and dropout mask and the
+        # backward is a constant 1/(1-p) scale on the kept positions
         y, mean, invstd = ext.bn_fwd_train(x, gamma, beta, running_mean,
-                                           running_var, momentum, eps, relu)
-        ctx.save_for_backward(x, gamma, mean, invstd, y if relu else None)
-        ctx.relu = relu
+                                           running_var, momentum, eps, relu,
+                                           dropout_p, seed)
+        save_y = y if (relu or dropout_p > 0.0) else None
+        ctx.save_for_backward(x, gamma, mean, invstd, save_y)
+        ctx.relu = relu or dropout_p > 0.0
+        ctx.dy_scale = 1.0 / (1.0 - dropout_p) if dropout_p > 0.0 else 1.0
         return y, mean, invstd
 
     @staticmethod
@@ -124,8 +130,8 @@ class _BatchNormAct(torch.autograd.Function):
         x, gamma, mean, invstd, y = ctx.saved_tensors
         ext = _C.ext()
         dx, dgamma, dbeta = ext.bn_bwd(x, dy.contiguous(), gamma, mean, invstd,
-                                       y if ctx.relu else None)
-        return dx, dgamma, dbeta, None, None, None, None, None
+                                       y if ctx.relu else None, ctx.dy_scale)
+        return (dx, dgamma, dbeta, None, None, None, None, None, None, None)
 
 
 def batch_norm_act(
@@ -138,20 +144,27 @@ def batch_norm_act(
     momentum: float = 0.1,
     eps: float = 1e-5,
     relu: bool = False,
+    dropout_p: float = 0.0,
 ):
-    """BatchNorm over NHWC channels-last with optional fused ReLU.
+    """BatchNorm over NHWC channels-last with optional fused ReLU and
+    (train-time) fused dropout.
 
     Returns y. Updates running stats in-place when training (fp32, unbiased
     variance, matching torch semantics). gamma/beta/running stats are fp32
     regardless of x dtype (reference batchnorm_layer.cpp:140-148).
+    dropout_p > 0 (requires relu) folds the dropout mask + 1/(1-p) scale
+    into the BN apply kernel -- one memory pass instead of three.
     """
     C = x.shape[-1]
     n = x.numel() // C
     if _use_hip(x):
         if training:
+            seed = 0
+            if dropout_p > 0.0:
+                seed = int(torch.randint(0, 2 ** 62, (1,)).item())
             y, _, _ = _BatchNormAct.apply(x.contiguous(), gamma, beta,
                                           running_mean, running_var, momentum,
-                                          eps, relu)
+                                          eps, relu, dropout_p, seed)
             return y
         ext = _C.ext()
         return ext.bn_fwd_infer(x.contiguous(), gamma, beta, running_mean, running_var,
@@ -171,6 +184,8 @@ def batch_norm_act(
     y = xhat * gamma + beta
     if relu:
         y = F.relu(y)
+    if dropout_p > 0.0 and training:
+        y = F.dropout(y, dropout_p, training=True)
     return y.to(x.dtype)
 
 
