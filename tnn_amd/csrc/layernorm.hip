@@ -80,8 +80,69 @@ __global__ void k_ln_bwd(const T* __restrict__ x, const T* __restrict__ dy,
     float gy = VecIO<T>::to_f32(dyr[c]);
     float xhat = (VecIO<T>::to_f32(xr[c]) - m) * is;
     dxr[c] = VecIO<T>::from_f32(is * (gy * gamma[c] - ma - xhat * mb));
-    atomicAdd(&dgamma[c], gy * xhat);
-    atomicAdd(&dbeta[c], gy);
+    if (dgamma) {  // ragged-cols fallback; the vec column kernel is used
+      atomicAdd(&dgamma[c], gy * xhat);  // when cols % V == 0
+      atomicAdd(&dbeta[c], gy);
+    }
+  }
+}
+
+// dgamma/dbeta column sums as a separate vectorized pass: the per-row
+// kernel's per-element global atomics serialized 4096 rows onto the same
+// 768 addresses (measured 131us for a 19MB op). Same structure as the BN
+// vec reduce but with per-ROW mean/invstd.
+template <typename T>
+__global__ void k_ln_bwd_col(const T* __restrict__ x, const T* __restrict__ dy,
+                             const float* __restrict__ mean,
+                             const float* __restrict__ invstd,
+                             float* __restrict__ dgamma,
+                             float* __restrict__ dbeta, int64_t rows,
+                             int cols) {
+  constexpr int V = 16 / sizeof(T);
+  struct alignas(16) P { T e[16 / sizeof(T)]; };
+  const int groups = cols / V;
+  const int gpb = min(groups, 256);
+  const int rows_per_iter = 256 / gpb;
+  const int g = threadIdx.x % gpb + blockIdx.x * gpb;
+  const int r_off = threadIdx.x / gpb;
+  __shared__ float sh[2 * 256 * V];
+  float* sh_g = sh;
+  float* sh_b = sh + 256 * V;
+  float sg[V] = {}, sb[V] = {};
+  if (g < groups && r_off < rows_per_iter) {
+    const int64_t r0 = rows * blockIdx.y / gridDim.y;
+    const int64_t r1 = rows * (blockIdx.y + 1) / gridDim.y;
+    for (int64_t r = r0 + r_off; r < r1; r += rows_per_iter) {
+      const float m = mean[r], is = invstd[r];
+      int64_t i = r * cols + (int64_t)g * V;
+      P vy = *(const P*)&dy[i];
+      P vx = *(const P*)&x[i];
+#pragma unroll
+      for (int j = 0; j < V; ++j) {
+        float gy = VecIO<T>::to_f32(vy.e[j]);
+        sg[j] += gy * (VecIO<T>::to_f32(vx.e[j]) - m) * is;
+        sb[j] += gy;
+      }
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < V; ++j) {
+    sh_g[threadIdx.x * V + j] = sg[j];
+    sh_b[threadIdx.x * V + j] = sb[j];
+  }
+  __syncthreads();
+  const int cpb_total = gpb * V;
+  for (int cl = threadIdx.x; cl < cpb_total; cl += 256) {
+    int gl = cl / V, j = cl % V;
+    float tg = 0.0f, tb = 0.0f;
+    for (int r = 0; r < rows_per_iter; ++r) {
+      tg += sh_g[(r * gpb + gl) * V + j];
+      tb += sh_b[(r * gpb + gl) * V + j];
+    }
+    int c = blockIdx.x * cpb_total + cl;
+    if (c >= cols) break;
+    atomicAdd(&dgamma[c], tg);
+    atomicAdd(&dbeta[c], tb);
   }
 }
 
@@ -102,14 +163,37 @@ void ln_bwd_launch(DT dt, const void* x, const void* dy, const float* gamma,
                    const float* mean, const float* invstd, void* dx,
                    float* dgamma, float* dbeta, int64_t rows, int cols,
                    hipStream_t s) {
+  int V = dt == DT::F32 ? 4 : 8;
+  bool vec = cols % V == 0 && (((uintptr_t)x & 15) == 0) &&
+             (((uintptr_t)dy & 15) == 0);
   if (dt == DT::F32)
     hipLaunchKernelGGL(k_ln_bwd<float>, dim3(rows), dim3(256), 0, s,
                        (const float*)x, (const float*)dy, gamma, mean, invstd,
-                       (float*)dx, dgamma, dbeta, cols);
+                       (float*)dx, vec ? nullptr : dgamma,
+                       vec ? nullptr : dbeta, cols);
   else
     hipLaunchKernelGGL(k_ln_bwd<bf16>, dim3(rows), dim3(256), 0, s,
                        (const bf16*)x, (const bf16*)dy, gamma, mean, invstd,
-                       (bf16*)dx, dgamma, dbeta, cols);
+                       (bf16*)dx, vec ? nullptr : dgamma, vec ? nullptr : dbeta,
+                       cols);
+  if (vec) {
+    int groups = cols / V;
+    int gpb = groups < 256 ? groups : 256;
+    int cblocks = (groups + gpb - 1) / gpb;
+    int rows_per_iter = 256 / gpb;
+    int64_t iters = (rows + rows_per_iter - 1) / rows_per_iter;
+    int rslices = (int)std::min<int64_t>(
+        std::max<int64_t>(2048 / cblocks, 1), std::max<int64_t>(iters / 8, 1));
+    dim3 grid(cblocks, rslices);
+    if (dt == DT::F32)
+      hipLaunchKernelGGL(k_ln_bwd_col<float>, grid, dim3(256), 0, s,
+                         (const float*)x, (const float*)dy, mean, invstd,
+                         dgamma, dbeta, rows, cols);
+    else
+      hipLaunchKernelGGL(k_ln_bwd_col<bf16>, grid, dim3(256), 0, s,
+                         (const bf16*)x, (const bf16*)dy, mean, invstd, dgamma,
+                         dbeta, rows, cols);
+  }
 }
 
 }  // namespace tnn
