@@ -66,7 +66,7 @@ def bench_conv(dtype=torch.bfloat16):
         report(f"conv_dgrad", f"{name} mb{N}", timeit(lambda: ext.conv2d_dgrad(
             dy, w, H, W, S, S, P, P)), fl)
         report(f"conv_wgrad", f"{name} mb{N}", timeit(lambda: ext.conv2d_wgrad(
-            x, dy, KS, KS, S, S, P, P)), fl)
+            x, dy, KS, KS, S, S, P, P, False)), fl)
 
 
 def bench_bn(dtype=torch.bfloat16):

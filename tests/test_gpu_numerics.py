@@ -110,7 +110,7 @@ def test_gemm_tn(dtype, M, N, K):
     torch.manual_seed(3)
     a = torch.randn(M, K, dtype=dtype, device=DEV)
     b = torch.randn(M, N, dtype=dtype, device=DEV)
-    c = ext.gemm_tn(a, b)  # fp32 out
+    c = ext.gemm_tn(a, b, False)  # fp32 out
     ref = a.float().t() @ b.float()
     tol = TOL[dtype] * max(1, M // 256)
     assert relerr(c, ref) < tol, relerr(c, ref)
@@ -180,7 +180,7 @@ def test_conv2d_wgrad(dtype, case):
     OW = (W + 2 * P - KW) // S + 1
     x = torch.randn(N, H, W, Ci, dtype=dtype, device=DEV)
     dy = torch.randn(N, OH, OW, Co, dtype=dtype, device=DEV)
-    dw = ext.conv2d_wgrad(x, dy, KH, KW, S, S, P, P)  # [KH,KW,Ci,Co] fp32
+    dw = ext.conv2d_wgrad(x, dy, KH, KW, S, S, P, P, False)  # [KH,KW,Ci,Co] fp32
     ref = torch.nn.grad.conv2d_weight(
         x.permute(0, 3, 1, 2).float().cpu(), (Co, Ci, KH, KW),
         dy.permute(0, 3, 1, 2).float().cpu(), stride=S, padding=P)

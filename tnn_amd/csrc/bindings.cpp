@@ -147,22 +147,26 @@ at::Tensor gemm_nt(const at::Tensor& a, const at::Tensor& b) {
   return c;
 }
 
-at::Tensor gemm_tn(const at::Tensor& a, const at::Tensor& b) {
-  // c[k, n] = sum_m a[m, k] * b[m, n]  (fp32 accumulate + output)
+at::Tensor gemm_tn(const at::Tensor& a, const at::Tensor& b, bool out_in_dt) {
+  // c[k, n] = sum_m a[m, k] * b[m, n]  (fp32 accumulate; output fp32, or
+  // a's dtype when out_in_dt -- the cast rides the split-K reduce)
   CHECK_IN(a);
   CHECK_IN(b);
   TORCH_CHECK(a.dim() == 2 && b.dim() == 2 && a.size(0) == b.size(0),
               "gemm_tn shapes ", a.sizes(), "^T @ ", b.sizes());
   int M = a.size(0), K = a.size(1), N = b.size(1);
-  auto c = at::empty({K, N}, a.options().dtype(at::kFloat));
+  DT out_dt = out_in_dt ? dt_of(a) : DT::F32;
+  auto c = at::empty({K, N}, out_dt == DT::F32
+                                 ? a.options().dtype(at::kFloat)
+                                 : a.options());
   int z = gemm_tn_zsplits(M, N, K);
   at::Tensor ws;
   float* wsp = nullptr;
-  if (z > 1) {
+  if (z > 1 || out_dt != DT::F32) {
     ws = at::empty({(int64_t)z * K * N}, a.options().dtype(at::kFloat));
     wsp = ws.data_ptr<float>();
   }
-  gemm_tn_launch(dt_of(a), a.data_ptr(), b.data_ptr(), c.data_ptr<float>(),
+  gemm_tn_launch(dt_of(a), a.data_ptr(), b.data_ptr(), c.data_ptr(), out_dt,
                  wsp, z, zero_page(a), M, N, K, cur_stream());
   return c;
 }
@@ -277,24 +281,25 @@ at::Tensor conv2d_dgrad(const at::Tensor& dy, const at::Tensor& w, int64_t H,
 
 at::Tensor conv2d_wgrad(const at::Tensor& x, const at::Tensor& dy, int64_t KH,
                         int64_t KW, int64_t sh, int64_t sw, int64_t ph,
-                        int64_t pw) {
+                        int64_t pw, bool out_in_dt) {
   CHECK_IN(x);
   CHECK_IN(dy);
   auto cs = conv_shape(x, x.size(3), dy.size(3), KH, KW, sh, sw, ph, pw);
   TORCH_CHECK(dy.size(1) == cs.OH && dy.size(2) == cs.OW, "wgrad shape");
+  DT out_dt = out_in_dt ? dt_of(x) : DT::F32;
   auto dw = at::empty({KH, KW, cs.Cin, cs.Cout},
-                      x.options().dtype(at::kFloat));
+                      out_dt == DT::F32 ? x.options().dtype(at::kFloat)
+                                        : x.options());
   int z = conv2d_wgrad_zsplits(cs);
   at::Tensor ws;
   float* wsp = nullptr;
-  if (z > 1) {
+  if (z > 1 || out_dt != DT::F32) {
     ws = at::empty({(int64_t)z * KH * KW * cs.Cin * cs.Cout},
                    x.options().dtype(at::kFloat));
     wsp = ws.data_ptr<float>();
   }
-  conv2d_wgrad_launch(dt_of(x), x.data_ptr(), dy.data_ptr(),
-                      dw.data_ptr<float>(), wsp, z, zero_page(x), cs,
-                      cur_stream());
+  conv2d_wgrad_launch(dt_of(x), x.data_ptr(), dy.data_ptr(), dw.data_ptr(),
+                      out_dt, wsp, z, zero_page(x), cs, cur_stream());
   return dw;
 }
 
@@ -601,6 +606,19 @@ void adam_step(at::Tensor param, at::Tensor master, const at::Tensor& grad,
                    adamw, has_master, cur_stream());
 }
 
+void adam_step_mt(const at::Tensor& desc, const at::Tensor& chunks,
+                  int64_t dt_p, int64_t dt_g, bool has_master, int64_t step,
+                  double lr, double beta1, double beta2, double eps,
+                  double weight_decay, bool adamw) {
+  CHECK_IN(desc);
+  CHECK_IN(chunks);
+  adam_mt_launch((DT)dt_p, (DT)dt_g, has_master,
+                 desc.data_ptr<int64_t>(), chunks.data_ptr<int64_t>(),
+                 (int)chunks.numel(), (int)step, (float)lr, (float)beta1,
+                 (float)beta2, (float)eps, (float)weight_decay, adamw,
+                 cur_stream());
+}
+
 }  // namespace tnn
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -635,4 +653,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_bwd", &tnn::attn_bwd);
   m.def("sgd_step", &tnn::sgd_step);
   m.def("adam_step", &tnn::adam_step);
+  m.def("adam_step_mt", &tnn::adam_step_mt);
 }

@@ -987,11 +987,14 @@ int conv2d_wgrad_zsplits(const ConvShape& cs) {
   return std::max(z, 1);
 }
 
-void conv2d_wgrad_launch(DT dt, const void* x, const void* dy, float* dw_f32,
-                         float* ws, int z, const void* zero16,
+void conv2d_wgrad_launch(DT dt, const void* x, const void* dy, void* dw_out,
+                         DT out_dt, float* ws, int z, const void* zero16,
                          const ConvShape& cs, hipStream_t s) {
   int Kout = cs.KH * cs.KW * cs.Cin;
-  float* target = z == 1 ? dw_f32 : ws;
+  // bf16 output always goes through the (casting) reduce; fp32 with z==1
+  // writes the slab target directly
+  bool direct = z == 1 && out_dt == DT::F32;
+  float* target = direct ? (float*)dw_out : ws;
   dim3 grid(ceil_div(Kout, BM), ceil_div(cs.Cout, BN), z);
   bool p2 = all_pow2(cs);
   if (dt == DT::F32) {
@@ -1019,8 +1022,8 @@ void conv2d_wgrad_launch(DT dt, const void* x, const void* dy, float* dw_f32,
                          (const bf16*)dy, target, cs);
     }
   }
-  if (z > 1)
-    splitk_reduce_launch(ws, dw_f32, z, (int64_t)Kout * cs.Cout, s);
+  if (!direct)
+    splitk_reduce_launch(ws, dw_out, out_dt, z, (int64_t)Kout * cs.Cout, s);
 }
 
 void transpose_w_launch(DT dt, const void* w, void* w_t, int KH, int KW,

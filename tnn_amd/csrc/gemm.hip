@@ -427,22 +427,24 @@ __global__ void k_gemm_tn_vec(const T* __restrict__ A, const T* __restrict__ B,
   });
 }
 
+template <typename OUT>
 __global__ void k_splitk_reduce(const float* __restrict__ ws,
-                                float* __restrict__ out, int z, int64_t n) {
+                                OUT* __restrict__ out, int z, int64_t n) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (; i < n; i += stride) {
     float acc = 0.0f;
     for (int s = 0; s < z; ++s) acc += ws[(int64_t)s * n + i];
-    out[i] = acc;
+    out[i] = VecIO<OUT>::from_f32(acc);
   }
 }
 
 // float4 variant with 4 independent slab accumulators: the scalar kernel is
 // latency-bound (one dependent 4B load per z iteration measured ~1.25 TB/s);
 // 16B loads x 4-deep MLP reach the HBM roofline.
+template <typename OUT>
 __global__ void k_splitk_reduce4(const float4* __restrict__ ws,
-                                 float4* __restrict__ out, int z, int64_t n4) {
+                                 OUT* __restrict__ out, int z, int64_t n4) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (; i < n4; i += stride) {
@@ -462,27 +464,38 @@ __global__ void k_splitk_reduce4(const float4* __restrict__ ws,
       float4 v = ws[(int64_t)s * n4 + i];
       a0.x += v.x; a0.y += v.y; a0.z += v.z; a0.w += v.w;
     }
-    float4 r;
-    r.x = (a0.x + a1.x) + (a2.x + a3.x);
-    r.y = (a0.y + a1.y) + (a2.y + a3.y);
-    r.z = (a0.z + a1.z) + (a2.z + a3.z);
-    r.w = (a0.w + a1.w) + (a2.w + a3.w);
-    out[i] = r;
+    float r[4];
+    r[0] = (a0.x + a1.x) + (a2.x + a3.x);
+    r[1] = (a0.y + a1.y) + (a2.y + a3.y);
+    r[2] = (a0.z + a1.z) + (a2.z + a3.z);
+    r[3] = (a0.w + a1.w) + (a2.w + a3.w);
+    struct alignas(4 * sizeof(OUT)) O4 { OUT e[4]; } o;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) o.e[j] = VecIO<OUT>::from_f32(r[j]);
+    ((O4*)out)[i] = o;
   }
 }
 
-void splitk_reduce_launch(const float* ws, float* out, int z, int64_t n,
-                          hipStream_t s) {
+void splitk_reduce_launch(const float* ws, void* out, DT out_dt, int z,
+                          int64_t n, hipStream_t s) {
   if ((n & 3) == 0) {
     int64_t n4 = n >> 2;
     int blocks = (int)std::min<int64_t>((n4 + 255) / 256, (int64_t)2048);
-    hipLaunchKernelGGL(k_splitk_reduce4, dim3(blocks), dim3(256), 0, s,
-                       (const float4*)ws, (float4*)out, z, n4);
+    if (out_dt == DT::F32)
+      hipLaunchKernelGGL(k_splitk_reduce4<float>, dim3(blocks), dim3(256), 0,
+                         s, (const float4*)ws, (float*)out, z, n4);
+    else
+      hipLaunchKernelGGL(k_splitk_reduce4<bf16>, dim3(blocks), dim3(256), 0,
+                         s, (const float4*)ws, (bf16*)out, z, n4);
     return;
   }
   int blocks = (int)std::min<int64_t>((n + 255) / 256, (int64_t)2048);
-  hipLaunchKernelGGL(k_splitk_reduce, dim3(blocks), dim3(256), 0, s, ws, out,
-                     z, n);
+  if (out_dt == DT::F32)
+    hipLaunchKernelGGL(k_splitk_reduce<float>, dim3(blocks), dim3(256), 0, s,
+                       ws, (float*)out, z, n);
+  else
+    hipLaunchKernelGGL(k_splitk_reduce<bf16>, dim3(blocks), dim3(256), 0, s,
+                       ws, (bf16*)out, z, n);
 }
 
 // ---------------------------------------------------------------------------
@@ -575,11 +588,12 @@ int gemm_tn_zsplits(int M, int N, int K) {
   return std::max(z, 1);
 }
 
-void gemm_tn_launch(DT dt, const void* a, const void* b, float* c_f32,
-                    float* ws, int z, const void* zero16, int M, int N, int K,
-                    hipStream_t s) {
+void gemm_tn_launch(DT dt, const void* a, const void* b, void* c_out,
+                    DT out_dt, float* ws, int z, const void* zero16, int M,
+                    int N, int K, hipStream_t s) {
   dim3 grid(ceil_div(K, BM), ceil_div(N, BN), z);
-  float* target = z == 1 ? c_f32 : ws;
+  bool direct = z == 1 && out_dt == DT::F32;
+  float* target = direct ? (float*)c_out : ws;
   if (dt == DT::F32) {
     if (K % 4 == 0 && N % 4 == 0 && (((uintptr_t)a & 15) == 0) &&
         (((uintptr_t)b & 15) == 0))
@@ -599,7 +613,8 @@ void gemm_tn_launch(DT dt, const void* a, const void* b, float* c_f32,
       hipLaunchKernelGGL(k_gemm_tn<bf16>, grid, dim3(THREADS), 0, s,
                          (const bf16*)a, (const bf16*)b, target, M, N, K);
   }
-  if (z > 1) splitk_reduce_launch(ws, c_f32, z, (int64_t)K * N, s);
+  if (!direct)
+    splitk_reduce_launch(ws, c_out, out_dt, z, (int64_t)K * N, s);
 }
 
 // ---------------------------------------------------------------------------

@@ -35,11 +35,11 @@ void gemv_nn_launch(DT dt, const void* x, const void* b, const void* bias,
                     void* y, float* ws, int ks, int N, int K, int act_kind,
                     hipStream_t s);
 int gemm_tn_zsplits(int M, int N, int K);
-void gemm_tn_launch(DT dt, const void* a, const void* b, float* c_f32,
-                    float* ws, int z, const void* zero16, int M, int N, int K,
-                    hipStream_t s);
-void splitk_reduce_launch(const float* ws, float* out, int z, int64_t n,
-                          hipStream_t s);
+void gemm_tn_launch(DT dt, const void* a, const void* b, void* c_out,
+                    DT out_dt, float* ws, int z, const void* zero16, int M,
+                    int N, int K, hipStream_t s);
+void splitk_reduce_launch(const float* ws, void* out, DT out_dt, int z,
+                          int64_t n, hipStream_t s);
 void mfma_selftest_launch(const void* a_bf16, const void* b_bf16, float* d,
                           hipStream_t s);
 void mfma_selftest_f32_launch(const float* a, const float* b, float* d,
@@ -83,8 +83,8 @@ void conv2d_dgrad_launch(DT dt, const void* dy, const void* w_t,
                          const void* w_t2d, void* dx, const void* zero16,
                          const ConvShape& cs, hipStream_t s);
 int conv2d_wgrad_zsplits(const ConvShape& cs);
-void conv2d_wgrad_launch(DT dt, const void* x, const void* dy, float* dw_f32,
-                         float* ws, int z, const void* zero16,
+void conv2d_wgrad_launch(DT dt, const void* x, const void* dy, void* dw_out,
+                         DT out_dt, float* ws, int z, const void* zero16,
                          const ConvShape& cs, hipStream_t s);
 void transpose_w_launch(DT dt, const void* w, void* w_t, int KH, int KW,
                         int Cin, int Cout, hipStream_t s);
@@ -171,5 +171,9 @@ void adam_step_launch(DT dt_p, const void* grad, DT dt_g, void* param,
                       float lr, float beta1, float beta2, float eps,
                       float weight_decay, bool adamw, bool has_master,
                       hipStream_t s);
+void adam_mt_launch(DT dt_p, DT dt_g, bool has_master, const int64_t* desc,
+                    const int64_t* chunks, int nchunks, int step, float lr,
+                    float beta1, float beta2, float eps, float weight_decay,
+                    bool adamw, hipStream_t s);
 
 }  // namespace tnn

@@ -50,6 +50,63 @@ __global__ void k_adam(TP* __restrict__ p, float* __restrict__ master,
   }
 }
 
+// Multi-tensor Adam: one launch for a whole parameter group. desc holds
+// 6 int64 per tensor (param, master, grad, m, v pointers + numel); chunks
+// maps each block to (tensor_idx << 32 | chunk_index) over CHUNK-element
+// spans. Collapses ~150 per-parameter launches into one.
+constexpr int MT_CHUNK = 16384;
+
+template <typename TP, typename TG, bool MASTER>
+__launch_bounds__(256)
+__global__ void k_adam_mt(const int64_t* __restrict__ desc,
+                          const int64_t* __restrict__ chunks, float lr,
+                          float beta1, float beta2, float eps, float wd,
+                          float bc1, float bc2, bool adamw) {
+  int64_t c = chunks[blockIdx.x];
+  int ti = (int)(c >> 32);
+  int64_t off = (int64_t)(c & 0xffffffff) * MT_CHUNK;
+  const int64_t* d = desc + (int64_t)ti * 6;
+  TP* p = (TP*)d[0];
+  float* master = (float*)d[1];
+  const TG* g = (const TG*)d[2];
+  float* m = (float*)d[3];
+  float* v = (float*)d[4];
+  int64_t n = d[5];
+  int64_t end = off + MT_CHUNK < n ? off + MT_CHUNK : n;
+  for (int64_t i = off + threadIdx.x; i < end; i += 256) {
+    float w = MASTER ? master[i] : VecIO<TP>::to_f32(p[i]);
+    float gr = VecIO<TG>::to_f32(g[i]);
+    if (!adamw) gr += wd * w;
+    float mi = m[i] = beta1 * m[i] + (1.0f - beta1) * gr;
+    float vi = v[i] = beta2 * v[i] + (1.0f - beta2) * gr * gr;
+    if (adamw) w *= (1.0f - lr * wd);
+    w -= lr / bc1 * mi / (sqrtf(vi / bc2) + eps);
+    if (MASTER) master[i] = w;
+    p[i] = VecIO<TP>::from_f32(w);
+  }
+}
+
+void adam_mt_launch(DT dt_p, DT dt_g, bool has_master, const int64_t* desc,
+                    const int64_t* chunks, int nchunks, int step, float lr,
+                    float beta1, float beta2, float eps, float weight_decay,
+                    bool adamw, hipStream_t s) {
+  float bc1 = 1.0f - powf(beta1, (float)step);
+  float bc2 = 1.0f - powf(beta2, (float)step);
+#define CASE(TP, TG, M)                                                         hipLaunchKernelGGL((k_adam_mt<TP, TG, M>), dim3(nchunks), dim3(256), 0, s,                       desc, chunks, lr, beta1, beta2, eps, weight_decay, bc1,                       bc2, adamw)
+  if (has_master) {
+    if (dt_p == DT::F32 && dt_g == DT::F32) CASE(float, float, true);
+    else if (dt_p == DT::BF16 && dt_g == DT::BF16) CASE(bf16, bf16, true);
+    else if (dt_p == DT::BF16 && dt_g == DT::F32) CASE(bf16, float, true);
+    else CASE(float, bf16, true);
+  } else {
+    if (dt_p == DT::F32 && dt_g == DT::F32) CASE(float, float, false);
+    else if (dt_p == DT::BF16 && dt_g == DT::BF16) CASE(bf16, bf16, false);
+    else if (dt_p == DT::BF16 && dt_g == DT::F32) CASE(bf16, float, false);
+    else CASE(float, bf16, false);
+  }
+#undef CASE
+}
+
 static inline int ob(int64_t n) {
   int64_t b = (n + 255) / 256;
   return (int)(b < 2048 ? b : 2048);

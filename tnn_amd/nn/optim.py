@@ -103,6 +103,64 @@ class Adam(Optimizer):
                  weight_decay=0.0):
         super().__init__(params, lr)
         self.beta1, self.beta2, self.eps, self.weight_decay = beta1, beta2, eps, weight_decay
+        self._mt_chunk_cache = {}
+
+    @torch.no_grad()
+    def step(self):
+        """Multi-tensor fused update on GPU: one kernel launch per
+        (param dtype, grad dtype, has-master) group instead of one per
+        parameter (~150 launches -> ~2 on GPT-2)."""
+        if not self.params or not self.params[0].is_cuda:
+            return super().step()
+        self.step_count += 1
+        ext = _C.ext()
+        groups = {}
+        for i, p in enumerate(self.params):
+            if p.grad is None:
+                continue
+            st = self.state.setdefault(i, {})
+            if p.dtype != torch.float32 and "master" not in st:
+                st["master"] = p.detach().float().clone()
+            if "m" not in st:
+                ref = st.get("master", p.data)
+                st["m"] = torch.zeros_like(ref, dtype=torch.float32)
+                st["v"] = torch.zeros_like(ref, dtype=torch.float32)
+            g = p.grad
+            if (g.dtype not in (torch.float32, torch.bfloat16)
+                    or not g.is_contiguous() or g.dtype != p.dtype):
+                self._update(p, g, st)
+                continue
+            groups.setdefault((p.dtype, g.dtype, "master" in st),
+                              []).append((p, g, st))
+        CH = 16384  # MT_CHUNK in optim.hip
+        for (pdt, gdt, has_master), items in groups.items():
+            if len(items) == 1:
+                p, g, st = items[0]
+                self._update(p, g, st)
+                continue
+            dev = items[0][0].device
+            # chunk map depends only on the numels (stable): cache it
+            numels = tuple(p.numel() for p, _, _ in items)
+            chunks_t = self._mt_chunk_cache.get(numels)
+            if chunks_t is None:
+                chunks = []
+                for ti, n in enumerate(numels):
+                    for c in range((n + CH - 1) // CH):
+                        chunks.append((ti << 32) | c)
+                chunks_t = torch.tensor(chunks, dtype=torch.int64, device=dev)
+                self._mt_chunk_cache[numels] = chunks_t
+            desc = []
+            for p, g, st in items:
+                desc += [p.data_ptr(),
+                         st["master"].data_ptr() if has_master else 0,
+                         g.data_ptr(), st["m"].data_ptr(),
+                         st["v"].data_ptr(), p.numel()]
+            desc_t = torch.tensor(desc, dtype=torch.int64, device=dev)
+            ext.adam_step_mt(desc_t, chunks_t,
+                             0 if pdt == torch.float32 else 1,
+                             0 if gdt == torch.float32 else 1, has_master,
+                             self.step_count, self.lr, self.beta1, self.beta2,
+                             self.eps, self.weight_decay, self._adamw)
 
     def _update(self, p, g, st):
         if "m" not in st:

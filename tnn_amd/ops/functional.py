@@ -73,8 +73,9 @@ class _Conv2dNHWC(torch.autograd.Function):
                                   ctx.stride[0], ctx.stride[1], ctx.padding[0], ctx.padding[1])
         if ctx.needs_input_grad[1]:
             dw = ext.conv2d_wgrad(x, dy, w.shape[0], w.shape[1],
-                                  ctx.stride[0], ctx.stride[1], ctx.padding[0], ctx.padding[1])
-            dw = dw.to(w.dtype)
+                                  ctx.stride[0], ctx.stride[1], ctx.padding[0],
+                                  ctx.padding[1], w.dtype == x.dtype)
+            dw = dw.to(w.dtype)  # no-op when the reduce already cast
         if ctx.has_bias and ctx.needs_input_grad[2]:
             db = ext.colsum(dy.reshape(-1, dy.shape[-1]))
         return dx, dw, db, None, None, None
@@ -228,14 +229,14 @@ class _Linear(torch.autograd.Function):
             if ctx.needs_input_grad[0]:
                 dx = ext.gemm_nt(dy_p, F.pad(w, (0, pad)))
             if ctx.needs_input_grad[1]:
-                dw = ext.gemm_tn(x, dy_p)[:, :N].contiguous()
-                dw = dw.to(w.dtype)
+                dw = ext.gemm_tn(x, dy_p, w.dtype == x.dtype)
+                dw = dw[:, :N].contiguous().to(w.dtype)
             return dx, dw, db, None
         if ctx.needs_input_grad[0]:
             dx = ext.gemm_nt(dy, w)       # [M,N] @ [K,N]^T -> [M,K]
         if ctx.needs_input_grad[1]:
-            dw = ext.gemm_tn(x, dy)       # [M,K]^T @ [M,N] -> [K,N]
-            dw = dw.to(w.dtype)
+            dw = ext.gemm_tn(x, dy, w.dtype == x.dtype)
+            dw = dw.to(w.dtype)  # no-op when the reduce already cast
         return dx, dw, db, None
 
 
