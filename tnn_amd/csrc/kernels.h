@@ -61,6 +61,7 @@ struct IDiv {
 struct ConvShape {
   int N, H, W, Cin, Cout, KH, KW, SH, SW, PH, PW, OH, OW;
   IDiv d_ohow, d_ow, d_cin, d_hw, d_w, d_cout;
+  int mkw16;  // ceil-reciprocal for kidx / KW: exact while kidx * KW < 2^16
   void init_fdiv() {
     d_ohow.set(OH * OW);
     d_ow.set(OW);
@@ -68,6 +69,7 @@ struct ConvShape {
     d_hw.set(H * W);
     d_w.set(W);
     d_cout.set(Cout);
+    mkw16 = 65536 / KW + 1;
   }
 };
 bool conv2d_fwd_wants_db(DT dt, const void* x, const ConvShape& cs);
