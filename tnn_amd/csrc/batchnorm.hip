@@ -76,6 +76,7 @@ __global__ void k_bn_partial_vec(const T* __restrict__ x,
   if (g < groups && r_off < rows_per_iter) {
     const int64_t r0 = rows * blockIdx.y / gridDim.y;
     const int64_t r1 = rows * (blockIdx.y + 1) / gridDim.y;
+#pragma unroll 4
     for (int64_t r = r0 + r_off; r < r1; r += rows_per_iter) {
       VecT v = *(const VecT*)&x[r * cols + (int64_t)g * V];
 #pragma unroll
@@ -135,6 +136,7 @@ __global__ void k_bn_bwd_reduce_vec(const T* __restrict__ x,
       m[j] = mean[g * V + j];
       is[j] = invstd[g * V + j];
     }
+#pragma unroll 2
     for (int64_t r = r0 + r_off; r < r1; r += rows_per_iter) {
       int64_t i = r * cols + (int64_t)g * V;
       VecT vg = *(const VecT*)&dy[i];

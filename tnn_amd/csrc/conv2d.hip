@@ -827,11 +827,15 @@ bool conv2d_fwd_wants_db(DT dt, const void* x, const ConvShape& cs) {
     return !(e && atoi(e) == 0);
   }();
   if (!db_on) return false;
+  static const int maxk = []() {
+    const char* e = getenv("TNN_DB_MAXK");
+    return e ? atoi(e) : 2304;
+  }();
   int v = dt == DT::F32 ? 4 : 8;
   // 2x LDS buffers cap occupancy at 3 blocks/CU: a win while the k-loop is
   // short (measured +8% at K<=1152, -15% at K=4608 where the single-buffer
   // kernel's 6 resident blocks hide latency better)
-  return cs.KH * cs.KW * cs.Cin <= 2304 && all_pow2(cs) &&
+  return cs.KH * cs.KW * cs.Cin <= maxk && all_pow2(cs) &&
          cs.Cin % v == 0 && (((uintptr_t)x & 15) == 0);
 }
 
@@ -896,8 +900,12 @@ bool conv2d_dgrad_wants_db(DT dt, const void* dy, const ConvShape& cs) {
     return !(e && atoi(e) == 0);
   }();
   if (!db_on) return false;
+  static const int maxk = []() {
+    const char* e = getenv("TNN_DB_MAXK_DGRAD");
+    return e ? atoi(e) : 4608;  // dgrad ties-or-wins with DB up to here
+  }();
   int v = dt == DT::F32 ? 4 : 8;
-  return cs.KH * cs.KW * cs.Cout <= 2304 && all_pow2(cs) &&
+  return cs.KH * cs.KW * cs.Cout <= maxk && all_pow2(cs) &&
          cs.Cout % v == 0 && (((uintptr_t)dy & 15) == 0);
 }
 

@@ -186,6 +186,7 @@ __global__ void k_colsum_vec(const T* __restrict__ x, float* __restrict__ out,
   if (g < groups && r_off < rows_per_iter) {
     const int64_t r0 = rows * blockIdx.y / gridDim.y;
     const int64_t r1 = rows * (blockIdx.y + 1) / gridDim.y;
+#pragma unroll 4
     for (int64_t r = r0 + r_off; r < r1; r += rows_per_iter) {
       P v = *(const P*)&x[r * cols + (int64_t)g * V];
 #pragma unroll

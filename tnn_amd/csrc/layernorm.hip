@@ -112,6 +112,7 @@ __global__ void k_ln_bwd_col(const T* __restrict__ x, const T* __restrict__ dy,
   if (g < groups && r_off < rows_per_iter) {
     const int64_t r0 = rows * blockIdx.y / gridDim.y;
     const int64_t r1 = rows * (blockIdx.y + 1) / gridDim.y;
+#pragma unroll 2
     for (int64_t r = r0 + r_off; r < r1; r += rows_per_iter) {
       const float m = mean[r], is = invstd[r];
       int64_t i = r * cols + (int64_t)g * V;
