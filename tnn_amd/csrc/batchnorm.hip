@@ -245,6 +245,41 @@ __global__ void k_bn_apply_drop(const T* __restrict__ x, const float* mean,
   }
 }
 
+// vectorized fused BN+ReLU+dropout: one 16B load/store per thread-iter,
+// two Philox draws per vector (the scalar form measured 1.65 TB/s vs ~5
+// for plain bn_apply)
+template <typename T>
+__global__ void k_bn_apply_drop_vec(const T* __restrict__ x, const float* mean,
+                                    const float* invstd, const float* gamma,
+                                    const float* beta, T* __restrict__ y,
+                                    int64_t nv, int groups, float p,
+                                    uint64_t seed) {
+  constexpr int V = 16 / sizeof(T);
+  using VecT = BnPack<T>;
+  const float scale = 1.0f / (1.0f - p);
+  Philox rng(seed);
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < nv; i += stride) {
+    int c0 = (int)((uint32_t)i % (uint32_t)groups) * V;
+    VecT vx = ((const VecT*)x)[i];
+    uint4 r0 = rng(2 * i);
+    uint4 r1 = rng(2 * i + 1);
+    unsigned int rs[8] = {r0.x, r0.y, r0.z, r0.w, r1.x, r1.y, r1.z, r1.w};
+    VecT vy;
+#pragma unroll
+    for (int j = 0; j < V; ++j) {
+      int c = c0 + j;
+      float v =
+          (VecIO<T>::to_f32(vx.e[j]) - mean[c]) * invstd[c] * gamma[c] + beta[c];
+      v = fmaxf(v, 0.0f);
+      bool keep = u32_to_uniform(rs[j]) > p;
+      vy.e[j] = VecIO<T>::from_f32(keep ? v * scale : 0.0f);
+    }
+    ((VecT*)y)[i] = vy;
+  }
+}
+
 template <typename T>
 __global__ void k_bn_infer(const T* __restrict__ x, const float* rmean,
                            const float* rvar, const float* gamma,
@@ -404,6 +439,20 @@ void bn_apply_drop_launch(DT dt, const void* x, const float* mean,
                           const float* beta, void* y, int64_t rows, int cols,
                           float p, uint64_t seed, hipStream_t s) {
   int64_t n = rows * cols;
+  int V = dt == DT::F32 ? 4 : 8;
+  if (cols % V == 0 && (((uintptr_t)x & 15) == 0) && n < (1ll << 34)) {
+    int64_t nv = n / V;
+    int blocks = (int)std::min<int64_t>((nv + 255) / 256, (int64_t)2048);
+    if (dt == DT::F32)
+      hipLaunchKernelGGL(k_bn_apply_drop_vec<float>, dim3(blocks), dim3(256),
+                         0, s, (const float*)x, mean, invstd, gamma, beta,
+                         (float*)y, nv, cols / V, p, seed);
+    else
+      hipLaunchKernelGGL(k_bn_apply_drop_vec<bf16>, dim3(blocks), dim3(256),
+                         0, s, (const bf16*)x, mean, invstd, gamma, beta,
+                         (bf16*)y, nv, cols / V, p, seed);
+    return;
+  }
   int blocks = (int)std::min<int64_t>(((n + 3) / 4 + 255) / 256, (int64_t)2048);
   if (dt == DT::F32)
     hipLaunchKernelGGL(k_bn_apply_drop<float>, dim3(blocks), dim3(256), 0, s,
