@@ -662,12 +662,28 @@ __global__ void k_conv_wgrad_vec(const T* __restrict__ X,
   using VecT = Pack16<T>;
 
   for (int k0 = m_begin; k0 < m_end; k0 += BK) {
+    // bf16: each thread owns one 8-row group x 4 consecutive m columns --
+    // after the loads an in-thread 8x4 transpose packs 4 m-values per row
+    // into one 8B ds_write_b64 (vs 8x ds_write_b16 per load; LDS
+    // instruction issue was wgrad's post-phase-separation bottleneck:
+    // 48 b16 writes vs 16 MFMAs per thread-chunk). The XOR swizzle flips
+    // byte bits 4-6 only, so the 8B alignment survives. fp32 keeps the
+    // generic scatter.
+    constexpr bool TR = sizeof(T) == 2;
+    const int a_rr = TR ? (threadIdx.x & (BM / V - 1)) * V : 0;
+    const int a_mm0 = TR ? (threadIdx.x / (BM / V)) * 4 : 0;
     const T* asrc[RA];
 #pragma unroll
     for (int it = 0; it < RA; ++it) {
-      int c = threadIdx.x + it * THREADS;
-      int mm = c / (BM / V);
-      int rr = (c % (BM / V)) * V;
+      int mm, rr;
+      if constexpr (TR) {
+        mm = a_mm0 + it;
+        rr = a_rr;
+      } else {
+        int c = threadIdx.x + it * THREADS;
+        mm = c / (BM / V);
+        rr = (c % (BM / V)) * V;
+      }
       int gm = k0 + mm, gr = r0 + rr;
       const T* p = zero16;
       if (gm < m_end && gr < Kout) {
@@ -701,13 +717,23 @@ __global__ void k_conv_wgrad_vec(const T* __restrict__ X,
     for (int it = 0; it < RA; ++it) va[it] = *(const VecT*)asrc[it];
 #pragma unroll
     for (int it = 0; it < RB; ++it) vb[it] = *(const VecT*)bsrc[it];
+    if constexpr (TR) {
 #pragma unroll
-    for (int it = 0; it < RA; ++it) {
-      int c = threadIdx.x + it * THREADS;
-      int mm = c / (BM / V);
-      int rr = (c % (BM / V)) * V;
+      for (int j = 0; j < V; ++j) {
+        struct alignas(8) H4 { T e[4]; } h;
 #pragma unroll
-      for (int j = 0; j < V; ++j) As[lds_off<T>(rr + j, mm)] = va[it].e[j];
+        for (int q = 0; q < 4; ++q) h.e[q] = va[q].e[j];
+        *(H4*)&As[lds_off<T>(a_rr + j, a_mm0)] = h;
+      }
+    } else {
+#pragma unroll
+      for (int it = 0; it < RA; ++it) {
+        int c = threadIdx.x + it * THREADS;
+        int mm = c / (BM / V);
+        int rr = (c % (BM / V)) * V;
+#pragma unroll
+        for (int j = 0; j < V; ++j) As[lds_off<T>(rr + j, mm)] = va[it].e[j];
+      }
     }
 #pragma unroll
     for (int it = 0; it < RB; ++it) {
