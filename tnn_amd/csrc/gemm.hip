@@ -377,12 +377,23 @@ __global__ void k_gemm_tn_vec(const T* __restrict__ A, const T* __restrict__ B,
   using VecT = Pack16<T>;
 
   for (int k0 = m_begin; k0 < m_end; k0 += BK) {
+    // bf16: (8-row group x 4 consecutive m) per thread + in-thread
+    // transpose -> ds_write_b64 (see k_conv_wgrad_vec)
+    constexpr bool TR = sizeof(T) == 2;
+    const int a_rr = TR ? (threadIdx.x & (BM / V - 1)) * V : 0;
+    const int a_mm0 = TR ? (threadIdx.x / (BM / V)) * 4 : 0;
     const T* asrc[RA];
 #pragma unroll
     for (int it = 0; it < RA; ++it) {
-      int c = threadIdx.x + it * THREADS;
-      int mm = c / (BM / V);
-      int rr = (c % (BM / V)) * V;
+      int mm, rr;
+      if constexpr (TR) {
+        mm = a_mm0 + it;
+        rr = a_rr;
+      } else {
+        int c = threadIdx.x + it * THREADS;
+        mm = c / (BM / V);
+        rr = (c % (BM / V)) * V;
+      }
       int gm = k0 + mm, gr = r0 + rr;
       asrc[it] = (gm < m_end && gr < K) ? &A[(int64_t)gm * K + gr] : zero16;
     }
@@ -400,13 +411,23 @@ __global__ void k_gemm_tn_vec(const T* __restrict__ A, const T* __restrict__ B,
     for (int it = 0; it < RA; ++it) va[it] = *(const VecT*)asrc[it];
 #pragma unroll
     for (int it = 0; it < RB; ++it) vb[it] = *(const VecT*)bsrc[it];
+    if constexpr (TR) {
 #pragma unroll
-    for (int it = 0; it < RA; ++it) {
-      int c = threadIdx.x + it * THREADS;
-      int mm = c / (BM / V);
-      int rr = (c % (BM / V)) * V;
+      for (int j = 0; j < V; ++j) {
+        struct alignas(8) H4 { T e[4]; } h;
 #pragma unroll
-      for (int j = 0; j < V; ++j) As[lds_off<T>(rr + j, mm)] = va[it].e[j];
+        for (int q = 0; q < 4; ++q) h.e[q] = va[q].e[j];
+        *(H4*)&As[lds_off<T>(a_rr + j, a_mm0)] = h;
+      }
+    } else {
+#pragma unroll
+      for (int it = 0; it < RA; ++it) {
+        int c = threadIdx.x + it * THREADS;
+        int mm = c / (BM / V);
+        int rr = (c % (BM / V)) * V;
+#pragma unroll
+        for (int j = 0; j < V; ++j) As[lds_off<T>(rr + j, mm)] = va[it].e[j];
+      }
     }
 #pragma unroll
     for (int it = 0; it < RB; ++it) {
