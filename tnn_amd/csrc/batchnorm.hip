@@ -371,7 +371,7 @@ static inline dim3 bn_reduce_grid_vec(int64_t rows, int cols) {
   int rows_per_iter = 256 / gpb;
   int64_t iters = (rows + rows_per_iter - 1) / rows_per_iter;
   int rslices = (int)std::min<int64_t>(std::max<int64_t>(2048 / cblocks, 1),
-                                       std::max<int64_t>(iters / 8, 1));
+                                       std::max<int64_t>(iters / 32, 1));
   return dim3(cblocks, rslices);
 }
 

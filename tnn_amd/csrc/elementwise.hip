@@ -224,7 +224,7 @@ void colsum_launch(DT dt, const void* x, void* out_f32, int64_t rows,
     int rows_per_iter = 256 / gpb;
     int64_t iters = (rows + rows_per_iter - 1) / rows_per_iter;
     int rslices = (int)std::min<int64_t>(
-        std::max<int64_t>(2048 / cblocks, 1), std::max<int64_t>(iters / 8, 1));
+        std::max<int64_t>(2048 / cblocks, 1), std::max<int64_t>(iters / 32, 1));
     dim3 grid(cblocks, rslices);
     if (dt == DT::F32)
       hipLaunchKernelGGL(k_colsum_vec<float>, grid, dim3(256), 0, s,

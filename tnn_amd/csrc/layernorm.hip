@@ -184,7 +184,7 @@ void ln_bwd_launch(DT dt, const void* x, const void* dy, const float* gamma,
     int rows_per_iter = 256 / gpb;
     int64_t iters = (rows + rows_per_iter - 1) / rows_per_iter;
     int rslices = (int)std::min<int64_t>(
-        std::max<int64_t>(2048 / cblocks, 1), std::max<int64_t>(iters / 8, 1));
+        std::max<int64_t>(2048 / cblocks, 1), std::max<int64_t>(iters / 32, 1));
     dim3 grid(cblocks, rslices);
     if (dt == DT::F32)
       hipLaunchKernelGGL(k_ln_bwd_col<float>, grid, dim3(256), 0, s,
