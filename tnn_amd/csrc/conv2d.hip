@@ -755,20 +755,6 @@ __global__ void k_conv_wgrad_vec(const T* __restrict__ X,
 }
 
 // ---------------------------------------------------------------------------
-template <typename T>
-__global__ void k_transpose_w(const T* __restrict__ W, T* __restrict__ WT,
-                              int KHW, int Cin, int Cout) {
-  // [khw, ci, co] -> [khw, co, ci]
-  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  int64_t n = (int64_t)KHW * Cin * Cout;
-  if (i >= n) return;
-  int co = i % Cout;
-  int ci = (i / Cout) % Cin;
-  int khw = i / ((int64_t)Cin * Cout);
-  WT[((int64_t)khw * Cout + co) * Cin + ci] = W[i];
-}
-
-// ---------------------------------------------------------------------------
 // batched LDS-tiled 2D transpose: out[b][c][r] = in[b][r][c]. 32x32 tiles,
 // coalesced on both sides (the naive elementwise form writes strided and
 // measured 2.6 TB/s; this hits ~5).
@@ -826,19 +812,6 @@ __global__ void k_transpose_w_dgrad_vec(const T* __restrict__ W,
   int64_t khw = i / ((int64_t)Cin * cov);
   ((P*)WT2D)[((int64_t)ci * KHW + khw) * cov + c] =
       ((const P*)W)[((int64_t)khw * Cin + ci) * cov + c];
-}
-
-// ---------------------------------------------------------------------------
-// [KH,KW,Ci,Co] -> [Co, KH*KW*Ci] (k-contiguous rows for the fwd B glds)
-template <typename T>
-__global__ void k_transpose_w_fwd(const T* __restrict__ W, T* __restrict__ WT2,
-                                  int KHW, int Cin, int Cout) {
-  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  int64_t n = (int64_t)KHW * Cin * Cout;
-  if (i >= n) return;
-  int co = i % Cout;
-  int64_t k = i / Cout;  // khw*Cin + ci
-  WT2[(int64_t)co * (KHW * Cin) + k] = W[i];
 }
 
 // ---------------------------------------------------------------------------
