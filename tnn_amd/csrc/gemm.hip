@@ -380,6 +380,10 @@ __global__ void k_gemm_tn_vec(const T* __restrict__ A, const T* __restrict__ B,
     // bf16: (8-row group x 4 consecutive m) per thread + in-thread
     // transpose -> ds_write_b64 (see k_conv_wgrad_vec)
     constexpr bool TR = sizeof(T) == 2;
+    // lane mapping: consecutive lanes share the m-quad and span the row
+    // groups: the 16 lanes' 16B global loads form one contiguous 256B run
+    // (the conflict-free same-row/mm-strided write mapping was tried and
+    // lost more to broken global coalescing than its bank relief won)
     const int a_rr = TR ? (threadIdx.x & (BM / V - 1)) * V : 0;
     const int a_mm0 = TR ? (threadIdx.x / (BM / V)) * 4 : 0;
     const T* asrc[RA];
