@@ -76,7 +76,7 @@ def bench_bn(dtype=torch.bfloat16):
         g = torch.ones(C, device=DEV)
         b = torch.zeros(C, device=DEV)
         nbytes = x.numel() * x.element_size()
-        secs = timeit(lambda: ext.bn_fwd_train(x, g, b, None, None, 0.1, 1e-5, True, 0.0, 0))
+        secs = timeit(lambda: ext.bn_fwd_train(x, g, b, None, None, 0.1, 1e-5, True, 0.0, 0, None))
         report("bn_fwd_train", f"rows={rows} C={C} {dtype}", secs,
                bytes_=3 * nbytes)
 

@@ -200,7 +200,7 @@ def test_bn_fwd_train(dtype, C):
     x = torch.randn(8, 6, 6, C, dtype=dtype, device=DEV) * 2 + 0.5
     gamma = torch.rand(C, device=DEV) + 0.5
     beta = torch.randn(C, device=DEV)
-    y, mean, invstd = ext.bn_fwd_train(x, gamma, beta, None, None, 0.1, 1e-5, False, 0.0, 0)
+    y, mean, invstd = ext.bn_fwd_train(x, gamma, beta, None, None, 0.1, 1e-5, False, 0.0, 0, None)
     xf = x.float().reshape(-1, C)
     rmean = xf.mean(0)
     rvar = xf.var(0, unbiased=False)
@@ -215,7 +215,7 @@ def test_bn_relu_and_infer(dtype):
     x = torch.randn(4, 5, 5, C, dtype=dtype, device=DEV)
     gamma = torch.ones(C, device=DEV)
     beta = torch.zeros(C, device=DEV)
-    y, _, _ = ext.bn_fwd_train(x, gamma, beta, None, None, 0.1, 1e-5, True, 0.0, 0)
+    y, _, _ = ext.bn_fwd_train(x, gamma, beta, None, None, 0.1, 1e-5, True, 0.0, 0, None)
     assert (y.float() >= 0).all()
     rm = torch.randn(C, device=DEV) * 0.1
     rv = torch.rand(C, device=DEV) + 0.5
@@ -241,7 +241,7 @@ def test_bn_bwd_matches_autograd(dtype):
     xdt = x.to(dtype)
     y_k, mean_k, invstd_k = ext.bn_fwd_train(xdt, gamma.detach(), beta.detach(),
                                              None, None, 0.1, 1e-5, False, 0.0,
-                                             0)
+                                             0, None)
     dx, dgamma, dbeta = ext.bn_bwd(xdt, dy.to(dtype), gamma.detach(), mean_k,
                                    invstd_k, None, 1.0)
     tol = TOL[dtype] * 4
@@ -603,3 +603,56 @@ def test_bn_relu_dropout_fused(dtype):
     assert relerr(xr.grad, xa.grad) < tol
     assert relerr(gr.grad, ga.grad) < tol
     assert relerr(br.grad, ba.grad) < tol
+
+
+@pytest.mark.parametrize("cfg", [(256, 16, 16, 128, 128), (256, 8, 8, 256, 512)])
+def test_conv_fwd_fused_bn_stats(cfg):
+    # conv epilogue per-channel sum/sumsq vs reductions of the conv output
+    N, H, W, Ci, Co = cfg
+    torch.manual_seed(12)
+    x = torch.randn(N, H, W, Ci, dtype=torch.bfloat16, device=DEV)
+    w = torch.randn(3, 3, Ci, Co, dtype=torch.bfloat16, device=DEV) * 0.05
+    bias = torch.randn(Co, dtype=torch.bfloat16, device=DEV)
+    y, stats = ext.conv2d_fwd_stats(x, w, bias, 1, 1, 1, 1)
+    assert stats.numel() == 2 * Co, "DB path expected for this shape"
+    yf = y.float().reshape(-1, Co)
+    assert relerr(stats[0], yf.sum(0)) < 2e-2
+    assert relerr(stats[1], (yf * yf).sum(0)) < 2e-2
+
+
+def test_wrn_block_fused_peepholes_match_unfused():
+    # Sequential.forward (peepholes active: conv->BN stats fusion) vs
+    # calling each layer directly (no peepholes). Stats come from fp32
+    # atomics in a different order, so compare with bf16-level tolerance.
+    from tnn_amd.nn import LayerBuilder
+    torch.manual_seed(13)
+    m = (LayerBuilder((16, 16, 128))
+         .batchnorm(relu=True, name="bn1")
+         .conv2d(128, 3, 3, 1, 1, 1, 1, True, "conv1")
+         .batchnorm(relu=True, name="bn2")
+         .conv2d(128, 3, 3, 1, 1, 1, 1, True, "conv2")
+         .build("seq_net"))
+    from tnn_amd.nn.layer import cast_compute_dtype
+    cast_compute_dtype(m, torch.bfloat16)
+    m = m.to(DEV)
+    m.train()
+    x = torch.randn(32, 16, 16, 128, dtype=torch.bfloat16, device=DEV)
+
+    y_fused = m(x)
+    loss = y_fused.float().square().mean()
+    loss.backward()
+    g_fused = [p.grad.clone() for p in m.parameters() if p.grad is not None]
+    for p in m.parameters():
+        p.grad = None
+
+    h = x
+    for layer in m.layers:   # direct layer calls skip the peepholes
+        h = layer(h)
+    loss2 = h.float().square().mean()
+    loss2.backward()
+    g_plain = [p.grad.clone() for p in m.parameters() if p.grad is not None]
+
+    assert relerr(y_fused, h) < 2e-2
+    assert len(g_fused) == len(g_plain)
+    for a, b in zip(g_fused, g_plain):
+        assert relerr(a, b) < 5e-2

@@ -418,6 +418,17 @@ void bn_stats_launch(DT dt, const void* x, float* mean, float* invstd,
                      cols, eps);
 }
 
+// finalize from precomputed sums (conv-epilogue fused stats): skips the
+// partial-reduce pass entirely
+void bn_finalize_launch(const float* sum, const float* sumsq, float* mean,
+                        float* invstd, float* rmean, float* rvar,
+                        float momentum, int64_t rows, int cols, float eps,
+                        hipStream_t s) {
+  hipLaunchKernelGGL(k_bn_finalize, dim3((cols + 255) / 256), dim3(256), 0, s,
+                     mean, invstd, sum, sumsq, rmean, rvar, momentum, rows,
+                     cols, eps);
+}
+
 void bn_apply_launch(DT dt, const void* x, const float* mean,
                      const float* invstd, const float* gamma, const float* beta,
                      void* y, int64_t rows, int cols, bool relu,

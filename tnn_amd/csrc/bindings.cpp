@@ -235,8 +235,38 @@ at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& w,
     wt2p = wt2.data_ptr();
   }
   conv2d_fwd_launch(dt_of(x), x.data_ptr(), w.data_ptr(), wt2p, bp,
-                    y.data_ptr(), zero_page(x), cs, relu, cur_stream());
+                    y.data_ptr(), zero_page(x), nullptr, cs, relu,
+                    cur_stream());
   return y;
+}
+
+// conv fwd + fused per-channel sum/sumsq for a following train-mode BN
+// (linear act). Returns {y, stats[2, Cout]}; stats is EMPTY when the
+// double-buffered kernel (the only one carrying the stats epilogue) is not
+// eligible -- callers fall back to the separate BN stats pass.
+std::vector<at::Tensor> conv2d_fwd_stats(const at::Tensor& x,
+                                         const at::Tensor& w,
+                                         const c10::optional<at::Tensor>& bias,
+                                         int64_t sh, int64_t sw, int64_t ph,
+                                         int64_t pw) {
+  CHECK_IN(x);
+  CHECK_IN(w);
+  auto cs = conv_shape(x, w.size(2), w.size(3), w.size(0), w.size(1), sh, sw,
+                       ph, pw);
+  auto y = at::empty({cs.N, cs.OH, cs.OW, cs.Cout}, x.options());
+  const void* bp = bias.has_value() ? bias->data_ptr() : nullptr;
+  if (!conv2d_fwd_wants_db(dt_of(x), x.data_ptr(), cs))
+    return {y, at::empty({0}, x.options().dtype(at::kFloat))};
+  auto wt2 = at::empty({(int64_t)cs.Cout, (int64_t)cs.KH * cs.KW * cs.Cin},
+                       w.options());
+  transpose_w_fwd_launch(dt_of(x), w.data_ptr(), wt2.data_ptr(),
+                         cs.KH * cs.KW, cs.Cin, cs.Cout, cur_stream());
+  auto stats = at::zeros({2, (int64_t)cs.Cout},
+                         x.options().dtype(at::kFloat));
+  conv2d_fwd_launch(dt_of(x), x.data_ptr(), w.data_ptr(), wt2.data_ptr(), bp,
+                    y.data_ptr(), zero_page(x), stats.data_ptr<float>(), cs,
+                    /*relu=*/false, cur_stream());
+  return {y, stats};
 }
 
 at::Tensor conv2d_dgrad(const at::Tensor& dy, const at::Tensor& w, int64_t H,
@@ -310,7 +340,8 @@ std::vector<at::Tensor> bn_fwd_train(const at::Tensor& x,
                                      const c10::optional<at::Tensor>& rmean,
                                      const c10::optional<at::Tensor>& rvar,
                                      double momentum, double eps, bool relu,
-                                     double dropout_p, int64_t seed) {
+                                     double dropout_p, int64_t seed,
+                                     const c10::optional<at::Tensor>& precomp) {
   CHECK_IN(x);
   TORCH_CHECK(dropout_p == 0.0 || relu, "fused BN dropout requires relu");
   int C = x.size(-1);
@@ -320,9 +351,18 @@ std::vector<at::Tensor> bn_fwd_train(const at::Tensor& x,
   auto y = at::empty_like(x);
   float* rm = rmean.has_value() ? rmean->data_ptr<float>() : nullptr;
   float* rv = rvar.has_value() ? rvar->data_ptr<float>() : nullptr;
-  bn_stats_launch(dt_of(x), x.data_ptr(), mean.data_ptr<float>(),
-                  invstd.data_ptr<float>(), rm, rv, (float)momentum, rows, C,
-                  (float)eps, cur_stream());
+  if (precomp.has_value()) {
+    // sums already produced by the conv epilogue ({2, C}: sum, sumsq)
+    TORCH_CHECK(precomp->numel() == 2 * C, "bad precomputed stats");
+    const float* ps = precomp->data_ptr<float>();
+    bn_finalize_launch(ps, ps + C, mean.data_ptr<float>(),
+                       invstd.data_ptr<float>(), rm, rv, (float)momentum,
+                       rows, C, (float)eps, cur_stream());
+  } else {
+    bn_stats_launch(dt_of(x), x.data_ptr(), mean.data_ptr<float>(),
+                    invstd.data_ptr<float>(), rm, rv, (float)momentum, rows, C,
+                    (float)eps, cur_stream());
+  }
   if (dropout_p > 0.0)
     bn_apply_drop_launch(dt_of(x), x.data_ptr(), mean.data_ptr<float>(),
                          invstd.data_ptr<float>(), gamma.data_ptr<float>(),
@@ -634,6 +674,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_selftest", &tnn::mfma_selftest);
   m.def("mfma_selftest_f32", &tnn::mfma_selftest_f32);
   m.def("conv2d_fwd", &tnn::conv2d_fwd);
+  m.def("conv2d_fwd_stats", &tnn::conv2d_fwd_stats);
   m.def("conv2d_dgrad", &tnn::conv2d_dgrad);
   m.def("conv2d_wgrad", &tnn::conv2d_wgrad);
   m.def("bn_fwd_train", &tnn::bn_fwd_train);

@@ -52,14 +52,14 @@ DEV void kdecode(int kidx, const ConvShape& cs, int& kh, int& kw) {
 // next chunk's DMAs overlap the current chunk's MFMAs. Counted
 // s_waitcnt vmcnt + raw s_barrier per the CDNA4 glds idiom -- a
 // __syncthreads() here would drain the in-flight next-chunk DMAs.
-template <typename T, bool POW2>
+template <typename T, bool POW2, bool STATS = false>
 __launch_bounds__(THREADS)
 __global__ void k_conv_fwd_db(const T* __restrict__ X,
                               const T* __restrict__ WT2,
                               const float* __restrict__ bias_f32,
                               const T* __restrict__ bias_t, T* __restrict__ Y,
                               const T* __restrict__ zero16, ConvShape cs,
-                              int act_kind) {
+                              int act_kind, float* __restrict__ stats) {
   __shared__ alignas(16) T As[2][BM * BK];
   __shared__ alignas(16) T Bs[2][BN * BK];
 
@@ -122,6 +122,46 @@ __global__ void k_conv_fwd_db(const T* __restrict__ X,
       Y[(int64_t)row * cs.Cout + col] = VecIO<T>::from_f32(v);
     }
   });
+
+  if constexpr (STATS) {
+    // per-channel sum/sumsq of the tile straight from the accumulators
+    // (post-bias, linear act only): saves the BN stats pass's full read
+    // of y. C/D fragment col = lane&15; lanes {l, l+16, l+32, l+48} share
+    // a column -> two shfl_xor folds, then one atomic per (block, col).
+    float* ssum = stats;
+    float* ssumsq = stats + cs.Cout;
+    const int cr = (wc.lane >> 4) * 4;
+    const int cc = wc.lane & 15;
+#pragma unroll
+    for (int fn = 0; fn < FN; ++fn) {
+      int col = n0 + wc.wcol0 + fn * 16 + cc;
+      float bias = 0.0f;
+      if (col < cs.Cout) {
+        if (bias_f32) bias = bias_f32[col];
+        if (bias_t) bias = VecIO<T>::to_f32(bias_t[col]);
+      }
+      float s = 0.0f, ss = 0.0f;
+#pragma unroll
+      for (int fm = 0; fm < FM; ++fm)
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          int row = m0 + wc.wrow0 + fm * 16 + cr + j;
+          if (row < M && col < cs.Cout) {
+            float v = acc[fm][fn][j] + bias;
+            s += v;
+            ss += v * v;
+          }
+        }
+      s += __shfl_xor(s, 16, 64);
+      ss += __shfl_xor(ss, 16, 64);
+      s += __shfl_xor(s, 32, 64);
+      ss += __shfl_xor(ss, 32, 64);
+      if (wc.lane < 16 && col < cs.Cout) {
+        atomicAdd(&ssum[col], s);
+        atomicAdd(&ssumsq[col], ss);
+      }
+    }
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -857,17 +897,20 @@ void transpose_w_fwd_launch(DT dt, const void* w, void* w_t2, int KHW, int Cin,
 
 void conv2d_fwd_launch(DT dt, const void* x, const void* w, const void* w_t2,
                        const void* bias, void* y, const void* zero16,
-                       const ConvShape& cs, bool relu, hipStream_t s) {
+                       float* stats, const ConvShape& cs, bool relu,
+                       hipStream_t s) {
   int M = cs.N * cs.OH * cs.OW;
   dim3 grid(ceil_div(M, BM), ceil_div(cs.Cout, BN));
   int act = relu ? ACT_RELU : ACT_LINEAR;
   bool p2 = all_pow2(cs);
   if (dt == DT::F32) {
     if (w_t2) {
-      hipLaunchKernelGGL((k_conv_fwd_db<float, true>), grid, dim3(THREADS), 0,
-                         s, (const float*)x, (const float*)w_t2,
-                         (const float*)bias, (const float*)nullptr, (float*)y,
-                         (const float*)zero16, cs, act);
+      auto kern = stats ? k_conv_fwd_db<float, true, true>
+                        : k_conv_fwd_db<float, true>;
+      hipLaunchKernelGGL(kern, grid, dim3(THREADS), 0, s, (const float*)x,
+                         (const float*)w_t2, (const float*)bias,
+                         (const float*)nullptr, (float*)y,
+                         (const float*)zero16, cs, act, stats);
       return;
     }
     bool g = p2 && cs.Cin % 4 == 0 && (((uintptr_t)x & 15) == 0);
@@ -880,10 +923,12 @@ void conv2d_fwd_launch(DT dt, const void* x, const void* w, const void* w_t2,
                        cs, act);
   } else {
     if (w_t2) {
-      hipLaunchKernelGGL((k_conv_fwd_db<bf16, true>), grid, dim3(THREADS), 0,
-                         s, (const bf16*)x, (const bf16*)w_t2,
-                         (const float*)nullptr, (const bf16*)bias, (bf16*)y,
-                         (const bf16*)zero16, cs, act);
+      auto kern = stats ? k_conv_fwd_db<bf16, true, true>
+                        : k_conv_fwd_db<bf16, true>;
+      hipLaunchKernelGGL(kern, grid, dim3(THREADS), 0, s, (const bf16*)x,
+                         (const bf16*)w_t2, (const float*)nullptr,
+                         (const bf16*)bias, (bf16*)y, (const bf16*)zero16, cs,
+                         act, stats);
       return;
     }
     bool g = p2 && cs.Cin % 8 == 0 && (((uintptr_t)x & 15) == 0);

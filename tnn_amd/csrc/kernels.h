@@ -77,7 +77,8 @@ void transpose_w_fwd_launch(DT dt, const void* w, void* w_t2, int KHW, int Cin,
                             int Cout, hipStream_t s);
 void conv2d_fwd_launch(DT dt, const void* x, const void* w, const void* w_t2,
                        const void* bias, void* y, const void* zero16,
-                       const ConvShape& cs, bool relu, hipStream_t s);
+                       float* stats, const ConvShape& cs, bool relu,
+                       hipStream_t s);
 bool conv2d_dgrad_wants_db(DT dt, const void* dy, const ConvShape& cs);
 void transpose_w_dgrad_launch(DT dt, const void* w, void* w_t2d, int KHW,
                               int Cin, int Cout, hipStream_t s);
@@ -95,6 +96,10 @@ void transpose_w_launch(DT dt, const void* w, void* w_t, int KH, int KW,
 void bn_stats_launch(DT dt, const void* x, float* mean, float* invstd,
                      float* rmean, float* rvar, float momentum,
                      int64_t rows, int cols, float eps, hipStream_t s);
+void bn_finalize_launch(const float* sum, const float* sumsq, float* mean,
+                        float* invstd, float* rmean, float* rvar,
+                        float momentum, int64_t rows, int cols, float eps,
+                        hipStream_t s);
 void bn_apply_launch(DT dt, const void* x, const float* mean,
                      const float* invstd, const float* gamma, const float* beta,
                      void* y, int64_t rows, int cols, bool relu, hipStream_t s);

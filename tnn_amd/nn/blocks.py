@@ -25,15 +25,37 @@ class Sequential(Layer):
         self.layers = nn.ModuleList(layers or [])
 
     def forward(self, x):
-        from .layers import BatchNorm, Dropout
+        from .layers import BatchNorm, Conv2D, Dropout
         n = len(self.layers)
         i = 0
         while i < n:
             layer = self.layers[i]
+            fuse = self.training and x.is_cuda
+            # peephole: Conv2D -> BatchNorm(train) computes the BN batch
+            # statistics in the conv epilogue (saves the stats pass's full
+            # read of y); a following Dropout merges into the BN apply too
+            if (fuse and isinstance(layer, Conv2D) and i + 1 < n
+                    and isinstance(self.layers[i + 1], BatchNorm)):
+                bn = self.layers[i + 1]
+                drop = None
+                if (i + 2 < n and bn.relu
+                        and isinstance(self.layers[i + 2], Dropout)
+                        and self.layers[i + 2].p > 0):
+                    drop = self.layers[i + 2]
+                y, stats = ops.conv2d_nhwc(x.to(layer.io_dtype), layer.weight,
+                                           layer.bias, layer.stride,
+                                           layer.padding, want_stats=True)
+                x = ops.batch_norm_act(y, bn.gamma, bn.beta, bn.running_mean,
+                                       bn.running_var, True, bn.momentum,
+                                       bn.eps, bn.relu,
+                                       drop.p if drop is not None else 0.0,
+                                       precomputed=stats)
+                i += 3 if drop is not None else 2
+                continue
             # peephole: BN(relu) directly followed by Dropout runs as one
             # fused kernel pass on GPU during training (ops.batch_norm_act
             # dropout_p) -- saves the dropout read/write passes entirely
-            if (i + 1 < n and self.training and x.is_cuda
+            if (fuse and i + 1 < n
                     and isinstance(layer, BatchNorm) and layer.relu
                     and isinstance(self.layers[i + 1], Dropout)
                     and self.layers[i + 1].p > 0):

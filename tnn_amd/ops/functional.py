@@ -52,7 +52,7 @@ def _use_hip(*tensors: torch.Tensor) -> bool:
 
 class _Conv2dNHWC(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, w, bias, stride, padding, fuse_relu):
+    def forward(ctx, x, w, bias, stride, padding, fuse_relu, want_stats=False):
         ext = _C.ext()
         # narrow-channel inputs (the RGB stem) fall off every vector/glds
         # path (measured 192us vs ~35 for the padded form): zero-pad Cin to
@@ -63,14 +63,25 @@ class _Conv2dNHWC(torch.autograd.Function):
         if Ci % V != 0 and Ci < V:
             x = F.pad(x, (0, V - Ci))
             w = F.pad(w, (0, 0, 0, V - Ci))
-        y = ext.conv2d_fwd(x, w, bias, stride[0], stride[1], padding[0], padding[1], fuse_relu)
+        ctx.want_stats = want_stats
+        if want_stats:
+            # fused per-channel sum/sumsq for a following train-mode BN;
+            # stats is empty when the fused kernel is not eligible
+            y, stats = ext.conv2d_fwd_stats(x, w, bias, stride[0], stride[1],
+                                            padding[0], padding[1])
+        else:
+            y = ext.conv2d_fwd(x, w, bias, stride[0], stride[1], padding[0],
+                               padding[1], fuse_relu)
         ctx.save_for_backward(x, w, y if fuse_relu else None)
         ctx.stride, ctx.padding, ctx.fuse_relu = stride, padding, fuse_relu
         ctx.has_bias = bias is not None
+        if want_stats:
+            ctx.mark_non_differentiable(stats)
+            return y, stats
         return y
 
     @staticmethod
-    def backward(ctx, dy):
+    def backward(ctx, dy, *unused_stats_grad):
         x, w, y = ctx.saved_tensors
         ext = _C.ext()
         dy = dy.contiguous()
@@ -91,7 +102,7 @@ class _Conv2dNHWC(torch.autograd.Function):
             dw = dw.to(w.dtype)  # no-op when the reduce already cast
         if ctx.has_bias and ctx.needs_input_grad[2]:
             db = ext.colsum(dy.reshape(-1, dy.shape[-1]))
-        return dx, dw, db, None, None, None
+        return dx, dw, db, None, None, None, None
 
 
 def conv2d_nhwc(
@@ -101,14 +112,27 @@ def conv2d_nhwc(
     stride: Tuple[int, int] = (1, 1),
     padding: Tuple[int, int] = (0, 0),
     fuse_relu: bool = False,
-) -> torch.Tensor:
-    """2-D convolution, NHWC activations, ``[KH,KW,Cin,Cout]`` weights."""
+    want_stats: bool = False,
+):
+    """2-D convolution, NHWC activations, ``[KH,KW,Cin,Cout]`` weights.
+
+    ``want_stats=True`` (GPU, linear act) additionally returns the fused
+    per-channel (sum, sumsq) for a following train-mode BatchNorm, or None
+    when the fused kernel is not eligible."""
     if _use_hip(x):
+        if want_stats:
+            y, stats = _Conv2dNHWC.apply(
+                x.contiguous(), w.contiguous(),
+                None if bias is None else bias.contiguous(), stride, padding,
+                False, True)
+            return y, (stats if stats.numel() else None)
         return _Conv2dNHWC.apply(x.contiguous(), w.contiguous(),
                                  None if bias is None else bias.contiguous(),
                                  stride, padding, fuse_relu)
     wn = w.permute(3, 2, 0, 1)  # -> [Cout, Cin, KH, KW]
     y = F.conv2d(_nhwc_to_nchw(x), wn, bias, stride=stride, padding=padding)
+    if want_stats:
+        return _nchw_to_nhwc(y), None
     y = _nchw_to_nhwc(y)
     if fuse_relu:
         y = F.relu(y)
@@ -124,7 +148,7 @@ def conv2d_nhwc(
 class _BatchNormAct(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, gamma, beta, running_mean, running_var, momentum,
-                eps, relu, dropout_p=0.0, seed=0):
+                eps, relu, dropout_p=0.0, seed=0, precomp=None):
         ext = _C.ext()
         # running stats update fused into the finalize kernel; dropout (if
         # any) fused into the apply kernel -- dropped positions write 0, so
@@ -132,7 +156,7 @@ class _BatchNormAct(torch.autograd.Function):
         # backward is a constant 1/(1-p) scale on the kept positions
         y, mean, invstd = ext.bn_fwd_train(x, gamma, beta, running_mean,
                                            running_var, momentum, eps, relu,
-                                           dropout_p, seed)
+                                           dropout_p, seed, precomp)
         save_y = y if (relu or dropout_p > 0.0) else None
         ctx.save_for_backward(x, gamma, mean, invstd, save_y)
         ctx.relu = relu or dropout_p > 0.0
@@ -145,7 +169,8 @@ class _BatchNormAct(torch.autograd.Function):
         ext = _C.ext()
         dx, dgamma, dbeta = ext.bn_bwd(x, dy.contiguous(), gamma, mean, invstd,
                                        y if ctx.relu else None, ctx.dy_scale)
-        return (dx, dgamma, dbeta, None, None, None, None, None, None, None)
+        return (dx, dgamma, dbeta, None, None, None, None, None, None, None,
+                None)
 
 
 def batch_norm_act(
@@ -159,6 +184,7 @@ def batch_norm_act(
     eps: float = 1e-5,
     relu: bool = False,
     dropout_p: float = 0.0,
+    precomputed: Optional[torch.Tensor] = None,
 ):
     """BatchNorm over NHWC channels-last with optional fused ReLU and
     (train-time) fused dropout.
@@ -178,7 +204,8 @@ def batch_norm_act(
                 seed = int(torch.randint(0, 2 ** 62, (1,)).item())
             y, _, _ = _BatchNormAct.apply(x.contiguous(), gamma, beta,
                                           running_mean, running_var, momentum,
-                                          eps, relu, dropout_p, seed)
+                                          eps, relu, dropout_p, seed,
+                                          precomputed)
             return y
         ext = _C.ext()
         return ext.bn_fwd_infer(x.contiguous(), gamma, beta, running_mean, running_var,
