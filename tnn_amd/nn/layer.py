@@ -28,6 +28,8 @@ _DTYPES = {
     "float64": torch.float64,
 }
 _DTYPE_NAMES = {v: k for k, v in _DTYPES.items()}
+_DTYPES.update({"fp32": torch.float32, "bf16": torch.bfloat16,
+                "fp16": torch.float16})  # CLI-style aliases
 
 
 def dtype_from_name(name: str) -> torch.dtype:

@@ -135,12 +135,13 @@ def train_model(model, train_loader, val_loader=None,
         if val_loader is not None:
             vstats = validate_model(model, val_loader, criterion, cfg)
             entry.update({f"val_{k}": v for k, v in vstats.items()})
-            if cfg.save_best and vstats["accuracy"] > best_val:
+            if vstats["accuracy"] > best_val:
                 best_val = vstats["accuracy"]
-                from ..utils.checkpoint import save_model
-                os.makedirs(cfg.snapshot_dir, exist_ok=True)
-                save_model(model, os.path.join(
-                    cfg.snapshot_dir, getattr(model, "name", "model")))
+                if cfg.save_best:
+                    from ..utils.checkpoint import save_model
+                    os.makedirs(cfg.snapshot_dir, exist_ok=True)
+                    save_model(model, os.path.join(
+                        cfg.snapshot_dir, getattr(model, "name", "model")))
         log.info("epoch %d done: %s", epoch,
                  {k: round(v, 4) if isinstance(v, float) else v
                   for k, v in entry.items()})
