@@ -9,12 +9,14 @@
 // GEMM views (row-major):
 //   fwd  : y[M=N*OH*OW, Cout] = A[M, K=KH*KW*Cin] @ w[K, Cout]
 //   dgrad: dx[M=N*H*W, Cin]   = A'[M, K=KH*KW*Cout] @ w_t[K, Cin]
-//   wgrad: dw[K=KH*KW*Cin, Cout] += im2col(x)^T @ dy  (split-M atomics)
+//   wgrad: dw[K=KH*KW*Cin, Cout] = im2col(x)^T @ dy  (deterministic
+//          split-M fp32 slabs + casting reduce -- no atomics)
 //
 // A-tiles are gathered on the fly: a 16B k-chunk stays within one (kh,kw)
 // slice whenever Cin (fwd) / Cout (dgrad) is a multiple of the vector
-// width, which holds for every hot layer; otherwise a scalar gather path
-// handles stem convs (Cin=3).
+// width -- which holds for every hot layer, and the python op zero-pads
+// narrow-channel inputs (the RGB stem) up to the vector width; a scalar
+// gather path remains for the rest.
 
 #include "common.h"
 #include "kernels.h"

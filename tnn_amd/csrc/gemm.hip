@@ -1,5 +1,6 @@
 // MFMA GEMM kernels: NN (fused bias+activation epilogue), NT (dgrad),
-// TN (wgrad, split-M atomic fp32 accumulate).
+// TN (wgrad: deterministic split-M fp32 slabs + casting reduce), plus
+// M=1 GEMV forms for the decode path.
 // Replaces the reference's cublasGemmEx / cuDNN-frontend matmul call sites
 // (src/math/cuda/gemm.cu:64, src/math/cuda/cudnn_gemm.cu:285,
 // src/nn/layers_impl/cuda/dense_ops.cu:17-117).
