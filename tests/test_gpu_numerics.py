@@ -129,6 +129,8 @@ CONV_CASES = [
     (4, 32, 32, 128, 256, 1, 1, 2, 0),  # 1x1 shortcut
     (2, 8, 8, 512, 512, 3, 3, 1, 1),    # deep layer
     (2, 9, 9, 24, 40, 3, 3, 2, 1),      # odd sizes
+    (2, 56, 56, 64, 64, 3, 3, 1, 1),    # non-pow2 spatial (imagenet-style)
+    (2, 28, 28, 96, 192, 3, 3, 2, 1),   # non-pow2 channels + stride
 ]
 
 

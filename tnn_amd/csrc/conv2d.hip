@@ -31,12 +31,15 @@ using namespace tile;
 template <bool POW2>
 DEV int idiv(int x, const IDiv& f) {
   if constexpr (POW2) return x >> f.lg;
-  else return x / f.d;
+  else if (f.magic)  // mul-shift reciprocal (exact for x < 2^26, see IDiv)
+    return (int)(((unsigned long long)(unsigned)x * f.magic) >> f.sh);
+  else
+    return x / f.d;
 }
 template <bool POW2>
 DEV int imod(int x, const IDiv& f) {
   if constexpr (POW2) return x & (f.d - 1);
-  else return x % f.d;
+  else return x - idiv<POW2>(x, f) * f.d;
 }
 
 // kidx -> (kh, kw) via mul-shift reciprocal (exact while kidx * KW < 2^16;

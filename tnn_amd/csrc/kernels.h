@@ -51,10 +51,22 @@ void mfma_selftest_f32_launch(const float* a, const float* b, float* d,
 // back to hardware division)
 struct IDiv {
   int d = 1;
-  int lg = 0;  // log2(d) if power of two, else -1
+  int lg = 0;        // log2(d) if power of two, else -1
+  int sh = 0;        // magic shift (non-pow2), 0 = use plain division
+  unsigned long long magic = 0;  // ceil(2^sh / d) + exactness headroom
   void set(int dd) {
     d = dd;
     lg = (dd & (dd - 1)) == 0 ? __builtin_ctz(dd) : -1;
+    magic = 0;
+    sh = 0;
+    // mul-shift reciprocal, exact for 0 <= x < 2^26 and 1 < d < 2^16
+    // (error term x/2^sh < 1/d strictly; covers every zoo shape --
+    // e.g. 224x224 ImageNet at batch 1024). Larger divisors fall back
+    // to hardware division.
+    if (lg < 0 && dd > 1 && dd < 65536) {
+      sh = dd < 1024 ? 36 : 42;
+      magic = ((1ull << sh) + dd - 1) / dd;
+    }
   }
 };
 
