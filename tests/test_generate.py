@@ -76,3 +76,29 @@ def test_generate_graphed_static_path_matches_recompute():
     # state fully restored: training-mode forward still works
     y = m(torch.randint(0, 64, (2, 16)))
     assert y.shape == (2, 16, 64)
+
+
+def test_generate_graphed_eot_truncation_and_capacity():
+    from tnn_amd.models.generate import generate_graphed
+    torch.manual_seed(0)
+    m = _tiny_gpt(seq=16)
+    # capacity: prompt of 12 in a 16-slot cache -> at most 4 new tokens
+    out = generate_graphed(m, list(range(12)), max_new_tokens=10, seq_len=16,
+                           eot_token=None, use_graph=False)
+    assert len(out) == 16
+    # eot: output truncates right after the token (parity with generate)
+    ref = generate(m, [1, 2, 3], max_new_tokens=10, seq_len=16, eot_token=None)
+    eot = ref[5]   # force an "eot" we know will be produced
+    a = generate(m, [1, 2, 3], max_new_tokens=10, seq_len=16, eot_token=eot)
+    b = generate_graphed(m, [1, 2, 3], max_new_tokens=10, seq_len=16,
+                         eot_token=eot, use_graph=False)
+    assert b == a
+
+
+def test_generate_graphed_full_prompt_noop():
+    from tnn_amd.models.generate import generate_graphed
+    torch.manual_seed(0)
+    m = _tiny_gpt(seq=8)
+    out = generate_graphed(m, list(range(8)), max_new_tokens=4, seq_len=8,
+                           eot_token=None, use_graph=False)
+    assert out == list(range(8))
