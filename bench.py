@@ -58,6 +58,7 @@ def main():
 
     model = models.create_model(args.model) if rank == 0 else None
     shapes = {
+        "mnist_cnn": ((28, 28, 1), 10),
         "cifar100_wrn16_8": ((32, 32, 3), 100),
         "cifar10_resnet9": ((32, 32, 3), 10),
         "cifar10_vgg": ((32, 32, 3), 10),
