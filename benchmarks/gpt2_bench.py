@@ -36,7 +36,7 @@ def main():
     model.to(dev)
 
     # ---- training-step throughput (tokens/s) ----
-    if args.train:
+    if args.train and args.steps > 0:
         model.train()
         crit = CrossEntropyLoss()
         opt = AdamW(model.parameters(), lr=1e-4)
