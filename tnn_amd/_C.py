@@ -6,8 +6,9 @@ tree. On a GPU machine the HIP kernels are the ONLY compute path for the hot
 ops — if the extension is missing there we raise loudly instead of silently
 falling back to eager PyTorch (that would invalidate every benchmark).
 
-On CPU-only machines (CI) ops use the pure-PyTorch reference implementations
-in :mod:`tnn_amd.ops.reference`, so the extension is optional there.
+On CPU-only machines (CI) ops use the pure-PyTorch fp32 reference branches
+inside :mod:`tnn_amd.ops.functional` (the numerics oracle the GPU tests
+compare against), so the extension is optional there.
 """
 
 from __future__ import annotations

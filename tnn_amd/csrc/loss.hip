@@ -34,7 +34,11 @@ __global__ void k_ce_fwd(const T* __restrict__ logits,
   if (threadIdx.x == 0) {
     float l = mx + __logf(se);
     lse[row] = l;
-    loss[row] = l - VecIO<T>::to_f32(x[targets[row]]);
+    // out-of-range target (corrupt batch / foreign ignore-index convention):
+    // no OOB read; NaN loss fails loudly instead of reading garbage
+    int64_t t = targets[row];
+    loss[row] = (t >= 0 && t < cols) ? l - VecIO<T>::to_f32(x[t])
+                                     : __int_as_float(0x7fc00000);
   }
 }
 

@@ -40,9 +40,12 @@ class CrossEntropyLoss(Loss):
     def __call__(self, pred, target):
         if target.dtype in (torch.int64, torch.int32):
             return softmax_cross_entropy(pred, target.long())
-        # one-hot / soft targets
-        return softmax_cross_entropy(pred, target.reshape(-1, target.shape[-1])
-                                     .argmax(-1))
+        # float targets (one-hot, label-smoothed or genuinely soft): exact
+        # soft cross-entropy -(t * log_softmax(p)).sum(-1) — an argmax here
+        # would silently discard non-one-hot probability mass
+        p2 = pred.reshape(-1, pred.shape[-1]).float()
+        t2 = target.reshape(-1, target.shape[-1]).float()
+        return -(t2 * F.log_softmax(p2, dim=-1)).sum(-1).mean()
 
 
 class MSELoss(Loss):

@@ -60,6 +60,30 @@ class Optimizer:
                     out.append((f"{i}.{k}", v))
         return out
 
+    def materialize_state_slot(self, name: str):
+        """Create (and return) the state tensor named ``"{param_idx}.{key}"``
+        so a checkpoint can be restored into a freshly constructed optimizer
+        whose lazy state dict is still empty. Returns None for names this
+        optimizer cannot place (unknown key or out-of-range param index)."""
+        idx_s, _, key = name.partition(".")
+        try:
+            i = int(idx_s)
+        except ValueError:
+            return None
+        if not key or i < 0 or i >= len(self.params):
+            return None
+        p = self.params[i]
+        st = self.state.setdefault(i, {})
+        if key in st and torch.is_tensor(st[key]):
+            return st[key]
+        if key == "master":
+            st[key] = p.detach().float().clone()
+        elif key in ("m", "v", "buf"):
+            st[key] = torch.zeros(p.shape, dtype=torch.float32, device=p.device)
+        else:
+            return None
+        return st[key]
+
 
 class SGD(Optimizer):
     _type = "sgd"

@@ -1,12 +1,14 @@
 """Checkpoint archive format (reference include/nn/graph.hpp:119-183,
 include/tensor/tensor.hpp:585-606).
 
-Same structural format as the reference: a self-describing JSON header
+Same structural *idea* as the reference — a self-describing JSON header
 (the model's full config) followed by raw per-tensor records in
-registration order, each record = ``{u32 dtype, u32 ndims, u64 shape[],
-bytes}``. We additionally store tensor names in the header (the reference
-relies purely on registration order) and, unlike the reference, can also
-checkpoint optimizer state (the reference does not — SURVEY §5).
+registration order — but not byte-compatible: the reference streams JSON
+unprefixed and writes ndims as u64, ours length-prefixes the header and
+uses ``{u32 dtype, u32 ndims, u64 shape[], bytes}`` records with its own
+dtype codes. We additionally store tensor names in the header (the
+reference relies purely on registration order) and, unlike the reference,
+can also checkpoint optimizer state (the reference does not — SURVEY §5).
 
 Layout:
     [u64 header_len][header JSON utf-8][record 0][record 1]...
@@ -135,6 +137,13 @@ def save_checkpoint(model, optimizer, path: str,
 
 
 def load_checkpoint(path: str, model, optimizer=None) -> Dict[str, Any]:
+    """Restore model weights + optimizer state. A freshly constructed
+    optimizer (the normal resume flow) has empty lazy state; its slots are
+    materialized from the record names before copying, so Adam m/v and fp32
+    master weights survive resume. Records that cannot be placed raise —
+    silently dropping optimizer moments would give oversized momentum-free
+    post-resume steps."""
+    unplaced = []
     with open(path, "rb") as f:
         header = _read_header(f)
         named = dict(_model_tensors(model))
@@ -147,8 +156,19 @@ def load_checkpoint(path: str, model, optimizer=None) -> Dict[str, Any]:
                 dst = named.get(name)
                 if dst is None:
                     dst = opt_named.get(name)
+                if dst is None and name.startswith("opt/") and optimizer is not None \
+                        and hasattr(optimizer, "materialize_state_slot"):
+                    dst = optimizer.materialize_state_slot(name[len("opt/"):])
                 if dst is not None:
                     dst.copy_(t.to(dst.dtype))
+                elif not (name.startswith("opt/") and optimizer is None):
+                    # opt/ records are skipped intentionally when no
+                    # optimizer was passed (model-only load)
+                    unplaced.append(name)
+    if unplaced:
+        raise RuntimeError(
+            f"checkpoint {path!r}: {len(unplaced)} tensor record(s) could not "
+            f"be placed: {unplaced[:8]}{'...' if len(unplaced) > 8 else ''}")
     if optimizer is not None:
         optimizer.step_count = header.get("step_count", 0)
     return header
