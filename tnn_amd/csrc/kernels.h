@@ -180,6 +180,24 @@ void attn_bwd_launch(const void* q, const void* k, const void* v,
                      const void* zero16, int BH, int S, int D, bool causal,
                      float scale, hipStream_t s);
 
+// ---- bmm.hip ---------------------------------------------------------------
+void bmm_launch(DT dt, const void* a, const void* b, void* c, int batch, int M,
+                int N, int K, int lda, int ldb, int ldc, int64_t sa, int64_t sb,
+                int64_t sc, bool ta, bool tb, hipStream_t s);
+
+// ---- softmax.hip -----------------------------------------------------------
+void smax_fwd_launch(DT dt, const void* x, void* y, int64_t rows, int cols,
+                     int mrows, int qoff, float scale, bool causal,
+                     hipStream_t s);
+void smax_bwd_launch(DT dt, const void* p, const void* dy, void* dx,
+                     int64_t rows, int cols, float scale, hipStream_t s);
+
+// ---- decode.hip (bf16; D in {64,128}) --------------------------------------
+void attn_decode_launch(const void* q, const void* k, const void* v, float* po,
+                        float* ml, void* out, const int64_t* pos_ptr,
+                        int len_static, int BH, int cap, int D, int splits,
+                        float scale, hipStream_t s);
+
 // ---- optim.hip -------------------------------------------------------------
 void sgd_step_launch(DT dt_p, const void* grad, DT dt_g, void* param,
                      float* master, float* momentum_buf, int64_t n, float lr,

@@ -271,12 +271,31 @@ class _MHABase(Layer):
         (the KV-cache fast path the reference lacks —
         examples/gpt2_inference.cpp recomputes the full sequence).
 
-        Static mode (s_new == 1 with a pos_t cache): k/v span the full
-        capacity and positions beyond pos_t are masked from pos_t itself,
-        keeping every shape fixed for hipGraph capture."""
+        GPU single-token steps run the fused decode-attention kernel
+        (csrc/decode.hip — one split-KV pass + combine, replacing the
+        matmul → mask → softmax → matmul chain); multi-token prefill runs
+        batched MFMA GEMMs + the fused causal-softmax kernel. The static
+        (pos_t) cache reads its live length from the device position
+        tensor inside the kernel, so every shape stays fixed for hipGraph
+        capture."""
         c = self._kv_cache
         t, s_new = k.shape[-2], q.shape[-2]
         scale = self.head_dim ** -0.5
+        if (q.is_cuda and q.dtype == torch.bfloat16 and s_new == 1
+                and c is not None and self.head_dim in (64, 128)
+                and not q.requires_grad):
+            b, h, _, hd = q.shape
+            kc, vc = c["k"], c["v"]  # full [B,H,cap,D] buffers
+            cap = kc.shape[2]
+            o = ops.attention_decode(
+                q.reshape(b * h, hd), kc.view(b * h, cap, hd),
+                vc.view(b * h, cap, hd), c.get("pos_t"), t, scale)
+            return o.view(b, h, 1, hd)
+        if q.is_cuda and not (c is not None and "pos_t" in c and s_new == 1):
+            # prefill / multi-token suffix: causal within the suffix with
+            # offset t - s_new, full attention over the cached prefix
+            return ops.sdpa_materialized(q, k, v, causal=self.causal,
+                                         qoff=t - s_new)
         scores = torch.matmul(q.float(), k.float().transpose(-1, -2)) * scale
         if c is not None and "pos_t" in c and s_new == 1:
             if "arange" not in c:
@@ -307,7 +326,10 @@ class _MHABase(Layer):
 
 @register_layer("attention_block")
 class AttentionBlock(_MHABase):
-    """Materialized-scores attention (reference AttentionBlock)."""
+    """Materialized-scores attention (reference AttentionBlock): batched
+    MFMA GEMMs + one fused scale/causal-mask softmax pass — no rocBLAS,
+    no separate mask fill (reference attention_block.cpp:144-147,
+    permute_heads.cu, causal_mask.cu, softmax.cu)."""
 
     def forward(self, x):
         b, s, _ = x.shape
@@ -315,13 +337,7 @@ class AttentionBlock(_MHABase):
         if self._kv_cache is not None:
             o = self._cached_attention(q, k, v)
             return self._merge(o, b, s)
-        scale = self.head_dim ** -0.5
-        scores = torch.matmul(q.float(), k.float().transpose(-1, -2)) * scale
-        if self.causal:
-            mask = torch.ones(s, s, dtype=torch.bool, device=x.device).triu(1)
-            scores = scores.masked_fill(mask, float("-inf"))
-        p = torch.softmax(scores, dim=-1).to(v.dtype)
-        o = torch.matmul(p, v)
+        o = ops.sdpa_materialized(q, k, v, causal=self.causal)
         return self._merge(o, b, s)
 
 

@@ -24,6 +24,9 @@ from .functional import (
     layer_norm,
     embedding,
     attention,
+    sdpa_materialized,
+    attention_decode,
+    scaled_softmax,
 )
 
 __all__ = [
@@ -39,4 +42,7 @@ __all__ = [
     "layer_norm",
     "embedding",
     "attention",
+    "sdpa_materialized",
+    "attention_decode",
+    "scaled_softmax",
 ]

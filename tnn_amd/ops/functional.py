@@ -306,13 +306,38 @@ def linear(x: torch.Tensor, w: torch.Tensor, bias: Optional[torch.Tensor] = None
     return y.reshape(*lead, w.shape[1])
 
 
+class _Bmm(torch.autograd.Function):
+    """Strided-batched MFMA GEMM (reference gemm_strided_batched_ex,
+    src/math/cuda/gemm.cu:84-105). Transpose views pass through as
+    strides — no materialized transposes anywhere in fwd or bwd."""
+
+    @staticmethod
+    def forward(ctx, a, b):
+        ext = _C.ext()
+        ctx.save_for_backward(a, b)
+        return ext.bmm(a, b)
+
+    @staticmethod
+    def backward(ctx, dc):
+        a, b = ctx.saved_tensors
+        ext = _C.ext()
+        da = db = None
+        if ctx.needs_input_grad[0]:
+            da = ext.bmm(dc, b.transpose(-1, -2))
+        if ctx.needs_input_grad[1]:
+            db = ext.bmm(a.transpose(-1, -2), dc)
+        return da, db
+
+
 def matmul(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
-    """Plain 2-D/batched matmul through the HIP GEMM on GPU."""
+    """Plain 2-D/batched matmul through the HIP MFMA GEMMs on GPU."""
     if _use_hip(a, b):
         if a.dim() == 2 and b.dim() == 2:
             return _Linear.apply(a.contiguous(), b.contiguous(), None, "linear")
-        ext = _C.ext()
-        return ext.bmm(a.contiguous(), b.contiguous())
+        lead = a.shape[:-2]
+        a3 = a.reshape(-1, a.shape[-2], a.shape[-1])
+        b3 = b.reshape(-1, b.shape[-2], b.shape[-1])
+        return _Bmm.apply(a3, b3).reshape(*lead, a.shape[-2], b.shape[-1])
     return a @ b
 
 
@@ -425,10 +450,52 @@ class _Activation(torch.autograd.Function):
         return ext.act_bwd(dy.contiguous(), x, y, ACT_KINDS[ctx.kind]), None
 
 
+class _ScaledSoftmax(torch.autograd.Function):
+    """Row softmax with fused scale + causal mask (replaces the reference's
+    causal_mask.cu fill + cudnnSoftmaxForward/Backward pair and the
+    softmax activation kernels, softmax_kernels.cu:11-346)."""
+
+    @staticmethod
+    def forward(ctx, x, scale, causal, mrows, qoff):
+        ext = _C.ext()
+        y = ext.smax_fwd(x, mrows, qoff, scale, causal)
+        ctx.save_for_backward(y)
+        ctx.scale = scale
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (y,) = ctx.saved_tensors
+        ext = _C.ext()
+        dx = ext.smax_bwd(y, dy.contiguous(), ctx.scale)
+        return dx, None, None, None, None
+
+
+def scaled_softmax(x: torch.Tensor, scale: float = 1.0, causal: bool = False,
+                   qoff: int = 0) -> torch.Tensor:
+    """softmax(x * scale) over the last dim; with ``causal``, row r of each
+    trailing [M, C] matrix may attend to columns <= r + qoff (suffix-causal
+    decode prefill passes qoff = kv_len - M). Fully masked rows are 0."""
+    if _use_hip(x):
+        mrows = x.shape[-2] if x.dim() >= 2 else 1
+        return _ScaledSoftmax.apply(x.contiguous(), scale, causal, mrows, qoff)
+    xf = x.float() * scale
+    if causal:
+        M, C = x.shape[-2], x.shape[-1]
+        pos_q = torch.arange(M, device=x.device).unsqueeze(-1) + qoff
+        pos_k = torch.arange(C, device=x.device)
+        xf = xf.masked_fill(pos_k > pos_q, float("-inf"))
+    y = F.softmax(xf, dim=-1)
+    y = torch.nan_to_num(y, nan=0.0)  # fully-masked rows -> 0 (kernel parity)
+    return y.to(x.dtype)
+
+
 def activation(x: torch.Tensor, kind: str) -> torch.Tensor:
     if kind == "linear":
         return x
     if kind == "softmax":
+        if _use_hip(x):
+            return scaled_softmax(x, 1.0)
         return F.softmax(x, dim=-1)
     if _use_hip(x):
         return _Activation.apply(x.contiguous(), kind)
@@ -561,3 +628,31 @@ def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
         s = s.masked_fill(mask, float("-inf"))
     p = F.softmax(s, dim=-1)
     return (p @ v.float()).to(q.dtype)
+
+
+def sdpa_materialized(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                      causal: bool = True, qoff: int = 0) -> torch.Tensor:
+    """Reference-AttentionBlock math with the S x S scores materialized:
+    batched MFMA GEMMs + one fused scale/causal-mask softmax pass on GPU
+    (reference attention_block.cpp:144-147, cuda/softmax.cu:47-79,
+    causal_mask.cu:13). q [*, M, D], k/v [*, T, D]; with ``qoff`` the
+    suffix-causal offset (kv_len - M) for KV-cache prefill."""
+    scale = q.shape[-1] ** -0.5
+    if _use_hip(q):
+        scores = matmul(q, k.transpose(-1, -2))
+        p = scaled_softmax(scores, scale, causal=causal, qoff=qoff)
+        return matmul(p, v)
+    s = (q.float() @ k.float().transpose(-1, -2))
+    p = scaled_softmax(s, scale, causal=causal, qoff=qoff)
+    return (p @ v.float()).to(q.dtype)
+
+
+def attention_decode(q: torch.Tensor, k_cache: torch.Tensor,
+                     v_cache: torch.Tensor, pos: Optional[torch.Tensor],
+                     length: int, scale: float) -> torch.Tensor:
+    """Fused single-token decode attention (inference only): q [BH, D]
+    against the full KV cache buffers [BH, cap, D]; live length from the
+    device ``pos`` tensor (len = pos+1, hipGraph-capturable) or ``length``."""
+    ext = _C.ext()
+    return ext.attn_decode(q.contiguous(), k_cache, v_cache, pos,
+                           length, scale)
