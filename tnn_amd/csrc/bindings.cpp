@@ -525,6 +525,69 @@ at::Tensor ce_bwd(const at::Tensor& logits, const at::Tensor& targets,
   return dl;
 }
 
+at::Tensor ptloss_fwd(const at::Tensor& pred, const at::Tensor& tgt,
+                      int64_t kind, double delta) {
+  CHECK_IN(pred);
+  CHECK_IN(tgt);
+  TORCH_CHECK(pred.sizes() == tgt.sizes(), "pred/target shape mismatch");
+  auto out = at::zeros({}, pred.options().dtype(at::kFloat));
+  ptloss_fwd_launch(dt_of(pred), pred.data_ptr(), tgt.data_ptr(),
+                    out.data_ptr<float>(), pred.numel(), (int)kind,
+                    (float)delta, cur_stream());
+  return out;
+}
+
+at::Tensor ptloss_bwd(const at::Tensor& pred, const at::Tensor& tgt,
+                      const at::Tensor& dloss, int64_t kind, double delta) {
+  CHECK_IN(pred);
+  auto dpred = at::empty_like(pred);
+  ptloss_bwd_launch(dt_of(pred), pred.data_ptr(), tgt.data_ptr(),
+                    dloss.contiguous().data_ptr<float>(), dpred.data_ptr(),
+                    pred.numel(), (int)kind, (float)delta, cur_stream());
+  return dpred;
+}
+
+// ---- group norm ------------------------------------------------------------
+// x [N, ..., C] NHWC-contiguous; gamma/beta fp32 [C]
+std::vector<at::Tensor> gn_fwd(const at::Tensor& x, const at::Tensor& gamma,
+                               const at::Tensor& beta, int64_t groups,
+                               double eps) {
+  CHECK_IN(x);
+  const int C = x.size(-1);
+  const int64_t N = x.size(0);
+  const int64_t HW = x.numel() / (N * C);
+  TORCH_CHECK(C % groups == 0, "C % groups != 0");
+  auto y = at::empty_like(x);
+  auto mean = at::empty({N * groups}, x.options().dtype(at::kFloat));
+  auto invstd = at::empty_like(mean);
+  gn_fwd_launch(dt_of(x), x.data_ptr(), gamma.data_ptr<float>(),
+                beta.data_ptr<float>(), y.data_ptr(), mean.data_ptr<float>(),
+                invstd.data_ptr<float>(), N, HW, C, (int)groups, (float)eps,
+                cur_stream());
+  return {y, mean, invstd};
+}
+
+std::vector<at::Tensor> gn_bwd(const at::Tensor& x, const at::Tensor& dy,
+                               const at::Tensor& mean, const at::Tensor& invstd,
+                               const at::Tensor& gamma, int64_t groups) {
+  CHECK_IN(x);
+  CHECK_IN(dy);
+  const int C = x.size(-1);
+  const int64_t N = x.size(0);
+  const int64_t HW = x.numel() / (N * C);
+  auto dx = at::empty_like(x);
+  auto s1 = at::empty({N * groups}, x.options().dtype(at::kFloat));
+  auto s2 = at::empty_like(s1);
+  auto dgamma = at::zeros({C}, x.options().dtype(at::kFloat));
+  auto dbeta = at::zeros({C}, x.options().dtype(at::kFloat));
+  gn_bwd_launch(dt_of(x), x.data_ptr(), dy.data_ptr(), mean.data_ptr<float>(),
+                invstd.data_ptr<float>(), gamma.data_ptr<float>(),
+                s1.data_ptr<float>(), s2.data_ptr<float>(), dx.data_ptr(),
+                dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), N, HW, C,
+                (int)groups, cur_stream());
+  return {dx, dgamma, dbeta};
+}
+
 // ---- layer norm ------------------------------------------------------------
 std::vector<at::Tensor> ln_fwd(const at::Tensor& x, const at::Tensor& gamma,
                                const at::Tensor& beta, double eps) {
@@ -773,6 +836,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("avgpool_bwd", &tnn::avgpool_bwd);
   m.def("ce_fwd", &tnn::ce_fwd);
   m.def("ce_bwd", &tnn::ce_bwd);
+  m.def("ptloss_fwd", &tnn::ptloss_fwd);
+  m.def("ptloss_bwd", &tnn::ptloss_bwd);
+  m.def("gn_fwd", &tnn::gn_fwd);
+  m.def("gn_bwd", &tnn::gn_bwd);
   m.def("ln_fwd", &tnn::ln_fwd);
   m.def("ln_bwd", &tnn::ln_bwd);
   m.def("embedding_fwd", &tnn::embedding_fwd);

@@ -153,6 +153,21 @@ void ce_fwd_launch(DT dt, const void* logits, const int64_t* targets,
 void ce_bwd_launch(DT dt, const void* logits, const int64_t* targets,
                    const float* lse, const float* dloss, void* dlogits,
                    int64_t rows, int cols, hipStream_t s);
+// kind: 0 = MSE, 1 = MAE, 2 = Huber
+void ptloss_fwd_launch(DT dt, const void* pred, const void* tgt, float* out,
+                       int64_t n, int kind, float delta, hipStream_t s);
+void ptloss_bwd_launch(DT dt, const void* pred, const void* tgt,
+                       const float* dloss, void* dpred, int64_t n, int kind,
+                       float delta, hipStream_t s);
+
+// ---- groupnorm.hip ---------------------------------------------------------
+void gn_fwd_launch(DT dt, const void* x, const float* gamma, const float* beta,
+                   void* y, float* mean, float* invstd, int64_t N, int64_t HW,
+                   int C, int G, float eps, hipStream_t s);
+void gn_bwd_launch(DT dt, const void* x, const void* dy, const float* mean,
+                   const float* invstd, const float* gamma, float* s1,
+                   float* s2, void* dx, float* dgamma, float* dbeta, int64_t N,
+                   int64_t HW, int C, int G, hipStream_t s);
 
 // ---- layernorm.hip ---------------------------------------------------------
 void ln_fwd_launch(DT dt, const void* x, const float* gamma, const float* beta,

@@ -60,6 +60,87 @@ __global__ void k_ce_bwd(const T* __restrict__ logits,
   }
 }
 
+// ---------------------------------------------------------------------------
+// Pointwise regression losses: MSE / MAE / Huber, mean reduction
+// (reference src/nn/loss_impl/cuda/loss_ops.cu:308-390). Forward
+// accumulates the scaled partial sums into one fp32 scalar; backward is
+// elementwise d(mean loss)/d(pred) * upstream.
+// ---------------------------------------------------------------------------
+
+enum PtLoss : int { PT_MSE = 0, PT_MAE = 1, PT_HUBER = 2 };
+
+template <typename T>
+__launch_bounds__(256)
+__global__ void k_ptloss_fwd(const T* __restrict__ pred,
+                             const T* __restrict__ tgt,
+                             float* __restrict__ out, int64_t n, int kind,
+                             float delta, float inv_n) {
+  __shared__ float scratch[8];
+  int64_t i = (int64_t)blockIdx.x * 256 + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * 256;
+  float acc = 0.0f;
+  for (; i < n; i += stride) {
+    const float d = VecIO<T>::to_f32(pred[i]) - VecIO<T>::to_f32(tgt[i]);
+    if (kind == PT_MSE) acc += d * d;
+    else if (kind == PT_MAE) acc += fabsf(d);
+    else {
+      const float a = fabsf(d);
+      acc += a <= delta ? 0.5f * d * d : delta * (a - 0.5f * delta);
+    }
+  }
+  acc = block_reduce_sum(acc, scratch);
+  if (threadIdx.x == 0) atomicAdd(out, acc * inv_n);
+}
+
+template <typename T>
+__launch_bounds__(256)
+__global__ void k_ptloss_bwd(const T* __restrict__ pred,
+                             const T* __restrict__ tgt,
+                             const float* __restrict__ dloss,
+                             T* __restrict__ dpred, int64_t n, int kind,
+                             float delta, float inv_n) {
+  const float g = dloss[0] * inv_n;
+  int64_t i = (int64_t)blockIdx.x * 256 + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * 256;
+  for (; i < n; i += stride) {
+    const float d = VecIO<T>::to_f32(pred[i]) - VecIO<T>::to_f32(tgt[i]);
+    float gd;
+    if (kind == PT_MSE) gd = 2.0f * d;
+    else if (kind == PT_MAE) gd = d > 0.0f ? 1.0f : (d < 0.0f ? -1.0f : 0.0f);
+    else gd = fmaxf(-delta, fminf(d, delta));
+    dpred[i] = VecIO<T>::from_f32(gd * g);
+  }
+}
+
+void ptloss_fwd_launch(DT dt, const void* pred, const void* tgt, float* out,
+                       int64_t n, int kind, float delta, hipStream_t s) {
+  const int blocks = (int)std::min<int64_t>((n + 255) / 256, (int64_t)2048);
+  const float inv_n = 1.0f / (float)n;
+  if (dt == DT::F32)
+    hipLaunchKernelGGL(k_ptloss_fwd<float>, dim3(blocks), dim3(256), 0, s,
+                       (const float*)pred, (const float*)tgt, out, n, kind,
+                       delta, inv_n);
+  else
+    hipLaunchKernelGGL(k_ptloss_fwd<bf16>, dim3(blocks), dim3(256), 0, s,
+                       (const bf16*)pred, (const bf16*)tgt, out, n, kind,
+                       delta, inv_n);
+}
+
+void ptloss_bwd_launch(DT dt, const void* pred, const void* tgt,
+                       const float* dloss, void* dpred, int64_t n, int kind,
+                       float delta, hipStream_t s) {
+  const int blocks = (int)std::min<int64_t>((n + 255) / 256, (int64_t)2048);
+  const float inv_n = 1.0f / (float)n;
+  if (dt == DT::F32)
+    hipLaunchKernelGGL(k_ptloss_bwd<float>, dim3(blocks), dim3(256), 0, s,
+                       (const float*)pred, (const float*)tgt, dloss,
+                       (float*)dpred, n, kind, delta, inv_n);
+  else
+    hipLaunchKernelGGL(k_ptloss_bwd<bf16>, dim3(blocks), dim3(256), 0, s,
+                       (const bf16*)pred, (const bf16*)tgt, dloss,
+                       (bf16*)dpred, n, kind, delta, inv_n);
+}
+
 void ce_fwd_launch(DT dt, const void* logits, const int64_t* targets,
                    float* loss, float* lse, int64_t rows, int cols,
                    hipStream_t s) {

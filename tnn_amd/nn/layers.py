@@ -156,13 +156,8 @@ class GroupNorm(Layer):
         self.beta = nn.Parameter(torch.zeros(num_channels, dtype=torch.float32))
 
     def forward(self, x):
-        n = x.shape[0]
-        xf = x.float().reshape(n, -1, self.num_groups,
-                               self.num_channels // self.num_groups)
-        mean = xf.mean(dim=(1, 3), keepdim=True)
-        var = xf.var(dim=(1, 3), unbiased=False, keepdim=True)
-        xhat = ((xf - mean) / torch.sqrt(var + self.eps)).reshape(*x.shape)
-        return (xhat * self.gamma + self.beta).to(x.dtype)
+        return ops.group_norm(x, self.gamma, self.beta, self.num_groups,
+                              self.eps)
 
     def extra_config(self):
         return {"num_groups": self.num_groups, "num_channels": self.num_channels,

@@ -15,7 +15,7 @@ from typing import Any, Dict
 import torch
 import torch.nn.functional as F
 
-from ..ops import softmax_cross_entropy
+from ..ops import pointwise_loss, softmax_cross_entropy
 
 
 class Loss:
@@ -49,17 +49,19 @@ class CrossEntropyLoss(Loss):
 
 
 class MSELoss(Loss):
+    """Fused GPU reduce kernel (reference loss_ops.cu:308-330)."""
+
     _type = "mse"
 
     def __call__(self, pred, target):
-        return F.mse_loss(pred.float(), target.float())
+        return pointwise_loss(pred, target, "mse")
 
 
 class MAELoss(Loss):
     _type = "mae"
 
     def __call__(self, pred, target):
-        return F.l1_loss(pred.float(), target.float())
+        return pointwise_loss(pred, target, "mae")
 
 
 class HuberLoss(Loss):
@@ -69,7 +71,7 @@ class HuberLoss(Loss):
         self.delta = delta
 
     def __call__(self, pred, target):
-        return F.huber_loss(pred.float(), target.float(), delta=self.delta)
+        return pointwise_loss(pred, target, "huber", self.delta)
 
     def extra_config(self):
         return {"delta": self.delta}

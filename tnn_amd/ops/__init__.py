@@ -27,6 +27,8 @@ from .functional import (
     sdpa_materialized,
     attention_decode,
     scaled_softmax,
+    group_norm,
+    pointwise_loss,
 )
 
 __all__ = [
@@ -45,4 +47,6 @@ __all__ = [
     "sdpa_materialized",
     "attention_decode",
     "scaled_softmax",
+    "group_norm",
+    "pointwise_loss",
 ]

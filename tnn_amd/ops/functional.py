@@ -580,6 +580,80 @@ def layer_norm(x: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
 
 
 # ---------------------------------------------------------------------------
+# group norm NHWC (reference groupnorm_ops.cu:46-171)
+# ---------------------------------------------------------------------------
+
+
+class _GroupNorm(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, gamma, beta, groups, eps):
+        ext = _C.ext()
+        y, mean, invstd = ext.gn_fwd(x, gamma, beta, groups, eps)
+        ctx.save_for_backward(x, gamma, mean, invstd)
+        ctx.groups = groups
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, gamma, mean, invstd = ctx.saved_tensors
+        ext = _C.ext()
+        dx, dgamma, dbeta = ext.gn_bwd(x, dy.contiguous(), mean, invstd,
+                                       gamma, ctx.groups)
+        return dx, dgamma, dbeta, None, None
+
+
+def group_norm(x: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
+               groups: int, eps: float = 1e-5) -> torch.Tensor:
+    """GroupNorm over channels-last [N, ..., C]; fp32 stats, fp32 gamma/beta."""
+    if _use_hip(x):
+        return _GroupNorm.apply(x.contiguous(), gamma, beta, groups, eps)
+    n, c = x.shape[0], x.shape[-1]
+    xf = x.float().reshape(n, -1, groups, c // groups)
+    mean = xf.mean(dim=(1, 3), keepdim=True)
+    var = xf.var(dim=(1, 3), unbiased=False, keepdim=True)
+    xhat = ((xf - mean) / torch.sqrt(var + eps)).reshape(*x.shape)
+    return (xhat * gamma.float() + beta.float()).to(x.dtype)
+
+
+# ---------------------------------------------------------------------------
+# pointwise regression losses (reference loss_ops.cu:308-390)
+# ---------------------------------------------------------------------------
+
+PT_LOSS_KINDS = {"mse": 0, "mae": 1, "huber": 2}
+
+
+class _PtLoss(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, pred, target, kind, delta):
+        ext = _C.ext()
+        loss = ext.ptloss_fwd(pred, target, PT_LOSS_KINDS[kind], delta)
+        ctx.save_for_backward(pred, target)
+        ctx.kind, ctx.delta = kind, delta
+        return loss
+
+    @staticmethod
+    def backward(ctx, dloss):
+        pred, target = ctx.saved_tensors
+        ext = _C.ext()
+        dpred = ext.ptloss_bwd(pred, target, dloss.contiguous().float(),
+                               PT_LOSS_KINDS[ctx.kind], ctx.delta)
+        return dpred, None, None, None
+
+
+def pointwise_loss(pred: torch.Tensor, target: torch.Tensor, kind: str,
+                   delta: float = 1.0) -> torch.Tensor:
+    """Mean MSE / MAE / Huber loss with a fused reduce on GPU."""
+    if _use_hip(pred):
+        return _PtLoss.apply(pred.contiguous(),
+                             target.to(pred.dtype).contiguous(), kind, delta)
+    if kind == "mse":
+        return F.mse_loss(pred.float(), target.float())
+    if kind == "mae":
+        return F.l1_loss(pred.float(), target.float())
+    return F.huber_loss(pred.float(), target.float(), delta=delta)
+
+
+# ---------------------------------------------------------------------------
 # embedding (reference embedding_ops.cu:17,48)
 # ---------------------------------------------------------------------------
 
