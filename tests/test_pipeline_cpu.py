@@ -168,3 +168,145 @@ def test_pipeline_4stage_matches_single_process(tmp_path):
         assert p.exitcode == 0
     losses = torch.load(tmp_path / "losses.pt", weights_only=False)
     assert losses == pytest.approx(ref_losses, rel=1e-4), (losses, ref_losses)
+
+
+@pytest.mark.timeout(420)
+def test_pipeline_8stage_matches_single_process(tmp_path):
+    """8 ranks (the full single-node MI355X shape) x 8 micro-batches; every
+    middle rank depth runs a different warmup/steady split. Exact parity
+    with single-process grad accumulation."""
+    ref_losses, _ = _reference_losses(M=8)
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_pipeline_worker_n,
+                         args=(r, 8, str(tmp_path), 8)) for r in range(8)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=360)
+        assert p.exitcode == 0
+    losses = torch.load(tmp_path / "losses.pt", weights_only=False)
+    assert losses == pytest.approx(ref_losses, rel=1e-4), (losses, ref_losses)
+
+
+@pytest.mark.timeout(300)
+def test_pipeline_3stage_m8_uneven(tmp_path):
+    """3 ranks, M=8: M is not a multiple of num_stages and the weighted
+    partitioner yields uneven stages (9 layers over 3 ranks)."""
+    ref_losses, _ = _reference_losses(M=8)
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_pipeline_worker_n,
+                         args=(r, 3, str(tmp_path), 8)) for r in range(3)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=240)
+        assert p.exitcode == 0
+    losses = torch.load(tmp_path / "losses.pt", weights_only=False)
+    assert losses == pytest.approx(ref_losses, rel=1e-4), (losses, ref_losses)
+
+
+# ---------------------------------------------------------------------------
+# P2P transcript pairing: every recv posted by rank r from peer p must be
+# matched, in order, by a send from p to r of the same shape — and in the
+# steady phase bidirectional traffic with one peer must travel as ONE
+# batched group (the RCCL deadlock-avoidance invariant; on RCCL a
+# mismatched or split pair hangs, which a gloo run would hide).
+# ---------------------------------------------------------------------------
+
+
+class _RecordingComm:
+    def __init__(self, comm, log):
+        self._c = comm
+        self._log = log
+        self.rank, self.world_size = comm.rank, comm.world_size
+
+    def isend(self, t, dst):
+        self._log.append(("send", dst, tuple(t.shape)))
+        return self._c.isend(t, dst)
+
+    def send(self, t, dst):
+        self._log.append(("send", dst, tuple(t.shape)))
+        self._c.send(t, dst)
+
+    def irecv(self, t, src):
+        self._log.append(("recv", src, tuple(t.shape)))
+        return self._c.irecv(t, src)
+
+    def recv(self, t, src):
+        self._log.append(("recv", src, tuple(t.shape)))
+        return self._c.recv(t, src)
+
+    def batch_p2p(self, ops):
+        import torch.distributed as dist
+        for op in ops:
+            kind = "send" if op.op is dist.isend else "recv"
+            self._log.append((f"batch_{kind}", op.peer, tuple(op.tensor.shape)))
+        return self._c.batch_p2p(ops)
+
+    def __getattr__(self, name):
+        return getattr(self._c, name)
+
+
+def _transcript_worker(rank, world, tmpdir, M):
+    os.environ.update({
+        "RANK": str(rank), "WORLD_SIZE": str(world), "LOCAL_RANK": str(rank),
+    })
+    import torch.distributed as dist
+    dist.init_process_group(
+        "gloo", init_method=f"file://{tmpdir}/pg_init", rank=rank,
+        world_size=world)
+    from tnn_amd.parallel import Communicator, PipelineEngine
+    log = []
+    comm = _RecordingComm(Communicator(), log)
+    model = _build_model() if rank == 0 else None
+    engine = PipelineEngine(model, comm, input_shape=(8, 8, 3),
+                            num_microbatches=M,
+                            optimizer_config={"type": "adamw", "lr": 1e-3},
+                            device=torch.device("cpu"), sync_weights=True)
+    x, y = _data(1)[0]
+    engine.train_batch(x, y)
+    torch.save(log, os.path.join(tmpdir, f"log_{rank}.pt"))
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_p2p_pairing_transcript(tmp_path):
+    world, M = 4, 8
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_transcript_worker,
+                         args=(r, world, str(tmp_path), M)) for r in range(world)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=240)
+        assert p.exitcode == 0
+    logs = {r: torch.load(tmp_path / f"log_{r}.pt", weights_only=False)
+            for r in range(world)}
+
+    def sends_to(r, peer):
+        return [(k, s) for k, p, s in logs[r] if p == peer and "send" in k]
+
+    def recvs_from(r, peer):
+        return [(k, s) for k, p, s in logs[r] if p == peer and "recv" in k]
+
+    for r in range(world - 1):
+        down = sends_to(r, r + 1)       # acts downstream
+        up_in = recvs_from(r + 1, r)    # matching receives
+        assert len(down) == len(up_in) == M, (r, len(down), len(up_in))
+        assert [s for _, s in down] == [s for _, s in up_in]
+        back = sends_to(r + 1, r)       # grads upstream
+        back_in = recvs_from(r, r + 1)
+        assert len(back) == len(back_in) == M
+        assert [s for _, s in back] == [s for _, s in back_in]
+
+    # steady-phase bidirectional ops must be batched: any rank with both a
+    # send and a recv towards the SAME peer between two compute phases
+    # issues them as batch_* entries (count check: middle ranks batch
+    # steady-state traffic in both directions)
+    for r in range(1, world - 1):
+        batched = [e for e in logs[r] if e[0].startswith("batch_")]
+        steady = M - min(world - 1 - r, M)
+        # each steady iteration produces one fwd/bwd batched pair with r+1
+        # and (except the last) one with r-1
+        assert len(batched) >= 2 * steady, (r, len(batched), steady)
