@@ -22,7 +22,8 @@ CSRC = os.path.join(ROOT, "tnn_amd", "csrc")
 # exclude them or a rebuild would compile every kernel twice.
 sources = sorted(p for p in glob.glob(os.path.join(CSRC, "*.hip"))
                  if not p.endswith("_hip.hip")) + [
-    os.path.join(CSRC, "bindings.cpp")
+    os.path.join(CSRC, "bindings.cpp"),
+    os.path.join(CSRC, "imagecodec.cpp"),  # host-only (stb_image analog)
 ]
 
 cxx_flags = ["-O3", "-std=c++17"]

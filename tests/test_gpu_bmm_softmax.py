@@ -131,7 +131,7 @@ def test_softmax_bwd(dtype):
 def test_scaled_softmax_autograd():
     torch.manual_seed(4)
     x = torch.randn(6, 32, 32, dtype=torch.bfloat16, device=DEV,
-                    requires_grad=True) * 1
+                    requires_grad=True)
     y = ops.scaled_softmax(x, 0.2, causal=True)
     g = torch.randn_like(y)
     y.backward(g)

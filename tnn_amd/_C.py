@@ -40,3 +40,16 @@ def ext():
             f"--inplace` (original import error: {_load_error!r})"
         )
     return _ext
+
+
+def ext_any():
+    """The extension module for host-side features (the C++ image codec)
+    — legitimately callable on CPU-only machines, same loud failure when
+    the extension is unbuilt."""
+    if _ext is None:
+        raise RuntimeError(
+            "tnn_amd._hip extension is not built (needed for the in-tree "
+            "image codec). Build it: `python setup.py build_ext --inplace` "
+            f"(original import error: {_load_error!r})"
+        )
+    return _ext

@@ -809,6 +809,10 @@ at::Tensor attn_decode(const at::Tensor& q, const at::Tensor& k,
   return out;
 }
 
+namespace imgcodec {
+at::Tensor decode_image(const py::bytes& data);  // imagecodec.cpp (host-only)
+}
+
 }  // namespace tnn
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -853,4 +857,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_step", &tnn::sgd_step);
   m.def("adam_step", &tnn::adam_step);
   m.def("adam_step_mt", &tnn::adam_step_mt);
+  m.def("decode_image", &tnn::imgcodec::decode_image);
 }

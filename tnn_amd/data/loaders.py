@@ -158,8 +158,12 @@ class CIFAR100Loader(CIFAR10Loader):
 
 
 class TinyImageNetLoader(BaseDataLoader):
-    """Tiny-ImageNet from pre-decoded .npy (stb_image decoding in the
-    reference; here images must be pre-converted — no PIL/stb in image)."""
+    """Tiny-ImageNet, loaded either from the RAW dataset layout
+    (``<path>/<split>/<class>/**/*.JPEG``, decoded by the in-tree C++
+    codec — reference src/data_loading/stb_image_impl.cpp analog) or
+    from pre-decoded ``{split}_{x,y}.npy`` when present (fast path)."""
+
+    image_size = 64
 
     def __init__(self, path: str, train: bool = True, **kw):
         super().__init__(**kw)
@@ -167,16 +171,25 @@ class TinyImageNetLoader(BaseDataLoader):
 
     def load_data(self):
         split = "train" if self.train else "val"
-        self.x = torch.from_numpy(
-            np.load(os.path.join(self.path, f"{split}_x.npy")).astype(np.float32) / 255.0)
-        self.y = torch.from_numpy(
-            np.load(os.path.join(self.path, f"{split}_y.npy")).astype(np.int64))
+        npx = os.path.join(self.path, f"{split}_x.npy")
+        if os.path.exists(npx):
+            self.x = torch.from_numpy(
+                np.load(npx).astype(np.float32) / 255.0)
+            self.y = torch.from_numpy(
+                np.load(os.path.join(self.path, f"{split}_y.npy"))
+                .astype(np.int64))
+            return
+        from .imageio import load_image_dir
+        x, self.y = load_image_dir(os.path.join(self.path, split),
+                                   size=self.image_size)
+        self.x = x.float() / 255.0
 
 
 class ImageNet100Loader(TinyImageNetLoader):
-    """ImageNet-100 from pre-decoded .npy splits (reference decodes JPEG
-    via stb_image; this image has no decoder, so images are converted
-    offline to {train,val}_{x,y}.npy exactly like TinyImageNetLoader)."""
+    """ImageNet-100: same dual raw-directory / .npy loading as
+    Tiny-ImageNet, 224x224 center-fit (reference decodes via stb_image)."""
+
+    image_size = 224
 
 
 class OpenWebTextLoader(BaseDataLoader):
