@@ -204,3 +204,45 @@ def test_attn_decode_pos_tensor():
     o = ext.attn_decode(q, k, v, pos, 0, scale)
     ref_o = ext.attn_decode(q, k, v, None, 300, scale)
     assert torch.equal(o, ref_o)
+
+
+# ---------------------------------------------------------------------------
+# fused-residual linear + single-launch decode GEMV
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("M", [1, 2, 128])
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_linear_residual_fused(M, dtype):
+    torch.manual_seed(8)
+    K, N = 768, 768
+    x = torch.randn(M, K, dtype=dtype, device=DEV, requires_grad=True)
+    w = torch.randn(K, N, dtype=dtype, device=DEV, requires_grad=True) * 0.02
+    b = torch.randn(N, dtype=dtype, device=DEV, requires_grad=True)
+    r = torch.randn(M, N, dtype=dtype, device=DEV, requires_grad=True)
+    y = ops.linear(x, w, b, residual=r)
+    g = torch.randn_like(y)
+    y.backward(g)
+
+    x2 = x.detach().float().requires_grad_()
+    w2 = w.detach().float().requires_grad_()
+    b2 = b.detach().float().requires_grad_()
+    r2 = r.detach().float().requires_grad_()
+    (x2 @ w2 + b2 + r2).backward(g.float())
+    t = tol(dtype, K)
+    assert maxerr(y, (x.detach().float() @ w.detach().float()
+                      + b.detach().float() + r.detach().float())) < t
+    assert maxerr(r.grad, r2.grad) < 1e-5
+    assert maxerr(x.grad, x2.grad) < t
+    assert maxerr(w.grad, w2.grad) < t * 2
+
+
+@pytest.mark.parametrize("N,K", [(768, 768), (3072, 768), (768, 3072),
+                                 (50257, 768), (65, 100)])
+def test_gemv_single_launch_m1(N, K):
+    torch.manual_seed(9)
+    x = torch.randn(1, K, dtype=torch.bfloat16, device=DEV)
+    w = torch.randn(K, N, dtype=torch.bfloat16, device=DEV) * 0.05
+    b = torch.randn(N, dtype=torch.bfloat16, device=DEV)
+    y = ext.gemm(x, w, b, 0)
+    ref = x.float() @ w.float() + b.float()
+    assert maxerr(y, ref) < tol(torch.bfloat16, K)

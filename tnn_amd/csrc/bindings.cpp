@@ -97,7 +97,8 @@ at::Tensor colsum(const at::Tensor& x) {
 
 // ---- gemm ------------------------------------------------------------------
 at::Tensor gemm(const at::Tensor& a, const at::Tensor& b,
-                const c10::optional<at::Tensor>& bias, int64_t act_kind) {
+                const c10::optional<at::Tensor>& bias, int64_t act_kind,
+                const c10::optional<at::Tensor>& residual = c10::nullopt) {
   CHECK_IN(a);
   CHECK_IN(b);
   TORCH_CHECK(a.dim() == 2 && b.dim() == 2 && a.size(1) == b.size(0),
@@ -107,7 +108,23 @@ at::Tensor gemm(const at::Tensor& a, const at::Tensor& b,
   int M = a.size(0), K = a.size(1), N = b.size(1);
   auto c = at::empty({M, N}, a.options());
   const void* bp = bias.has_value() ? bias->data_ptr() : nullptr;
-  if (M == 1) {  // decode-path matvec: K-split partials + finalize
+  const void* rp = nullptr;
+  if (residual.has_value()) {
+    TORCH_CHECK(residual->is_contiguous() &&
+                residual->scalar_type() == a.scalar_type() &&
+                residual->numel() == (int64_t)M * N,
+                "gemm residual mismatch");
+    rp = residual->data_ptr();
+  }
+  if (M == 1) {
+    if (K <= GEMV1_MAX_K_DECL) {
+      // single-launch matvec, fused bias/act/residual epilogue
+      gemv_nn1_launch(dt_of(a), a.data_ptr(), b.data_ptr(), bp, rp,
+                      c.data_ptr(), N, K, (int)act_kind, cur_stream());
+      return c;
+    }
+    // huge-K fallback: K-split partials + finalize (no residual fusion)
+    TORCH_CHECK(rp == nullptr, "gemm residual unsupported for K > 8192 M=1");
     int ks = gemv_nn_ksplits(N, K);
     auto ws = at::empty({(int64_t)ks * N}, a.options().dtype(at::kFloat));
     gemv_nn_launch(dt_of(a), a.data_ptr(), b.data_ptr(), bp, c.data_ptr(),
@@ -124,12 +141,12 @@ at::Tensor gemm(const at::Tensor& a, const at::Tensor& b,
     transpose_w_launch(dt_of(b), b.data_ptr(), bt.data_ptr(), 1, 1, K, N,
                        cur_stream());
     gemm_launch(dt_of(a), a.data_ptr(), bt.data_ptr(), bp, c.data_ptr(),
-                zero_page(a), M, N, K, /*trans_b=*/true, (int)act_kind,
+                zero_page(a), rp, M, N, K, /*trans_b=*/true, (int)act_kind,
                 cur_stream());
     return c;
   }
   gemm_launch(dt_of(a), a.data_ptr(), b.data_ptr(), bp, c.data_ptr(),
-              zero_page(a), M, N, K, /*trans_b=*/false, (int)act_kind,
+              zero_page(a), rp, M, N, K, /*trans_b=*/false, (int)act_kind,
               cur_stream());
   return c;
 }
@@ -143,7 +160,8 @@ at::Tensor gemm_nt(const at::Tensor& a, const at::Tensor& b) {
   int M = a.size(0), K = a.size(1), N = b.size(0);
   auto c = at::empty({M, N}, a.options());
   gemm_launch(dt_of(a), a.data_ptr(), b.data_ptr(), nullptr, c.data_ptr(),
-              zero_page(a), M, N, K, /*trans_b=*/true, 0, cur_stream());
+              zero_page(a), /*resid=*/nullptr, M, N, K, /*trans_b=*/true, 0,
+              cur_stream());
   return c;
 }
 
@@ -791,7 +809,10 @@ at::Tensor attn_decode(const at::Tensor& q, const at::Tensor& k,
   const int BH = q.size(0), D = q.size(1), cap = k.size(1);
   TORCH_CHECK(D == 64 || D == 128, "attn_decode supports D in {64,128}");
   TORCH_CHECK(k.size(0) == BH && k.size(2) == D && v.sizes() == k.sizes());
-  const int splits = std::min<int>(8, std::max<int>(1, cap / 512));
+  // split KV so the grid approaches the CU count (decode is latency-
+  // bound; 12 heads x 1 split measured 22us/call): static from (BH, cap)
+  // for hipGraph capture
+  int splits = std::max<int>(1, std::min<int>(16, 256 / BH));
   auto po = at::empty({(int64_t)BH * splits, D},
                       q.options().dtype(at::kFloat));
   auto ml = at::empty({(int64_t)BH * splits, 2},
@@ -822,7 +843,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dropout_fwd", &tnn::dropout_fwd);
   m.def("dropout_bwd", &tnn::dropout_bwd);
   m.def("colsum", &tnn::colsum);
-  m.def("gemm", &tnn::gemm);
+  m.def("gemm", &tnn::gemm, py::arg("a"), py::arg("b"), py::arg("bias"),
+        py::arg("act_kind"), py::arg("residual") = c10::nullopt);
   m.def("gemm_nt", &tnn::gemm_nt);
   m.def("gemm_tn", &tnn::gemm_tn);
   m.def("mfma_selftest", &tnn::mfma_selftest);

@@ -24,7 +24,8 @@ __launch_bounds__(THREADS)
 __global__ void k_gemm(const T* __restrict__ A, const T* __restrict__ B,
                        const float* __restrict__ bias_f32,
                        const T* __restrict__ bias_t, T* __restrict__ C,
-                       const T* __restrict__ zero16, int M, int N, int K,
+                       const T* __restrict__ zero16,
+                       const T* __restrict__ resid, int M, int N, int K,
                        int act_kind) {
   constexpr int V = 16 / sizeof(T);  // elems per 16B vector
   __shared__ alignas(16) T As[BM * BK];
@@ -122,12 +123,13 @@ __global__ void k_gemm(const T* __restrict__ A, const T* __restrict__ B,
     __syncthreads();
   }
 
-  // ---- epilogue: bias + activation + cast store ----
+  // ---- epilogue: bias + activation + residual + cast store ----
   epilogue_visit(wc, acc, m0, n0, [&](int row, int col, float v) {
     if (row < M && col < N) {
       if (bias_f32) v += bias_f32[col];
       if (bias_t) v += VecIO<T>::to_f32(bias_t[col]);
       if (act_kind != ACT_LINEAR) v = act_apply(v, act_kind);
+      if (resid) v += VecIO<T>::to_f32(resid[(int64_t)row * N + col]);
       C[(int64_t)row * N + col] = VecIO<T>::from_f32(v);
     }
   });
@@ -226,7 +228,8 @@ __launch_bounds__(THREADS)
 __global__ void k_gemm_nt_db(const T* __restrict__ A, const T* __restrict__ B,
                              const float* __restrict__ bias_f32,
                              const T* __restrict__ bias_t, T* __restrict__ C,
-                             const T* __restrict__ zero16, int M, int N, int K,
+                             const T* __restrict__ zero16,
+                             const T* __restrict__ resid, int M, int N, int K,
                              int act_kind) {
   __shared__ alignas(16) T As[2][BM * BK];
   __shared__ alignas(16) T Bs[2][BN * BK];
@@ -270,6 +273,7 @@ __global__ void k_gemm_nt_db(const T* __restrict__ A, const T* __restrict__ B,
       if (bias_f32) v += bias_f32[col];
       if (bias_t) v += VecIO<T>::to_f32(bias_t[col]);
       if (act_kind != ACT_LINEAR) v = act_apply(v, act_kind);
+      if (resid) v += VecIO<T>::to_f32(resid[(int64_t)row * N + col]);
       C[(int64_t)row * N + col] = VecIO<T>::from_f32(v);
     }
   });
@@ -315,10 +319,66 @@ __global__ void k_gemv_nt(const T* __restrict__ x, const T* __restrict__ B,
   }
 }
 
-// M=1 NN matvec: out[N] = x[K] @ B[K,N] with B row-major over N (the
-// layout every linear layer stores). Split K over grid.y into fp32
-// partials, then a finalize pass adds bias/activation. Two tiny launches
-// beat one M=1 MFMA tile kernel ~6x on gpt2 decode shapes.
+// Single-launch M=1 NN matvec: out[N] = x[K] @ B[K,N], x cached in LDS,
+// bias/activation/residual fused into the epilogue. The two-kernel
+// K-split form below measured 23us per decode linear (its finalize is a
+// 48-deep serial slab reduce on 3 blocks); this is one latency-bound
+// pass (~2-3us). Blocks own 64 n-columns; the 4 thread-rows split K and
+// fold through LDS. K <= GEMV1_MAX_K (x LDS cache); larger K falls back.
+constexpr int GEMV1_MAX_K = 8192;
+
+template <typename T>
+__launch_bounds__(256)
+__global__ void k_gemv_nn1(const T* __restrict__ x, const T* __restrict__ B,
+                           const float* __restrict__ bias_f32,
+                           const T* __restrict__ bias_t,
+                           const T* __restrict__ res, T* __restrict__ y,
+                           int N, int K, int act_kind) {
+  __shared__ float xs[GEMV1_MAX_K];
+  __shared__ float red[4][64];
+  const int c = threadIdx.x & 63;
+  const int ks = threadIdx.x >> 6;
+  const int n = blockIdx.x * 64 + c;
+  for (int k = threadIdx.x; k < K; k += 256)
+    xs[k] = VecIO<T>::to_f32(x[k]);
+  __syncthreads();
+  float acc = 0.0f;
+  if (n < N) {
+    int k = ks;
+#pragma unroll 8
+    for (; k < K; k += 4)
+      acc += xs[k] * VecIO<T>::to_f32(B[(int64_t)k * N + n]);
+  }
+  red[ks][c] = acc;
+  __syncthreads();
+  if (ks == 0 && n < N) {
+    float a = red[0][c] + red[1][c] + red[2][c] + red[3][c];
+    if (bias_f32) a += bias_f32[n];
+    if (bias_t) a += VecIO<T>::to_f32(bias_t[n]);
+    if (act_kind != ACT_LINEAR) a = act_apply(a, act_kind);
+    if (res) a += VecIO<T>::to_f32(res[n]);
+    y[n] = VecIO<T>::from_f32(a);
+  }
+}
+
+void gemv_nn1_launch(DT dt, const void* x, const void* b, const void* bias,
+                     const void* res, void* y, int N, int K, int act_kind,
+                     hipStream_t s) {
+  dim3 g(ceil_div(N, 64));
+  if (dt == DT::F32)
+    hipLaunchKernelGGL(k_gemv_nn1<float>, g, dim3(256), 0, s,
+                       (const float*)x, (const float*)b, (const float*)bias,
+                       (const float*)nullptr, (const float*)res, (float*)y, N,
+                       K, act_kind);
+  else
+    hipLaunchKernelGGL(k_gemv_nn1<bf16>, g, dim3(256), 0, s, (const bf16*)x,
+                       (const bf16*)b, (const float*)nullptr,
+                       (const bf16*)bias, (const bf16*)res, (bf16*)y, N, K,
+                       act_kind);
+}
+
+// Legacy two-kernel K-split matvec (kept for K > GEMV1_MAX_K): fp32
+// partials then a finalize pass.
 template <typename T>
 __launch_bounds__(256)
 __global__ void k_gemv_nn_part(const T* __restrict__ x, const T* __restrict__ B,
@@ -529,8 +589,8 @@ void splitk_reduce_launch(const float* ws, void* out, DT out_dt, int z,
 // ---------------------------------------------------------------------------
 
 void gemm_launch(DT dt, const void* a, const void* b, const void* bias,
-                 void* c, const void* zero16, int M, int N, int K,
-                 bool trans_b, int act_kind, hipStream_t s) {
+                 void* c, const void* zero16, const void* resid, int M, int N,
+                 int K, bool trans_b, int act_kind, hipStream_t s) {
   if (M == 1 && trans_b) {  // decode-path matvec
     dim3 gg(ceil_div(N, 4));
     if (dt == DT::F32)
@@ -551,7 +611,8 @@ void gemm_launch(DT dt, const void* a, const void* b, const void* bias,
       hipLaunchKernelGGL(k_gemm_nt_db<float>, grid, blk, 0, s,
                          (const float*)a, (const float*)b, (const float*)bias,
                          (const float*)nullptr, (float*)c,
-                         (const float*)zero16, M, N, K, act_kind);
+                         (const float*)zero16, (const float*)resid, M, N, K,
+                         act_kind);
       return;
     }
     auto kern = trans_b ? (g ? k_gemm<float, true, true>
@@ -560,14 +621,15 @@ void gemm_launch(DT dt, const void* a, const void* b, const void* bias,
                              : k_gemm<float, false, false>);
     hipLaunchKernelGGL(kern, grid, blk, 0, s, (const float*)a, (const float*)b,
                        (const float*)bias, (const float*)nullptr, (float*)c,
-                       (const float*)zero16, M, N, K, act_kind);
+                       (const float*)zero16, (const float*)resid, M, N, K,
+                       act_kind);
   } else {
     bool g = K % 8 == 0 && (((uintptr_t)a & 15) == 0);
     if (trans_b && g && K <= 3072 && (((uintptr_t)b & 15) == 0)) {
       hipLaunchKernelGGL(k_gemm_nt_db<bf16>, grid, blk, 0, s, (const bf16*)a,
                          (const bf16*)b, (const float*)nullptr,
-                         (const bf16*)bias, (bf16*)c, (const bf16*)zero16, M,
-                         N, K, act_kind);
+                         (const bf16*)bias, (bf16*)c, (const bf16*)zero16,
+                         (const bf16*)resid, M, N, K, act_kind);
       return;
     }
     auto kern = trans_b ? (g ? k_gemm<bf16, true, true>
@@ -576,7 +638,8 @@ void gemm_launch(DT dt, const void* a, const void* b, const void* bias,
                              : k_gemm<bf16, false, false>);
     hipLaunchKernelGGL(kern, grid, blk, 0, s, (const bf16*)a, (const bf16*)b,
                        (const float*)nullptr, (const bf16*)bias, (bf16*)c,
-                       (const bf16*)zero16, M, N, K, act_kind);
+                       (const bf16*)zero16, (const bf16*)resid, M, N, K,
+                       act_kind);
   }
 }
 

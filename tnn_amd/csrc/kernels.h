@@ -28,9 +28,13 @@ void cast_f32_launch(DT dt_out, const float* x, void* y, int64_t n,
 
 // ---- gemm.hip --------------------------------------------------------------
 void gemm_launch(DT dt, const void* a, const void* b, const void* bias,
-                 void* c, const void* zero16, int M, int N, int K,
-                 bool trans_b, int act_kind, hipStream_t s);
+                 void* c, const void* zero16, const void* resid, int M, int N,
+                 int K, bool trans_b, int act_kind, hipStream_t s);
 int gemv_nn_ksplits(int N, int K);
+void gemv_nn1_launch(DT dt, const void* x, const void* b, const void* bias,
+                     const void* res, void* y, int N, int K, int act_kind,
+                     hipStream_t s);
+constexpr int GEMV1_MAX_K_DECL = 8192;
 void gemv_nn_launch(DT dt, const void* x, const void* b, const void* bias,
                     void* y, float* ws, int ks, int N, int K, int act_kind,
                     hipStream_t s);

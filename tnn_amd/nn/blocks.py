@@ -309,9 +309,11 @@ class _MHABase(Layer):
         p = torch.softmax(scores, dim=-1).to(v.dtype)
         return torch.matmul(p, v)
 
-    def _merge(self, o, b, s):
+    def _merge(self, o, b, s, residual=None):
         o = o.transpose(1, 2).reshape(b, s, self.dim)
-        return ops.linear(o, self.wo, self.bo)
+        # the transformer-block residual add rides the o-projection GEMM
+        # epilogue when supplied (one fused kernel, no at::native add pass)
+        return ops.linear(o, self.wo, self.bo, residual=residual)
 
     def output_shape(self, in_shape):
         return tuple(in_shape)
@@ -331,14 +333,14 @@ class AttentionBlock(_MHABase):
     no separate mask fill (reference attention_block.cpp:144-147,
     permute_heads.cu, causal_mask.cu, softmax.cu)."""
 
-    def forward(self, x):
+    def forward(self, x, residual=None):
         b, s, _ = x.shape
         q, k, v = self._project(x)
         if self._kv_cache is not None:
             o = self._cached_attention(q, k, v)
-            return self._merge(o, b, s)
+            return self._merge(o, b, s, residual)
         o = ops.sdpa_materialized(q, k, v, causal=self.causal)
-        return self._merge(o, b, s)
+        return self._merge(o, b, s, residual)
 
 
 @register_layer("flash_attention_block")
@@ -346,15 +348,15 @@ class FlashAttentionBlock(_MHABase):
     """Flash attention via the hand-written CDNA4 kernel on GPU
     (reference FlashAttentionBlock / cudnn SDPA graph)."""
 
-    def forward(self, x):
+    def forward(self, x, residual=None):
         b, s, _ = x.shape
         q, k, v = self._project(x)
         if self._kv_cache is not None:
             o = self._cached_attention(q, k, v)
-            return self._merge(o, b, s)
+            return self._merge(o, b, s, residual)
         o = ops.attention(q.contiguous(), k.contiguous(), v.contiguous(),
                           causal=self.causal)
-        return self._merge(o, b, s)
+        return self._merge(o, b, s, residual)
 
 
 @register_layer("gpt_block")
@@ -382,12 +384,12 @@ class GPTBlock(Layer):
         self.drop = Dropout(dropout, name=f"{name}_drop", dtype=dtype) if dropout > 0 else None
 
     def forward(self, x):
-        x = x + self.attn(self.ln1(x))
+        x = self.attn(self.ln1(x), residual=x)
         h = ops.linear(self.ln2(x), self.w1, self.b1, act="gelu")
-        h = ops.linear(h, self.w2, self.b2)
         if self.drop is not None:
-            h = self.drop(h)
-        return x + h
+            h = self.drop(ops.linear(h, self.w2, self.b2))
+            return x + h
+        return ops.linear(h, self.w2, self.b2, residual=x)
 
     def flops_per_item(self, in_shape):
         s, d = in_shape

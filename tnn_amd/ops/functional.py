@@ -236,13 +236,20 @@ def batch_norm_act(
 
 
 class _Linear(torch.autograd.Function):
+    """y = act(x @ w + b) [+ residual] — the residual add rides the GEMM
+    epilogue (one fused kernel instead of a separate at::native
+    elementwise pass; reference dense_ops.cu has no analog)."""
+
     @staticmethod
-    def forward(ctx, x, w, bias, act):
+    def forward(ctx, x, w, bias, act, residual=None):
         ext = _C.ext()
-        y = ext.gemm(x, w, bias, ACT_KINDS[act])
+        assert residual is None or act == "linear", \
+            "fused residual requires act='linear' (relu mask would include it)"
+        y = ext.gemm(x, w, bias, ACT_KINDS[act], residual)
         ctx.save_for_backward(x, w, y if act != "linear" else None)
         ctx.act = act
         ctx.has_bias = bias is not None
+        ctx.has_res = residual is not None
         return y
 
     @staticmethod
@@ -250,6 +257,7 @@ class _Linear(torch.autograd.Function):
         x, w, y = ctx.saved_tensors
         ext = _C.ext()
         dy = dy.contiguous()
+        dres = dy if ctx.has_res else None
         if ctx.act == "relu":
             dy = ext.relu_bwd_mask(dy, y)
         elif ctx.act != "linear":
@@ -271,28 +279,36 @@ class _Linear(torch.autograd.Function):
             if ctx.needs_input_grad[1]:
                 dw = ext.gemm_tn(x, dy_p, w.dtype == x.dtype)
                 dw = dw[:, :N].contiguous().to(w.dtype)
-            return dx, dw, db, None
+            return dx, dw, db, None, dres
         if ctx.needs_input_grad[0]:
             dx = ext.gemm_nt(dy, w)       # [M,N] @ [K,N]^T -> [M,K]
         if ctx.needs_input_grad[1]:
             dw = ext.gemm_tn(x, dy, w.dtype == x.dtype)
             dw = dw.to(w.dtype)  # no-op when the reduce already cast
-        return dx, dw, db, None
+        return dx, dw, db, None, dres
 
 
 def linear(x: torch.Tensor, w: torch.Tensor, bias: Optional[torch.Tensor] = None,
-           act: str = "linear") -> torch.Tensor:
-    """y = act(x @ w + bias); x ``[..., K]``, w ``[K, N]`` row-major."""
+           act: str = "linear",
+           residual: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """y = act(x @ w + bias) [+ residual]; x ``[..., K]``, w ``[K, N]``
+    row-major. ``residual`` (shape of y) fuses the add into the GEMM
+    epilogue on GPU (act must be 'linear')."""
     lead = x.shape[:-1]
     x2 = x.reshape(-1, x.shape[-1])
     if _use_hip(x):
         # relu fuses into the GEMM epilogue (mask-replay backward); other
         # activations need the pre-activation saved, so run them unfused
         fused = act if act in ("linear", "relu") else "linear"
+        r2 = (residual.reshape(-1, residual.shape[-1]).contiguous()
+              if residual is not None else None)
         y = _Linear.apply(x2.contiguous(), w.contiguous(),
-                          None if bias is None else bias.contiguous(), fused)
+                          None if bias is None else bias.contiguous(), fused,
+                          r2 if fused == act else None)
         if fused != act:
             y = _Activation.apply(y, act)
+            if residual is not None:
+                y = y + r2
     else:
         y = x2 @ w
         if bias is not None:
@@ -303,6 +319,8 @@ def linear(x: torch.Tensor, w: torch.Tensor, bias: Optional[torch.Tensor] = None
             y = F.gelu(y, approximate="tanh")
         elif act != "linear":
             y = activation(y, act)
+        if residual is not None:
+            y = y + residual.reshape(-1, residual.shape[-1])
     return y.reshape(*lead, w.shape[1])
 
 
