@@ -216,7 +216,7 @@ def test_linear_residual_fused(M, dtype):
     torch.manual_seed(8)
     K, N = 768, 768
     x = torch.randn(M, K, dtype=dtype, device=DEV, requires_grad=True)
-    w = torch.randn(K, N, dtype=dtype, device=DEV, requires_grad=True) * 0.02
+    w = (torch.randn(K, N, dtype=dtype, device=DEV) * 0.02).requires_grad_()
     b = torch.randn(N, dtype=dtype, device=DEV, requires_grad=True)
     r = torch.randn(M, N, dtype=dtype, device=DEV, requires_grad=True)
     y = ops.linear(x, w, b, residual=r)

@@ -98,7 +98,10 @@ at::Tensor colsum(const at::Tensor& x) {
 // ---- gemm ------------------------------------------------------------------
 at::Tensor gemm(const at::Tensor& a, const at::Tensor& b,
                 const c10::optional<at::Tensor>& bias, int64_t act_kind,
-                const c10::optional<at::Tensor>& residual = c10::nullopt) {
+                const c10::optional<at::Tensor>& residual = c10::nullopt,
+                const c10::optional<at::Tensor>& ln_gamma = c10::nullopt,
+                const c10::optional<at::Tensor>& ln_beta = c10::nullopt,
+                double ln_eps = 1e-5) {
   CHECK_IN(a);
   CHECK_IN(b);
   TORCH_CHECK(a.dim() == 2 && b.dim() == 2 && a.size(1) == b.size(0),
@@ -116,11 +119,30 @@ at::Tensor gemm(const at::Tensor& a, const at::Tensor& b,
                 "gemm residual mismatch");
     rp = residual->data_ptr();
   }
+  const float* lng = nullptr;
+  const float* lnb = nullptr;
+  if (ln_gamma.has_value()) {
+    TORCH_CHECK(M == 1 && K <= GEMV1_MAX_K_DECL && ln_beta.has_value() &&
+                ln_gamma->scalar_type() == at::kFloat &&
+                ln_beta->scalar_type() == at::kFloat,
+                "fused ln: decode GEMV only, fp32 gamma/beta");
+    lng = ln_gamma->data_ptr<float>();
+    lnb = ln_beta->data_ptr<float>();
+  }
   if (M == 1) {
     if (K <= GEMV1_MAX_K_DECL) {
-      // single-launch matvec, fused bias/act/residual epilogue
+      // LDS-cached matvec; K-split across blocks to fill the chip (fp32
+      // partials + a parallel finalize) when N alone is too narrow
+      int ns = gemv_nn1_nsplit(N, K);
+      at::Tensor part;
+      float* pp = nullptr;
+      if (ns > 1) {
+        part = at::empty({(int64_t)ns * N}, a.options().dtype(at::kFloat));
+        pp = part.data_ptr<float>();
+      }
       gemv_nn1_launch(dt_of(a), a.data_ptr(), b.data_ptr(), bp, rp,
-                      c.data_ptr(), N, K, (int)act_kind, cur_stream());
+                      c.data_ptr(), pp, ns, lng, lnb, (float)ln_eps, N, K,
+                      (int)act_kind, cur_stream());
       return c;
     }
     // huge-K fallback: K-split partials + finalize (no residual fusion)
@@ -844,7 +866,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dropout_bwd", &tnn::dropout_bwd);
   m.def("colsum", &tnn::colsum);
   m.def("gemm", &tnn::gemm, py::arg("a"), py::arg("b"), py::arg("bias"),
-        py::arg("act_kind"), py::arg("residual") = c10::nullopt);
+        py::arg("act_kind"), py::arg("residual") = c10::nullopt,
+        py::arg("ln_gamma") = c10::nullopt, py::arg("ln_beta") = c10::nullopt,
+        py::arg("ln_eps") = 1e-5);
   m.def("gemm_nt", &tnn::gemm_nt);
   m.def("gemm_tn", &tnn::gemm_tn);
   m.def("mfma_selftest", &tnn::mfma_selftest);

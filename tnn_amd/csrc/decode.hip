@@ -114,9 +114,11 @@ __global__ void k_attn_decode_fin(const float* __restrict__ po,
   const int d = threadIdx.x;
   if (d >= D) return;
   float m = -INFINITY;
+#pragma unroll 4
   for (int s = 0; s < splits; ++s)
     m = fmaxf(m, ml[((int64_t)bh * splits + s) * 2]);
   float l = 0.0f, o = 0.0f;
+#pragma unroll 4
   for (int s = 0; s < splits; ++s) {
     const float ms = ml[((int64_t)bh * splits + s) * 2];
     const float ls = ml[((int64_t)bh * splits + s) * 2 + 1];

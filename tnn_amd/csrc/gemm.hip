@@ -333,26 +333,77 @@ __global__ void k_gemv_nn1(const T* __restrict__ x, const T* __restrict__ B,
                            const float* __restrict__ bias_f32,
                            const T* __restrict__ bias_t,
                            const T* __restrict__ res, T* __restrict__ y,
+                           float* __restrict__ part,
+                           const float* __restrict__ ln_g,
+                           const float* __restrict__ ln_b, float ln_eps,
                            int N, int K, int act_kind) {
   __shared__ float xs[GEMV1_MAX_K];
   __shared__ float red[4][64];
   const int c = threadIdx.x & 63;
   const int ks = threadIdx.x >> 6;
   const int n = blockIdx.x * 64 + c;
-  for (int k = threadIdx.x; k < K; k += 256)
-    xs[k] = VecIO<T>::to_f32(x[k]);
-  __syncthreads();
+  // this block's K slice (grid.y = nsplit; 4-aligned so the thread-row
+  // interleave stays uniform)
+  const int nsplit = gridDim.y;
+  const int k_lo = (int)(((int64_t)K * blockIdx.y / nsplit) & ~3LL);
+  const int k_hi = blockIdx.y + 1 == nsplit
+                       ? K
+                       : (int)(((int64_t)K * (blockIdx.y + 1) / nsplit) & ~3LL);
+  if (ln_g) {
+    // fused LayerNorm on the x vector (decode: saves the separate ln
+    // kernel + its memory round trip; every split block recomputes the
+    // row stats — trivial next to the per-kernel floor)
+    for (int k = threadIdx.x; k < K; k += 256)
+      xs[k] = VecIO<T>::to_f32(x[k]);
+    __syncthreads();
+    float sm = 0.0f, sq = 0.0f;
+    for (int k = threadIdx.x; k < K; k += 256) {
+      const float v = xs[k];
+      sm += v;
+      sq += v * v;
+    }
+    sm = block_reduce_sum(sm, red[0]);
+    __shared__ float bmean;
+    if (threadIdx.x == 0) bmean = sm / (float)K;
+    __syncthreads();
+    sq = block_reduce_sum(sq, red[0]);
+    __shared__ float binv;
+    if (threadIdx.x == 0)
+      binv = rsqrtf(fmaxf(sq / (float)K - bmean * bmean, 0.0f) + ln_eps);
+    __syncthreads();
+    const float mu = bmean, is = binv;
+    for (int k = k_lo + (int)threadIdx.x; k < k_hi; k += 256)
+      xs[k] = (xs[k] - mu) * is * ln_g[k] + ln_b[k];
+    __syncthreads();
+  } else {
+    for (int k = k_lo + (int)threadIdx.x; k < k_hi; k += 256)
+      xs[k] = VecIO<T>::to_f32(x[k]);
+    __syncthreads();
+  }
   float acc = 0.0f;
   if (n < N) {
-    int k = ks;
-#pragma unroll 8
-    for (; k < K; k += 4)
+    int k = k_lo + ks;
+    // batched 8-deep loads, then the FMAs: keeps 8 loads in flight per
+    // thread instead of a serialized load-fma chain
+    for (; k + 32 <= k_hi; k += 32) {
+      float bv[8];
+#pragma unroll
+      for (int u = 0; u < 8; ++u)
+        bv[u] = VecIO<T>::to_f32(B[(int64_t)(k + 4 * u) * N + n]);
+#pragma unroll
+      for (int u = 0; u < 8; ++u) acc += xs[k + 4 * u] * bv[u];
+    }
+    for (; k < k_hi; k += 4)
       acc += xs[k] * VecIO<T>::to_f32(B[(int64_t)k * N + n]);
   }
   red[ks][c] = acc;
   __syncthreads();
   if (ks == 0 && n < N) {
     float a = red[0][c] + red[1][c] + red[2][c] + red[3][c];
+    if (part) {
+      part[(int64_t)blockIdx.y * N + n] = a;
+      return;
+    }
     if (bias_f32) a += bias_f32[n];
     if (bias_t) a += VecIO<T>::to_f32(bias_t[n]);
     if (act_kind != ACT_LINEAR) a = act_apply(a, act_kind);
@@ -361,20 +412,68 @@ __global__ void k_gemv_nn1(const T* __restrict__ x, const T* __restrict__ B,
   }
 }
 
+// parallel finalize over n: sum <= 16 fp32 partial rows + fused epilogue
+template <typename T>
+__launch_bounds__(256)
+__global__ void k_gemv_fin2(const float* __restrict__ part,
+                            const float* __restrict__ bias_f32,
+                            const T* __restrict__ bias_t,
+                            const T* __restrict__ res, T* __restrict__ y,
+                            int N, int nsplit, int act_kind) {
+  const int n = blockIdx.x * 256 + threadIdx.x;
+  if (n >= N) return;
+  float a0 = 0.0f, a1 = 0.0f, a2 = 0.0f, a3 = 0.0f;
+  int s = 0;
+  for (; s + 4 <= nsplit; s += 4) {
+    a0 += part[(int64_t)s * N + n];
+    a1 += part[(int64_t)(s + 1) * N + n];
+    a2 += part[(int64_t)(s + 2) * N + n];
+    a3 += part[(int64_t)(s + 3) * N + n];
+  }
+  for (; s < nsplit; ++s) a0 += part[(int64_t)s * N + n];
+  float a = (a0 + a1) + (a2 + a3);
+  if (bias_f32) a += bias_f32[n];
+  if (bias_t) a += VecIO<T>::to_f32(bias_t[n]);
+  if (act_kind != ACT_LINEAR) a = act_apply(a, act_kind);
+  if (res) a += VecIO<T>::to_f32(res[n]);
+  y[n] = VecIO<T>::from_f32(a);
+}
+
+int gemv_nn1_nsplit(int N, int K) {
+  // fill the chip: blocks = (N/64) * nsplit ~ 256; keep slices >= 64 rows
+  int ncb = ceil_div(N, 64);
+  int ns = std::max(1, std::min(16, 256 / ncb));
+  ns = std::min(ns, std::max(1, K / 64));
+  return ns;
+}
+
 void gemv_nn1_launch(DT dt, const void* x, const void* b, const void* bias,
-                     const void* res, void* y, int N, int K, int act_kind,
-                     hipStream_t s) {
-  dim3 g(ceil_div(N, 64));
-  if (dt == DT::F32)
+                     const void* res, void* y, float* part, int nsplit,
+                     const float* ln_g, const float* ln_b, float ln_eps, int N,
+                     int K, int act_kind, hipStream_t s) {
+  dim3 g(ceil_div(N, 64), nsplit);
+  dim3 fg(ceil_div(N, 256));
+  if (dt == DT::F32) {
     hipLaunchKernelGGL(k_gemv_nn1<float>, g, dim3(256), 0, s,
                        (const float*)x, (const float*)b, (const float*)bias,
-                       (const float*)nullptr, (const float*)res, (float*)y, N,
-                       K, act_kind);
-  else
+                       (const float*)nullptr, (const float*)res, (float*)y,
+                       nsplit > 1 ? part : nullptr, ln_g, ln_b, ln_eps, N, K,
+                       act_kind);
+    if (nsplit > 1)
+      hipLaunchKernelGGL(k_gemv_fin2<float>, fg, dim3(256), 0, s, part,
+                         (const float*)bias, (const float*)nullptr,
+                         (const float*)res, (float*)y, N, nsplit, act_kind);
+  } else {
     hipLaunchKernelGGL(k_gemv_nn1<bf16>, g, dim3(256), 0, s, (const bf16*)x,
                        (const bf16*)b, (const float*)nullptr,
-                       (const bf16*)bias, (const bf16*)res, (bf16*)y, N, K,
+                       (const bf16*)bias, (const bf16*)res, (bf16*)y,
+                       nsplit > 1 ? part : nullptr, ln_g, ln_b, ln_eps, N, K,
                        act_kind);
+    if (nsplit > 1)
+      hipLaunchKernelGGL(k_gemv_fin2<bf16>, fg, dim3(256), 0, s, part,
+                         (const float*)nullptr, (const bf16*)bias,
+                         (const bf16*)res, (bf16*)y, N, nsplit, act_kind);
+  }
 }
 
 // Legacy two-kernel K-split matvec (kept for K > GEMV1_MAX_K): fp32

@@ -31,9 +31,11 @@ void gemm_launch(DT dt, const void* a, const void* b, const void* bias,
                  void* c, const void* zero16, const void* resid, int M, int N,
                  int K, bool trans_b, int act_kind, hipStream_t s);
 int gemv_nn_ksplits(int N, int K);
+int gemv_nn1_nsplit(int N, int K);
 void gemv_nn1_launch(DT dt, const void* x, const void* b, const void* bias,
-                     const void* res, void* y, int N, int K, int act_kind,
-                     hipStream_t s);
+                     const void* res, void* y, float* part, int nsplit,
+                     const float* ln_g, const float* ln_b, float ln_eps, int N,
+                     int K, int act_kind, hipStream_t s);
 constexpr int GEMV1_MAX_K_DECL = 8192;
 void gemv_nn_launch(DT dt, const void* x, const void* b, const void* bias,
                     void* y, float* ws, int ks, int N, int K, int act_kind,

@@ -221,9 +221,19 @@ class _MHABase(Layer):
         self.bo = nn.Parameter(torch.zeros(dim, dtype=dtype))
 
     _kv_cache = None  # dict cache state; None = training mode
+    _wqkv = None      # concatenated projection weights (decode fast path)
 
     def reset_cache(self):
         self._kv_cache = None
+        self._wqkv = self._bqkv = None
+
+    def _build_qkv_merge(self):
+        """One [D, 3D] GEMV instead of three per decode step (the per-
+        kernel floor dominates batch-1 decode; built once per cache
+        session, dropped on reset)."""
+        with torch.no_grad():
+            self._wqkv = torch.cat([self.wq, self.wk, self.wv], 1).contiguous()
+            self._bqkv = torch.cat([self.bq, self.bk, self.bv]).contiguous()
 
     def enable_cache(self, max_len: int = 1024):
         # buffers allocated lazily at prefill (batch size unknown here);
@@ -240,12 +250,22 @@ class _MHABase(Layer):
         self._kv_cache = {"k": None, "v": None, "len": 0, "cap": max_len,
                           "pos_t": pos_t}
 
-    def _project(self, x):
+    def _project(self, x, ln=None):
         b, s, d = x.shape
         h, hd = self.num_heads, self.head_dim
-        q = ops.linear(x, self.wq, self.bq).view(b, s, h, hd).transpose(1, 2)
-        k = ops.linear(x, self.wk, self.bk).view(b, s, h, hd).transpose(1, 2)
-        v = ops.linear(x, self.wv, self.bv).view(b, s, h, hd).transpose(1, 2)
+        if (self._kv_cache is not None and s == 1 and x.is_cuda
+                and not x.requires_grad):
+            if self._wqkv is None:
+                self._build_qkv_merge()
+            qkv = ops.linear(x, self._wqkv, self._bqkv, ln=ln)
+            q, k, v = (t.view(b, s, h, hd).transpose(1, 2)
+                       for t in qkv.split(d, dim=-1))
+        else:
+            if ln is not None:
+                x = ops.layer_norm(x, *ln)
+            q = ops.linear(x, self.wq, self.bq).view(b, s, h, hd).transpose(1, 2)
+            k = ops.linear(x, self.wk, self.bk).view(b, s, h, hd).transpose(1, 2)
+            v = ops.linear(x, self.wv, self.bv).view(b, s, h, hd).transpose(1, 2)
         c = self._kv_cache
         if c is not None:
             if c["k"] is None:
@@ -333,9 +353,9 @@ class AttentionBlock(_MHABase):
     no separate mask fill (reference attention_block.cpp:144-147,
     permute_heads.cu, causal_mask.cu, softmax.cu)."""
 
-    def forward(self, x, residual=None):
+    def forward(self, x, residual=None, ln=None):
         b, s, _ = x.shape
-        q, k, v = self._project(x)
+        q, k, v = self._project(x, ln=ln)
         if self._kv_cache is not None:
             o = self._cached_attention(q, k, v)
             return self._merge(o, b, s, residual)
@@ -348,9 +368,9 @@ class FlashAttentionBlock(_MHABase):
     """Flash attention via the hand-written CDNA4 kernel on GPU
     (reference FlashAttentionBlock / cudnn SDPA graph)."""
 
-    def forward(self, x, residual=None):
+    def forward(self, x, residual=None, ln=None):
         b, s, _ = x.shape
-        q, k, v = self._project(x)
+        q, k, v = self._project(x, ln=ln)
         if self._kv_cache is not None:
             o = self._cached_attention(q, k, v)
             return self._merge(o, b, s, residual)
@@ -384,12 +404,21 @@ class GPTBlock(Layer):
         self.drop = Dropout(dropout, name=f"{name}_drop", dtype=dtype) if dropout > 0 else None
 
     def forward(self, x):
+        if (x.is_cuda and x.shape[1] == 1 and not torch.is_grad_enabled()
+                and self.attn._kv_cache is not None):
+            # decode fast path: both LayerNorms ride the projection GEMVs
+            x = self.attn(x, residual=x,
+                          ln=(self.ln1.gamma, self.ln1.beta, self.ln1.eps))
+            h = ops.linear(x, self.w1, self.b1, act="gelu",
+                           ln=(self.ln2.gamma, self.ln2.beta, self.ln2.eps))
+            return ops.linear(h, self.w2, self.b2, residual=x)
         x = self.attn(self.ln1(x), residual=x)
         h = ops.linear(self.ln2(x), self.w1, self.b1, act="gelu")
         if self.drop is not None:
             h = self.drop(ops.linear(h, self.w2, self.b2))
             return x + h
         return ops.linear(h, self.w2, self.b2, residual=x)
+
 
     def flops_per_item(self, in_shape):
         s, d = in_shape

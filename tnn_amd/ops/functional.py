@@ -290,16 +290,33 @@ class _Linear(torch.autograd.Function):
 
 def linear(x: torch.Tensor, w: torch.Tensor, bias: Optional[torch.Tensor] = None,
            act: str = "linear",
-           residual: Optional[torch.Tensor] = None) -> torch.Tensor:
-    """y = act(x @ w + bias) [+ residual]; x ``[..., K]``, w ``[K, N]``
+           residual: Optional[torch.Tensor] = None,
+           ln: Optional[tuple] = None) -> torch.Tensor:
+    """y = act(ln?(x) @ w + bias) [+ residual]; x ``[..., K]``, w ``[K, N]``
     row-major. ``residual`` (shape of y) fuses the add into the GEMM
-    epilogue on GPU (act must be 'linear')."""
+    epilogue on GPU (act must be 'linear'). ``ln=(gamma, beta, eps)``
+    fuses a LayerNorm of x into the decode GEMV's staging (M=1,
+    inference only)."""
     lead = x.shape[:-1]
     x2 = x.reshape(-1, x.shape[-1])
+    if ln is not None:
+        gamma, beta, eps = ln
+        if (_use_hip(x) and x2.shape[0] == 1
+                and not torch.is_grad_enabled()):
+            ext = _C.ext()
+            y = ext.gemm(x2.contiguous(), w, bias, ACT_KINDS[act],
+                         residual.reshape(1, -1).contiguous()
+                         if residual is not None else None,
+                         gamma, beta, eps)
+            return y.reshape(*lead, w.shape[1])
+        x = layer_norm(x, gamma, beta, eps)
+        x2 = x.reshape(-1, x.shape[-1])
     if _use_hip(x):
         # relu fuses into the GEMM epilogue (mask-replay backward); other
-        # activations need the pre-activation saved, so run them unfused
-        fused = act if act in ("linear", "relu") else "linear"
+        # activations need the pre-activation saved, so run them unfused —
+        # except under no_grad (decode), where any activation fuses
+        fused = act if (act in ("linear", "relu")
+                        or not torch.is_grad_enabled()) else "linear"
         r2 = (residual.reshape(-1, residual.shape[-1]).contiguous()
               if residual is not None else None)
         y = _Linear.apply(x2.contiguous(), w.contiguous(),
