@@ -246,3 +246,34 @@ def test_gemv_single_launch_m1(N, K):
     y = ext.gemm(x, w, b, 0)
     ref = x.float() @ w.float() + b.float()
     assert maxerr(y, ref) < tol(torch.bfloat16, K)
+
+
+@pytest.mark.parametrize("act", ["linear", "relu", "tanh"])
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_add_act_fused(act, dtype):
+    torch.manual_seed(10)
+    a = torch.randn(1000, 37, dtype=dtype, device=DEV, requires_grad=True)
+    b = torch.randn(1000, 37, dtype=dtype, device=DEV, requires_grad=True)
+    y = ops.add_act(a, b, act)
+    g = torch.randn_like(y)
+    y.backward(g)
+    a2 = a.detach().float().requires_grad_()
+    b2 = b.detach().float().requires_grad_()
+    z = a2 + b2
+    z = {"linear": z, "relu": torch.relu(z), "tanh": torch.tanh(z)}[act]
+    z.backward(g.float())
+    t = 1e-5 if dtype == torch.float32 else 2e-2
+    assert maxerr(y, z.detach()) < t
+    assert maxerr(a.grad, a2.grad) < t
+    assert maxerr(b.grad, b2.grad) < t
+
+
+def test_gemm_nt_huge_k_split():
+    """vocab-head dgrad shape: K-split fp32 slabs path"""
+    torch.manual_seed(11)
+    M, K, N = 256, 16384, 128
+    a = torch.randn(M, K, dtype=torch.bfloat16, device=DEV)
+    b = torch.randn(N, K, dtype=torch.bfloat16, device=DEV)
+    c = ext.gemm_nt(a, b)
+    ref = a.float() @ b.float().t()
+    assert maxerr(c, ref) < tol(torch.bfloat16, K)

@@ -52,6 +52,26 @@ at::Tensor act_bwd(const at::Tensor& dy, const at::Tensor& x,
   return dx;
 }
 
+at::Tensor add_act_fwd(const at::Tensor& a, const at::Tensor& b,
+                       int64_t kind) {
+  CHECK_IN(a);
+  CHECK_IN(b);
+  TORCH_CHECK(a.sizes() == b.sizes() && a.scalar_type() == b.scalar_type());
+  auto y = at::empty_like(a);
+  add_act_fwd_launch(dt_of(a), a.data_ptr(), b.data_ptr(), y.data_ptr(),
+                     a.numel(), (int)kind, cur_stream());
+  return y;
+}
+
+at::Tensor add_act_bwd(const at::Tensor& dy, const at::Tensor& y,
+                       int64_t kind) {
+  CHECK_IN(dy);
+  auto g = at::empty_like(dy);
+  add_act_bwd_launch(dt_of(dy), dy.data_ptr(), y.data_ptr(), g.data_ptr(),
+                     dy.numel(), (int)kind, cur_stream());
+  return g;
+}
+
 at::Tensor relu_bwd_mask(const at::Tensor& dy, const at::Tensor& y) {
   CHECK_IN(dy);
   CHECK_IN(y);
@@ -181,6 +201,16 @@ at::Tensor gemm_nt(const at::Tensor& a, const at::Tensor& b) {
               "gemm_nt shapes ", a.sizes(), " @ ", b.sizes(), "^T");
   int M = a.size(0), K = a.size(1), N = b.size(0);
   auto c = at::empty({M, N}, a.options());
+  int z = gemm_nt_zsplits(dt_of(a), M, N, K);
+  if (z > 1 && ((uintptr_t)a.data_ptr() & 15) == 0 &&
+      ((uintptr_t)b.data_ptr() & 15) == 0) {
+    // huge-K underfilled shape (vocab-head dgrad): K-split fp32 slabs
+    auto ws = at::empty({(int64_t)z * M * N}, a.options().dtype(at::kFloat));
+    gemm_nt_z_launch(dt_of(a), a.data_ptr(), b.data_ptr(),
+                     ws.data_ptr<float>(), c.data_ptr(), dt_of(a), z,
+                     zero_page(a), M, N, K, cur_stream());
+    return c;
+  }
   gemm_launch(dt_of(a), a.data_ptr(), b.data_ptr(), nullptr, c.data_ptr(),
               zero_page(a), /*resid=*/nullptr, M, N, K, /*trans_b=*/true, 0,
               cur_stream());
@@ -862,6 +892,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("act_fwd", &tnn::act_fwd);
   m.def("act_bwd", &tnn::act_bwd);
   m.def("relu_bwd_mask", &tnn::relu_bwd_mask);
+  m.def("add_act_fwd", &tnn::add_act_fwd);
+  m.def("add_act_bwd", &tnn::add_act_bwd);
   m.def("dropout_fwd", &tnn::dropout_fwd);
   m.def("dropout_bwd", &tnn::dropout_bwd);
   m.def("colsum", &tnn::colsum);

@@ -42,6 +42,84 @@ __global__ void k_act_bwd(const T* dy, const T* x, const T* y, T* dx,
                                         VecIO<T>::to_f32(y[i]), kind));
 }
 
+// ---- fused residual add + activation (ResidualBlock / MSequential add
+// join: one kernel instead of the at::native add + relu pair; backward
+// is a single mask pass shared by both inputs) -------------------------------
+template <typename T>
+__global__ void k_add_act_fwd(const T* __restrict__ a, const T* __restrict__ b,
+                              T* __restrict__ y, int64_t n, int kind) {
+  struct alignas(16) P { T e[16 / sizeof(T)]; };
+  constexpr int V = 16 / sizeof(T);
+  const int64_t n_v = n / V;
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const P* av = (const P*)a;
+  const P* bv = (const P*)b;
+  P* yv = (P*)y;
+  for (; i < n_v; i += stride) {
+    P x = av[i], z = bv[i], o;
+#pragma unroll
+    for (int j = 0; j < V; ++j)
+      o.e[j] = VecIO<T>::from_f32(act_apply(
+          VecIO<T>::to_f32(x.e[j]) + VecIO<T>::to_f32(z.e[j]), kind));
+    yv[i] = o;
+  }
+  for (int64_t k = n_v * V + (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       k < n; k += stride)
+    y[k] = VecIO<T>::from_f32(act_apply(
+        VecIO<T>::to_f32(a[k]) + VecIO<T>::to_f32(b[k]), kind));
+}
+
+template <typename T>
+__global__ void k_add_act_bwd(const T* __restrict__ dy, const T* __restrict__ y,
+                              T* __restrict__ g, int64_t n, int kind) {
+  struct alignas(16) P { T e[16 / sizeof(T)]; };
+  constexpr int V = 16 / sizeof(T);
+  const int64_t n_v = n / V;
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const P* dv = (const P*)dy;
+  const P* yv = (const P*)y;
+  P* gv = (P*)g;
+  for (; i < n_v; i += stride) {
+    P d = dv[i], yy = yv[i], o;
+#pragma unroll
+    for (int j = 0; j < V; ++j)
+      o.e[j] = VecIO<T>::from_f32(
+          act_grad(VecIO<T>::to_f32(d.e[j]), 0.0f,
+                   VecIO<T>::to_f32(yy.e[j]), kind));
+    gv[i] = o;
+  }
+  for (int64_t k = n_v * V + (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       k < n; k += stride)
+    g[k] = VecIO<T>::from_f32(act_grad(VecIO<T>::to_f32(dy[k]), 0.0f,
+                                       VecIO<T>::to_f32(y[k]), kind));
+}
+
+void add_act_fwd_launch(DT dt, const void* a, const void* b, void* y,
+                        int64_t n, int kind, hipStream_t s) {
+  if (dt == DT::F32)
+    hipLaunchKernelGGL(k_add_act_fwd<float>, dim3(ew_blocks(n / 4 + 1)),
+                       dim3(256), 0, s, (const float*)a, (const float*)b,
+                       (float*)y, n, kind);
+  else
+    hipLaunchKernelGGL(k_add_act_fwd<bf16>, dim3(ew_blocks(n / 8 + 1)),
+                       dim3(256), 0, s, (const bf16*)a, (const bf16*)b,
+                       (bf16*)y, n, kind);
+}
+
+void add_act_bwd_launch(DT dt, const void* dy, const void* y, void* g,
+                        int64_t n, int kind, hipStream_t s) {
+  if (dt == DT::F32)
+    hipLaunchKernelGGL(k_add_act_bwd<float>, dim3(ew_blocks(n / 4 + 1)),
+                       dim3(256), 0, s, (const float*)dy, (const float*)y,
+                       (float*)g, n, kind);
+  else
+    hipLaunchKernelGGL(k_add_act_bwd<bf16>, dim3(ew_blocks(n / 8 + 1)),
+                       dim3(256), 0, s, (const bf16*)dy, (const bf16*)y,
+                       (bf16*)g, n, kind);
+}
+
 void act_fwd_launch(DT dt, const void* x, void* y, int64_t n, int kind,
                     hipStream_t s) {
   if (dt == DT::F32)

@@ -319,6 +319,88 @@ __global__ void k_gemv_nt(const T* __restrict__ x, const T* __restrict__ B,
   }
 }
 
+// K-split NT gemm for huge-K underfilled shapes (the GPT-2 vocab-head
+// dgrad: dx[4096,768] = dy[4096,50264] @ W^T — 384 blocks and a 785-
+// chunk k-loop on the plain kernel, 313 TF). grid.z slices K; each
+// slice writes an fp32 slab, the split-K reduce (shared with wgrad)
+// sums and casts. Double-buffered glds staging (both operands' rows are
+// k-contiguous).
+template <typename T>
+__launch_bounds__(THREADS)
+__global__ void k_gemm_nt_z(const T* __restrict__ A, const T* __restrict__ B,
+                            float* __restrict__ Cws,
+                            const T* __restrict__ zero16, int M, int N,
+                            int K) {
+  __shared__ alignas(16) T As[2][BM * BK];
+  __shared__ alignas(16) T Bs[2][BN * BK];
+  const int m0 = blockIdx.x * BM;
+  const int n0 = blockIdx.y * BN;
+  const int Z = gridDim.z;
+  const int k_lo = (int)(((int64_t)K * blockIdx.z / Z) / BK * BK);
+  const int k_hi = blockIdx.z + 1 == Z
+                       ? K
+                       : (int)(((int64_t)K * (blockIdx.z + 1) / Z) / BK * BK);
+  const WaveCoord wc;
+  f32x4 acc[FM][FN] = {};
+
+  auto stage = [&](int kk0, int which) {
+    glds_stage<T, BM>(As[which], wc, [&](int rl, int kk) -> const T* {
+      int gm = m0 + rl, gk = kk0 + kk;
+      if (gm >= M || gk >= k_hi) return zero16;
+      return &A[(int64_t)gm * K + gk];
+    });
+    glds_stage<T, BN>(Bs[which], wc, [&](int rl, int kk) -> const T* {
+      int gn = n0 + rl, gk = kk0 + kk;
+      if (gn >= N || gk >= k_hi) return zero16;
+      return &B[(int64_t)gn * K + gk];
+    });
+  };
+  constexpr int NPER = glds_count<T, BM>() + glds_count<T, BN>();
+
+  const int nch = (k_hi - k_lo + BK - 1) / BK;
+  stage(k_lo, 0);
+  for (int t = 0; t < nch; ++t) {
+    const int cur = t & 1;
+    if (t + 1 < nch) {
+      stage(k_lo + (t + 1) * BK, cur ^ 1);
+      wait_vmcnt<NPER>();
+    } else {
+      wait_vmcnt<0>();
+    }
+    __builtin_amdgcn_s_barrier();
+    mfma_compute_tile(As[cur], Bs[cur], wc, acc);
+    __builtin_amdgcn_s_barrier();
+  }
+
+  float* out = Cws + (int64_t)blockIdx.z * M * N;
+  epilogue_visit(wc, acc, m0, n0, [&](int row, int col, float v) {
+    if (row < M && col < N) out[(int64_t)row * N + col] = v;
+  });
+}
+
+int gemm_nt_zsplits(DT dt, int M, int N, int K) {
+  const int base = ceil_div(M, BM) * ceil_div(N, BN);
+  if (base >= 1024 || K < 8192) return 1;
+  const bool g = dt == DT::F32 ? K % 4 == 0 : K % 8 == 0;
+  if (!g) return 1;
+  return std::min(ceil_div(K, 4 * BK), std::max(1, 1024 / base));
+}
+
+void gemm_nt_z_launch(DT dt, const void* a, const void* b, float* ws,
+                      void* c_out, DT out_dt, int z, const void* zero16,
+                      int M, int N, int K, hipStream_t s) {
+  dim3 grid(ceil_div(M, BM), ceil_div(N, BN), z);
+  if (dt == DT::F32)
+    hipLaunchKernelGGL(k_gemm_nt_z<float>, grid, dim3(THREADS), 0, s,
+                       (const float*)a, (const float*)b, ws,
+                       (const float*)zero16, M, N, K);
+  else
+    hipLaunchKernelGGL(k_gemm_nt_z<bf16>, grid, dim3(THREADS), 0, s,
+                       (const bf16*)a, (const bf16*)b, ws,
+                       (const bf16*)zero16, M, N, K);
+  splitk_reduce_launch(ws, c_out, out_dt, z, (int64_t)M * N, s);
+}
+
 // Single-launch M=1 NN matvec: out[N] = x[K] @ B[K,N], x cached in LDS,
 // bias/activation/residual fused into the epilogue. The two-kernel
 // K-split form below measured 23us per decode linear (its finalize is a

@@ -15,6 +15,10 @@ void act_fwd_launch(DT dt, const void* x, void* y, int64_t n, int kind,
                     hipStream_t s);
 void act_bwd_launch(DT dt, const void* dy, const void* x, const void* y,
                     void* dx, int64_t n, int kind, hipStream_t s);
+void add_act_fwd_launch(DT dt, const void* a, const void* b, void* y,
+                        int64_t n, int kind, hipStream_t s);
+void add_act_bwd_launch(DT dt, const void* dy, const void* y, void* g,
+                        int64_t n, int kind, hipStream_t s);
 void relu_bwd_mask_launch(DT dt, const void* dy, const void* y, void* dx,
                           int64_t n, hipStream_t s);
 void dropout_fwd_launch(DT dt, const void* x, void* y, uint8_t* mask,
@@ -40,6 +44,10 @@ constexpr int GEMV1_MAX_K_DECL = 8192;
 void gemv_nn_launch(DT dt, const void* x, const void* b, const void* bias,
                     void* y, float* ws, int ks, int N, int K, int act_kind,
                     hipStream_t s);
+int gemm_nt_zsplits(DT dt, int M, int N, int K);
+void gemm_nt_z_launch(DT dt, const void* a, const void* b, float* ws,
+                      void* c_out, DT out_dt, int z, const void* zero16,
+                      int M, int N, int K, hipStream_t s);
 int gemm_tn_zsplits(int M, int N, int K);
 void gemm_tn_launch(DT dt, const void* a, const void* b, void* c_out,
                     DT out_dt, float* ws, int z, const void* zero16, int M,

@@ -120,10 +120,7 @@ class ResidualBlock(Layer):
     def forward(self, x):
         y = self.main(x)
         s = self.shortcut(x) if self.shortcut is not None else x
-        y = y + s
-        if self.final_activation not in ("linear", "none"):
-            y = ops.activation(y, self.final_activation)
-        return y
+        return ops.add_act(y, s, self.final_activation)
 
     def output_shape(self, in_shape):
         return self.main.output_shape(in_shape)
@@ -172,6 +169,11 @@ class MSequential(Layer):
         outs = [b(x) for b in self.branches]
         if self.join == "concat":
             return torch.cat(outs, dim=-1)
+        if self.join == "add":
+            y = outs[0]
+            for o in outs[1:]:
+                y = ops.add_act(y, o)
+            return y
         fn = self.JOINS[self.join]
         y = outs[0]
         for o in outs[1:]:

@@ -29,6 +29,7 @@ from .functional import (
     scaled_softmax,
     group_norm,
     pointwise_loss,
+    add_act,
 )
 
 __all__ = [
@@ -49,4 +50,5 @@ __all__ = [
     "scaled_softmax",
     "group_norm",
     "pointwise_loss",
+    "add_act",
 ]

@@ -525,6 +525,39 @@ def scaled_softmax(x: torch.Tensor, scale: float = 1.0, causal: bool = False,
     return y.to(x.dtype)
 
 
+class _AddAct(torch.autograd.Function):
+    """y = act(a + b) in one pass (ResidualBlock join; backward is one
+    shared mask pass for both inputs). Only y-derivable activations
+    (relu/sigmoid/tanh/linear)."""
+
+    @staticmethod
+    def forward(ctx, a, b, kind):
+        ext = _C.ext()
+        y = ext.add_act_fwd(a, b, ACT_KINDS[kind])
+        ctx.save_for_backward(y)
+        ctx.kind = kind
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (y,) = ctx.saved_tensors
+        ext = _C.ext()
+        g = ext.add_act_bwd(dy.contiguous(), y, ACT_KINDS[ctx.kind])
+        return g, g, None
+
+
+def add_act(a: torch.Tensor, b: torch.Tensor,
+            act: str = "linear") -> torch.Tensor:
+    """Fused act(a + b) (residual joins)."""
+    if act == "none":
+        act = "linear"
+    if (_use_hip(a, b) and a.shape == b.shape and a.dtype == b.dtype
+            and act in ("linear", "relu", "sigmoid", "tanh")):
+        return _AddAct.apply(a.contiguous(), b.contiguous(), act)
+    y = a + b
+    return activation(y, act) if act != "linear" else y
+
+
 def activation(x: torch.Tensor, kind: str) -> torch.Tensor:
     if kind == "linear":
         return x
