@@ -222,7 +222,9 @@ template <typename T>
 __global__ void k_bn_apply_drop(const T* __restrict__ x, const float* mean,
                                 const float* invstd, const float* gamma,
                                 const float* beta, T* __restrict__ y,
-                                int64_t n, int cols, float p, uint64_t seed) {
+                                int64_t n, int cols, float p, uint64_t seed,
+                                const int64_t* __restrict__ ctr) {
+  if (ctr) seed ^= (uint64_t)(*ctr) * 0x9E3779B97F4A7C15ull;
   const float scale = 1.0f / (1.0f - p);
   Philox rng(seed);
   int64_t i4 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -253,7 +255,9 @@ __global__ void k_bn_apply_drop_vec(const T* __restrict__ x, const float* mean,
                                     const float* invstd, const float* gamma,
                                     const float* beta, T* __restrict__ y,
                                     int64_t nv, int groups, float p,
-                                    uint64_t seed) {
+                                    uint64_t seed,
+                                    const int64_t* __restrict__ ctr) {
+  if (ctr) seed ^= (uint64_t)(*ctr) * 0x9E3779B97F4A7C15ull;
   constexpr int V = 16 / sizeof(T);
   using VecT = BnPack<T>;
   const float scale = 1.0f / (1.0f - p);
@@ -448,7 +452,8 @@ void bn_apply_launch(DT dt, const void* x, const float* mean,
 void bn_apply_drop_launch(DT dt, const void* x, const float* mean,
                           const float* invstd, const float* gamma,
                           const float* beta, void* y, int64_t rows, int cols,
-                          float p, uint64_t seed, hipStream_t s) {
+                          float p, uint64_t seed, const int64_t* ctr,
+                          hipStream_t s) {
   int64_t n = rows * cols;
   int V = dt == DT::F32 ? 4 : 8;
   if (cols % V == 0 && (((uintptr_t)x & 15) == 0) && n < (1ll << 34)) {
@@ -457,22 +462,22 @@ void bn_apply_drop_launch(DT dt, const void* x, const float* mean,
     if (dt == DT::F32)
       hipLaunchKernelGGL(k_bn_apply_drop_vec<float>, dim3(blocks), dim3(256),
                          0, s, (const float*)x, mean, invstd, gamma, beta,
-                         (float*)y, nv, cols / V, p, seed);
+                         (float*)y, nv, cols / V, p, seed, ctr);
     else
       hipLaunchKernelGGL(k_bn_apply_drop_vec<bf16>, dim3(blocks), dim3(256),
                          0, s, (const bf16*)x, mean, invstd, gamma, beta,
-                         (bf16*)y, nv, cols / V, p, seed);
+                         (bf16*)y, nv, cols / V, p, seed, ctr);
     return;
   }
   int blocks = (int)std::min<int64_t>(((n + 3) / 4 + 255) / 256, (int64_t)2048);
   if (dt == DT::F32)
     hipLaunchKernelGGL(k_bn_apply_drop<float>, dim3(blocks), dim3(256), 0, s,
                        (const float*)x, mean, invstd, gamma, beta, (float*)y, n,
-                       cols, p, seed);
+                       cols, p, seed, ctr);
   else
     hipLaunchKernelGGL(k_bn_apply_drop<bf16>, dim3(blocks), dim3(256), 0, s,
                        (const bf16*)x, mean, invstd, gamma, beta, (bf16*)y, n,
-                       cols, p, seed);
+                       cols, p, seed, ctr);
 }
 
 void bn_infer_launch(DT dt, const void* x, const float* rmean,

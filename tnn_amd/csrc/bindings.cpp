@@ -82,13 +82,15 @@ at::Tensor relu_bwd_mask(const at::Tensor& dy, const at::Tensor& y) {
 }
 
 std::vector<at::Tensor> dropout_fwd(const at::Tensor& x, double p,
-                                    int64_t seed) {
+                                    int64_t seed,
+                                    const c10::optional<at::Tensor>& ctr) {
   CHECK_IN(x);
   auto y = at::empty_like(x);
   auto mask = at::empty(x.sizes(), x.options().dtype(at::kByte));
+  const int64_t* ctrp = ctr.has_value() ? ctr->data_ptr<int64_t>() : nullptr;
   dropout_fwd_launch(dt_of(x), x.data_ptr(), y.data_ptr(),
                      mask.data_ptr<uint8_t>(), x.numel(), (float)p,
-                     (uint64_t)seed, cur_stream());
+                     (uint64_t)seed, ctrp, cur_stream());
   return {y, mask};
 }
 
@@ -325,8 +327,16 @@ std::vector<at::Tensor> conv2d_fwd_stats(const at::Tensor& x,
                        ph, pw);
   auto y = at::empty({cs.N, cs.OH, cs.OW, cs.Cout}, x.options());
   const void* bp = bias.has_value() ? bias->data_ptr() : nullptr;
-  if (!conv2d_fwd_wants_db(dt_of(x), x.data_ptr(), cs))
+  if (!conv2d_fwd_wants_db(dt_of(x), x.data_ptr(), cs)) {
+    // stats fusion needs the db kernel; ineligible shapes (e.g. K > 2304)
+    // still must COMPUTE y via the plain path. (Round-1 latent bug: this
+    // early return handed back uninitialized y — garbage activations for
+    // every 512-channel conv under BN training.)
+    conv2d_fwd_launch(dt_of(x), x.data_ptr(), w.data_ptr(), nullptr, bp,
+                      y.data_ptr(), zero_page(x), nullptr, cs,
+                      /*relu=*/false, cur_stream());
     return {y, at::empty({0}, x.options().dtype(at::kFloat))};
+  }
   auto wt2 = at::empty({(int64_t)cs.Cout, (int64_t)cs.KH * cs.KW * cs.Cin},
                        w.options());
   transpose_w_fwd_launch(dt_of(x), w.data_ptr(), wt2.data_ptr(),
@@ -411,7 +421,9 @@ std::vector<at::Tensor> bn_fwd_train(const at::Tensor& x,
                                      const c10::optional<at::Tensor>& rvar,
                                      double momentum, double eps, bool relu,
                                      double dropout_p, int64_t seed,
-                                     const c10::optional<at::Tensor>& precomp) {
+                                     const c10::optional<at::Tensor>& precomp,
+                                     const c10::optional<at::Tensor>& ctr =
+                                         c10::nullopt) {
   CHECK_IN(x);
   TORCH_CHECK(dropout_p == 0.0 || relu, "fused BN dropout requires relu");
   int C = x.size(-1);
@@ -437,7 +449,9 @@ std::vector<at::Tensor> bn_fwd_train(const at::Tensor& x,
     bn_apply_drop_launch(dt_of(x), x.data_ptr(), mean.data_ptr<float>(),
                          invstd.data_ptr<float>(), gamma.data_ptr<float>(),
                          beta.data_ptr<float>(), y.data_ptr(), rows, C,
-                         (float)dropout_p, (uint64_t)seed, cur_stream());
+                         (float)dropout_p, (uint64_t)seed,
+                         ctr.has_value() ? ctr->data_ptr<int64_t>() : nullptr,
+                         cur_stream());
   else
     bn_apply_launch(dt_of(x), x.data_ptr(), mean.data_ptr<float>(),
                     invstd.data_ptr<float>(), gamma.data_ptr<float>(),
@@ -769,25 +783,31 @@ void sgd_step(at::Tensor param, at::Tensor master, const at::Tensor& grad,
 void adam_step(at::Tensor param, at::Tensor master, const at::Tensor& grad,
                at::Tensor m, at::Tensor v, int64_t step, double lr,
                double beta1, double beta2, double eps, double weight_decay,
-               bool adamw) {
+               bool adamw,
+               const c10::optional<at::Tensor>& step_dev = c10::nullopt) {
   CHECK_IN(param);
   bool has_master = master.data_ptr() != param.data_ptr();
+  const int64_t* sd =
+      step_dev.has_value() ? step_dev->data_ptr<int64_t>() : nullptr;
   adam_step_launch(dt_of(param), grad.data_ptr(), dt_of(grad), param.data_ptr(),
                    master.data_ptr<float>(), m.data_ptr<float>(),
-                   v.data_ptr<float>(), param.numel(), (int)step, (float)lr,
-                   (float)beta1, (float)beta2, (float)eps, (float)weight_decay,
-                   adamw, has_master, cur_stream());
+                   v.data_ptr<float>(), param.numel(), (int)step, sd,
+                   (float)lr, (float)beta1, (float)beta2, (float)eps,
+                   (float)weight_decay, adamw, has_master, cur_stream());
 }
 
 void adam_step_mt(const at::Tensor& desc, const at::Tensor& chunks,
                   int64_t dt_p, int64_t dt_g, bool has_master, int64_t step,
                   double lr, double beta1, double beta2, double eps,
-                  double weight_decay, bool adamw) {
+                  double weight_decay, bool adamw,
+                  const c10::optional<at::Tensor>& step_dev = c10::nullopt) {
   CHECK_IN(desc);
   CHECK_IN(chunks);
+  const int64_t* sd =
+      step_dev.has_value() ? step_dev->data_ptr<int64_t>() : nullptr;
   adam_mt_launch((DT)dt_p, (DT)dt_g, has_master,
                  desc.data_ptr<int64_t>(), chunks.data_ptr<int64_t>(),
-                 (int)chunks.numel(), (int)step, (float)lr, (float)beta1,
+                 (int)chunks.numel(), (int)step, sd, (float)lr, (float)beta1,
                  (float)beta2, (float)eps, (float)weight_decay, adamw,
                  cur_stream());
 }
@@ -894,7 +914,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("relu_bwd_mask", &tnn::relu_bwd_mask);
   m.def("add_act_fwd", &tnn::add_act_fwd);
   m.def("add_act_bwd", &tnn::add_act_bwd);
-  m.def("dropout_fwd", &tnn::dropout_fwd);
+  m.def("dropout_fwd", &tnn::dropout_fwd, py::arg("x"), py::arg("p"),
+        py::arg("seed"), py::arg("ctr") = c10::nullopt);
   m.def("dropout_bwd", &tnn::dropout_bwd);
   m.def("colsum", &tnn::colsum);
   m.def("gemm", &tnn::gemm, py::arg("a"), py::arg("b"), py::arg("bias"),
@@ -909,7 +930,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_fwd_stats", &tnn::conv2d_fwd_stats);
   m.def("conv2d_dgrad", &tnn::conv2d_dgrad);
   m.def("conv2d_wgrad", &tnn::conv2d_wgrad);
-  m.def("bn_fwd_train", &tnn::bn_fwd_train);
+  m.def("bn_fwd_train", &tnn::bn_fwd_train, py::arg("x"), py::arg("gamma"),
+        py::arg("beta"), py::arg("rmean"), py::arg("rvar"),
+        py::arg("momentum"), py::arg("eps"), py::arg("relu"),
+        py::arg("dropout_p"), py::arg("seed"), py::arg("precomp"),
+        py::arg("ctr") = c10::nullopt);
   m.def("bn_fwd_infer", &tnn::bn_fwd_infer);
   m.def("bn_bwd", &tnn::bn_bwd);
   m.def("maxpool_fwd", &tnn::maxpool_fwd);
@@ -933,7 +958,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("smax_bwd", &tnn::smax_bwd);
   m.def("attn_decode", &tnn::attn_decode);
   m.def("sgd_step", &tnn::sgd_step);
-  m.def("adam_step", &tnn::adam_step);
-  m.def("adam_step_mt", &tnn::adam_step_mt);
+  m.def("adam_step", &tnn::adam_step, py::arg("param"), py::arg("master"),
+        py::arg("grad"), py::arg("m"), py::arg("v"), py::arg("step"),
+        py::arg("lr"), py::arg("beta1"), py::arg("beta2"), py::arg("eps"),
+        py::arg("weight_decay"), py::arg("adamw"),
+        py::arg("step_dev") = c10::nullopt);
+  m.def("adam_step_mt", &tnn::adam_step_mt, py::arg("desc"),
+        py::arg("chunks"), py::arg("dt_p"), py::arg("dt_g"),
+        py::arg("has_master"), py::arg("step"), py::arg("lr"),
+        py::arg("beta1"), py::arg("beta2"), py::arg("eps"),
+        py::arg("weight_decay"), py::arg("adamw"),
+        py::arg("step_dev") = c10::nullopt);
   m.def("decode_image", &tnn::imgcodec::decode_image);
 }

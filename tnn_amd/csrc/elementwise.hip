@@ -164,7 +164,11 @@ void relu_bwd_mask_launch(DT dt, const void* dy, const void* y, void* dx,
 // ---- dropout (Philox 4x32; reference dropout.cu:33 vectorized form) --------
 template <typename T>
 __global__ void k_dropout_fwd(const T* x, T* y, uint8_t* mask, int64_t n,
-                              float p, uint64_t seed) {
+                              float p, uint64_t seed,
+                              const int64_t* __restrict__ ctr) {
+  // hipGraph-captured step: the per-call salt is baked at capture, the
+  // device counter varies per replay -> fresh Philox stream each step
+  if (ctr) seed ^= (uint64_t)(*ctr) * 0x9E3779B97F4A7C15ull;
   const float scale = 1.0f / (1.0f - p);
   Philox rng(seed);
   int64_t i4 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -197,14 +201,15 @@ __global__ void k_dropout_bwd(const T* dy, const uint8_t* mask, T* dx,
 }
 
 void dropout_fwd_launch(DT dt, const void* x, void* y, uint8_t* mask,
-                        int64_t n, float p, uint64_t seed, hipStream_t s) {
+                        int64_t n, float p, uint64_t seed,
+                        const int64_t* ctr, hipStream_t s) {
   int blocks = ew_blocks((n + 3) / 4);
   if (dt == DT::F32)
     hipLaunchKernelGGL(k_dropout_fwd<float>, dim3(blocks), dim3(256), 0, s,
-                       (const float*)x, (float*)y, mask, n, p, seed);
+                       (const float*)x, (float*)y, mask, n, p, seed, ctr);
   else
     hipLaunchKernelGGL(k_dropout_fwd<bf16>, dim3(blocks), dim3(256), 0, s,
-                       (const bf16*)x, (bf16*)y, mask, n, p, seed);
+                       (const bf16*)x, (bf16*)y, mask, n, p, seed, ctr);
 }
 
 void dropout_bwd_launch(DT dt, const void* dy, const uint8_t* mask, void* dx,

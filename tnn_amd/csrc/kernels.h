@@ -22,7 +22,8 @@ void add_act_bwd_launch(DT dt, const void* dy, const void* y, void* g,
 void relu_bwd_mask_launch(DT dt, const void* dy, const void* y, void* dx,
                           int64_t n, hipStream_t s);
 void dropout_fwd_launch(DT dt, const void* x, void* y, uint8_t* mask,
-                        int64_t n, float p, uint64_t seed, hipStream_t s);
+                        int64_t n, float p, uint64_t seed,
+                        const int64_t* ctr, hipStream_t s);
 void dropout_bwd_launch(DT dt, const void* dy, const uint8_t* mask, void* dx,
                         int64_t n, float p, hipStream_t s);
 void colsum_launch(DT dt, const void* x, void* out_f32, int64_t rows,
@@ -132,7 +133,8 @@ void bn_apply_launch(DT dt, const void* x, const float* mean,
 void bn_apply_drop_launch(DT dt, const void* x, const float* mean,
                           const float* invstd, const float* gamma,
                           const float* beta, void* y, int64_t rows, int cols,
-                          float p, uint64_t seed, hipStream_t s);
+                          float p, uint64_t seed, const int64_t* ctr,
+                          hipStream_t s);
 void bn_infer_launch(DT dt, const void* x, const float* rmean,
                      const float* rvar, const float* gamma, const float* beta,
                      void* y, int64_t rows, int cols, float eps, bool relu,
@@ -234,12 +236,13 @@ void sgd_step_launch(DT dt_p, const void* grad, DT dt_g, void* param,
                      bool has_master, hipStream_t s);
 void adam_step_launch(DT dt_p, const void* grad, DT dt_g, void* param,
                       float* master, float* m, float* v, int64_t n, int step,
-                      float lr, float beta1, float beta2, float eps,
-                      float weight_decay, bool adamw, bool has_master,
-                      hipStream_t s);
+                      const int64_t* step_dev, float lr, float beta1,
+                      float beta2, float eps, float weight_decay, bool adamw,
+                      bool has_master, hipStream_t s);
 void adam_mt_launch(DT dt_p, DT dt_g, bool has_master, const int64_t* desc,
-                    const int64_t* chunks, int nchunks, int step, float lr,
-                    float beta1, float beta2, float eps, float weight_decay,
-                    bool adamw, hipStream_t s);
+                    const int64_t* chunks, int nchunks, int step,
+                    const int64_t* step_dev, float lr, float beta1,
+                    float beta2, float eps, float weight_decay, bool adamw,
+                    hipStream_t s);
 
 }  // namespace tnn

@@ -34,7 +34,15 @@ __global__ void k_adam(TP* __restrict__ p, float* __restrict__ master,
                        const TG* __restrict__ g, float* __restrict__ m,
                        float* __restrict__ v, int64_t n, float lr, float beta1,
                        float beta2, float eps, float wd, float bc1, float bc2,
-                       bool adamw, bool has_master) {
+                       bool adamw, bool has_master,
+                       const int64_t* __restrict__ step_dev) {
+  if (step_dev) {
+    // hipGraph-captured step: bias correction from the device step
+    // counter (incremented in-graph), not a baked-in host value
+    const float t = (float)*step_dev;
+    bc1 = 1.0f - powf(beta1, t);  // precise powf: bitwise-matches the
+    bc2 = 1.0f - powf(beta2, t);  // host-computed eager bias correction
+  }
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (; i < n; i += stride) {
@@ -61,7 +69,13 @@ __launch_bounds__(256)
 __global__ void k_adam_mt(const int64_t* __restrict__ desc,
                           const int64_t* __restrict__ chunks, float lr,
                           float beta1, float beta2, float eps, float wd,
-                          float bc1, float bc2, bool adamw) {
+                          float bc1, float bc2, bool adamw,
+                          const int64_t* __restrict__ step_dev) {
+  if (step_dev) {
+    const float t = (float)*step_dev;
+    bc1 = 1.0f - powf(beta1, t);  // precise powf (see k_adam)
+    bc2 = 1.0f - powf(beta2, t);
+  }
   int64_t c = chunks[blockIdx.x];
   int ti = (int)(c >> 32);
   int64_t off = (int64_t)(c & 0xffffffff) * MT_CHUNK;
@@ -87,12 +101,13 @@ __global__ void k_adam_mt(const int64_t* __restrict__ desc,
 }
 
 void adam_mt_launch(DT dt_p, DT dt_g, bool has_master, const int64_t* desc,
-                    const int64_t* chunks, int nchunks, int step, float lr,
-                    float beta1, float beta2, float eps, float weight_decay,
-                    bool adamw, hipStream_t s) {
+                    const int64_t* chunks, int nchunks, int step,
+                    const int64_t* step_dev, float lr, float beta1,
+                    float beta2, float eps, float weight_decay, bool adamw,
+                    hipStream_t s) {
   float bc1 = 1.0f - powf(beta1, (float)step);
   float bc2 = 1.0f - powf(beta2, (float)step);
-#define CASE(TP, TG, M)                                                         hipLaunchKernelGGL((k_adam_mt<TP, TG, M>), dim3(nchunks), dim3(256), 0, s,                       desc, chunks, lr, beta1, beta2, eps, weight_decay, bc1,                       bc2, adamw)
+#define CASE(TP, TG, M)                                                         hipLaunchKernelGGL((k_adam_mt<TP, TG, M>), dim3(nchunks), dim3(256), 0, s,                       desc, chunks, lr, beta1, beta2, eps, weight_decay, bc1,                       bc2, adamw, step_dev)
   if (has_master) {
     if (dt_p == DT::F32 && dt_g == DT::F32) CASE(float, float, true);
     else if (dt_p == DT::BF16 && dt_g == DT::BF16) CASE(bf16, bf16, true);
@@ -129,15 +144,16 @@ void sgd_step_launch(DT dt_p, const void* grad, DT dt_g, void* param,
 
 void adam_step_launch(DT dt_p, const void* grad, DT dt_g, void* param,
                       float* master, float* m, float* v, int64_t n, int step,
-                      float lr, float beta1, float beta2, float eps,
-                      float weight_decay, bool adamw, bool has_master,
-                      hipStream_t s) {
+                      const int64_t* step_dev, float lr, float beta1,
+                      float beta2, float eps, float weight_decay, bool adamw,
+                      bool has_master, hipStream_t s) {
   float bc1 = 1.0f - powf(beta1, (float)step);
   float bc2 = 1.0f - powf(beta2, (float)step);
 #define CASE(TP, TG)                                                          \
   hipLaunchKernelGGL((k_adam<TP, TG>), dim3(ob(n)), dim3(256), 0, s,          \
                      (TP*)param, master, (const TG*)grad, m, v, n, lr, beta1, \
-                     beta2, eps, weight_decay, bc1, bc2, adamw, has_master)
+                     beta2, eps, weight_decay, bc1, bc2, adamw, has_master,   \
+                     step_dev)
   if (dt_p == DT::F32 && dt_g == DT::F32) CASE(float, float);
   else if (dt_p == DT::BF16 && dt_g == DT::BF16) CASE(bf16, bf16);
   else if (dt_p == DT::BF16 && dt_g == DT::F32) CASE(bf16, float);

@@ -128,6 +128,10 @@ class Adam(Optimizer):
         super().__init__(params, lr)
         self.beta1, self.beta2, self.eps, self.weight_decay = beta1, beta2, eps, weight_decay
         self._mt_chunk_cache = {}
+        self._mt_desc_cache = {}
+        # device step counter for hipGraph-captured steps (bias correction
+        # read on device instead of baked in at capture)
+        self._step_dev = None
 
     @torch.no_grad()
     def step(self):
@@ -179,12 +183,23 @@ class Adam(Optimizer):
                          st["master"].data_ptr() if has_master else 0,
                          g.data_ptr(), st["m"].data_ptr(),
                          st["v"].data_ptr(), p.numel()]
-            desc_t = torch.tensor(desc, dtype=torch.int64, device=dev)
+            # pointer-stable across steps: cache the device copy per dtype
+            # group, revalidated against the pointer tuple (also keeps H2D
+            # pageable copies out of hipGraph capture)
+            dkey = tuple(desc)
+            ck = (pdt, gdt, has_master, numels)
+            ent = self._mt_desc_cache.get(ck)
+            if ent is not None and ent[0] == dkey:
+                desc_t = ent[1]
+            else:
+                desc_t = torch.tensor(desc, dtype=torch.int64, device=dev)
+                self._mt_desc_cache[ck] = (dkey, desc_t)
             ext.adam_step_mt(desc_t, chunks_t,
                              0 if pdt == torch.float32 else 1,
                              0 if gdt == torch.float32 else 1, has_master,
                              self.step_count, self.lr, self.beta1, self.beta2,
-                             self.eps, self.weight_decay, self._adamw)
+                             self.eps, self.weight_decay, self._adamw,
+                             step_dev=self._step_dev)
 
     def _update(self, p, g, st):
         if "m" not in st:
@@ -196,7 +211,8 @@ class Adam(Optimizer):
             master = st.get("master", p.data)
             ext.adam_step(p.data, master, g, st["m"], st["v"], self.step_count,
                           self.lr, self.beta1, self.beta2, self.eps,
-                          self.weight_decay, self._adamw)
+                          self.weight_decay, self._adamw,
+                          step_dev=self._step_dev)
             return
         master = st.get("master", p.data)
         gf = g.float()

@@ -45,6 +45,24 @@ def _use_hip(*tensors: torch.Tensor) -> bool:
     return any(t is not None and t.is_cuda for t in tensors)
 
 
+# hipGraph-captured training steps: RNG consumers combine a per-call salt
+# (baked at capture) with this device counter (incremented in-graph), so
+# every replay draws a fresh Philox stream with zero host involvement.
+_graph_ctr: Optional[torch.Tensor] = None
+
+
+def set_graph_seed_ctr(t: Optional[torch.Tensor]):
+    global _graph_ctr
+    _graph_ctr = t
+
+
+def _draw_seed() -> int:
+    if _graph_ctr is not None:
+        import random
+        return random.getrandbits(62)  # salt only; no GPU sync in capture
+    return int(torch.randint(0, 2 ** 62, (1,)).item())
+
+
 # ---------------------------------------------------------------------------
 # conv2d NHWC (implicit GEMM on GPU; reference legacy_conv2d_layer.cpp:137)
 # ---------------------------------------------------------------------------
@@ -148,7 +166,7 @@ def conv2d_nhwc(
 class _BatchNormAct(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, gamma, beta, running_mean, running_var, momentum,
-                eps, relu, dropout_p=0.0, seed=0, precomp=None):
+                eps, relu, dropout_p=0.0, seed=0, precomp=None, ctr=None):
         ext = _C.ext()
         # running stats update fused into the finalize kernel; dropout (if
         # any) fused into the apply kernel -- dropped positions write 0, so
@@ -156,7 +174,7 @@ class _BatchNormAct(torch.autograd.Function):
         # backward is a constant 1/(1-p) scale on the kept positions
         y, mean, invstd = ext.bn_fwd_train(x, gamma, beta, running_mean,
                                            running_var, momentum, eps, relu,
-                                           dropout_p, seed, precomp)
+                                           dropout_p, seed, precomp, ctr)
         save_y = y if (relu or dropout_p > 0.0) else None
         ctx.save_for_backward(x, gamma, mean, invstd, save_y)
         ctx.relu = relu or dropout_p > 0.0
@@ -170,7 +188,7 @@ class _BatchNormAct(torch.autograd.Function):
         dx, dgamma, dbeta = ext.bn_bwd(x, dy.contiguous(), gamma, mean, invstd,
                                        y if ctx.relu else None, ctx.dy_scale)
         return (dx, dgamma, dbeta, None, None, None, None, None, None, None,
-                None)
+                None, None)
 
 
 def batch_norm_act(
@@ -199,13 +217,11 @@ def batch_norm_act(
     n = x.numel() // C
     if _use_hip(x):
         if training:
-            seed = 0
-            if dropout_p > 0.0:
-                seed = int(torch.randint(0, 2 ** 62, (1,)).item())
+            seed = _draw_seed() if dropout_p > 0.0 else 0
             y, _, _ = _BatchNormAct.apply(x.contiguous(), gamma, beta,
                                           running_mean, running_var, momentum,
                                           eps, relu, dropout_p, seed,
-                                          precomputed)
+                                          precomputed, _graph_ctr)
             return y
         ext = _C.ext()
         return ext.bn_fwd_infer(x.contiguous(), gamma, beta, running_mean, running_var,
@@ -441,9 +457,9 @@ def avg_pool2d_nhwc(x, kernel, stride=None, padding=(0, 0)):
 
 class _Dropout(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, p, seed):
+    def forward(ctx, x, p, seed, ctr=None):
         ext = _C.ext()
-        y, mask = ext.dropout_fwd(x, p, seed)
+        y, mask = ext.dropout_fwd(x, p, seed, ctr)
         ctx.save_for_backward(mask)
         ctx.p = p
         return y
@@ -452,15 +468,15 @@ class _Dropout(torch.autograd.Function):
     def backward(ctx, dy):
         (mask,) = ctx.saved_tensors
         ext = _C.ext()
-        return ext.dropout_bwd(dy.contiguous(), mask, ctx.p), None, None
+        return (ext.dropout_bwd(dy.contiguous(), mask, ctx.p), None, None,
+                None)
 
 
 def dropout(x: torch.Tensor, p: float, training: bool) -> torch.Tensor:
     if not training or p <= 0.0:
         return x
     if _use_hip(x):
-        seed = int(torch.randint(0, 2 ** 62, (1,)).item())
-        return _Dropout.apply(x.contiguous(), p, seed)
+        return _Dropout.apply(x.contiguous(), p, _draw_seed(), _graph_ctr)
     return F.dropout(x, p, training=True)
 
 

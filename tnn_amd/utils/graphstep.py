@@ -39,3 +39,68 @@ class GraphedInference:
         self.static_in.copy_(x)
         self.graph.replay()
         return self.static_out
+
+
+class GraphedTrainStep:
+    """Whole-training-step hipGraph capture: grad-zeroing, forward, loss,
+    backward and the fused optimizer update replay as ONE graph launch
+    (the round-1 blockers are solved device-side: dropout Philox streams
+    combine a capture-baked salt with an in-graph device counter, and the
+    Adam bias correction reads the same counter instead of a baked-in
+    host step — csrc/optim.hip, elementwise.hip, batchnorm.hip).
+
+    Constraints: fixed shapes, constant LR (schedulers would freeze), the
+    optimizer's python ``step_count`` stays at its capture value (the
+    device counter is the truth; checkpoint via ``sync_step_count()``).
+    """
+
+    def __init__(self, model, criterion, optimizer, example_x, example_y,
+                 warmup: int = 3):
+        from ..ops.functional import set_graph_seed_ctr
+        assert example_x.is_cuda
+        self.model, self.criterion, self.optimizer = model, criterion, optimizer
+        self.static_x = example_x.clone()
+        self.static_y = example_y.clone()
+        dev = example_x.device
+        self.step_ctr = torch.zeros(1, dtype=torch.int64, device=dev)
+        model.train()
+        optimizer._step_dev = self.step_ctr
+        set_graph_seed_ctr(self.step_ctr)
+        try:
+            stream = torch.cuda.Stream()
+            stream.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(stream):
+                for _ in range(warmup):
+                    self._step_body()
+            torch.cuda.current_stream().wait_stream(stream)
+            self.graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self.graph):
+                self.static_loss = self._step_body()
+        finally:
+            set_graph_seed_ctr(None)
+
+    def _step_body(self):
+        self.step_ctr.add_(1)
+        # in-place zeroing (set-to-None would reallocate grads per replay)
+        for p in self.optimizer.params:
+            if p.grad is not None:
+                p.grad.zero_()
+        out = self.model(self.static_x)
+        loss = self.criterion(out, self.static_y)
+        loss.backward()
+        self.optimizer.step()
+        return loss
+
+    def __call__(self, x=None, y=None) -> torch.Tensor:
+        """Run one training step; returns the (device) loss tensor."""
+        if x is not None:
+            self.static_x.copy_(x)
+        if y is not None:
+            self.static_y.copy_(y)
+        self.graph.replay()
+        return self.static_loss
+
+    def sync_step_count(self):
+        """Pull the device step counter back into the optimizer (one host
+        sync; call before checkpointing)."""
+        self.optimizer.step_count = int(self.step_ctr.item())
