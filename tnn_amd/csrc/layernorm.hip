@@ -183,8 +183,9 @@ void ln_bwd_launch(DT dt, const void* x, const void* dy, const float* gamma,
     int cblocks = (groups + gpb - 1) / gpb;
     int rows_per_iter = 256 / gpb;
     int64_t iters = (rows + rows_per_iter - 1) / rows_per_iter;
+    // ~512 blocks (the iters/32 cap starved 768-col reduces at 64 blocks)
     int rslices = (int)std::min<int64_t>(
-        std::max<int64_t>(2048 / cblocks, 1), std::max<int64_t>(iters / 32, 1));
+        std::max<int64_t>(512 / cblocks, 1), std::max<int64_t>(iters / 4, 1));
     dim3 grid(cblocks, rslices);
     if (dt == DT::F32)
       hipLaunchKernelGGL(k_ln_bwd_col<float>, grid, dim3(256), 0, s,

@@ -87,6 +87,43 @@ __global__ void k_adam_mt(const int64_t* __restrict__ desc,
   float* v = (float*)d[4];
   int64_t n = d[5];
   int64_t end = off + MT_CHUNK < n ? off + MT_CHUNK : n;
+  // 4-wide vector path: m/v/master float4, p/g 4-element packs (the
+  // scalar loop measured ~1.5x off the update's memory roofline)
+  struct alignas(4 * sizeof(TP)) P4 { TP e[4]; };
+  struct alignas(4 * sizeof(TG)) G4 { TG e[4]; };
+  const bool vec4 = (end - off) % 4 == 0 && ((uintptr_t)p & 15) == 0 &&
+                    ((uintptr_t)g & 15) == 0;
+  if (vec4) {
+    for (int64_t i4 = off / 4 + threadIdx.x; i4 * 4 < end; i4 += 256) {
+      float4 mi4 = ((float4*)m)[i4];
+      float4 vi4 = ((float4*)v)[i4];
+      float4 w4;
+      if (MASTER) w4 = ((float4*)master)[i4];
+      P4 pp;
+      if (!MASTER) pp = ((P4*)p)[i4];
+      G4 gg = ((const G4*)g)[i4];
+      float* wv = (float*)&w4;
+      float* mv = (float*)&mi4;
+      float* vv = (float*)&vi4;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        float w = MASTER ? wv[j] : VecIO<TP>::to_f32(pp.e[j]);
+        float gr = VecIO<TG>::to_f32(gg.e[j]);
+        if (!adamw) gr += wd * w;
+        float mi = mv[j] = beta1 * mv[j] + (1.0f - beta1) * gr;
+        float vi = vv[j] = beta2 * vv[j] + (1.0f - beta2) * gr * gr;
+        if (adamw) w *= (1.0f - lr * wd);
+        w -= lr / bc1 * mi / (sqrtf(vi / bc2) + eps);
+        wv[j] = w;
+        pp.e[j] = VecIO<TP>::from_f32(w);
+      }
+      ((float4*)m)[i4] = mi4;
+      ((float4*)v)[i4] = vi4;
+      if (MASTER) ((float4*)master)[i4] = w4;
+      ((P4*)p)[i4] = pp;
+    }
+    return;
+  }
   for (int64_t i = off + threadIdx.x; i < end; i += 256) {
     float w = MASTER ? master[i] : VecIO<TP>::to_f32(p[i]);
     float gr = VecIO<TG>::to_f32(g[i]);
