@@ -108,3 +108,27 @@ def test_hwinfo_probe():
     info = hwinfo.HardwareInfo.probe()
     assert info.cpu_count >= 1
     assert isinstance(info.gpus, list)
+
+
+def test_hwinfo_topology_parse():
+    """xGMI matrix parsing (both rocm-smi output forms) + chain order."""
+    from tnn_amd.utils.hwinfo import HardwareInfo, _parse_topo_matrix
+    table = (
+        "        GPU0  GPU1  GPU2\n"
+        "GPU0    0     XGMI  PCIE\n"
+        "GPU1    XGMI  0     XGMI\n"
+        "GPU2    PCIE  XGMI  0\n")
+    links = _parse_topo_matrix(table)
+    assert links[(0, 1)] == "XGMI" and links[(0, 2)] == "PCIE"
+    pairform = ("(Topology) Link type between DRM devices 0 and 1: XGMI\n"
+                "(Topology) Link type between DRM devices 0 and 2: PCIE\n")
+    links2 = _parse_topo_matrix(pairform)
+    assert links2[(1, 0)] == "XGMI" and links2[(2, 0)] == "PCIE"
+
+    hw = HardwareInfo(cpu_count=8)
+    from tnn_amd.utils.hwinfo import GPUInfo
+    hw.gpus = [GPUInfo(i, "MI355X", 0, 256) for i in range(3)]
+    hw.link_type = links
+    assert hw.xgmi_peers(1) == [0, 2]
+    order = hw.pipeline_order()
+    assert order == [0, 1, 2]  # chain follows xGMI neighbors
