@@ -429,6 +429,21 @@ __global__ void k_attn_bwd(const bf16* __restrict__ Q, const bf16* __restrict__ 
   stage_tile<BKV, D, NW>(v + (int64_t)kv0 * D, v_lds, S - kv0, zero16);
   stage_tile_t<BKV, D, BWD_THREADS>(k + (int64_t)kv0 * D, kt_lds, S - kv0);
 
+  // hoist this wave's K/V MFMA A-fragments into registers for the whole
+  // q-tile loop (re-read from LDS every tile otherwise). D==64 only:
+  // at D==128 the +32 VGPR would push past the 256 cap and spill.
+  bf16x8 k_frag[D == 64 ? D / 32 : 1], v_frag[D == 64 ? D / 32 : 1];
+  if constexpr (D == 64) {
+    __syncthreads();
+    const int r = lane & 15;
+#pragma unroll
+    for (int ks = 0; ks < D / 32; ++ks) {
+      const int kb = ks * 32 + (lane >> 4) * 8;
+      k_frag[ks] = *(const bf16x8*)&k_lds[aoff<bf16, D>(wrow + r, kb)];
+      v_frag[ks] = *(const bf16x8*)&v_lds[aoff<bf16, D>(wrow + r, kb)];
+    }
+  }
+
   f32x4 dk_acc[FO] = {}, dv_acc[FO] = {};
 
   // T14 split on the transposed q/dO stagings: tile t+1's register loads
@@ -460,7 +475,10 @@ __global__ void k_attn_bwd(const bf16* __restrict__ Q, const bf16* __restrict__ 
 
     // S^T[kv][q] = K Q^T ; P^T = exp(S^T*scale - lse[q]) — in place
     f32x4 st_acc[FN] = {};
-    mma_nt<D, FN>(k_lds, q_lds, wrow, lane, st_acc);
+    if constexpr (D == 64)
+      mma_nt_areg<D, FN>(k_frag, q_lds, lane, st_acc);
+    else
+      mma_nt<D, FN>(k_lds, q_lds, wrow, lane, st_acc);
     const int cr = (lane >> 4) * 4;
     const int cc = lane & 15;
 #pragma unroll
@@ -480,7 +498,10 @@ __global__ void k_attn_bwd(const bf16* __restrict__ Q, const bf16* __restrict__ 
 
     // dP^T[kv][q] = V dO^T
     f32x4 dpt_acc[FN] = {};
-    mma_nt<D, FN>(v_lds, dot_lds, wrow, lane, dpt_acc);
+    if constexpr (D == 64)
+      mma_nt_areg<D, FN>(v_frag, dot_lds, lane, dpt_acc);
+    else
+      mma_nt<D, FN>(v_lds, dot_lds, wrow, lane, dpt_acc);
 
     // dS^T = P^T * (dP^T - Di[q]) * scale — in place into dpt_acc
 #pragma unroll

@@ -7,6 +7,150 @@
 
 namespace tnn {
 
+
+// dual block reduce: both sums in one LDS round trip
+DEV void block_reduce_sum2(float& a, float& b, float* scratch16) {
+  const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    a += __shfl_down(a, off, 64);
+    b += __shfl_down(b, off, 64);
+  }
+  if (lane == 0) {
+    scratch16[wid] = a;
+    scratch16[8 + wid] = b;
+  }
+  __syncthreads();
+  const int nw = blockDim.x >> 6;
+  a = (threadIdx.x < nw) ? scratch16[threadIdx.x] : 0.0f;
+  b = (threadIdx.x < nw) ? scratch16[8 + threadIdx.x] : 0.0f;
+  if (wid == 0) {
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+      a += __shfl_down(a, off, 64);
+      b += __shfl_down(b, off, 64);
+    }
+  }
+}
+
+// vectorized one-block-per-row LN fwd: x held in registers across the
+// stats + write phases (the scalar form re-read global 3x and paid two
+// serial block reduces; measured 10.8us for a 6.3MB pass, ~4x roofline).
+// PK = 16B packs per thread; requires cols % V == 0, cols <= PK*256*V.
+template <typename T, int PK>
+__launch_bounds__(256)
+__global__ void k_ln_fwd_vec(const T* __restrict__ x, const float* gamma,
+                             const float* beta, T* __restrict__ y,
+                             float* __restrict__ mean,
+                             float* __restrict__ invstd, int cols, float eps) {
+  constexpr int V = 16 / sizeof(T);
+  struct alignas(16) P { T e[16 / sizeof(T)]; };
+  const int64_t row = blockIdx.x;
+  const T* xr = x + row * cols;
+  const int npk = cols / V;
+  __shared__ float scratch[16];
+  __shared__ float s_m, s_is;
+
+  P v[PK];
+  float s = 0.0f, ss = 0.0f;
+#pragma unroll
+  for (int k = 0; k < PK; ++k) {
+    const int pk = threadIdx.x + k * 256;
+    if (pk < npk) {
+      v[k] = ((const P*)xr)[pk];
+#pragma unroll
+      for (int j = 0; j < V; ++j) {
+        const float f = VecIO<T>::to_f32(v[k].e[j]);
+        s += f;
+        ss += f * f;
+      }
+    }
+  }
+  block_reduce_sum2(s, ss, scratch);
+  if (threadIdx.x == 0) {
+    const float m = s / cols;
+    s_m = m;
+    s_is = rsqrtf(fmaxf(ss / cols - m * m, 0.0f) + eps);
+    mean[row] = m;
+    invstd[row] = s_is;
+  }
+  __syncthreads();
+  const float m = s_m, is = s_is;
+  T* yr = y + row * cols;
+#pragma unroll
+  for (int k = 0; k < PK; ++k) {
+    const int pk = threadIdx.x + k * 256;
+    if (pk < npk) {
+      P o;
+#pragma unroll
+      for (int j = 0; j < V; ++j) {
+        const int c = pk * V + j;
+        const float xhat = (VecIO<T>::to_f32(v[k].e[j]) - m) * is;
+        o.e[j] = VecIO<T>::from_f32(xhat * gamma[c] + beta[c]);
+      }
+      ((P*)yr)[pk] = o;
+    }
+  }
+}
+
+template <typename T, int PK>
+__launch_bounds__(256)
+__global__ void k_ln_bwd_vec(const T* __restrict__ x, const T* __restrict__ dy,
+                             const float* gamma, const float* mean,
+                             const float* invstd, T* __restrict__ dx,
+                             int cols) {
+  constexpr int V = 16 / sizeof(T);
+  struct alignas(16) P { T e[16 / sizeof(T)]; };
+  const int64_t row = blockIdx.x;
+  const T* xr = x + row * cols;
+  const T* dyr = dy + row * cols;
+  const float m = mean[row], is = invstd[row];
+  const int npk = cols / V;
+  __shared__ float scratch[16];
+  __shared__ float s_a, s_b;
+
+  P vx[PK], vd[PK];
+  float sa = 0.0f, sb = 0.0f;
+#pragma unroll
+  for (int k = 0; k < PK; ++k) {
+    const int pk = threadIdx.x + k * 256;
+    if (pk < npk) {
+      vx[k] = ((const P*)xr)[pk];
+      vd[k] = ((const P*)dyr)[pk];
+#pragma unroll
+      for (int j = 0; j < V; ++j) {
+        const float g = VecIO<T>::to_f32(vd[k].e[j]) * gamma[pk * V + j];
+        const float xhat = (VecIO<T>::to_f32(vx[k].e[j]) - m) * is;
+        sa += g;
+        sb += g * xhat;
+      }
+    }
+  }
+  block_reduce_sum2(sa, sb, scratch);
+  if (threadIdx.x == 0) {
+    s_a = sa / cols;
+    s_b = sb / cols;
+  }
+  __syncthreads();
+  const float ma = s_a, mb = s_b;
+  T* dxr = dx + row * cols;
+#pragma unroll
+  for (int k = 0; k < PK; ++k) {
+    const int pk = threadIdx.x + k * 256;
+    if (pk < npk) {
+      P o;
+#pragma unroll
+      for (int j = 0; j < V; ++j) {
+        const float gy = VecIO<T>::to_f32(vd[k].e[j]);
+        const float xhat = (VecIO<T>::to_f32(vx[k].e[j]) - m) * is;
+        o.e[j] = VecIO<T>::from_f32(
+            is * (gy * gamma[pk * V + j] - ma - xhat * mb));
+      }
+      ((P*)dxr)[pk] = o;
+    }
+  }
+}
+
 template <typename T>
 __global__ void k_ln_fwd(const T* __restrict__ x, const float* gamma,
                          const float* beta, T* __restrict__ y,
@@ -150,6 +294,25 @@ __global__ void k_ln_bwd_col(const T* __restrict__ x, const T* __restrict__ dy,
 void ln_fwd_launch(DT dt, const void* x, const float* gamma, const float* beta,
                    void* y, float* mean, float* invstd, int64_t rows, int cols,
                    float eps, hipStream_t s) {
+  const int V = dt == DT::F32 ? 4 : 8;
+  const bool vec = cols % V == 0 && (((uintptr_t)x & 15) == 0);
+  const int npk = cols / V;
+#define LNF(T, PK)                                                           \
+  hipLaunchKernelGGL((k_ln_fwd_vec<T, PK>), dim3(rows), dim3(256), 0, s,     \
+                     (const T*)x, gamma, beta, (T*)y, mean, invstd, cols, eps)
+  if (vec && npk <= 4 * 256) {
+    if (dt == DT::F32) {
+      if (npk <= 256) LNF(float, 1);
+      else if (npk <= 512) LNF(float, 2);
+      else LNF(float, 4);
+    } else {
+      if (npk <= 256) LNF(bf16, 1);
+      else if (npk <= 512) LNF(bf16, 2);
+      else LNF(bf16, 4);
+    }
+    return;
+  }
+#undef LNF
   if (dt == DT::F32)
     hipLaunchKernelGGL(k_ln_fwd<float>, dim3(rows), dim3(256), 0, s,
                        (const float*)x, gamma, beta, (float*)y, mean, invstd,
@@ -167,7 +330,20 @@ void ln_bwd_launch(DT dt, const void* x, const void* dy, const float* gamma,
   int V = dt == DT::F32 ? 4 : 8;
   bool vec = cols % V == 0 && (((uintptr_t)x & 15) == 0) &&
              (((uintptr_t)dy & 15) == 0);
-  if (dt == DT::F32)
+  const int npk = cols / V;
+#define LNB(T, PK)                                                           \
+  hipLaunchKernelGGL((k_ln_bwd_vec<T, PK>), dim3(rows), dim3(256), 0, s,     \
+                     (const T*)x, (const T*)dy, gamma, mean, invstd, (T*)dx, \
+                     cols)
+  if (vec && npk <= 2 * 256) {  // PK<=2: x+dy packs stay in registers
+    if (dt == DT::F32) {
+      if (npk <= 256) LNB(float, 1);
+      else LNB(float, 2);
+    } else {
+      if (npk <= 256) LNB(bf16, 1);
+      else LNB(bf16, 2);
+    }
+  } else if (dt == DT::F32)
     hipLaunchKernelGGL(k_ln_bwd<float>, dim3(rows), dim3(256), 0, s,
                        (const float*)x, (const float*)dy, gamma, mean, invstd,
                        (float*)dx, vec ? nullptr : dgamma,
@@ -177,6 +353,7 @@ void ln_bwd_launch(DT dt, const void* x, const void* dy, const float* gamma,
                        (const bf16*)x, (const bf16*)dy, gamma, mean, invstd,
                        (bf16*)dx, vec ? nullptr : dgamma, vec ? nullptr : dbeta,
                        cols);
+#undef LNB
   if (vec) {
     int groups = cols / V;
     int gpb = groups < 256 ? groups : 256;
