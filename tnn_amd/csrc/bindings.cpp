@@ -837,9 +837,9 @@ at::Tensor bmm(const at::Tensor& a_in, const at::Tensor& b_in) {
   TORCH_CHECK(lda <= INT32_MAX && ldb <= INT32_MAX && M <= INT32_MAX &&
               N <= INT32_MAX && K <= INT32_MAX, "bmm dims exceed int32");
   auto c = at::empty({batch, M, N}, a.options());
-  bmm_launch(dt_of(a), a.data_ptr(), b.data_ptr(), c.data_ptr(), (int)batch,
-             (int)M, (int)N, (int)K, (int)lda, (int)ldb, (int)N, a.stride(0),
-             b.stride(0), M * N, ta, tb, cur_stream());
+  bmm_launch(dt_of(a), a.data_ptr(), b.data_ptr(), c.data_ptr(), zero_page(a),
+             (int)batch, (int)M, (int)N, (int)K, (int)lda, (int)ldb, (int)N,
+             a.stride(0), b.stride(0), M * N, ta, tb, cur_stream());
   return c;
 }
 

@@ -137,11 +137,87 @@ __global__ void k_bmm(const T* __restrict__ A, const T* __restrict__ B,
   });
 }
 
-void bmm_launch(DT dt, const void* a, const void* b, void* c, int batch, int M,
-                int N, int K, int lda, int ldb, int ldc, int64_t sa, int64_t sb,
-                int64_t sc, bool ta, bool tb, hipStream_t s) {
+// double-buffered glds variant for the NT layout (both operands' rows
+// k-contiguous — the QK^T / dO@V^T attention shapes): tile t+1's
+// LDS-DMAs overlap tile t's MFMAs (same structure as k_gemm_nt_db).
+template <typename T>
+__launch_bounds__(THREADS)
+__global__ void k_bmm_nt_db(const T* __restrict__ A, const T* __restrict__ B,
+                            T* __restrict__ C, const T* __restrict__ zero16,
+                            int M, int N, int K, int lda, int ldb, int ldc,
+                            int64_t sa, int64_t sb, int64_t sc) {
+  __shared__ alignas(16) T As[2][BM * BK];
+  __shared__ alignas(16) T Bs[2][BN * BK];
+  const T* a = A + blockIdx.z * sa;
+  const T* b = B + blockIdx.z * sb;
+  T* c = C + blockIdx.z * sc;
+  const int m0 = blockIdx.x * BM;
+  const int n0 = blockIdx.y * BN;
+  const WaveCoord wc;
+  f32x4 acc[FM][FN] = {};
+
+  auto stage = [&](int t, int which) {
+    const int kk0 = t * BK;
+    glds_stage<T, BM>(As[which], wc, [&](int rl, int kk) -> const T* {
+      int gm = m0 + rl, gk = kk0 + kk;
+      if (gm >= M || gk >= K) return zero16;
+      return &a[(int64_t)gm * lda + gk];
+    });
+    glds_stage<T, BN>(Bs[which], wc, [&](int rl, int kk) -> const T* {
+      int gn = n0 + rl, gk = kk0 + kk;
+      if (gn >= N || gk >= K) return zero16;
+      return &b[(int64_t)gn * ldb + gk];
+    });
+  };
+  constexpr int NPER = glds_count<T, BM>() + glds_count<T, BN>();
+
+  const int nch = (K + BK - 1) / BK;
+  stage(0, 0);
+  for (int t = 0; t < nch; ++t) {
+    const int cur = t & 1;
+    if (t + 1 < nch) {
+      stage(t + 1, cur ^ 1);
+      wait_vmcnt<NPER>();
+    } else {
+      wait_vmcnt<0>();
+    }
+    __builtin_amdgcn_s_barrier();
+    mfma_compute_tile(As[cur], Bs[cur], wc, acc);
+    __builtin_amdgcn_s_barrier();
+  }
+
+  epilogue_visit(wc, acc, m0, n0, [&](int row, int col, float v) {
+    if (row < M && col < N)
+      c[(int64_t)row * ldc + col] = VecIO<T>::from_f32(v);
+  });
+}
+
+void bmm_launch(DT dt, const void* a, const void* b, void* c,
+                const void* zero16, int batch, int M, int N, int K, int lda,
+                int ldb, int ldc, int64_t sa, int64_t sb, int64_t sc, bool ta,
+                bool tb, hipStream_t s) {
   dim3 grid(ceil_div(M, BM), ceil_div(N, BN), batch);
   dim3 blk(THREADS);
+  // NT fast path: rows of both operands k-contiguous and 16B-clean
+  if (!ta && tb) {
+    const int V = dt == DT::F32 ? 4 : 8;
+    const bool clean = K % V == 0 && lda % V == 0 && ldb % V == 0 &&
+                       sa % V == 0 && sb % V == 0 &&
+                       (((uintptr_t)a & 15) == 0) && (((uintptr_t)b & 15) == 0);
+    if (clean && K <= 3072) {
+      if (dt == DT::F32)
+        hipLaunchKernelGGL(k_bmm_nt_db<float>, grid, blk, 0, s,
+                           (const float*)a, (const float*)b, (float*)c,
+                           (const float*)zero16, M, N, K, lda, ldb, ldc, sa,
+                           sb, sc);
+      else
+        hipLaunchKernelGGL(k_bmm_nt_db<bf16>, grid, blk, 0, s,
+                           (const bf16*)a, (const bf16*)b, (bf16*)c,
+                           (const bf16*)zero16, M, N, K, lda, ldb, ldc, sa,
+                           sb, sc);
+      return;
+    }
+  }
 #define LAUNCH_T(T)                                                          \
   do {                                                                       \
     auto kern = ta ? (tb ? k_bmm<T, true, true> : k_bmm<T, true, false>)     \

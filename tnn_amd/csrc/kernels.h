@@ -212,9 +212,10 @@ void attn_bwd_launch(const void* q, const void* k, const void* v,
                      float scale, hipStream_t s);
 
 // ---- bmm.hip ---------------------------------------------------------------
-void bmm_launch(DT dt, const void* a, const void* b, void* c, int batch, int M,
-                int N, int K, int lda, int ldb, int ldc, int64_t sa, int64_t sb,
-                int64_t sc, bool ta, bool tb, hipStream_t s);
+void bmm_launch(DT dt, const void* a, const void* b, void* c,
+                const void* zero16, int batch, int M, int N, int K, int lda,
+                int ldb, int ldc, int64_t sa, int64_t sb, int64_t sc, bool ta,
+                bool tb, hipStream_t s);
 
 // ---- softmax.hip -----------------------------------------------------------
 void smax_fwd_launch(DT dt, const void* x, void* y, int64_t rows, int cols,
