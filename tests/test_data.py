@@ -66,3 +66,22 @@ def test_tokenizer_roundtrip(tmp_path):
     assert t2.vocab_size == 10
     ids = t2.encode("hello world")
     assert t2.decode(ids) == "hello world"
+
+
+def test_csv_loader(tmp_path):
+    """Generic CSV loader (reference wifi_data_loader.hpp shape)."""
+    import numpy as np
+    from tnn_amd.data.loaders import CSVLoader
+    p = tmp_path / "w.csv"
+    p.write_text("a,b,c,y\n" + "\n".join(
+        f"{i},{i*2},{i%3},{i*0.5}" for i in range(40)))
+    dl = CSVLoader(str(p), batch_size=8)
+    dl.load_data()
+    assert dl.x.shape == (40, 3) and dl.y.shape == (40, 1)
+    assert abs(dl.x.mean().item()) < 1e-5  # normalized
+    xb, yb = next(iter(dl))
+    assert xb.shape == (8, 3)
+    # classification mode: int64 labels, unnormalized targets
+    dl2 = CSVLoader(str(p), regression=False, target_cols=[2], batch_size=8)
+    dl2.load_data()
+    assert dl2.y.dtype.is_floating_point is False

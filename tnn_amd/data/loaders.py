@@ -192,6 +192,55 @@ class ImageNet100Loader(TinyImageNetLoader):
     image_size = 224
 
 
+class CSVLoader(BaseDataLoader):
+    """Generic CSV feature/target loader with column ranges, optional
+    header skip and z-score normalization (reference
+    include/data_loading/wifi_data_loader.hpp — the WiFi RSSI dataset is
+    one instance of this shape)."""
+
+    def __init__(self, path: str, feature_cols=None, target_cols=None,
+                 has_header: bool = True, regression: bool = True,
+                 normalize: bool = True, **kw):
+        super().__init__(**kw)
+        self.path = path
+        self.feature_cols, self.target_cols = feature_cols, target_cols
+        self.has_header, self.regression = has_header, regression
+        self.normalize = normalize
+
+    def load_data(self):
+        rows = []
+        with open(self.path) as f:
+            for i, line in enumerate(f):
+                if i == 0 and self.has_header:
+                    continue
+                cells = line.strip().split(",")
+                if cells and any(cells):
+                    rows.append([float(c) for c in cells if c != ""])
+        if not rows:
+            raise ValueError(f"no rows in {self.path}")
+        arr = np.asarray(rows, dtype=np.float32)
+        ncol = arr.shape[1]
+        fcols = (list(self.feature_cols) if self.feature_cols is not None
+                 else list(range(ncol - 1)))
+        tcols = (list(self.target_cols) if self.target_cols is not None
+                 else [ncol - 1])
+        x = arr[:, fcols]
+        t = arr[:, tcols]
+        if self.normalize:
+            mu, sd = x.mean(0), x.std(0)
+            sd[sd == 0] = 1.0
+            x = (x - mu) / sd
+            self.feature_stats = (mu, sd)
+            if self.regression:
+                tmu, tsd = t.mean(0), t.std(0)
+                tsd[tsd == 0] = 1.0
+                t = (t - tmu) / tsd
+                self.target_stats = (tmu, tsd)
+        self.x = torch.from_numpy(x)
+        self.y = (torch.from_numpy(t) if self.regression
+                  else torch.from_numpy(t[:, 0].astype(np.int64)))
+
+
 class OpenWebTextLoader(BaseDataLoader):
     """mmap'd uint16 GPT-2 token file, random windows
     (reference include/data_loading/open_webtext_data_loader.hpp:11-95).
@@ -240,6 +289,8 @@ class DataLoaderFactory:
         "cifar100": CIFAR100Loader,
         "tiny_imagenet": TinyImageNetLoader,
         "imagenet_100": ImageNet100Loader,
+        "csv": CSVLoader,
+        "wifi": CSVLoader,
         "openwebtext": OpenWebTextLoader,
     }
 
