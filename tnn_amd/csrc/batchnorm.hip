@@ -359,6 +359,70 @@ __global__ void k_bn_bwd_apply(const T* __restrict__ x, const T* __restrict__ dy
   }
 }
 
+// vectorized train-apply: V channel-contiguous elems per thread
+template <typename T>
+__global__ void k_bn_apply_vec(const T* __restrict__ x, const float* mean,
+                               const float* invstd, const float* gamma,
+                               const float* beta, T* __restrict__ y,
+                               int64_t nv, int groups, bool relu) {
+  constexpr int V = 16 / sizeof(T);
+  struct alignas(16) P { T e[16 / sizeof(T)]; };
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < nv; i += stride) {
+    const int c0 = (int)(i % groups) * V;
+    P vx = ((const P*)x)[i];
+    P o;
+#pragma unroll
+    for (int j = 0; j < V; ++j) {
+      const int c = c0 + j;
+      float v = (VecIO<T>::to_f32(vx.e[j]) - mean[c]) * invstd[c] * gamma[c] +
+                beta[c];
+      if (relu) v = fmaxf(v, 0.0f);
+      o.e[j] = VecIO<T>::from_f32(v);
+    }
+    ((P*)y)[i] = o;
+  }
+}
+
+// vectorized backward apply: V channel-contiguous elems per thread, one
+// division per vector (the scalar form's per-element i%cols + 2B loads
+// measured 2.2 TB/s on a ~5 TB/s pass)
+template <typename T>
+__global__ void k_bn_bwd_apply_vec(const T* __restrict__ x,
+                                   const T* __restrict__ dy,
+                                   const T* __restrict__ y_relu,
+                                   const float* mean, const float* invstd,
+                                   const float* gamma, const float* sum_dy,
+                                   const float* sum_dy_xhat,
+                                   T* __restrict__ dx, int64_t nv, int groups,
+                                   int64_t rows, float dy_scale) {
+  constexpr int V = 16 / sizeof(T);
+  struct alignas(16) P { T e[16 / sizeof(T)]; };
+  const float inv_n = 1.0f / (float)rows;
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < nv; i += stride) {
+    const int c0 = (int)(i % groups) * V;
+    P vx = ((const P*)x)[i];
+    P vd = ((const P*)dy)[i];
+    P vr;
+    if (y_relu) vr = ((const P*)y_relu)[i];
+    P o;
+#pragma unroll
+    for (int j = 0; j < V; ++j) {
+      const int c = c0 + j;
+      float g = VecIO<T>::to_f32(vd.e[j]) * dy_scale;
+      if (y_relu && VecIO<T>::to_f32(vr.e[j]) <= 0.0f) g = 0.0f;
+      const float xhat = (VecIO<T>::to_f32(vx.e[j]) - mean[c]) * invstd[c];
+      o.e[j] = VecIO<T>::from_f32(
+          gamma[c] * invstd[c] *
+          (g - inv_n * (sum_dy[c] + xhat * sum_dy_xhat[c])));
+    }
+    ((P*)dx)[i] = o;
+  }
+}
+
 // ---------------------------------------------------------------------------
 template <typename T>
 static inline bool bn_vec_ok(const void* x, int cols) {
@@ -441,14 +505,27 @@ void bn_apply_launch(DT dt, const void* x, const float* mean,
                      hipStream_t s) {
   int64_t n = rows * cols;
   int blocks = (int)std::min<int64_t>((n + 255) / 256, (int64_t)2048);
-  if (dt == DT::F32)
+  if (dt == DT::F32) {
+    if (bn_vec_ok<float>(x, cols)) {
+      hipLaunchKernelGGL(k_bn_apply_vec<float>, dim3(blocks), dim3(256), 0, s,
+                         (const float*)x, mean, invstd, gamma, beta, (float*)y,
+                         n / 4, cols / 4, relu);
+      return;
+    }
     hipLaunchKernelGGL(k_bn_apply<float>, dim3(blocks), dim3(256), 0, s,
                        (const float*)x, mean, invstd, gamma, beta, (float*)y, n,
                        cols, relu);
-  else
+  } else {
+    if (bn_vec_ok<bf16>(x, cols)) {
+      hipLaunchKernelGGL(k_bn_apply_vec<bf16>, dim3(blocks), dim3(256), 0, s,
+                         (const bf16*)x, mean, invstd, gamma, beta, (bf16*)y,
+                         n / 8, cols / 8, relu);
+      return;
+    }
     hipLaunchKernelGGL(k_bn_apply<bf16>, dim3(blocks), dim3(256), 0, s,
                        (const bf16*)x, mean, invstd, gamma, beta, (bf16*)y, n,
                        cols, relu);
+  }
 }
 
 void bn_apply_drop_launch(DT dt, const void* x, const float* mean,
@@ -537,16 +614,35 @@ void bn_bwd_apply_launch(DT dt, const void* x, const void* dy,
                          hipStream_t s) {
   int64_t n = rows * cols;
   int blocks = (int)std::min<int64_t>((n + 255) / 256, (int64_t)2048);
-  if (dt == DT::F32)
+  const bool vec = dt == DT::F32 ? bn_vec_ok<float>(x, cols)
+                                 : bn_vec_ok<bf16>(x, cols);
+  if (dt == DT::F32) {
+    if (vec && ((uintptr_t)dy & 15) == 0) {
+      hipLaunchKernelGGL(k_bn_bwd_apply_vec<float>, dim3(blocks), dim3(256),
+                         0, s, (const float*)x, (const float*)dy,
+                         (const float*)y_relu, mean, invstd, gamma, sum_dy,
+                         sum_dy_xhat, (float*)dx, n / 4, cols / 4, rows,
+                         dy_scale);
+      return;
+    }
     hipLaunchKernelGGL(k_bn_bwd_apply<float>, dim3(blocks), dim3(256), 0, s,
                        (const float*)x, (const float*)dy, (const float*)y_relu,
                        mean, invstd, gamma, sum_dy, sum_dy_xhat, (float*)dx, n,
                        rows, cols, dy_scale);
-  else
+  } else {
+    if (vec && ((uintptr_t)dy & 15) == 0) {
+      hipLaunchKernelGGL(k_bn_bwd_apply_vec<bf16>, dim3(blocks), dim3(256),
+                         0, s, (const bf16*)x, (const bf16*)dy,
+                         (const bf16*)y_relu, mean, invstd, gamma, sum_dy,
+                         sum_dy_xhat, (bf16*)dx, n / 8, cols / 8, rows,
+                         dy_scale);
+      return;
+    }
     hipLaunchKernelGGL(k_bn_bwd_apply<bf16>, dim3(blocks), dim3(256), 0, s,
                        (const bf16*)x, (const bf16*)dy, (const bf16*)y_relu,
                        mean, invstd, gamma, sum_dy, sum_dy_xhat, (bf16*)dx, n,
                        rows, cols, dy_scale);
+  }
 }
 
 }  // namespace tnn
