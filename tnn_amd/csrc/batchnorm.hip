@@ -103,8 +103,10 @@ __global__ void k_bn_partial_vec(const T* __restrict__ x,
     }
     int c = blockIdx.x * cpb_total + cl;
     if (c >= cols) break;  // last block may cover fewer than gpb groups
-    atomicAdd(&sum[c], ts);
-    atomicAdd(&sumsq[c], tss);
+    // per-slice slab [y][2][cols] (sum/sumsq are a contiguous {2,C}
+    // buffer); k_slab_fin folds the slices — no atomics, no zero-init
+    sum[(int64_t)blockIdx.y * 2 * cols + c] = ts;
+    sum[(int64_t)blockIdx.y * 2 * cols + cols + c] = tss;
   }
 }
 
@@ -176,8 +178,9 @@ __global__ void k_bn_bwd_reduce_vec(const T* __restrict__ x,
     }
     int c = blockIdx.x * cpb_total + cl;
     if (c >= cols) break;  // last block may cover fewer than gpb groups
-    atomicAdd(&sum_dy[c], ts);
-    atomicAdd(&sum_dy_xhat[c], tsx);
+    // per-slice slab [y][2][cols] (sum_dy/sum_dy_xhat contiguous {2,C})
+    sum_dy[(int64_t)blockIdx.y * 2 * cols + c] = ts;
+    sum_dy[(int64_t)blockIdx.y * 2 * cols + cols + c] = tsx;
   }
 }
 
@@ -457,32 +460,48 @@ static inline dim3 bn_reduce_grid(int64_t rows, int cols) {
   return dim3(cblocks, rslices);
 }
 
+int64_t bn_stats_ws_floats(DT dt, const void* x, int64_t rows, int cols) {
+  // vec path: [slices][2C] slabs + a [2C] finalize region; 0 = scalar path
+  const bool vec = dt == DT::F32 ? bn_vec_ok<float>((void*)x, cols)
+                                 : bn_vec_ok<bf16>((void*)x, cols);
+  if (!vec) return 0;
+  dim3 g = dt == DT::F32 ? bn_reduce_grid_vec<float>(rows, cols)
+                         : bn_reduce_grid_vec<bf16>(rows, cols);
+  return ((int64_t)g.y + 1) * 2 * cols;
+}
+
 void bn_stats_launch(DT dt, const void* x, float* mean, float* invstd,
-                     float* rmean, float* rvar, float momentum,
+                     float* rmean, float* rvar, float momentum, float* ws,
                      int64_t rows, int cols, float eps, hipStream_t s) {
-  // mean/invstd double as the scratch sum/sumsq buffers (finalized in place);
-  // they must be zeroed first.
+  if (ws) {
+    // vec path: per-slice slabs (plain stores) + slab finalize — no
+    // atomics, no zero-init
+    dim3 g = dt == DT::F32 ? bn_reduce_grid_vec<float>(rows, cols)
+                           : bn_reduce_grid_vec<bf16>(rows, cols);
+    float* sums = ws + (int64_t)g.y * 2 * cols;
+    if (dt == DT::F32)
+      hipLaunchKernelGGL(k_bn_partial_vec<float>, g, dim3(256), 0, s,
+                         (const float*)x, ws, nullptr, rows, cols);
+    else
+      hipLaunchKernelGGL(k_bn_partial_vec<bf16>, g, dim3(256), 0, s,
+                         (const bf16*)x, ws, nullptr, rows, cols);
+    slab_fin_launch(ws, sums, (int)g.y, 2 * cols, s);
+    hipLaunchKernelGGL(k_bn_finalize, dim3((cols + 255) / 256), dim3(256), 0,
+                       s, mean, invstd, sums, sums + cols, rmean, rvar,
+                       momentum, rows, cols, eps);
+    return;
+  }
+  // scalar fallback: mean/invstd double as zeroed atomic sum/sumsq
   hipMemsetAsync(mean, 0, cols * sizeof(float), s);
   hipMemsetAsync(invstd, 0, cols * sizeof(float), s);
-  if (dt == DT::F32) {
-    if (bn_vec_ok<float>(x, cols))
-      hipLaunchKernelGGL(k_bn_partial_vec<float>,
-                         bn_reduce_grid_vec<float>(rows, cols), dim3(256), 0, s,
-                         (const float*)x, mean, invstd, rows, cols);
-    else
-      hipLaunchKernelGGL(k_bn_partial<float>, bn_reduce_grid(rows, cols),
-                         dim3(256), 0, s, (const float*)x, mean, invstd, rows,
-                         cols);
-  } else {
-    if (bn_vec_ok<bf16>(x, cols))
-      hipLaunchKernelGGL(k_bn_partial_vec<bf16>,
-                         bn_reduce_grid_vec<bf16>(rows, cols), dim3(256), 0, s,
-                         (const bf16*)x, mean, invstd, rows, cols);
-    else
-      hipLaunchKernelGGL(k_bn_partial<bf16>, bn_reduce_grid(rows, cols),
-                         dim3(256), 0, s, (const bf16*)x, mean, invstd, rows,
-                         cols);
-  }
+  if (dt == DT::F32)
+    hipLaunchKernelGGL(k_bn_partial<float>, bn_reduce_grid(rows, cols),
+                       dim3(256), 0, s, (const float*)x, mean, invstd, rows,
+                       cols);
+  else
+    hipLaunchKernelGGL(k_bn_partial<bf16>, bn_reduce_grid(rows, cols),
+                       dim3(256), 0, s, (const bf16*)x, mean, invstd, rows,
+                       cols);
   hipLaunchKernelGGL(k_bn_finalize, dim3((cols + 255) / 256), dim3(256), 0, s,
                      mean, invstd, mean, invstd, rmean, rvar, momentum, rows,
                      cols, eps);
@@ -575,35 +594,51 @@ void bn_infer_launch(DT dt, const void* x, const float* rmean,
                        cols, eps, relu);
 }
 
+int64_t bn_bwd_ws_floats(DT dt, const void* x, const void* dy, int64_t rows,
+                         int cols) {
+  const bool vec = dt == DT::F32
+                       ? (bn_vec_ok<float>((void*)x, cols) &&
+                          bn_vec_ok<float>((void*)dy, cols))
+                       : (bn_vec_ok<bf16>((void*)x, cols) &&
+                          bn_vec_ok<bf16>((void*)dy, cols));
+  if (!vec) return 0;
+  dim3 g = dt == DT::F32 ? bn_reduce_grid_vec<float>(rows, cols)
+                         : bn_reduce_grid_vec<bf16>(rows, cols);
+  return (int64_t)g.y * 2 * cols;
+}
+
 void bn_bwd_reduce_launch(DT dt, const void* x, const void* dy,
                           const void* y_relu, const float* mean,
                           const float* invstd, float* sum_dy,
-                          float* sum_dy_xhat, int64_t rows, int cols,
-                          float dy_scale, hipStream_t s) {
-  if (dt == DT::F32) {
-    if (bn_vec_ok<float>(x, cols) && bn_vec_ok<float>(dy, cols))
-      hipLaunchKernelGGL(k_bn_bwd_reduce_vec<float>,
-                         bn_reduce_grid_vec<float>(rows, cols), dim3(256), 0, s,
+                          float* sum_dy_xhat, float* ws, int64_t rows,
+                          int cols, float dy_scale, hipStream_t s) {
+  if (ws) {
+    // vec path: slabs (plain stores) + slab finalize into the contiguous
+    // {2, C} sums buffer (sum_dy_xhat == sum_dy + C)
+    dim3 g = dt == DT::F32 ? bn_reduce_grid_vec<float>(rows, cols)
+                           : bn_reduce_grid_vec<bf16>(rows, cols);
+    if (dt == DT::F32)
+      hipLaunchKernelGGL(k_bn_bwd_reduce_vec<float>, g, dim3(256), 0, s,
                          (const float*)x, (const float*)dy,
-                         (const float*)y_relu, mean, invstd, sum_dy,
-                         sum_dy_xhat, rows, cols, dy_scale);
+                         (const float*)y_relu, mean, invstd, ws, nullptr,
+                         rows, cols, dy_scale);
     else
-      hipLaunchKernelGGL(k_bn_bwd_reduce<float>, bn_reduce_grid(rows, cols),
-                         dim3(256), 0, s, (const float*)x, (const float*)dy,
-                         (const float*)y_relu, mean, invstd, sum_dy,
-                         sum_dy_xhat, rows, cols, dy_scale);
-  } else {
-    if (bn_vec_ok<bf16>(x, cols) && bn_vec_ok<bf16>(dy, cols))
-      hipLaunchKernelGGL(k_bn_bwd_reduce_vec<bf16>,
-                         bn_reduce_grid_vec<bf16>(rows, cols), dim3(256), 0, s,
+      hipLaunchKernelGGL(k_bn_bwd_reduce_vec<bf16>, g, dim3(256), 0, s,
                          (const bf16*)x, (const bf16*)dy, (const bf16*)y_relu,
-                         mean, invstd, sum_dy, sum_dy_xhat, rows, cols, dy_scale);
-    else
-      hipLaunchKernelGGL(k_bn_bwd_reduce<bf16>, bn_reduce_grid(rows, cols),
-                         dim3(256), 0, s, (const bf16*)x, (const bf16*)dy,
-                         (const bf16*)y_relu, mean, invstd, sum_dy,
-                         sum_dy_xhat, rows, cols, dy_scale);
+                         mean, invstd, ws, nullptr, rows, cols, dy_scale);
+    slab_fin_launch(ws, sum_dy, (int)g.y, 2 * cols, s);
+    return;
   }
+  if (dt == DT::F32)
+    hipLaunchKernelGGL(k_bn_bwd_reduce<float>, bn_reduce_grid(rows, cols),
+                       dim3(256), 0, s, (const float*)x, (const float*)dy,
+                       (const float*)y_relu, mean, invstd, sum_dy,
+                       sum_dy_xhat, rows, cols, dy_scale);
+  else
+    hipLaunchKernelGGL(k_bn_bwd_reduce<bf16>, bn_reduce_grid(rows, cols),
+                       dim3(256), 0, s, (const bf16*)x, (const bf16*)dy,
+                       (const bf16*)y_relu, mean, invstd, sum_dy,
+                       sum_dy_xhat, rows, cols, dy_scale);
 }
 
 void bn_bwd_apply_launch(DT dt, const void* x, const void* dy,

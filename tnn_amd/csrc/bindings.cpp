@@ -105,9 +105,22 @@ at::Tensor dropout_bwd(const at::Tensor& dy, const at::Tensor& mask, double p) {
 at::Tensor colsum(const at::Tensor& x) {
   CHECK_IN(x);
   TORCH_CHECK(x.dim() == 2, "colsum wants [rows, cols]");
-  auto out = at::zeros({x.size(1)}, x.options().dtype(at::kFloat));
-  colsum_launch(dt_of(x), x.data_ptr(), out.data_ptr(), x.size(0), x.size(1),
-                cur_stream());
+  const int slices = colsum_ws_slices(dt_of(x), x.data_ptr(), x.size(0),
+                                      x.size(1));
+  // vec path writes per-slice slabs + a finalize (no zero-init needed);
+  // scalar fallback atomically accumulates into a zeroed output
+  auto out = slices >= 1
+                 ? at::empty({x.size(1)}, x.options().dtype(at::kFloat))
+                 : at::zeros({x.size(1)}, x.options().dtype(at::kFloat));
+  at::Tensor ws;
+  float* wp = nullptr;
+  if (slices > 1) {
+    ws = at::empty({(int64_t)slices * x.size(1)},
+                   x.options().dtype(at::kFloat));
+    wp = ws.data_ptr<float>();
+  }
+  colsum_launch(dt_of(x), x.data_ptr(), out.data_ptr(), wp, slices, x.size(0),
+                x.size(1), cur_stream());
   if (x.scalar_type() != at::kFloat) {
     auto out_t = at::empty({x.size(1)}, x.options());
     cast_f32_launch(dt_of(x), out.data_ptr<float>(), out_t.data_ptr(),
@@ -441,9 +454,16 @@ std::vector<at::Tensor> bn_fwd_train(const at::Tensor& x,
                        invstd.data_ptr<float>(), rm, rv, (float)momentum,
                        rows, C, (float)eps, cur_stream());
   } else {
+    const int64_t wsn = bn_stats_ws_floats(dt_of(x), x.data_ptr(), rows, C);
+    at::Tensor ws;
+    float* wp = nullptr;
+    if (wsn) {
+      ws = at::empty({wsn}, x.options().dtype(at::kFloat));
+      wp = ws.data_ptr<float>();
+    }
     bn_stats_launch(dt_of(x), x.data_ptr(), mean.data_ptr<float>(),
-                    invstd.data_ptr<float>(), rm, rv, (float)momentum, rows, C,
-                    (float)eps, cur_stream());
+                    invstd.data_ptr<float>(), rm, rv, (float)momentum, wp,
+                    rows, C, (float)eps, cur_stream());
   }
   if (dropout_p > 0.0)
     bn_apply_drop_launch(dt_of(x), x.data_ptr(), mean.data_ptr<float>(),
@@ -483,16 +503,26 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& x, const at::Tensor& dy,
   CHECK_IN(dy);
   int C = x.size(-1);
   int64_t rows = x.numel() / C;
-  // one zeroed slab for both channel sums: halves the per-BN fill launches
-  auto sums = at::zeros({2, C}, x.options().dtype(at::kFloat));
+  const int64_t wsn = bn_bwd_ws_floats(dt_of(x), x.data_ptr(),
+                                       dy.data_ptr(), rows, C);
+  // slab path (vec) needs no zero-init; scalar fallback atomically
+  // accumulates into zeroed sums
+  auto sums = wsn ? at::empty({2, C}, x.options().dtype(at::kFloat))
+                  : at::zeros({2, C}, x.options().dtype(at::kFloat));
   auto sum_dy = sums[0];
   auto sum_dy_xhat = sums[1];
+  at::Tensor ws;
+  float* wp = nullptr;
+  if (wsn) {
+    ws = at::empty({wsn}, x.options().dtype(at::kFloat));
+    wp = ws.data_ptr<float>();
+  }
   auto dx = at::empty_like(x);
   const void* yr = y_relu.has_value() ? y_relu->data_ptr() : nullptr;
   bn_bwd_reduce_launch(dt_of(x), x.data_ptr(), dy.data_ptr(), yr,
                        mean.data_ptr<float>(), invstd.data_ptr<float>(),
                        sum_dy.data_ptr<float>(), sum_dy_xhat.data_ptr<float>(),
-                       rows, C, (float)dy_scale, cur_stream());
+                       wp, rows, C, (float)dy_scale, cur_stream());
   bn_bwd_apply_launch(dt_of(x), x.data_ptr(), dy.data_ptr(), yr,
                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
                       gamma.data_ptr<float>(), sum_dy.data_ptr<float>(),

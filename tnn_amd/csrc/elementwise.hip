@@ -223,6 +223,36 @@ void dropout_bwd_launch(DT dt, const void* dy, const uint8_t* mask, void* dx,
 }
 
 // ---- column sum (bias gradient; reference run_bgrad_kernel_ex) -------------
+// slab finalize: sum S slabs of C2 fp32 columns (written per row-slice by
+// the column-reduce kernels below — plain stores into an at::empty ws,
+// replacing per-(block,col) atomics whose same-address serialization cost
+// a fixed ~20-50us at 512 row slices, and the at::zeros the atomics
+// needed). Block owns 32 columns; 8 thread-rows split the S range.
+__global__ void k_slab_fin(const float* __restrict__ ws,
+                           float* __restrict__ out, int S, int C2) {
+  __shared__ float sh[8][32];
+  const int cl = threadIdx.x & 31;
+  const int sl = threadIdx.x >> 5;
+  const int c = blockIdx.x * 32 + cl;
+  float a = 0.0f;
+  if (c < C2)
+    for (int s = sl; s < S; s += 8) a += ws[(int64_t)s * C2 + c];
+  sh[sl][cl] = a;
+  __syncthreads();
+  if (sl == 0 && c < C2) {
+    float t = 0.0f;
+#pragma unroll
+    for (int r = 0; r < 8; ++r) t += sh[r][cl];
+    out[c] = t;
+  }
+}
+
+void slab_fin_launch(const float* ws, float* out, int slices, int c2,
+                     hipStream_t s) {
+  hipLaunchKernelGGL(k_slab_fin, dim3((c2 + 31) / 32), dim3(256), 0, s, ws,
+                     out, slices, c2);
+}
+
 // x: [rows, cols] -> out_f32[cols]; each block owns a col-chunk x row-slice,
 // partials combined with one atomic per (block, col).
 template <typename T>
@@ -288,42 +318,48 @@ __global__ void k_colsum_vec(const T* __restrict__ x, float* __restrict__ out,
     if (c >= cols) break;  // last block may cover fewer than gpb groups
     if (gridDim.y == 1)
       out[c] = t;
-    else
-      atomicAdd(&out[c], t);
+    else  // per-slice slab (out = ws base); k_slab_fin sums the slices
+      out[(int64_t)blockIdx.y * cols + c] = t;
   }
 }
 
-void colsum_launch(DT dt, const void* x, void* out_f32, int64_t rows,
-                   int64_t cols, hipStream_t s) {
-  int vf32 = 4, vbf = 8;
-  bool vec = (((uintptr_t)x & 15) == 0) &&
-             ((dt == DT::F32 && cols % vf32 == 0) ||
-              (dt == DT::BF16 && cols % vbf == 0));
-  if (vec) {
-    int V = dt == DT::F32 ? vf32 : vbf;
+int colsum_ws_slices(DT dt, const void* x, int64_t rows, int64_t cols) {
+  int V = dt == DT::F32 ? 4 : 8;
+  bool vec = (((uintptr_t)x & 15) == 0) && cols % V == 0;
+  if (!vec) return 0;  // atomic fallback (binding must zero the output)
+  int groups = (int)(cols / V);
+  int gpb = groups < 256 ? groups : 256;
+  int cblocks = (groups + gpb - 1) / gpb;
+  int rows_per_iter = 256 / gpb;
+  int64_t iters = (rows + rows_per_iter - 1) / rows_per_iter;
+  return (int)std::min<int64_t>(std::max<int64_t>(512 / cblocks, 1),
+                                std::max<int64_t>(iters / 4, 1));
+}
+
+void colsum_launch(DT dt, const void* x, void* out_f32, float* ws,
+                   int rslices, int64_t rows, int64_t cols, hipStream_t s) {
+  if (rslices >= 1) {
+    int V = dt == DT::F32 ? 4 : 8;
     int groups = (int)(cols / V);
     int gpb = groups < 256 ? groups : 256;
     int cblocks = (groups + gpb - 1) / gpb;
-    int rows_per_iter = 256 / gpb;
-    int64_t iters = (rows + rows_per_iter - 1) / rows_per_iter;
-    // fill the chip: ~512 blocks (the old iters/32 cap left a 768-col
-    // bias-grad reduce on 64 blocks = 13us, 13x off roofline)
-    int rslices = (int)std::min<int64_t>(
-        std::max<int64_t>(512 / cblocks, 1), std::max<int64_t>(iters / 4, 1));
     dim3 grid(cblocks, rslices);
+    float* target = rslices > 1 ? ws : (float*)out_f32;
     if (dt == DT::F32)
       hipLaunchKernelGGL(k_colsum_vec<float>, grid, dim3(256), 0, s,
-                         (const float*)x, (float*)out_f32, rows, (int)cols);
+                         (const float*)x, target, rows, (int)cols);
     else
       hipLaunchKernelGGL(k_colsum_vec<bf16>, grid, dim3(256), 0, s,
-                         (const bf16*)x, (float*)out_f32, rows, (int)cols);
+                         (const bf16*)x, target, rows, (int)cols);
+    if (rslices > 1)
+      slab_fin_launch(ws, (float*)out_f32, rslices, (int)cols, s);
     return;
   }
   // fill the chip: one block per ~64-row slice, capped so atomics stay cheap
   int cblocks = (int)((cols + 63) / 64);
   int cap = std::max(1, 2048 / cblocks);
-  int rslices = (int)std::min<int64_t>((rows + 63) / 64, (int64_t)cap);
-  dim3 grid(cblocks, rslices);
+  int rsl = (int)std::min<int64_t>((rows + 63) / 64, (int64_t)cap);
+  dim3 grid(cblocks, rsl);
   if (dt == DT::F32)
     hipLaunchKernelGGL(k_colsum<float>, grid, dim3(256), 0, s, (const float*)x,
                        (float*)out_f32, rows, (int)cols);

@@ -26,8 +26,11 @@ void dropout_fwd_launch(DT dt, const void* x, void* y, uint8_t* mask,
                         const int64_t* ctr, hipStream_t s);
 void dropout_bwd_launch(DT dt, const void* dy, const uint8_t* mask, void* dx,
                         int64_t n, float p, hipStream_t s);
-void colsum_launch(DT dt, const void* x, void* out_f32, int64_t rows,
-                   int64_t cols, hipStream_t s);
+int colsum_ws_slices(DT dt, const void* x, int64_t rows, int64_t cols);
+void colsum_launch(DT dt, const void* x, void* out_f32, float* ws,
+                   int rslices, int64_t rows, int64_t cols, hipStream_t s);
+void slab_fin_launch(const float* ws, float* out, int slices, int c2,
+                     hipStream_t s);
 void cast_f32_launch(DT dt_out, const float* x, void* y, int64_t n,
                      hipStream_t s);
 
@@ -120,8 +123,9 @@ void transpose_w_launch(DT dt, const void* w, void* w_t, int KH, int KW,
                         int Cin, int Cout, hipStream_t s);
 
 // ---- batchnorm.hip ---------------------------------------------------------
+int64_t bn_stats_ws_floats(DT dt, const void* x, int64_t rows, int cols);
 void bn_stats_launch(DT dt, const void* x, float* mean, float* invstd,
-                     float* rmean, float* rvar, float momentum,
+                     float* rmean, float* rvar, float momentum, float* ws,
                      int64_t rows, int cols, float eps, hipStream_t s);
 void bn_finalize_launch(const float* sum, const float* sumsq, float* mean,
                         float* invstd, float* rmean, float* rvar,
@@ -139,9 +143,11 @@ void bn_infer_launch(DT dt, const void* x, const float* rmean,
                      const float* rvar, const float* gamma, const float* beta,
                      void* y, int64_t rows, int cols, float eps, bool relu,
                      hipStream_t s);
+int64_t bn_bwd_ws_floats(DT dt, const void* x, const void* dy, int64_t rows,
+                         int cols);
 void bn_bwd_reduce_launch(DT dt, const void* x, const void* dy, const void* y_relu,
                           const float* mean, const float* invstd, float* sum_dy,
-                          float* sum_dy_xhat, int64_t rows, int cols,
+                          float* sum_dy_xhat, float* ws, int64_t rows, int cols,
                           float dy_scale, hipStream_t s);
 void bn_bwd_apply_launch(DT dt, const void* x, const void* dy, const void* y_relu,
                          const float* mean, const float* invstd,
