@@ -228,29 +228,42 @@ void dropout_bwd_launch(DT dt, const void* dy, const uint8_t* mask, void* dx,
 // replacing per-(block,col) atomics whose same-address serialization cost
 // a fixed ~20-50us at 512 row slices, and the at::zeros the atomics
 // needed). Block owns 32 columns; 8 thread-rows split the S range.
+template <int CPB>  // columns per block; 256/CPB threads fold slices per col
 __global__ void k_slab_fin(const float* __restrict__ ws,
                            float* __restrict__ out, int S, int C2) {
-  __shared__ float sh[8][32];
-  const int cl = threadIdx.x & 31;
-  const int sl = threadIdx.x >> 5;
-  const int c = blockIdx.x * 32 + cl;
+  constexpr int RPB = 256 / CPB;
+  __shared__ float sh[256];
+  const int cl = threadIdx.x % CPB;
+  const int sl = threadIdx.x / CPB;
+  const int c = blockIdx.x * CPB + cl;
   float a = 0.0f;
   if (c < C2)
-    for (int s = sl; s < S; s += 8) a += ws[(int64_t)s * C2 + c];
-  sh[sl][cl] = a;
+    for (int s = sl; s < S; s += RPB) a += ws[(int64_t)s * C2 + c];
+  sh[sl * CPB + cl] = a;
   __syncthreads();
-  if (sl == 0 && c < C2) {
-    float t = 0.0f;
-#pragma unroll
-    for (int r = 0; r < 8; ++r) t += sh[r][cl];
-    out[c] = t;
+  // tree-fold the slice rows (RPB is a power of two)
+  for (int h = RPB / 2; h > 0; h >>= 1) {
+    if (sl < h) sh[sl * CPB + cl] += sh[(sl + h) * CPB + cl];
+    __syncthreads();
   }
+  if (sl == 0 && c < C2) out[c] = sh[cl];
 }
 
 void slab_fin_launch(const float* ws, float* out, int slices, int c2,
                      hipStream_t s) {
-  hipLaunchKernelGGL(k_slab_fin, dim3((c2 + 31) / 32), dim3(256), 0, s, ws,
-                     out, slices, c2);
+  // pick columns-per-block so ~128+ blocks launch regardless of C: the
+  // slice fold is latency-bound and needs the parallelism spread over
+  // slices, not columns
+#define SLAB_CASE(CPB)                                                       \
+  hipLaunchKernelGGL(k_slab_fin<CPB>, dim3((c2 + CPB - 1) / CPB), dim3(256), \
+                     0, s, ws, out, slices, c2)
+  if (c2 >= 128 * 32) SLAB_CASE(32);
+  else if (c2 >= 128 * 16) SLAB_CASE(16);
+  else if (c2 >= 128 * 8) SLAB_CASE(8);
+  else if (c2 >= 128 * 4) SLAB_CASE(4);
+  else if (c2 >= 128 * 2) SLAB_CASE(2);
+  else SLAB_CASE(1);
+#undef SLAB_CASE
 }
 
 // x: [rows, cols] -> out_f32[cols]; each block owns a col-chunk x row-slice,

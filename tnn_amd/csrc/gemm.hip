@@ -743,8 +743,73 @@ __global__ void k_splitk_reduce4(const float4* __restrict__ ws,
   }
 }
 
+// tall fold for small outputs with many K-slices (e.g. early-conv wgrad:
+// n4 a few hundred, z in the hundreds): the flat kernel gets <16 blocks and
+// one serial z-chain per thread. Here 256/CPB thread-rows split z per
+// column, then an LDS tree folds the rows — same shape as k_slab_fin.
+template <typename OUT, int CPB>
+__global__ void k_splitk_fold(const float4* __restrict__ ws,
+                              OUT* __restrict__ out, int z, int64_t n4) {
+  constexpr int RPB = 256 / CPB;
+  __shared__ float4 sh[256];
+  const int cl = threadIdx.x % CPB;
+  const int r = threadIdx.x / CPB;
+  const int64_t c = (int64_t)blockIdx.x * CPB + cl;
+  float4 a{0.0f, 0.0f, 0.0f, 0.0f};
+  if (c < n4)
+    for (int sl = r; sl < z; sl += RPB) {
+      float4 v = ws[(int64_t)sl * n4 + c];
+      a.x += v.x; a.y += v.y; a.z += v.z; a.w += v.w;
+    }
+  sh[r * CPB + cl] = a;
+  __syncthreads();
+  for (int h = RPB / 2; h > 0; h >>= 1) {
+    if (r < h) {
+      float4 o = sh[(r + h) * CPB + cl];
+      float4 m = sh[r * CPB + cl];
+      m.x += o.x; m.y += o.y; m.z += o.z; m.w += o.w;
+      sh[r * CPB + cl] = m;
+    }
+    __syncthreads();
+  }
+  if (r == 0 && c < n4) {
+    float4 m = sh[cl];
+    struct alignas(4 * sizeof(OUT)) O4 { OUT e[4]; } o;
+    o.e[0] = VecIO<OUT>::from_f32(m.x);
+    o.e[1] = VecIO<OUT>::from_f32(m.y);
+    o.e[2] = VecIO<OUT>::from_f32(m.z);
+    o.e[3] = VecIO<OUT>::from_f32(m.w);
+    ((O4*)out)[c] = o;
+  }
+}
+
+template <typename OUT>
+static void splitk_fold_launch(const float4* ws, OUT* out, int z, int64_t n4,
+                               hipStream_t s) {
+#define FOLD_CASE(CPB)                                                    \
+  hipLaunchKernelGGL((k_splitk_fold<OUT, CPB>),                           \
+                     dim3((unsigned)((n4 + CPB - 1) / CPB)), dim3(256), 0, \
+                     s, ws, out, z, n4)
+  if (n4 >= 128 * 32) FOLD_CASE(32);
+  else if (n4 >= 128 * 16) FOLD_CASE(16);
+  else if (n4 >= 128 * 8) FOLD_CASE(8);
+  else if (n4 >= 128 * 4) FOLD_CASE(4);
+  else if (n4 >= 128 * 2) FOLD_CASE(2);
+  else FOLD_CASE(1);
+#undef FOLD_CASE
+}
+
 void splitk_reduce_launch(const float* ws, void* out, DT out_dt, int z,
                           int64_t n, hipStream_t s) {
+  if ((n & 3) == 0 && n < 4 * 32768 && z >= 8) {
+    // small output x deep z: parallelize over z (see k_splitk_fold)
+    int64_t n4 = n >> 2;
+    if (out_dt == DT::F32)
+      splitk_fold_launch((const float4*)ws, (float*)out, z, n4, s);
+    else
+      splitk_fold_launch((const float4*)ws, (bf16*)out, z, n4, s);
+    return;
+  }
   if ((n & 3) == 0) {
     int64_t n4 = n >> 2;
     int blocks = (int)std::min<int64_t>((n4 + 255) / 256, (int64_t)2048);
