@@ -58,7 +58,7 @@ DEV int aoff(int row, int col) {
 // tile::glds_stage; see tile_gemm.h). NW = waves in the block.
 template <int ROWS, int D, int NW>
 DEV void stage_tile(const bf16* __restrict__ g, bf16* lds, int nrows,
-                    const bf16* __restrict__ zero16) {
+                    const bf16* __restrict__ zero16, int64_t rs = D) {
   constexpr int RB = D * 2;            // bytes per image row
   constexpr int RPK = 1024 / RB;       // rows per 1 KiB region
   constexpr int LPR = RB / 16;         // lanes per row
@@ -73,7 +73,7 @@ DEV void stage_tile(const bf16* __restrict__ g, bf16* lds, int nrows,
     const int rl = RPK * j + lane / LPR;
     const int byte_in_row = (lane % LPR) * 16;
     const int col = (byte_in_row ^ ((((rl >> 3) ^ rl) & 7) << 4)) / 2;
-    const bf16* src = rl < nrows ? &g[rl * D + col] : zero16;
+    const bf16* src = rl < nrows ? &g[rl * rs + col] : zero16;
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) unsigned int*)src,
         (__attribute__((address_space(3))) unsigned int*)&lds[j * 512],
@@ -94,7 +94,7 @@ struct TStage {
   static constexpr int PER = (CH + NT - 1) / NT;
   Pack16<bf16> v[PER][4];
 
-  DEV void load(const bf16* __restrict__ g, int nrows) {
+  DEV void load(const bf16* __restrict__ g, int nrows, int64_t rs = D) {
 #pragma unroll
     for (int i = 0; i < PER; ++i) {
       const int c = threadIdx.x + i * NT;
@@ -104,7 +104,7 @@ struct TStage {
 #pragma unroll
       for (int q = 0; q < 4; ++q) {
         if (kv0 + q < nrows)
-          v[i][q] = *(const Pack16<bf16>*)&g[(kv0 + q) * D + d0];
+          v[i][q] = *(const Pack16<bf16>*)&g[(kv0 + q) * rs + d0];
         else
           v[i][q] = {};
       }
@@ -131,9 +131,10 @@ struct TStage {
 };
 
 template <int ROWS, int D, int NT>
-DEV void stage_tile_t(const bf16* __restrict__ g, bf16* lds, int nrows) {
+DEV void stage_tile_t(const bf16* __restrict__ g, bf16* lds, int nrows,
+                      int64_t rs = D) {
   TStage<ROWS, D, NT> st;
-  st.load(g, nrows);
+  st.load(g, nrows, rs);
   st.write(lds);
 }
 
@@ -226,7 +227,9 @@ __launch_bounds__(NW * 64)
 __global__ void k_attn_fwd(const bf16* __restrict__ Q, const bf16* __restrict__ K,
                            const bf16* __restrict__ V, bf16* __restrict__ O,
                            float* __restrict__ LSE,
-                           const bf16* __restrict__ zero16, int S,
+                           const bf16* __restrict__ zero16, int S, int H,
+                           int64_t i_rs, int64_t i_hs, int64_t i_bs,
+                           int64_t o_rs, int64_t o_hs, int64_t o_bs,
                            float scale) {
   constexpr int BQ = NW * 16;    // q rows per block
   constexpr int THREADS = NW * 64;
@@ -249,14 +252,18 @@ __global__ void k_attn_fwd(const bf16* __restrict__ Q, const bf16* __restrict__ 
 
   const int q0 = blockIdx.x * BQ;
   const int64_t bh = blockIdx.y;
-  const bf16* q = Q + bh * S * D;
-  const bf16* k = K + bh * S * D;
-  const bf16* v = V + bh * S * D;
+  // q/k/v share one stride tuple (BHSD-contiguous, a BSHD view, or three
+  // slices of a merged [B,S,3*H*D] QKV buffer -- bases differ, strides
+  // don't); rows stay d-contiguous in every layout
+  const int64_t ibase = (bh / H) * i_bs + (bh % H) * i_hs;
+  const bf16* q = Q + ibase;
+  const bf16* k = K + ibase;
+  const bf16* v = V + ibase;
   const int wid = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int wrow = wid * 16;                 // wave's q-row offset in tile
 
-  stage_tile<BQ, D, NW>(q + (int64_t)q0 * D, q_lds, S - q0, zero16);
+  stage_tile<BQ, D, NW>(q + (int64_t)q0 * i_rs, q_lds, S - q0, zero16, i_rs);
   // constant ones/zero tail rows of both V^T buffers (written once)
   for (int i = threadIdx.x; i < 16 * BKV; i += THREADS) {
     const int rr = i / BKV, col = i % BKV;
@@ -287,8 +294,8 @@ __global__ void k_attn_fwd(const bf16* __restrict__ Q, const bf16* __restrict__ 
   TStage<BKV, D, THREADS> vst;
 
   // prologue: stage tile 0 into buffer 0
-  stage_tile<BKV, D, NW>(k, k_lds[0], S, zero16);
-  vst.load(v, S);
+  stage_tile<BKV, D, NW>(k, k_lds[0], S, zero16, i_rs);
+  vst.load(v, S, i_rs);
   vst.write(vt_lds[0]);
   __syncthreads();
 
@@ -299,9 +306,9 @@ __global__ void k_attn_fwd(const bf16* __restrict__ Q, const bf16* __restrict__ 
       // issue tile t+1's K LDS-DMAs and V loads; they stay in flight
       // under this tile's compute (no ordinary-load USE until the
       // write at the bottom, so hipcc keeps them outstanding)
-      stage_tile<BKV, D, NW>(k + (int64_t)(kv0 + BKV) * D, k_lds[cur ^ 1],
-                             S - kv0 - BKV, zero16);
-      vst.load(v + (int64_t)(kv0 + BKV) * D, S - kv0 - BKV);
+      stage_tile<BKV, D, NW>(k + (int64_t)(kv0 + BKV) * i_rs, k_lds[cur ^ 1],
+                             S - kv0 - BKV, zero16, i_rs);
+      vst.load(v + (int64_t)(kv0 + BKV) * i_rs, S - kv0 - BKV, i_rs);
     }
 
     f32x4 s_acc[FN] = {};
@@ -367,8 +374,8 @@ __global__ void k_attn_fwd(const bf16* __restrict__ Q, const bf16* __restrict__ 
     float inv_l = l > 0.0f ? 1.0f / l : 0.0f;
 #pragma unroll
     for (int fo = 0; fo < FO; ++fo)
-      O[bh * S * D + (int64_t)grow * D + fo * 16 + cc] =
-          f2bf(o_acc[fo][j] * inv_l);
+      O[(bh / H) * o_bs + (bh % H) * o_hs + (int64_t)grow * o_rs + fo * 16 +
+        cc] = f2bf(o_acc[fo][j] * inv_l);
     if (cc == 0)
       LSE[bh * S + grow] = l > 0.0f ? m_run[j] + __logf(l) : -INFINITY;
   }
@@ -378,13 +385,17 @@ __global__ void k_attn_fwd(const bf16* __restrict__ Q, const bf16* __restrict__ 
 // Di = rowsum(dO * O) per (b,h,row)
 __global__ void k_attn_dot(const bf16* __restrict__ dO,
                            const bf16* __restrict__ O, float* __restrict__ Di,
-                           int64_t rows, int D) {
+                           int64_t rows, int S, int H, int D,
+                           int64_t do_rs, int64_t do_hs, int64_t do_bs,
+                           int64_t o_rs, int64_t o_hs, int64_t o_bs) {
   int64_t row = (int64_t)blockIdx.x * blockDim.x / 64 + (threadIdx.x >> 6);
   if (row >= rows) return;
   const int lane = threadIdx.x & 63;
+  const int64_t bh = row / S, sr = row % S;
+  const bf16* dop = dO + (bh / H) * do_bs + (bh % H) * do_hs + sr * do_rs;
+  const bf16* op = O + (bh / H) * o_bs + (bh % H) * o_hs + sr * o_rs;
   float acc = 0.0f;
-  for (int d = lane; d < D; d += 64)
-    acc += bf2f(dO[row * D + d]) * bf2f(O[row * D + d]);
+  for (int d = lane; d < D; d += 64) acc += bf2f(dop[d]) * bf2f(op[d]);
   acc = wave_reduce_sum(acc);
   if (lane == 0) Di[row] = acc;
 }
@@ -398,7 +409,10 @@ __global__ void k_attn_bwd(const bf16* __restrict__ Q, const bf16* __restrict__ 
                            const float* __restrict__ Di,
                            float* __restrict__ dQw, bf16* __restrict__ dK,
                            bf16* __restrict__ dV,
-                           const bf16* __restrict__ zero16, int S,
+                           const bf16* __restrict__ zero16, int S, int H,
+                           int64_t i_rs, int64_t i_hs, int64_t i_bs,
+                           int64_t do_rs, int64_t do_hs, int64_t do_bs,
+                           int64_t w_rs, int64_t w_hs, int64_t w_bs,
                            float scale) {
   constexpr int FN = BWD_BQ / 16;  // 4 q-col fragments
   constexpr int FO = D / 16;
@@ -417,17 +431,25 @@ __global__ void k_attn_bwd(const bf16* __restrict__ Q, const bf16* __restrict__ 
 
   const int kv0 = blockIdx.x * BKV;
   const int64_t bh = blockIdx.y;
-  const bf16* q = Q + bh * S * D;
-  const bf16* k = K + bh * S * D;
-  const bf16* v = V + bh * S * D;
-  const bf16* dout = dO + bh * S * D;
+  const int64_t ibase = (bh / H) * i_bs + (bh % H) * i_hs;
+  const bf16* q = Q + ibase;
+  const bf16* k = K + ibase;
+  const bf16* v = V + ibase;
+  const bf16* dout = dO + (bh / H) * do_bs + (bh % H) * do_hs;
+  // dK/dV land strided (w_* tuple: BHSD-dense, a BSHD buffer, or slices
+  // of one merged [B,S,3*H*D] dQKV buffer)
+  bf16* dkp = dK + (bh / H) * w_bs + (bh % H) * w_hs;
+  bf16* dvp = dV + (bh / H) * w_bs + (bh % H) * w_hs;
   const int wid = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int wrow = wid * 16;                 // wave's kv-row offset
 
-  stage_tile<BKV, D, NW>(k + (int64_t)kv0 * D, k_lds, S - kv0, zero16);
-  stage_tile<BKV, D, NW>(v + (int64_t)kv0 * D, v_lds, S - kv0, zero16);
-  stage_tile_t<BKV, D, BWD_THREADS>(k + (int64_t)kv0 * D, kt_lds, S - kv0);
+  stage_tile<BKV, D, NW>(k + (int64_t)kv0 * i_rs, k_lds, S - kv0, zero16,
+                         i_rs);
+  stage_tile<BKV, D, NW>(v + (int64_t)kv0 * i_rs, v_lds, S - kv0, zero16,
+                         i_rs);
+  stage_tile_t<BKV, D, BWD_THREADS>(k + (int64_t)kv0 * i_rs, kt_lds, S - kv0,
+                                    i_rs);
 
   // hoist this wave's K/V MFMA A-fragments into registers for the whole
   // q-tile loop (re-read from LDS every tile otherwise). D==64 only:
@@ -453,13 +475,15 @@ __global__ void k_attn_bwd(const bf16* __restrict__ Q, const bf16* __restrict__ 
   TStage<BWD_BQ, D, BWD_THREADS> qtst, dotst;
   const int q_start = CAUSAL ? (kv0 / BWD_BQ) * BWD_BQ : 0;
   if (q_start < S) {
-    qtst.load(q + (int64_t)q_start * D, S - q_start);
-    dotst.load(dout + (int64_t)q_start * D, S - q_start);
+    qtst.load(q + (int64_t)q_start * i_rs, S - q_start, i_rs);
+    dotst.load(dout + (int64_t)q_start * do_rs, S - q_start, do_rs);
   }
   for (int qt = q_start; qt < S; qt += BWD_BQ) {
     __syncthreads();
-    stage_tile<BWD_BQ, D, NW>(q + (int64_t)qt * D, q_lds, S - qt, zero16);
-    stage_tile<BWD_BQ, D, NW>(dout + (int64_t)qt * D, dot_lds, S - qt, zero16);
+    stage_tile<BWD_BQ, D, NW>(q + (int64_t)qt * i_rs, q_lds, S - qt, zero16,
+                              i_rs);
+    stage_tile<BWD_BQ, D, NW>(dout + (int64_t)qt * do_rs, dot_lds, S - qt,
+                              zero16, do_rs);
     qtst.write(qt_lds);
     dotst.write(dott_lds);
     for (int i = threadIdx.x; i < BWD_BQ; i += BWD_THREADS) {
@@ -469,8 +493,9 @@ __global__ void k_attn_bwd(const bf16* __restrict__ Q, const bf16* __restrict__ 
     }
     __syncthreads();
     if (qt + BWD_BQ < S) {
-      qtst.load(q + (int64_t)(qt + BWD_BQ) * D, S - qt - BWD_BQ);
-      dotst.load(dout + (int64_t)(qt + BWD_BQ) * D, S - qt - BWD_BQ);
+      qtst.load(q + (int64_t)(qt + BWD_BQ) * i_rs, S - qt - BWD_BQ, i_rs);
+      dotst.load(dout + (int64_t)(qt + BWD_BQ) * do_rs, S - qt - BWD_BQ,
+                 do_rs);
     }
 
     // S^T[kv][q] = K Q^T ; P^T = exp(S^T*scale - lse[q]) — in place
@@ -553,8 +578,8 @@ __global__ void k_attn_bwd(const bf16* __restrict__ Q, const bf16* __restrict__ 
     if (grow >= S) continue;
 #pragma unroll
     for (int fo = 0; fo < FO; ++fo) {
-      dK[bh * S * D + (int64_t)grow * D + fo * 16 + cc] = f2bf(dk_acc[fo][j]);
-      dV[bh * S * D + (int64_t)grow * D + fo * 16 + cc] = f2bf(dv_acc[fo][j]);
+      dkp[(int64_t)grow * w_rs + fo * 16 + cc] = f2bf(dk_acc[fo][j]);
+      dvp[(int64_t)grow * w_rs + fo * 16 + cc] = f2bf(dv_acc[fo][j]);
     }
   }
 }
@@ -567,7 +592,8 @@ namespace tnn {
 using namespace attn;
 
 void attn_fwd_launch(const void* q, const void* k, const void* v, void* o,
-                     float* lse, const void* zero16, int BH, int S, int D,
+                     float* lse, const void* zero16, int B, int H, int S,
+                     int D, const int64_t* i_str, const int64_t* o_str,
                      bool causal, float scale, hipStream_t s) {
   // wave-count selection: TNN_ATTN_WAVES overrides (4 or 8); default 8
   // (with the double-buffered staging + MFMA-carried softmax sum the
@@ -578,10 +604,11 @@ void attn_fwd_launch(const void* q, const void* k, const void* v, void* o,
   }();
 #define LAUNCH(DD, CC, NWV)                                                  \
   hipLaunchKernelGGL((k_attn_fwd<DD, CC, NWV>),                              \
-                     dim3((S + NWV * 16 - 1) / (NWV * 16), BH),              \
+                     dim3((S + NWV * 16 - 1) / (NWV * 16), B * H),           \
                      dim3(NWV * 64), 0, s, (const bf16*)q, (const bf16*)k,   \
                      (const bf16*)v, (bf16*)o, lse, (const bf16*)zero16, S,  \
-                     scale)
+                     H, i_str[0], i_str[1], i_str[2], o_str[0], o_str[1],    \
+                     o_str[2], scale)
 #define PICK(DD, CC)                                                         \
   do {                                                                       \
     if (nw == 8 && S >= 128) LAUNCH(DD, CC, 8); /* short seqs (ViT S=65) */  \
@@ -593,23 +620,49 @@ void attn_fwd_launch(const void* q, const void* k, const void* v, void* o,
 #undef LAUNCH
 }
 
+// dense fp32 [B,H,S,D] workspace -> strided bf16 dq (w_* tuple)
+__global__ void k_cast_dq(const float* __restrict__ ws, bf16* __restrict__ dq,
+                          int64_t n, int S, int H, int D, int64_t w_rs,
+                          int64_t w_hs, int64_t w_bs) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    const int d = (int)(i % D);
+    const int64_t r = i / D;
+    const int sr = (int)(r % S);
+    const int64_t bh = r / S;
+    dq[(bh / H) * w_bs + (bh % H) * w_hs + (int64_t)sr * w_rs + d] =
+        f2bf(ws[i]);
+  }
+}
+
 void attn_bwd_launch(const void* q, const void* k, const void* v,
                      const void* o, const void* dout, const float* lse,
-                     float* di, float* dq_ws, void* dk, void* dv,
-                     const void* zero16, int BH, int S, int D, bool causal,
+                     float* di, float* dq_ws, void* dq, void* dk, void* dv,
+                     const void* zero16, int B, int H, int S, int D,
+                     const int64_t* i_str, const int64_t* o_str,
+                     const int64_t* do_str, const int64_t* w_str, bool causal,
                      float scale, hipStream_t s) {
-  int64_t rows = (int64_t)BH * S;
+  int64_t rows = (int64_t)B * H * S;
   hipLaunchKernelGGL(k_attn_dot, dim3((rows * 64 + 255) / 256), dim3(256), 0,
-                     s, (const bf16*)dout, (const bf16*)o, di, rows, D);
-  dim3 grid((S + BKV - 1) / BKV, BH);
+                     s, (const bf16*)dout, (const bf16*)o, di, rows, S, H, D,
+                     do_str[0], do_str[1], do_str[2], o_str[0], o_str[1],
+                     o_str[2]);
+  dim3 grid((S + BKV - 1) / BKV, B * H);
 #define LAUNCH(DD, CC)                                                     \
   hipLaunchKernelGGL((k_attn_bwd<DD, CC>), grid, dim3(BWD_THREADS), 0, s,  \
                      (const bf16*)q, (const bf16*)k, (const bf16*)v,       \
                      (const bf16*)dout, lse, di, dq_ws, (bf16*)dk,         \
-                     (bf16*)dv, (const bf16*)zero16, S, scale)
+                     (bf16*)dv, (const bf16*)zero16, S, H, i_str[0],       \
+                     i_str[1], i_str[2], do_str[0], do_str[1], do_str[2],  \
+                     w_str[0], w_str[1], w_str[2], scale)
   if (D == 64) { if (causal) LAUNCH(64, true); else LAUNCH(64, false); }
   else if (D == 128) { if (causal) LAUNCH(128, true); else LAUNCH(128, false); }
 #undef LAUNCH
+  const int64_t n = (int64_t)B * H * S * D;
+  int blocks = (int)std::min<int64_t>((n + 255) / 256, (int64_t)4096);
+  hipLaunchKernelGGL(k_cast_dq, dim3(blocks), dim3(256), 0, s, dq_ws,
+                     (bf16*)dq, n, S, H, D, w_str[0], w_str[1], w_str[2]);
 }
 
 }  // namespace tnn

@@ -4,6 +4,7 @@
 // pointers + shapes + the stream. Every entry point CHECKs device/layout —
 // a CPU tensor reaching these is a dispatch bug in the python layer.
 
+#include <array>
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 
@@ -758,41 +759,89 @@ at::Tensor embedding_bwd(const at::Tensor& ids, const at::Tensor& dy,
 }
 
 // ---- flash attention -------------------------------------------------------
-std::vector<at::Tensor> attn_fwd(const at::Tensor& q, const at::Tensor& k,
-                                 const at::Tensor& v, bool causal) {
-  CHECK_IN(q); CHECK_IN(k); CHECK_IN(v);
+// q/k/v/o/dout are [B,H,S,D] views that may be non-contiguous as long as
+// d stays contiguous: BHSD-dense tensors, transposed views of BSHD
+// buffers, or head-slices of one merged [B,S,3*H*D] QKV buffer. The
+// kernels read rows through {row, head, batch} stride tuples, so no
+// transpose/contiguous copies happen anywhere in the attention path.
+static bool attn_view_ok(const at::Tensor& t, bool need_align) {
+  if (t.stride(3) != 1) return false;
+  if (!need_align) return true;  // scalar reads/writes: any row alignment
+  // glds stages 16-B vectors: rows must stay 16-B aligned across s and h
+  return t.stride(2) % 8 == 0 && t.stride(1) % 8 == 0 &&
+         ((uintptr_t)t.data_ptr() & 15) == 0;
+}
+
+static std::array<int64_t, 3> attn_strides(const at::Tensor& t) {
+  return {t.stride(2), t.stride(1), t.stride(0)};
+}
+
+std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
+                                 bool causal) {
+  TORCH_CHECK(q.is_cuda(), "q must be on GPU");
   TORCH_CHECK(q.scalar_type() == at::kBFloat16, "flash attention is bf16");
   TORCH_CHECK(q.dim() == 4, "q must be [B,H,S,D]");
   int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
   TORCH_CHECK(D == 64 || D == 128, "head dim must be 64 or 128, got ", D);
-  auto o = at::empty_like(q);
+  const bool strided_ok = attn_view_ok(q, true) && attn_view_ok(k, true) &&
+                          attn_view_ok(v, true) &&
+                          q.strides() == k.strides() &&
+                          k.strides() == v.strides();
+  if (!strided_ok) {
+    q = q.contiguous(); k = k.contiguous(); v = v.contiguous();
+  }
+  // o lands BSHD-contiguous so the caller's transpose+reshape to
+  // [B,S,H*D] is a free view
+  auto o_buf = at::empty({B, S, (int64_t)H * D}, q.options());
+  auto o = o_buf.view({B, S, H, D}).permute({0, 2, 1, 3});
   auto lse = at::empty({B, H, S}, q.options().dtype(at::kFloat));
   float scale = 1.0f / std::sqrt((float)D);
-  attn_fwd_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
-                  lse.data_ptr<float>(), zero_page(q), B * H, S, D, causal,
+  attn_fwd_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(), o_buf.data_ptr(),
+                  lse.data_ptr<float>(), zero_page(q), B, H, S, D,
+                  attn_strides(q).data(), attn_strides(o).data(), causal,
                   scale, cur_stream());
   return {o, lse};
 }
 
-std::vector<at::Tensor> attn_bwd(const at::Tensor& q, const at::Tensor& k,
-                                 const at::Tensor& v, const at::Tensor& o,
-                                 const at::Tensor& dout, const at::Tensor& lse,
-                                 bool causal) {
-  CHECK_IN(q); CHECK_IN(dout);
+std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
+                                 const at::Tensor& o, at::Tensor dout,
+                                 const at::Tensor& lse, bool causal,
+                                 c10::optional<at::Tensor> dq_out,
+                                 c10::optional<at::Tensor> dk_out,
+                                 c10::optional<at::Tensor> dv_out) {
+  TORCH_CHECK(q.is_cuda() && dout.is_cuda(), "attn_bwd wants GPU tensors");
   int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
+  const bool in_ok = attn_view_ok(q, true) && attn_view_ok(k, true) &&
+                     attn_view_ok(v, true) && q.strides() == k.strides() &&
+                     k.strides() == v.strides();
+  if (!in_ok) { q = q.contiguous(); k = k.contiguous(); v = v.contiguous(); }
+  if (!attn_view_ok(dout, true)) dout = dout.contiguous();
+  TORCH_CHECK(attn_view_ok(o, false), "o must be d-contiguous");
+  at::Tensor dq, dk, dv;
+  if (dq_out.has_value()) {
+    // caller-provided grad views (e.g. three slices of one merged dQKV
+    // buffer: the fused-QKV backward writes it in place, no cat/pad)
+    dq = *dq_out; dk = *dk_out; dv = *dv_out;
+    TORCH_CHECK(dq.strides() == dk.strides() && dk.strides() == dv.strides()
+                    && dq.stride(3) == 1,
+                "grad views must share a d-contiguous stride tuple");
+  } else {
+    auto mk = [&]() {
+      return at::empty({B, S, (int64_t)H * D}, q.options())
+          .view({B, S, H, D}).permute({0, 2, 1, 3});
+    };
+    dq = mk(); dk = mk(); dv = mk();
+  }
   auto dqw = at::zeros({B, H, S, D}, q.options().dtype(at::kFloat));
   auto di = at::empty({B, H, S}, q.options().dtype(at::kFloat));
-  auto dk = at::empty_like(k);
-  auto dv = at::empty_like(v);
   float scale = 1.0f / std::sqrt((float)D);
   attn_bwd_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
-                  dout.contiguous().data_ptr(), lse.data_ptr<float>(),
-                  di.data_ptr<float>(), dqw.data_ptr<float>(), dk.data_ptr(),
-                  dv.data_ptr(), zero_page(q), B * H, S, D, causal, scale,
-                  cur_stream());
-  auto dq = at::empty_like(q);
-  cast_f32_launch(DT::BF16, dqw.data_ptr<float>(), dq.data_ptr(), dq.numel(),
-                  cur_stream());
+                  dout.data_ptr(), lse.data_ptr<float>(), di.data_ptr<float>(),
+                  dqw.data_ptr<float>(), dq.data_ptr(), dk.data_ptr(),
+                  dv.data_ptr(), zero_page(q), B, H, S, D,
+                  attn_strides(q).data(), attn_strides(o).data(),
+                  attn_strides(dout).data(), attn_strides(dq).data(), causal,
+                  scale, cur_stream());
   return {dq, dk, dv};
 }
 
@@ -982,7 +1031,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("embedding_fwd", &tnn::embedding_fwd);
   m.def("embedding_bwd", &tnn::embedding_bwd);
   m.def("attn_fwd", &tnn::attn_fwd);
-  m.def("attn_bwd", &tnn::attn_bwd);
+  m.def("attn_bwd", &tnn::attn_bwd, py::arg("q"), py::arg("k"),
+        py::arg("v"), py::arg("o"), py::arg("dout"), py::arg("lse"),
+        py::arg("causal"), py::arg("dq_out") = py::none(),
+        py::arg("dk_out") = py::none(), py::arg("dv_out") = py::none());
   m.def("bmm", &tnn::bmm);
   m.def("smax_fwd", &tnn::smax_fwd);
   m.def("smax_bwd", &tnn::smax_bwd);

@@ -208,13 +208,18 @@ void embedding_bwd_launch(DT dt, const int64_t* ids, const void* dy,
                           hipStream_t s);
 
 // ---- attention.hip (bf16 only; D in {64,128}) ------------------------------
+// i_str/o_str/do_str/w_str are {row(S), head, batch} element strides of
+// the [B,H,S,D]-shaped (possibly non-contiguous, d-contiguous) views
 void attn_fwd_launch(const void* q, const void* k, const void* v, void* o,
-                     float* lse, const void* zero16, int BH, int S, int D,
+                     float* lse, const void* zero16, int B, int H, int S,
+                     int D, const int64_t* i_str, const int64_t* o_str,
                      bool causal, float scale, hipStream_t s);
 void attn_bwd_launch(const void* q, const void* k, const void* v,
                      const void* o, const void* dout, const float* lse,
-                     float* di, float* dq_ws, void* dk, void* dv,
-                     const void* zero16, int BH, int S, int D, bool causal,
+                     float* di, float* dq_ws, void* dq, void* dk, void* dv,
+                     const void* zero16, int B, int H, int S, int D,
+                     const int64_t* i_str, const int64_t* o_str,
+                     const int64_t* do_str, const int64_t* w_str, bool causal,
                      float scale, hipStream_t s);
 
 // ---- bmm.hip ---------------------------------------------------------------

@@ -376,8 +376,7 @@ class FlashAttentionBlock(_MHABase):
         if self._kv_cache is not None:
             o = self._cached_attention(q, k, v)
             return self._merge(o, b, s, residual)
-        o = ops.attention(q.contiguous(), k.contiguous(), v.contiguous(),
-                          causal=self.causal)
+        o = ops.attention(q, k, v, causal=self.causal)
         return self._merge(o, b, s, residual)
 
 

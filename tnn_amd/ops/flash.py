@@ -14,6 +14,12 @@ from .. import _C
 
 
 class _FlashAttention(torch.autograd.Function):
+    """q/k/v may be non-contiguous [B,H,S,D] views (transposed BSHD
+    projections); the kernels read them through stride tuples and o and
+    the grads come back as transposed views of BSHD buffers, so the whole
+    attention round trip does zero layout copies (the round-2 profile
+    showed ~8 activation-sized copies per layer per step without this)."""
+
     @staticmethod
     def forward(ctx, q, k, v, causal):
         ext = _C.ext()
@@ -26,7 +32,7 @@ class _FlashAttention(torch.autograd.Function):
     def backward(ctx, do):
         q, k, v, o, lse = ctx.saved_tensors
         ext = _C.ext()
-        dq, dk, dv = ext.attn_bwd(q, k, v, o, do.contiguous(), lse, ctx.causal)
+        dq, dk, dv = ext.attn_bwd(q, k, v, o, do, lse, ctx.causal)
         return dq, dk, dv, None
 
 
@@ -45,6 +51,5 @@ def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                     causal: bool = True) -> torch.Tensor:
     if (q.dtype == torch.bfloat16 and q.shape[-1] in (64, 128)
             and q.is_cuda):
-        return _FlashAttention.apply(q.contiguous(), k.contiguous(),
-                                     v.contiguous(), causal)
+        return _FlashAttention.apply(q, k, v, causal)
     return _composed(q, k, v, causal)

@@ -179,6 +179,9 @@ class _BatchNormAct(torch.autograd.Function):
         ctx.save_for_backward(x, gamma, mean, invstd, save_y)
         ctx.relu = relu or dropout_p > 0.0
         ctx.dy_scale = 1.0 / (1.0 - dropout_p) if dropout_p > 0.0 else 1.0
+        # without this autograd materializes zero grads for mean/invstd on
+        # every backward (2 fill launches per BN per step)
+        ctx.mark_non_differentiable(mean, invstd)
         return y, mean, invstd
 
     @staticmethod
