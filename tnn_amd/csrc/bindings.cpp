@@ -198,6 +198,34 @@ at::Tensor gemm(const at::Tensor& a, const at::Tensor& b,
     auto bt = at::empty({(int64_t)N, (int64_t)K}, b.options());
     transpose_w_launch(dt_of(b), b.data_ptr(), bt.data_ptr(), 1, 1, K, N,
                        cur_stream());
+    static const bool use32 = [] {
+      const char* e = getenv("TNN_GEMM32");
+      return !(e && e[0] == '0');
+    }();
+    if (use32 && dt_of(a) == DT::BF16) {
+      // 32x32x16-MFMA tile core (peak issue at this kernel's 2
+      // waves/SIMD residency; the 16x16x32 core needs 3-4)
+      int z = N % 4 == 0 ? gemm_nt32_zsplits(M, N, K) : 1;
+      at::Tensor ws;
+      float* wp = nullptr;
+      if (z > 1) {
+        ws = at::empty({(int64_t)z * M * N}, a.options().dtype(at::kFloat));
+        wp = ws.data_ptr<float>();
+      }
+      gemm_nt32_launch(a.data_ptr(), bt.data_ptr(), wp, nullptr, bp, rp,
+                       c.data_ptr(), z, zero_page(a), M, N, K, (int)act_kind,
+                       cur_stream());
+      return c;
+    }
+    const int zf = N % 4 == 0 ? gemm_nt_fsplits(dt_of(a), M, N, K) : 1;
+    if (zf > 1 && K <= 3072 && ((uintptr_t)bt.data_ptr() & 15) == 0) {
+      auto ws = at::empty({(int64_t)zf * M * N},
+                          a.options().dtype(at::kFloat));
+      gemm_nt_zf_launch(dt_of(a), a.data_ptr(), bt.data_ptr(),
+                        ws.data_ptr<float>(), bp, rp, c.data_ptr(), zf,
+                        zero_page(a), M, N, K, (int)act_kind, cur_stream());
+      return c;
+    }
     gemm_launch(dt_of(a), a.data_ptr(), bt.data_ptr(), bp, c.data_ptr(),
                 zero_page(a), rp, M, N, K, /*trans_b=*/true, (int)act_kind,
                 cur_stream());
@@ -217,6 +245,25 @@ at::Tensor gemm_nt(const at::Tensor& a, const at::Tensor& b) {
               "gemm_nt shapes ", a.sizes(), " @ ", b.sizes(), "^T");
   int M = a.size(0), K = a.size(1), N = b.size(0);
   auto c = at::empty({M, N}, a.options());
+  static const bool use32nt = [] {
+    const char* e = getenv("TNN_GEMM32");
+    return !(e && e[0] == '0');
+  }();
+  if (use32nt && dt_of(a) == DT::BF16 && K % 8 == 0 &&
+      ((uintptr_t)a.data_ptr() & 15) == 0 &&
+      ((uintptr_t)b.data_ptr() & 15) == 0) {
+    int z32 = N % 4 == 0 ? gemm_nt32_zsplits(M, N, K) : 1;
+    at::Tensor ws;
+    float* wp = nullptr;
+    if (z32 > 1) {
+      ws = at::empty({(int64_t)z32 * M * N}, a.options().dtype(at::kFloat));
+      wp = ws.data_ptr<float>();
+    }
+    gemm_nt32_launch(a.data_ptr(), b.data_ptr(), wp, nullptr, nullptr,
+                     nullptr, c.data_ptr(), z32, zero_page(a), M, N, K, 0,
+                     cur_stream());
+    return c;
+  }
   int z = gemm_nt_zsplits(dt_of(a), M, N, K);
   if (z > 1 && ((uintptr_t)a.data_ptr() & 15) == 0 &&
       ((uintptr_t)b.data_ptr() & 15) == 0) {

@@ -277,9 +277,19 @@ __global__ void k_colsum(const T* __restrict__ x, float* __restrict__ out,
   const int64_t r0 = rows * rslice / nrs, r1 = rows * (rslice + 1) / nrs;
   float acc = 0.0f;
   if (col < cols) {
-    // 4 waves stride over the block's row slice; lanes = consecutive cols
-    for (int64_t r = r0 + (threadIdx.x >> 6); r < r1; r += 4)
+    // 4 waves stride over the block's row slice; lanes = consecutive
+    // cols; 4 independent accumulators keep 4 loads in flight per wave
+    // (single-chain form measured 1.7 TB/s on the odd-vocab head bias)
+    float a1 = 0.0f, a2 = 0.0f, a3 = 0.0f;
+    int64_t r = r0 + (threadIdx.x >> 6);
+    for (; r + 12 < r1; r += 16) {
       acc += VecIO<T>::to_f32(x[r * cols + col]);
+      a1 += VecIO<T>::to_f32(x[(r + 4) * cols + col]);
+      a2 += VecIO<T>::to_f32(x[(r + 8) * cols + col]);
+      a3 += VecIO<T>::to_f32(x[(r + 12) * cols + col]);
+    }
+    for (; r < r1; r += 4) acc += VecIO<T>::to_f32(x[r * cols + col]);
+    acc += (a1 + a2) + a3;
   }
   __shared__ float sh[4][64];
   sh[threadIdx.x >> 6][threadIdx.x & 63] = acc;
@@ -370,7 +380,7 @@ void colsum_launch(DT dt, const void* x, void* out_f32, float* ws,
   }
   // fill the chip: one block per ~64-row slice, capped so atomics stay cheap
   int cblocks = (int)((cols + 63) / 64);
-  int cap = std::max(1, 2048 / cblocks);
+  int cap = std::max(1, 8192 / cblocks);
   int rsl = (int)std::min<int64_t>((rows + 63) / 64, (int64_t)cap);
   dim3 grid(cblocks, rsl);
   if (dt == DT::F32)

@@ -378,12 +378,222 @@ __global__ void k_gemm_nt_z(const T* __restrict__ A, const T* __restrict__ B,
   });
 }
 
-int gemm_nt_zsplits(DT dt, int M, int N, int K) {
+// ---------------------------------------------------------------------------
+// 32x32x16-MFMA NT GEMM (bf16): the 16x16x32 tile core needs 3-4 waves
+// per SIMD to cover its ~17 cyc/SIMD single-wave issue floor, but
+// v_mfma_f32_32x32x16_bf16 sustains PEAK issue at ONE wave/SIMD
+// (MI355X_MICROARCH.md cycle table) and touches half the LDS bytes per
+// FLOP (A/B reuse 32 wide, not 16). Block tile 128x128 (2x2 waves of
+// 64x64, FM=FN=2 fragments of 32x32), BK=64, double-buffered glds
+// staging, optional K-split over blockIdx.z into fp32 slabs (finalized
+// by k_splitk_fin_ep with the bias/act/residual epilogue).
+namespace nt32 {
+constexpr int BM2 = 128, BN2 = 64, BK2 = 64;
+
+// C/D map for 32x32x16 (guide: col = lane&31, row = (reg&3) + 8*(reg>>2)
+// + 4*(lane>>5), reg in [0,16))
+template <typename T>
+__launch_bounds__(256)
+__global__ void k_gemm_nt32(const T* __restrict__ A, const T* __restrict__ B,
+                            const float* __restrict__ bias_f32,
+                            const T* __restrict__ bias_t, T* __restrict__ C,
+                            float* __restrict__ Cws,
+                            const T* __restrict__ zero16,
+                            const T* __restrict__ resid, int M, int N, int K,
+                            int act_kind) {
+  using namespace tile;
+  __shared__ alignas(16) T As[2][BM2 * BK2];
+  __shared__ alignas(16) T Bs[2][BN2 * BK2];
+  const int m0 = blockIdx.x * BM2;
+  const int n0 = blockIdx.y * BN2;
+  const int Z = gridDim.z;
+  const int k_lo = Z == 1 ? 0 : (int)(((int64_t)K * blockIdx.z / Z) / BK2 * BK2);
+  const int k_hi = (Z == 1 || blockIdx.z + 1 == Z)
+                       ? K
+                       : (int)(((int64_t)K * (blockIdx.z + 1) / Z) / BK2 * BK2);
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int wrow0 = wid * 32;   // 4 M-stacked waves: 32 rows x 64 cols each
+  const WaveCoord wc;  // only for glds_stage lane bookkeeping
+
+  f32x16 acc[2] = {};
+
+  auto stage = [&](int kk0, int which) {
+    glds_stage<T, BM2>(As[which], wc, [&](int rl, int kk) -> const T* {
+      int gm = m0 + rl, gk = kk0 + kk;
+      if (gm >= M || gk >= k_hi) return zero16;
+      return &A[(int64_t)gm * K + gk];
+    });
+    glds_stage<T, BN2>(Bs[which], wc, [&](int rl, int kk) -> const T* {
+      int gn = n0 + rl, gk = kk0 + kk;
+      if (gn >= N || gk >= k_hi) return zero16;
+      return &B[(int64_t)gn * K + gk];
+    });
+  };
+  constexpr int NPER = glds_count<T, BM2>() + glds_count<T, BN2>();
+
+  const int nch = (k_hi - k_lo + BK2 - 1) / BK2;
+  stage(k_lo, 0);
+  for (int t = 0; t < nch; ++t) {
+    const int cur = t & 1;
+    if (t + 1 < nch) {
+      stage(k_lo + (t + 1) * BK2, cur ^ 1);
+      wait_vmcnt<NPER>();
+    } else {
+      wait_vmcnt<0>();
+    }
+    __builtin_amdgcn_s_barrier();
+    // a-frag: lane supplies row (lane&31), k (lane>>5)*8..+8
+    const int ar = lane & 31;
+    const int kh = (lane >> 5) * 8;
+#pragma unroll
+    for (int ks = 0; ks < BK2 / 16; ++ks) {
+      const int kb = ks * 16 + kh;
+      bf16x8 a, b[2];
+      a = *(const bf16x8*)&As[cur][lds_off<T>(wrow0 + ar, kb)];
+#pragma unroll
+      for (int fn = 0; fn < 2; ++fn)
+        b[fn] = *(const bf16x8*)&Bs[cur][lds_off<T>(fn * 32 + ar, kb)];
+#pragma unroll
+      for (int fn = 0; fn < 2; ++fn)
+        acc[fn] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            a, b[fn], acc[fn], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_barrier();
+  }
+
+  // epilogue
+  const int cc = lane & 31;
+  const int r4 = (lane >> 5) * 4;
+#pragma unroll
+    for (int fn = 0; fn < 2; ++fn)
+#pragma unroll
+      for (int reg = 0; reg < 16; ++reg) {
+        const int row = m0 + wrow0 + (reg & 3) + 8 * (reg >> 2) + r4;
+        const int col = n0 + fn * 32 + cc;
+        if (row >= M || col >= N) continue;
+        float vv = acc[fn][reg];
+        if (Z > 1) {
+          Cws[(int64_t)blockIdx.z * M * N + (int64_t)row * N + col] = vv;
+        } else {
+          if (bias_f32) vv += bias_f32[col];
+          else if (bias_t) vv += VecIO<T>::to_f32(bias_t[col]);
+          if (resid) vv += VecIO<T>::to_f32(resid[(int64_t)row * N + col]);
+          C[(int64_t)row * N + col] = VecIO<T>::from_f32(act_apply(vv, act_kind));
+        }
+      }
+}
+}  // namespace nt32
+
+// split-K for moderately underfilled FORWARD NT shapes (e.g. the
+// transformer's [B*S, 768] projections: 384 blocks on 256 CUs ran at
+// ~170 TF while full-grid launches of the same kernel reach ~2.3x
+// that): fill to ~3 blocks/CU, epilogue moves into the fused finalize.
+int gemm_nt_fsplits(DT dt, int M, int N, int K) {
   const int base = ceil_div(M, BM) * ceil_div(N, BN);
-  if (base >= 1024 || K < 8192) return 1;
+  if (base >= 768 || K < 4 * BK) return 1;
   const bool g = dt == DT::F32 ? K % 4 == 0 : K % 8 == 0;
   if (!g) return 1;
-  return std::min(ceil_div(K, 4 * BK), std::max(1, 1024 / base));
+  return std::min(ceil_div(768, base), K / (2 * BK));
+}
+
+// fused split-K finalize: fold the Z fp32 slabs, add bias / residual,
+// apply the activation, cast to T (the bias/act/resid epilogue the
+// direct kernels do, relocated here for the z-split route)
+template <typename T>
+__launch_bounds__(256)
+__global__ void k_splitk_fin_ep(const float4* __restrict__ ws,
+                                const float* __restrict__ bias_f32,
+                                const T* __restrict__ bias_t,
+                                const T* __restrict__ resid,
+                                T* __restrict__ out, int z, int N,
+                                int64_t n4, int act_kind) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  struct alignas(4 * sizeof(T)) T4 { T e[4]; };
+  for (; i < n4; i += stride) {
+    float4 a{0.0f, 0.0f, 0.0f, 0.0f};
+    for (int sl = 0; sl < z; ++sl) {
+      float4 v = ws[(int64_t)sl * n4 + i];
+      a.x += v.x; a.y += v.y; a.z += v.z; a.w += v.w;
+    }
+    float r[4] = {a.x, a.y, a.z, a.w};
+    const int col0 = (int)((i * 4) % N);
+    T4 o;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      if (bias_f32) r[j] += bias_f32[col0 + j];
+      else if (bias_t) r[j] += VecIO<T>::to_f32(bias_t[col0 + j]);
+      if (resid) r[j] += VecIO<T>::to_f32(resid[i * 4 + j]);
+      o.e[j] = VecIO<T>::from_f32(act_apply(r[j], act_kind));
+    }
+    ((T4*)out)[i] = o;
+  }
+}
+
+void gemm_nt_zf_launch(DT dt, const void* a, const void* b, float* ws,
+                       const void* bias_f32_or_t, const void* resid, void* c,
+                       int z, const void* zero16, int M, int N, int K,
+                       int act_kind, hipStream_t s) {
+  dim3 grid(ceil_div(M, BM), ceil_div(N, BN), z);
+  const int64_t n4 = (int64_t)M * N / 4;  // N % 4 == 0 enforced by caller
+  int blocks = (int)std::min<int64_t>((n4 + 255) / 256, (int64_t)2048);
+  if (dt == DT::F32) {
+    hipLaunchKernelGGL(k_gemm_nt_z<float>, grid, dim3(THREADS), 0, s,
+                       (const float*)a, (const float*)b, ws,
+                       (const float*)zero16, M, N, K);
+    hipLaunchKernelGGL(k_splitk_fin_ep<float>, dim3(blocks), dim3(256), 0, s,
+                       (const float4*)ws, (const float*)bias_f32_or_t,
+                       (const float*)nullptr, (const float*)resid, (float*)c,
+                       z, N, n4, act_kind);
+  } else {
+    hipLaunchKernelGGL(k_gemm_nt_z<bf16>, grid, dim3(THREADS), 0, s,
+                       (const bf16*)a, (const bf16*)b, ws,
+                       (const bf16*)zero16, M, N, K);
+    hipLaunchKernelGGL(k_splitk_fin_ep<bf16>, dim3(blocks), dim3(256), 0, s,
+                       (const float4*)ws, (const float*)nullptr,
+                       (const bf16*)bias_f32_or_t, (const bf16*)resid,
+                       (bf16*)c, z, N, n4, act_kind);
+  }
+}
+
+// K-splits for the 32-wide NT kernel: fill to ~2 blocks/CU (its LDS
+// footprint caps residency at 2)
+int gemm_nt32_zsplits(int M, int N, int K) {
+  using namespace nt32;
+  const int base = ceil_div(M, BM2) * ceil_div(N, BN2);
+  if (base >= 512 || K < 4 * BK2) return 1;
+  if (K % 8 != 0) return 1;
+  return std::min(ceil_div(512, base), K / (2 * BK2));
+}
+
+void gemm_nt32_launch(const void* a, const void* b, float* ws,
+                      const void* bias_f32, const void* bias_t,
+                      const void* resid, void* c, int z, const void* zero16,
+                      int M, int N, int K, int act_kind, hipStream_t s) {
+  using namespace nt32;
+  dim3 grid(ceil_div(M, BM2), ceil_div(N, BN2), z);
+  hipLaunchKernelGGL(k_gemm_nt32<bf16>, grid, dim3(256), 0, s, (const bf16*)a,
+                     (const bf16*)b, (const float*)bias_f32,
+                     (const bf16*)bias_t, (bf16*)c, ws, (const bf16*)zero16,
+                     (const bf16*)resid, M, N, K, act_kind);
+  if (z > 1) {
+    const int64_t n4 = (int64_t)M * N / 4;
+    int blocks = (int)std::min<int64_t>((n4 + 255) / 256, (int64_t)2048);
+    hipLaunchKernelGGL(k_splitk_fin_ep<bf16>, dim3(blocks), dim3(256), 0, s,
+                       (const float4*)ws, (const float*)bias_f32,
+                       (const bf16*)bias_t, (const bf16*)resid, (bf16*)c, z,
+                       N, n4, act_kind);
+  }
+}
+
+
+int gemm_nt_zsplits(DT dt, int M, int N, int K) {
+  const int base = ceil_div(M, BM) * ceil_div(N, BN);
+  if (base >= 768 || K < 4 * BK) return 1;
+  const bool g = dt == DT::F32 ? K % 4 == 0 : K % 8 == 0;
+  if (!g) return 1;
+  return std::min(ceil_div(K, 4 * BK), ceil_div(1024, base));
 }
 
 void gemm_nt_z_launch(DT dt, const void* a, const void* b, float* ws,

@@ -49,6 +49,16 @@ void gemv_nn_launch(DT dt, const void* x, const void* b, const void* bias,
                     void* y, float* ws, int ks, int N, int K, int act_kind,
                     hipStream_t s);
 int gemm_nt_zsplits(DT dt, int M, int N, int K);
+int gemm_nt_fsplits(DT dt, int M, int N, int K);
+int gemm_nt32_zsplits(int M, int N, int K);
+void gemm_nt32_launch(const void* a, const void* b, float* ws,
+                      const void* bias_f32, const void* bias_t,
+                      const void* resid, void* c, int z, const void* zero16,
+                      int M, int N, int K, int act_kind, hipStream_t s);
+void gemm_nt_zf_launch(DT dt, const void* a, const void* b, float* ws,
+                       const void* bias_f32_or_t, const void* resid, void* c,
+                       int z, const void* zero16, int M, int N, int K,
+                       int act_kind, hipStream_t s);
 void gemm_nt_z_launch(DT dt, const void* a, const void* b, float* ws,
                       void* c_out, DT out_dt, int z, const void* zero16,
                       int M, int N, int K, hipStream_t s);

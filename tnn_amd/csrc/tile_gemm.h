@@ -116,13 +116,15 @@ constexpr int glds_count() {
 // double-buffered loops wait an explicit partial count + raw s_barrier.
 template <int N>
 DEV void wait_vmcnt() {
-  static_assert(N == 0 || N == 4 || N == 6 || N == 8 || N == 12 || N == 24,
+  static_assert(N == 0 || N == 4 || N == 6 || N == 8 || N == 12 ||
+                N == 16 || N == 24,
                 "add an asm literal for this count");
   if constexpr (N == 0) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   else if constexpr (N == 4) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
   else if constexpr (N == 6) asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
   else if constexpr (N == 8) asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
   else if constexpr (N == 12) asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
+  else if constexpr (N == 16) asm volatile("s_waitcnt vmcnt(16)" ::: "memory");
   else if constexpr (N == 24) asm volatile("s_waitcnt vmcnt(24)" ::: "memory");
 }
 
