@@ -773,13 +773,25 @@ std::vector<at::Tensor> ln_bwd(const at::Tensor& x, const at::Tensor& dy,
   int cols = x.size(-1);
   int64_t rows = x.numel() / cols;
   auto dx = at::empty_like(x);
-  auto dgamma = at::zeros({cols}, x.options().dtype(at::kFloat));
-  auto dbeta = at::zeros({cols}, x.options().dtype(at::kFloat));
+  const int64_t wsn = ln_bwd_ws_floats(dt_of(x), x.data_ptr(), dy.data_ptr(),
+                                       rows, cols);
+  const bool vec = (dt_of(x) == DT::F32 ? cols % 4 == 0 : cols % 8 == 0) &&
+                   ((uintptr_t)x.data_ptr() & 15) == 0 &&
+                   ((uintptr_t)dy.data_ptr() & 15) == 0;
+  // vec path single-writes (or slab-folds) the [2, cols] buffer; the
+  // scalar fallback atomically accumulates into zeros
+  auto sums = vec ? at::empty({2, cols}, x.options().dtype(at::kFloat))
+                  : at::zeros({2, cols}, x.options().dtype(at::kFloat));
+  at::Tensor ws;
+  float* wp = nullptr;
+  if (wsn) {
+    ws = at::empty({wsn}, x.options().dtype(at::kFloat));
+    wp = ws.data_ptr<float>();
+  }
   ln_bwd_launch(dt_of(x), x.data_ptr(), dy.data_ptr(), gamma.data_ptr<float>(),
                 mean.data_ptr<float>(), invstd.data_ptr<float>(), dx.data_ptr(),
-                dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), rows, cols,
-                cur_stream());
-  return {dx, dgamma, dbeta};
+                sums.data_ptr<float>(), wp, rows, cols, cur_stream());
+  return {dx, sums[0], sums[1]};
 }
 
 // ---- embedding -------------------------------------------------------------

@@ -205,9 +205,11 @@ void gn_bwd_launch(DT dt, const void* x, const void* dy, const float* mean,
 void ln_fwd_launch(DT dt, const void* x, const float* gamma, const float* beta,
                    void* y, float* mean, float* invstd, int64_t rows, int cols,
                    float eps, hipStream_t s);
+int64_t ln_bwd_ws_floats(DT dt, const void* x, const void* dy, int64_t rows,
+                         int cols);
 void ln_bwd_launch(DT dt, const void* x, const void* dy, const float* gamma,
                    const float* mean, const float* invstd, void* dx,
-                   float* dgamma, float* dbeta, int64_t rows, int cols,
+                   float* dgamma2, float* ws, int64_t rows, int cols,
                    hipStream_t s);
 
 // ---- embedding.hip ---------------------------------------------------------

@@ -239,9 +239,11 @@ template <typename T>
 __global__ void k_ln_bwd_col(const T* __restrict__ x, const T* __restrict__ dy,
                              const float* __restrict__ mean,
                              const float* __restrict__ invstd,
-                             float* __restrict__ dgamma,
-                             float* __restrict__ dbeta, int64_t rows,
+                             float* __restrict__ out, int64_t rows,
                              int cols) {
+  // out: gridDim.y == 1 -> the [2, cols] dgamma|dbeta buffer (single
+  // writer, no zero-init); else per-slice slabs [slice][2*cols] folded
+  // by k_slab_fin (no same-address atomic serialization)
   constexpr int V = 16 / sizeof(T);
   struct alignas(16) P { T e[16 / sizeof(T)]; };
   const int groups = cols / V;
@@ -286,8 +288,9 @@ __global__ void k_ln_bwd_col(const T* __restrict__ x, const T* __restrict__ dy,
     }
     int c = blockIdx.x * cpb_total + cl;
     if (c >= cols) break;
-    atomicAdd(&dgamma[c], tg);
-    atomicAdd(&dbeta[c], tb);
+    float* base = out + (int64_t)blockIdx.y * 2 * cols;
+    base[c] = tg;
+    base[cols + c] = tb;
   }
 }
 
@@ -323,9 +326,28 @@ void ln_fwd_launch(DT dt, const void* x, const float* gamma, const float* beta,
                        cols, eps);
 }
 
+// workspace floats for the vectorized dgamma/dbeta column reduce
+// ((slices+1) * 2*cols incl. finalize target); 0 = scalar fallback (the
+// binding must zero dgamma/dbeta for its atomics)
+int64_t ln_bwd_ws_floats(DT dt, const void* x, const void* dy, int64_t rows,
+                         int cols) {
+  int V = dt == DT::F32 ? 4 : 8;
+  bool vec = cols % V == 0 && (((uintptr_t)x & 15) == 0) &&
+             (((uintptr_t)dy & 15) == 0);
+  if (!vec) return 0;
+  int groups = cols / V;
+  int gpb = groups < 256 ? groups : 256;
+  int cblocks = (groups + gpb - 1) / gpb;
+  int rows_per_iter = 256 / gpb;
+  int64_t iters = (rows + rows_per_iter - 1) / rows_per_iter;
+  int rslices = (int)std::min<int64_t>(
+      std::max<int64_t>(512 / cblocks, 1), std::max<int64_t>(iters / 4, 1));
+  return rslices > 1 ? (int64_t)rslices * 2 * cols : 0;
+}
+
 void ln_bwd_launch(DT dt, const void* x, const void* dy, const float* gamma,
                    const float* mean, const float* invstd, void* dx,
-                   float* dgamma, float* dbeta, int64_t rows, int cols,
+                   float* dgamma2, float* ws, int64_t rows, int cols,
                    hipStream_t s) {
   int V = dt == DT::F32 ? 4 : 8;
   bool vec = cols % V == 0 && (((uintptr_t)x & 15) == 0) &&
@@ -346,13 +368,13 @@ void ln_bwd_launch(DT dt, const void* x, const void* dy, const float* gamma,
   } else if (dt == DT::F32)
     hipLaunchKernelGGL(k_ln_bwd<float>, dim3(rows), dim3(256), 0, s,
                        (const float*)x, (const float*)dy, gamma, mean, invstd,
-                       (float*)dx, vec ? nullptr : dgamma,
-                       vec ? nullptr : dbeta, cols);
+                       (float*)dx, vec ? nullptr : dgamma2,
+                       vec ? nullptr : dgamma2 + cols, cols);
   else
     hipLaunchKernelGGL(k_ln_bwd<bf16>, dim3(rows), dim3(256), 0, s,
                        (const bf16*)x, (const bf16*)dy, gamma, mean, invstd,
-                       (bf16*)dx, vec ? nullptr : dgamma, vec ? nullptr : dbeta,
-                       cols);
+                       (bf16*)dx, vec ? nullptr : dgamma2,
+                       vec ? nullptr : dgamma2 + cols, cols);
 #undef LNB
   if (vec) {
     int groups = cols / V;
@@ -364,14 +386,16 @@ void ln_bwd_launch(DT dt, const void* x, const void* dy, const float* gamma,
     int rslices = (int)std::min<int64_t>(
         std::max<int64_t>(512 / cblocks, 1), std::max<int64_t>(iters / 4, 1));
     dim3 grid(cblocks, rslices);
+    float* target = rslices > 1 ? ws : dgamma2;
     if (dt == DT::F32)
       hipLaunchKernelGGL(k_ln_bwd_col<float>, grid, dim3(256), 0, s,
                          (const float*)x, (const float*)dy, mean, invstd,
-                         dgamma, dbeta, rows, cols);
+                         target, rows, cols);
     else
       hipLaunchKernelGGL(k_ln_bwd_col<bf16>, grid, dim3(256), 0, s,
-                         (const bf16*)x, (const bf16*)dy, mean, invstd, dgamma,
-                         dbeta, rows, cols);
+                         (const bf16*)x, (const bf16*)dy, mean, invstd,
+                         target, rows, cols);
+    if (rslices > 1) slab_fin_launch(ws, dgamma2, rslices, 2 * cols, s);
   }
 }
 
