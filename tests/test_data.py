@@ -85,3 +85,33 @@ def test_csv_loader(tmp_path):
     dl2 = CSVLoader(str(p), regression=False, target_cols=[2], batch_size=8)
     dl2.load_data()
     assert dl2.y.dtype.is_floating_point is False
+
+
+def test_prepare_tokens_roundtrip(tmp_path):
+    """tools/prepare_tokens.py builds a reference-format vocab.bin +
+    uint16 train.bin that OpenWebTextLoader and Tokenizer can read."""
+    import subprocess, sys, os
+    import numpy as np
+    from tnn_amd.data.tokenizer import Tokenizer
+    from tnn_amd.data.loaders import OpenWebTextLoader
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    corpus = tmp_path / "corpus.txt"
+    corpus.write_text("the quick brown fox jumps over the lazy dog " * 200)
+    vocab = tmp_path / "vocab.bin"
+    train = tmp_path / "train.bin"
+    tool = os.path.join(root, "tools", "prepare_tokens.py")
+    subprocess.run([sys.executable, tool, "build-vocab", str(corpus),
+                    "--out", str(vocab), "--vocab-size", "300"], check=True)
+    subprocess.run([sys.executable, tool, "tokenize", str(corpus),
+                    "--vocab", str(vocab), "--out", str(train)], check=True)
+    tok = Tokenizer().load(str(vocab))
+    assert 256 < tok.vocab_size <= 300
+    ids = np.fromfile(train, dtype=np.uint16)
+    assert len(ids) > 100 and ids.max() < tok.vocab_size
+    # decode returns the original text (byte-level BPE is lossless)
+    text = tok.decode([int(i) for i in ids[:50]])
+    assert "quick" in text or "fox" in text
+    dl = OpenWebTextLoader(str(train), seq_len=32, samples_per_epoch=64,
+                           batch_size=8)
+    x, y = next(iter(dl))
+    assert x.shape == (8, 32) and y.shape == (8, 32)
