@@ -133,3 +133,49 @@ def test_nary_and_mbroadcast():
     assert len(outs) == 3
     sum(o.sum() for o in outs).backward()
     assert torch.allclose(x.grad, torch.full_like(x, 3.0))
+
+
+def test_msequential_exec_order_preserves_join_semantics():
+    """Memory-aware scheduling may run branches out of declaration order,
+    but non-commutative joins must still combine outputs in declaration
+    order (reference msequential.cpp compute_execution_order)."""
+    import torch
+    from tnn_amd.nn.blocks import MSequential
+    from tnn_amd.nn.layer import Layer
+
+    class Scale(Layer):
+        def __init__(self, f):
+            super().__init__(f"scale{f}", torch.float32)
+            self.f = f
+
+        def forward(self, x):
+            return x * self.f
+
+        def output_shape(self, in_shape):
+            return tuple(in_shape)
+
+    m = MSequential([Scale(3.0), Scale(2.0)], join="sub")
+    x = torch.ones(2, 4)
+    y = m(x)
+    assert torch.allclose(y, torch.full((2, 4), 1.0))  # 3x - 2x, not 2x - 3x
+    # force a reversed execution order: declaration-order join must hold
+    m2 = MSequential([Scale(3.0), Scale(2.0)], join="sub")
+    m2._exec_order = [1, 0]
+    assert torch.allclose(m2(x), torch.full((2, 4), 1.0))
+
+
+@pytest.mark.gpu
+def test_msequential_memory_order_measured_on_gpu():
+    import torch
+    from tnn_amd.nn.blocks import MSequential
+    from tnn_amd.nn.layers import Dense
+    wide = Dense(64, 512, name="wide")    # big transient, big retained
+    slim = Dense(64, 8, name="slim")
+    m = MSequential([slim, wide], join="concat").to("cuda")
+    x = torch.randn(16, 64, device="cuda")
+    y = m(x)
+    assert y.shape == (16, 520)
+    assert sorted(m._exec_order) == [0, 1]  # an order was measured
+    # joins still concat in declaration order
+    ref = torch.cat([m.branches[0](x), m.branches[1](x)], dim=-1)
+    assert torch.allclose(y, ref)

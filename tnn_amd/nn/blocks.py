@@ -151,9 +151,14 @@ class ResidualBlock(Layer):
 class MSequential(Layer):
     """Parallel branches + elementwise join (reference msequential.cpp:40-89).
 
-    The reference's memory-aware branch ordering (sort by peak-workspace −
-    retained bytes) is a workspace-allocator concern; with the caching
-    allocator we run branches in declaration order.
+    Memory-aware branch scheduling (reference measure_sequence_memory /
+    compute_execution_order): each branch is trial-run once on its first
+    GPU forward while the caching allocator's peak/retained deltas are
+    recorded; branches then execute sorted by priority = peak-workspace −
+    retained bytes (descending), so high-transient / low-retained
+    branches run while free memory is largest. The join always combines
+    outputs in DECLARATION order (sub/div are not commutative), only the
+    execution order changes.
     """
 
     JOINS = {"add": torch.add, "sub": torch.sub, "mul": torch.mul,
@@ -164,9 +169,34 @@ class MSequential(Layer):
         super().__init__(name, dtype)
         self.branches = nn.ModuleList(branches)
         self.join = join
+        self._exec_order: Optional[List[int]] = None
+
+    def _measure_order(self, x) -> List[int]:
+        """Reference SequenceMemInfo: one no-grad trial per branch,
+        priority = cycling_cost − output_size, stable sort descending."""
+        infos = []
+        dev = x.device
+        for i, b in enumerate(self.branches):
+            torch.cuda.synchronize(dev)
+            torch.cuda.reset_peak_memory_stats(dev)
+            before = torch.cuda.memory_allocated(dev)
+            with torch.no_grad():
+                out = b(x)
+            torch.cuda.synchronize(dev)
+            peak = torch.cuda.max_memory_allocated(dev) - before
+            retained = out.numel() * out.element_size()
+            del out
+            infos.append((peak - retained, i))
+        infos.sort(key=lambda t: (-t[0], t[1]))
+        return [i for _, i in infos]
 
     def forward(self, x):
-        outs = [b(x) for b in self.branches]
+        if self._exec_order is None:
+            self._exec_order = (self._measure_order(x) if x.is_cuda
+                                else list(range(len(self.branches))))
+        outs: List[Optional[torch.Tensor]] = [None] * len(self.branches)
+        for i in self._exec_order:
+            outs[i] = self.branches[i](x)
         if self.join == "concat":
             return torch.cat(outs, dim=-1)
         if self.join == "add":
