@@ -150,3 +150,45 @@ def test_attn_strided_views(causal):
     assert maxerr(dq, rq) < 2e-3
     assert maxerr(dk, rk) == 0.0
     assert maxerr(dv, rv) == 0.0
+
+
+@pytest.mark.gpu
+def test_ln_passthrough_residual_grads():
+    """layer_norm_res: the residual junction grad join happens inside
+    ln_bwd; grads must match the unfused two-consumer formulation."""
+    from tnn_amd import ops
+    torch.manual_seed(7)
+    x = torch.randn(4, 96, 768, dtype=torch.bfloat16, device=DEV)
+    g = torch.randn(768, device=DEV)
+    b = torch.randn(768, device=DEV)
+    w = (torch.randn(768, 768, dtype=torch.bfloat16, device=DEV) * 0.02)
+
+    xa = x.clone().requires_grad_()
+    y, xr = ops.layer_norm_res(xa, g, b, 1e-5)
+    out = ops.linear(y, w, None, residual=xr)
+    out.float().square().mean().backward()
+
+    xb = x.clone().requires_grad_()
+    y2 = ops.layer_norm(xb, g, b, 1e-5)
+    out2 = ops.linear(y2, w, None, residual=xb)
+    out2.float().square().mean().backward()
+
+    assert maxerr(out, out2) == 0.0
+    assert maxerr(xa.grad, xb.grad) < 2e-3
+
+
+@pytest.mark.gpu
+def test_gpt_block_train_matches_cpu_path():
+    """GPU GPTBlock (passthrough-residual structure) vs the CPU
+    composition on the same weights."""
+    from tnn_amd.nn.blocks import GPTBlock
+    torch.manual_seed(8)
+    blk = GPTBlock(128, 4, flash=True, dropout=0.0)
+    x = torch.randn(2, 64, 128)
+    ref = blk(x)
+    blk_g = blk.to(DEV)
+    y = blk_g(x.to(DEV))
+    assert maxerr(y.cpu(), ref) < 5e-2
+    y.square().mean().backward()
+    for n, p in blk_g.named_parameters():
+        assert p.grad is None or torch.isfinite(p.grad.float()).all(), n

@@ -767,9 +767,18 @@ std::vector<at::Tensor> ln_fwd(const at::Tensor& x, const at::Tensor& gamma,
 
 std::vector<at::Tensor> ln_bwd(const at::Tensor& x, const at::Tensor& dy,
                                const at::Tensor& gamma, const at::Tensor& mean,
-                               const at::Tensor& invstd) {
+                               const at::Tensor& invstd,
+                               const c10::optional<at::Tensor>& resid =
+                                   c10::nullopt) {
   CHECK_IN(x);
   CHECK_IN(dy);
+  const void* rp = nullptr;
+  if (resid.has_value()) {
+    TORCH_CHECK(resid->is_contiguous() &&
+                resid->scalar_type() == x.scalar_type() &&
+                resid->numel() == x.numel(), "ln_bwd resid mismatch");
+    rp = resid->data_ptr();
+  }
   int cols = x.size(-1);
   int64_t rows = x.numel() / cols;
   auto dx = at::empty_like(x);
@@ -790,7 +799,7 @@ std::vector<at::Tensor> ln_bwd(const at::Tensor& x, const at::Tensor& dy,
   }
   ln_bwd_launch(dt_of(x), x.data_ptr(), dy.data_ptr(), gamma.data_ptr<float>(),
                 mean.data_ptr<float>(), invstd.data_ptr<float>(), dx.data_ptr(),
-                sums.data_ptr<float>(), wp, rows, cols, cur_stream());
+                rp, sums.data_ptr<float>(), wp, rows, cols, cur_stream());
   return {dx, sums[0], sums[1]};
 }
 
@@ -1086,7 +1095,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gn_fwd", &tnn::gn_fwd);
   m.def("gn_bwd", &tnn::gn_bwd);
   m.def("ln_fwd", &tnn::ln_fwd);
-  m.def("ln_bwd", &tnn::ln_bwd);
+  m.def("ln_bwd", &tnn::ln_bwd, py::arg("x"), py::arg("dy"),
+        py::arg("gamma"), py::arg("mean"), py::arg("invstd"),
+        py::arg("resid") = py::none());
   m.def("embedding_fwd", &tnn::embedding_fwd);
   m.def("embedding_bwd", &tnn::embedding_bwd);
   m.def("attn_fwd", &tnn::attn_fwd);

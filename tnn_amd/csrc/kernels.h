@@ -209,8 +209,8 @@ int64_t ln_bwd_ws_floats(DT dt, const void* x, const void* dy, int64_t rows,
                          int cols);
 void ln_bwd_launch(DT dt, const void* x, const void* dy, const float* gamma,
                    const float* mean, const float* invstd, void* dx,
-                   float* dgamma2, float* ws, int64_t rows, int cols,
-                   hipStream_t s);
+                   const void* resid, float* dgamma2, float* ws, int64_t rows,
+                   int cols, hipStream_t s);
 
 // ---- embedding.hip ---------------------------------------------------------
 void embedding_fwd_launch(DT dt, const int64_t* ids, const void* table,

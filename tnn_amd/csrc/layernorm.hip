@@ -98,7 +98,9 @@ __launch_bounds__(256)
 __global__ void k_ln_bwd_vec(const T* __restrict__ x, const T* __restrict__ dy,
                              const float* gamma, const float* mean,
                              const float* invstd, T* __restrict__ dx,
-                             int cols) {
+                             const T* __restrict__ resid, int cols) {
+  // resid: passthrough-residual grad added into dx (the pre-LN block's
+  // residual junction — fused here so autograd never runs a join add)
   constexpr int V = 16 / sizeof(T);
   struct alignas(16) P { T e[16 / sizeof(T)]; };
   const int64_t row = blockIdx.x;
@@ -134,17 +136,21 @@ __global__ void k_ln_bwd_vec(const T* __restrict__ x, const T* __restrict__ dy,
   __syncthreads();
   const float ma = s_a, mb = s_b;
   T* dxr = dx + row * cols;
+  const T* rr = resid ? resid + row * cols : nullptr;
 #pragma unroll
   for (int k = 0; k < PK; ++k) {
     const int pk = threadIdx.x + k * 256;
     if (pk < npk) {
       P o;
+      P vr;
+      if (rr) vr = ((const P*)rr)[pk];
 #pragma unroll
       for (int j = 0; j < V; ++j) {
         const float gy = VecIO<T>::to_f32(vd[k].e[j]);
         const float xhat = (VecIO<T>::to_f32(vx[k].e[j]) - m) * is;
-        o.e[j] = VecIO<T>::from_f32(
-            is * (gy * gamma[pk * V + j] - ma - xhat * mb));
+        float v = is * (gy * gamma[pk * V + j] - ma - xhat * mb);
+        if (rr) v += VecIO<T>::to_f32(vr.e[j]);
+        o.e[j] = VecIO<T>::from_f32(v);
       }
       ((P*)dxr)[pk] = o;
     }
@@ -194,6 +200,7 @@ template <typename T>
 __global__ void k_ln_bwd(const T* __restrict__ x, const T* __restrict__ dy,
                          const float* gamma, const float* mean,
                          const float* invstd, T* __restrict__ dx,
+                         const T* __restrict__ resid,
                          float* __restrict__ dgamma, float* __restrict__ dbeta,
                          int cols) {
   const int64_t row = blockIdx.x;
@@ -220,10 +227,13 @@ __global__ void k_ln_bwd(const T* __restrict__ x, const T* __restrict__ dy,
   }
   __syncthreads();
   const float ma = s_a, mb = s_b;
+  const T* rr = resid ? resid + row * cols : nullptr;
   for (int c = threadIdx.x; c < cols; c += blockDim.x) {
     float gy = VecIO<T>::to_f32(dyr[c]);
     float xhat = (VecIO<T>::to_f32(xr[c]) - m) * is;
-    dxr[c] = VecIO<T>::from_f32(is * (gy * gamma[c] - ma - xhat * mb));
+    float v = is * (gy * gamma[c] - ma - xhat * mb);
+    if (rr) v += VecIO<T>::to_f32(rr[c]);
+    dxr[c] = VecIO<T>::from_f32(v);
     if (dgamma) {  // ragged-cols fallback; the vec column kernel is used
       atomicAdd(&dgamma[c], gy * xhat);  // when cols % V == 0
       atomicAdd(&dbeta[c], gy);
@@ -347,8 +357,8 @@ int64_t ln_bwd_ws_floats(DT dt, const void* x, const void* dy, int64_t rows,
 
 void ln_bwd_launch(DT dt, const void* x, const void* dy, const float* gamma,
                    const float* mean, const float* invstd, void* dx,
-                   float* dgamma2, float* ws, int64_t rows, int cols,
-                   hipStream_t s) {
+                   const void* resid, float* dgamma2, float* ws, int64_t rows,
+                   int cols, hipStream_t s) {
   int V = dt == DT::F32 ? 4 : 8;
   bool vec = cols % V == 0 && (((uintptr_t)x & 15) == 0) &&
              (((uintptr_t)dy & 15) == 0);
@@ -356,7 +366,7 @@ void ln_bwd_launch(DT dt, const void* x, const void* dy, const float* gamma,
 #define LNB(T, PK)                                                           \
   hipLaunchKernelGGL((k_ln_bwd_vec<T, PK>), dim3(rows), dim3(256), 0, s,     \
                      (const T*)x, (const T*)dy, gamma, mean, invstd, (T*)dx, \
-                     cols)
+                     (const T*)resid, cols)
   if (vec && npk <= 2 * 256) {  // PK<=2: x+dy packs stay in registers
     if (dt == DT::F32) {
       if (npk <= 256) LNB(float, 1);
@@ -368,12 +378,14 @@ void ln_bwd_launch(DT dt, const void* x, const void* dy, const float* gamma,
   } else if (dt == DT::F32)
     hipLaunchKernelGGL(k_ln_bwd<float>, dim3(rows), dim3(256), 0, s,
                        (const float*)x, (const float*)dy, gamma, mean, invstd,
-                       (float*)dx, vec ? nullptr : dgamma2,
+                       (float*)dx, (const float*)resid,
+                       vec ? nullptr : dgamma2,
                        vec ? nullptr : dgamma2 + cols, cols);
   else
     hipLaunchKernelGGL(k_ln_bwd<bf16>, dim3(rows), dim3(256), 0, s,
                        (const bf16*)x, (const bf16*)dy, gamma, mean, invstd,
-                       (bf16*)dx, vec ? nullptr : dgamma2,
+                       (bf16*)dx, (const bf16*)resid,
+                       vec ? nullptr : dgamma2,
                        vec ? nullptr : dgamma2 + cols, cols);
 #undef LNB
   if (vec) {

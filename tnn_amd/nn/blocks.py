@@ -443,6 +443,21 @@ class GPTBlock(Layer):
             h = ops.linear(x, self.w1, self.b1, act="gelu",
                            ln=(self.ln2.gamma, self.ln2.beta, self.ln2.eps))
             return ops.linear(h, self.w2, self.b2, residual=x)
+        if x.is_cuda:
+            # residual branches ride the LayerNorm passthrough outputs:
+            # x/a keep ONE consumer each, so the junction grad joins run
+            # inside ln_bwd (autograd fan-in adds removed — 2 activation
+            # adds per block per step)
+            ln1_out, xr = ops.layer_norm_res(x, self.ln1.gamma,
+                                             self.ln1.beta, self.ln1.eps)
+            a = self.attn(ln1_out, residual=xr)
+            ln2_out, ar = ops.layer_norm_res(a, self.ln2.gamma,
+                                             self.ln2.beta, self.ln2.eps)
+            h = ops.linear(ln2_out, self.w1, self.b1, act="gelu")
+            if self.drop is not None:
+                h = self.drop(ops.linear(h, self.w2, self.b2))
+                return ops.add_act(ar, h)
+            return ops.linear(h, self.w2, self.b2, residual=ar)
         x = self.attn(self.ln1(x), residual=x)
         h = ops.linear(self.ln2(x), self.w1, self.b1, act="gelu")
         if self.drop is not None:

@@ -21,7 +21,7 @@ from .functional import (
     dropout,
     activation,
     softmax_cross_entropy,
-    layer_norm,
+    layer_norm, layer_norm_res,
     embedding,
     attention,
     sdpa_materialized,

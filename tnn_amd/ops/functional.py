@@ -655,6 +655,47 @@ class _LayerNorm(torch.autograd.Function):
         return dx, dgamma, dbeta, None
 
 
+class _LayerNormRes(torch.autograd.Function):
+    """LayerNorm with a residual PASSTHROUGH output: ``y, xr = f(x)`` where
+    xr aliases x. A pre-LN transformer block routes its residual branch
+    through xr, so x has ONE consumer and the junction's grad join
+    (d_x = ln_grad + d_residual) happens inside the ln_bwd kernel instead
+    of an autograd fan-in add pass (2 activation-sized adds per block per
+    step without this)."""
+
+    @staticmethod
+    def forward(ctx, x, gamma, beta, eps):
+        ext = _C.ext()
+        y, mean, invstd = ext.ln_fwd(x, gamma, beta, eps)
+        ctx.save_for_backward(x, gamma, mean, invstd)
+        ctx.set_materialize_grads(False)
+        return y, x.view_as(x)
+
+    @staticmethod
+    def backward(ctx, dy, dres):
+        x, gamma, mean, invstd = ctx.saved_tensors
+        ext = _C.ext()
+        if dy is None:  # residual-only consumer (degenerate use)
+            return dres, None, None, None
+        if dres is not None:
+            dres = dres.contiguous()
+        dx, dgamma, dbeta = ext.ln_bwd(x, dy.contiguous(), gamma, mean,
+                                       invstd, resid=dres)
+        return dx, dgamma, dbeta, None
+
+
+def layer_norm_res(x: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
+                   eps: float = 1e-5):
+    """(layer_norm(x), residual passthrough of x) — see _LayerNormRes."""
+    if _use_hip(x):
+        lead = x.shape[:-1]
+        y, xr = _LayerNormRes.apply(
+            x.reshape(-1, x.shape[-1]).contiguous(), gamma, beta, eps)
+        return (y.reshape(*lead, x.shape[-1]),
+                xr.reshape(*lead, x.shape[-1]))
+    return layer_norm(x, gamma, beta, eps), x
+
+
 def layer_norm(x: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
                eps: float = 1e-5) -> torch.Tensor:
     if _use_hip(x):
