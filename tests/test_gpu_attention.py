@@ -192,3 +192,41 @@ def test_gpt_block_train_matches_cpu_path():
     y.square().mean().backward()
     for n, p in blk_g.named_parameters():
         assert p.grad is None or torch.isfinite(p.grad.float()).all(), n
+
+
+@pytest.mark.gpu
+def test_merged_qkv_training_path_matches_separate():
+    """linear_qkv + flash_attention_qkv (one GEMM, one dQKV buffer) vs
+    three separate projections + flash_attention: same fwd and grads."""
+    from tnn_amd import ops
+    torch.manual_seed(9)
+    B, S, H, D = 2, 128, 4, 64
+    dim = H * D
+    x = torch.randn(B, S, dim, dtype=torch.bfloat16, device=DEV)
+    ws = [torch.randn(dim, dim, dtype=torch.bfloat16, device=DEV) * 0.02
+          for _ in range(3)]
+    bs = [torch.randn(dim, dtype=torch.bfloat16, device=DEV) * 0.01
+          for _ in range(3)]
+
+    def run(merged):
+        wq, wk, wv = [w.clone().requires_grad_() for w in ws]
+        bq, bk, bv = [b.clone().requires_grad_() for b in bs]
+        xa = x.clone().requires_grad_()
+        if merged:
+            qkv = ops.linear_qkv(xa, wq, wk, wv, bq, bk, bv)
+            o = ops.flash_attention_qkv(qkv, H, True)
+        else:
+            q = ops.linear(xa, wq, bq).view(B, S, H, D).transpose(1, 2)
+            k = ops.linear(xa, wk, bk).view(B, S, H, D).transpose(1, 2)
+            v = ops.linear(xa, wv, bv).view(B, S, H, D).transpose(1, 2)
+            o = ops.attention(q, k, v, causal=True)
+            o = o.transpose(1, 2).reshape(B, S, dim)
+        o.float().square().mean().backward()
+        return o, xa.grad, wq.grad, bq.grad
+
+    o1, dx1, dwq1, dbq1 = run(True)
+    o2, dx2, dwq2, dbq2 = run(False)
+    assert maxerr(o1, o2) < 3e-3
+    assert maxerr(dx1, dx2) < 3e-3
+    assert maxerr(dwq1, dwq2) < 3e-3
+    assert maxerr(dbq1.float(), dbq2.float()) < 3e-3

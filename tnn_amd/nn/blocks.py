@@ -402,10 +402,22 @@ class FlashAttentionBlock(_MHABase):
 
     def forward(self, x, residual=None, ln=None):
         b, s, _ = x.shape
-        q, k, v = self._project(x, ln=ln)
         if self._kv_cache is not None:
+            q, k, v = self._project(x, ln=ln)
             o = self._cached_attention(q, k, v)
             return self._merge(o, b, s, residual)
+        if (x.is_cuda and x.dtype == torch.bfloat16
+                and self.head_dim in (64, 128)):
+            # merged-QKV training path: one projection GEMM, attention
+            # reads the head slices as strided views, dqkv comes back in
+            # one buffer (no fan-in adds, no slice backwards)
+            if ln is not None:
+                x = ops.layer_norm(x, *ln)
+            qkv = ops.linear_qkv(x, self.wq, self.wk, self.wv,
+                                 self.bq, self.bk, self.bv)
+            o = ops.flash_attention_qkv(qkv, self.num_heads, self.causal)
+            return ops.linear(o, self.wo, self.bo, residual=residual)
+        q, k, v = self._project(x, ln=ln)
         o = ops.attention(q, k, v, causal=self.causal)
         return self._merge(o, b, s, residual)
 

@@ -11,6 +11,7 @@ Every op has two implementations behind one functional API:
   eager fallback).
 """
 
+from .flash import flash_attention_qkv
 from .functional import (
     conv2d_nhwc,
     batch_norm_act,
@@ -21,7 +22,7 @@ from .functional import (
     dropout,
     activation,
     softmax_cross_entropy,
-    layer_norm, layer_norm_res,
+    layer_norm, layer_norm_res, linear_qkv,
     embedding,
     attention,
     sdpa_materialized,
@@ -45,6 +46,7 @@ __all__ = [
     "layer_norm",
     "embedding",
     "attention",
+    "flash_attention_qkv",
     "sdpa_materialized",
     "attention_decode",
     "scaled_softmax",

@@ -36,6 +36,53 @@ class _FlashAttention(torch.autograd.Function):
         return dq, dk, dv, None
 
 
+class _FlashQKV(torch.autograd.Function):
+    """Flash attention straight off the merged projection buffer:
+    qkv [B, S, 3*H*D] -> o [B, S, H*D]. q/k/v are head-slice VIEWS read
+    through the kernels' stride tuples, and backward writes dq/dk/dv into
+    ONE dqkv buffer (attn_bwd's dq_out/dk_out/dv_out) — without this the
+    three slice consumers cost autograd 3 activation-size zero-fills +
+    copies + 2 adds per layer per step."""
+
+    @staticmethod
+    def _views(t, H, D):
+        HD = H * D
+        return tuple(s.unflatten(-1, (H, D)).permute(0, 2, 1, 3)
+                     for s in t.split(HD, dim=-1))
+
+    @staticmethod
+    def forward(ctx, qkv, H, causal):
+        B, S, threeHD = qkv.shape
+        D = threeHD // (3 * H)
+        q, k, v = _FlashQKV._views(qkv, H, D)
+        ext = _C.ext()
+        o, lse = ext.attn_fwd(q, k, v, causal)
+        ctx.save_for_backward(qkv, o, lse)
+        ctx.H, ctx.D, ctx.causal = H, D, causal
+        # o is a [B,H,S,D] view of a BSHD-contiguous buffer: this reshape
+        # is free
+        return o.transpose(1, 2).reshape(B, S, H * D)
+
+    @staticmethod
+    def backward(ctx, do):
+        qkv, o, lse = ctx.saved_tensors
+        H, D, causal = ctx.H, ctx.D, ctx.causal
+        q, k, v = _FlashQKV._views(qkv, H, D)
+        dqkv = torch.empty_like(qkv)
+        dq_o, dk_o, dv_o = _FlashQKV._views(dqkv, H, D)
+        dov = do.contiguous().unflatten(-1, (H, D)).permute(0, 2, 1, 3)
+        ext = _C.ext()
+        ext.attn_bwd(q, k, v, o, dov, lse, causal,
+                     dq_out=dq_o, dk_out=dk_o, dv_out=dv_o)
+        return dqkv, None, None
+
+
+def flash_attention_qkv(qkv: torch.Tensor, num_heads: int,
+                        causal: bool = True) -> torch.Tensor:
+    """o [B,S,H*D] from a merged qkv [B,S,3*H*D] (see _FlashQKV)."""
+    return _FlashQKV.apply(qkv, num_heads, causal)
+
+
 def _composed(q, k, v, causal):
     scale = q.shape[-1] ** -0.5
     s = (q.float() @ k.float().transpose(-1, -2)) * scale

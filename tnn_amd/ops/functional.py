@@ -307,6 +307,54 @@ class _Linear(torch.autograd.Function):
         return dx, dw, db, None, dres
 
 
+class _LinearQKV(torch.autograd.Function):
+    """The three attention projections as ONE GEMM: wqkv = cat(wq,wk,wv)
+    rebuilt per call (a weight-sized copy, ~1% of the GEMM it replaces);
+    3x fewer GEMM/colsum launches and the shared input x keeps a single
+    consumer (no autograd fan-in adds). Weight/bias grads come back as
+    contiguous slices of the single dW/db reductions."""
+
+    @staticmethod
+    def forward(ctx, x2, wq, wk, wv, bq, bk, bv):
+        ext = _C.ext()
+        wqkv = torch.cat([wq, wk, wv], dim=1)
+        has_b = bq is not None
+        bqkv = torch.cat([bq, bk, bv]) if has_b else None
+        qkv = ext.gemm(x2, wqkv, bqkv, ACT_KINDS["linear"])
+        ctx.save_for_backward(x2, wqkv)
+        ctx.has_b = has_b
+        return qkv
+
+    @staticmethod
+    def backward(ctx, dy):
+        x2, wqkv = ctx.saved_tensors
+        ext = _C.ext()
+        dy = dy.contiguous()
+        N = wqkv.shape[1] // 3
+        dx = dwq = dwk = dwv = dbq = dbk = dbv = None
+        if ctx.has_b and ctx.needs_input_grad[4]:
+            db = ext.colsum(dy)
+            dbq, dbk, dbv = db[:N], db[N:2 * N], db[2 * N:]
+        if ctx.needs_input_grad[0]:
+            dx = ext.gemm_nt(dy, wqkv)
+        if ctx.needs_input_grad[1]:
+            dw = ext.gemm_tn(x2, dy, wqkv.dtype == x2.dtype).to(wqkv.dtype)
+            # contiguous slices: the fused multi-tensor optimizer reads
+            # grads as flat buffers
+            dwq = dw[:, :N].contiguous()
+            dwk = dw[:, N:2 * N].contiguous()
+            dwv = dw[:, 2 * N:].contiguous()
+        return dx, dwq, dwk, dwv, dbq, dbk, dbv
+
+
+def linear_qkv(x: torch.Tensor, wq, wk, wv, bq=None, bk=None, bv=None):
+    """Merged attention projection: [.., K] -> [.., 3N] (see _LinearQKV)."""
+    lead = x.shape[:-1]
+    x2 = x.reshape(-1, x.shape[-1])
+    y = _LinearQKV.apply(x2.contiguous(), wq, wk, wv, bq, bk, bv)
+    return y.reshape(*lead, y.shape[-1])
+
+
 def linear(x: torch.Tensor, w: torch.Tensor, bias: Optional[torch.Tensor] = None,
            act: str = "linear",
            residual: Optional[torch.Tensor] = None,
