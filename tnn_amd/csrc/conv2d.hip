@@ -70,8 +70,18 @@ __global__ void k_conv_fwd_db(const T* __restrict__ X,
 
   const int M = cs.N * cs.OH * cs.OW;
   const int K = cs.KH * cs.KW * cs.Cin;
-  const int m0 = blockIdx.x * BM;
-  const int n0 = blockIdx.y * BN;
+  int tm_, tn_;
+  // grouped tile ordering when the gathered activation stream exceeds
+  // the 256 MiB L3 across its Cout/BN re-reads (see tile_gemm.h)
+  if ((int64_t)cs.N * cs.OH * cs.OW * (cs.KH * cs.KW * cs.Cin) >
+      (int64_t)32 * 1024 * 1024)
+    tile::tile_remap<8>(tm_, tn_);
+  else {
+    tm_ = blockIdx.x;
+    tn_ = blockIdx.y;
+  }
+  const int m0 = tm_ * BM;
+  const int n0 = tn_ * BN;
   const WaveCoord wc;
   f32x4 acc[FM][FN] = {};
 
@@ -183,8 +193,18 @@ __global__ void k_conv_fwd(const T* __restrict__ X, const T* __restrict__ Wt,
 
   const int M = cs.N * cs.OH * cs.OW;
   const int K = cs.KH * cs.KW * cs.Cin;
-  const int m0 = blockIdx.x * BM;
-  const int n0 = blockIdx.y * BN;
+  int tm_, tn_;
+  // grouped tile ordering when the gathered activation stream exceeds
+  // the 256 MiB L3 across its Cout/BN re-reads (see tile_gemm.h)
+  if ((int64_t)cs.N * cs.OH * cs.OW * (cs.KH * cs.KW * cs.Cin) >
+      (int64_t)32 * 1024 * 1024)
+    tile::tile_remap<8>(tm_, tn_);
+  else {
+    tm_ = blockIdx.x;
+    tn_ = blockIdx.y;
+  }
+  const int m0 = tm_ * BM;
+  const int n0 = tn_ * BN;
   const WaveCoord wc;
   f32x4 acc[FM][FN] = {};
   using VecT = Pack16<T>;
@@ -319,8 +339,18 @@ __global__ void k_conv_dgrad(const T* __restrict__ DY, const T* __restrict__ WT,
   const int Wc = S2 ? (cs.W >> 1) : cs.W;
   const int M = cs.N * Hc * Wc;
   const int K = nkh * nkw * cs.Cout;
-  const int m0 = blockIdx.x * BM;
-  const int n0 = blockIdx.y * BN;
+  int tm_, tn_;
+  // grouped tile ordering when the gathered activation stream exceeds
+  // the 256 MiB L3 across its Cout/BN re-reads (see tile_gemm.h)
+  if ((int64_t)cs.N * cs.OH * cs.OW * (cs.KH * cs.KW * cs.Cin) >
+      (int64_t)32 * 1024 * 1024)
+    tile::tile_remap<8>(tm_, tn_);
+  else {
+    tm_ = blockIdx.x;
+    tn_ = blockIdx.y;
+  }
+  const int m0 = tm_ * BM;
+  const int n0 = tn_ * BN;
   const WaveCoord wc;
   f32x4 acc[FM][FN] = {};
   using VecT = Pack16<T>;
@@ -485,8 +515,18 @@ __global__ void k_conv_dgrad_db(const T* __restrict__ DY,
   const int M = cs.N * Hc * Wc;
   const int K = nkh * nkw * cs.Cout;
   const int KFULL = cs.KH * cs.KW * cs.Cout;
-  const int m0 = blockIdx.x * BM;
-  const int n0 = blockIdx.y * BN;
+  int tm_, tn_;
+  // grouped tile ordering when the gathered activation stream exceeds
+  // the 256 MiB L3 across its Cout/BN re-reads (see tile_gemm.h)
+  if ((int64_t)cs.N * cs.OH * cs.OW * (cs.KH * cs.KW * cs.Cin) >
+      (int64_t)32 * 1024 * 1024)
+    tile::tile_remap<8>(tm_, tn_);
+  else {
+    tm_ = blockIdx.x;
+    tn_ = blockIdx.y;
+  }
+  const int m0 = tm_ * BM;
+  const int n0 = tn_ * BN;
   const WaveCoord wc;
   f32x4 acc[FM][FN] = {};
 
@@ -591,8 +631,18 @@ __global__ void k_conv_wgrad(const T* __restrict__ X, const T* __restrict__ DY,
 
   const int M = cs.N * cs.OH * cs.OW;         // reduction dim
   const int Kout = cs.KH * cs.KW * cs.Cin;    // output rows
-  const int r0 = blockIdx.x * BM;
-  const int n0 = blockIdx.y * BN;
+  int tr_, tn_;
+  // same grouped ordering: the gathered-x A operand [Kout, M] is
+  // re-read gridDim.y times without it
+  if ((int64_t)(cs.KH * cs.KW * cs.Cin) * cs.N * cs.OH * cs.OW >
+      (int64_t)32 * 1024 * 1024)
+    tile::tile_remap<8>(tr_, tn_);
+  else {
+    tr_ = blockIdx.x;
+    tn_ = blockIdx.y;
+  }
+  const int r0 = tr_ * BM;
+  const int n0 = tn_ * BN;
   const int m_begin = (int)((int64_t)M * blockIdx.z / gridDim.z);
   const int m_end = (int)((int64_t)M * (blockIdx.z + 1) / gridDim.z);
   const WaveCoord wc;
@@ -698,8 +748,18 @@ __global__ void k_conv_wgrad_vec(const T* __restrict__ X,
   __shared__ alignas(16) T Bs[BN * BK];  // rows = co, k = m
   const int M = cs.N * cs.OH * cs.OW;
   const int Kout = cs.KH * cs.KW * cs.Cin;
-  const int r0 = blockIdx.x * BM;
-  const int n0 = blockIdx.y * BN;
+  int tr_, tn_;
+  // same grouped ordering: the gathered-x A operand [Kout, M] is
+  // re-read gridDim.y times without it
+  if ((int64_t)(cs.KH * cs.KW * cs.Cin) * cs.N * cs.OH * cs.OW >
+      (int64_t)32 * 1024 * 1024)
+    tile::tile_remap<8>(tr_, tn_);
+  else {
+    tr_ = blockIdx.x;
+    tn_ = blockIdx.y;
+  }
+  const int r0 = tr_ * BM;
+  const int n0 = tn_ * BN;
   const int m_begin = (int)((int64_t)M * blockIdx.z / gridDim.z);
   const int m_end = (int)((int64_t)M * (blockIdx.z + 1) / gridDim.z);
   const WaveCoord wc;

@@ -58,6 +58,24 @@ template <typename T> struct LDSBytes {
   static constexpr int total = A + B;
 };
 
+// L2-locality tile remap (grouped ordering): dispatch-linear block ids
+// sweep GM m-tiles x ALL n-tiles in panels, so a panel's A rows stay
+// L2-resident while every B column streams once per panel — without
+// this an [M,N,K] GEMM re-reads A (N/BN) times from HBM (measured: the
+// fc1-shaped GEMM's whole runtime equals that traffic at 8 TB/s).
+template <int GM>
+DEV void tile_remap(int& tm, int& tn) {
+  const int nbx = gridDim.x, nby = gridDim.y;
+  const int id = blockIdx.y * nbx + blockIdx.x;  // dispatch order, x fastest
+  const int per_group = GM * nby;
+  const int group = id / per_group;
+  const int first_m = group * GM;
+  const int gm = min(nbx - first_m, GM);
+  const int rem = id - group * per_group;
+  tm = first_m + rem % gm;
+  tn = rem / gm;
+}
+
 struct WaveCoord {
   int wid, lane, wrow0, wcol0;
   DEV WaveCoord() {
