@@ -75,7 +75,7 @@ __global__ void k_conv_fwd_db(const T* __restrict__ X,
   // the 256 MiB L3 across its Cout/BN re-reads (see tile_gemm.h)
   if ((int64_t)cs.N * cs.OH * cs.OW * (cs.KH * cs.KW * cs.Cin) >
       (int64_t)32 * 1024 * 1024)
-    tile::tile_remap<8>(tm_, tn_);
+    tile::tile_remap_xcd<4>(tm_, tn_);
   else {
     tm_ = blockIdx.x;
     tn_ = blockIdx.y;
@@ -198,7 +198,7 @@ __global__ void k_conv_fwd(const T* __restrict__ X, const T* __restrict__ Wt,
   // the 256 MiB L3 across its Cout/BN re-reads (see tile_gemm.h)
   if ((int64_t)cs.N * cs.OH * cs.OW * (cs.KH * cs.KW * cs.Cin) >
       (int64_t)32 * 1024 * 1024)
-    tile::tile_remap<8>(tm_, tn_);
+    tile::tile_remap_xcd<4>(tm_, tn_);
   else {
     tm_ = blockIdx.x;
     tn_ = blockIdx.y;
@@ -344,7 +344,7 @@ __global__ void k_conv_dgrad(const T* __restrict__ DY, const T* __restrict__ WT,
   // the 256 MiB L3 across its Cout/BN re-reads (see tile_gemm.h)
   if ((int64_t)cs.N * cs.OH * cs.OW * (cs.KH * cs.KW * cs.Cin) >
       (int64_t)32 * 1024 * 1024)
-    tile::tile_remap<8>(tm_, tn_);
+    tile::tile_remap_xcd<4>(tm_, tn_);
   else {
     tm_ = blockIdx.x;
     tn_ = blockIdx.y;
@@ -520,7 +520,7 @@ __global__ void k_conv_dgrad_db(const T* __restrict__ DY,
   // the 256 MiB L3 across its Cout/BN re-reads (see tile_gemm.h)
   if ((int64_t)cs.N * cs.OH * cs.OW * (cs.KH * cs.KW * cs.Cin) >
       (int64_t)32 * 1024 * 1024)
-    tile::tile_remap<8>(tm_, tn_);
+    tile::tile_remap_xcd<4>(tm_, tn_);
   else {
     tm_ = blockIdx.x;
     tn_ = blockIdx.y;
@@ -636,7 +636,7 @@ __global__ void k_conv_wgrad(const T* __restrict__ X, const T* __restrict__ DY,
   // re-read gridDim.y times without it
   if ((int64_t)(cs.KH * cs.KW * cs.Cin) * cs.N * cs.OH * cs.OW >
       (int64_t)32 * 1024 * 1024)
-    tile::tile_remap<8>(tr_, tn_);
+    tile::tile_remap_xcd<4>(tr_, tn_);
   else {
     tr_ = blockIdx.x;
     tn_ = blockIdx.y;
@@ -753,7 +753,7 @@ __global__ void k_conv_wgrad_vec(const T* __restrict__ X,
   // re-read gridDim.y times without it
   if ((int64_t)(cs.KH * cs.KW * cs.Cin) * cs.N * cs.OH * cs.OW >
       (int64_t)32 * 1024 * 1024)
-    tile::tile_remap<8>(tr_, tn_);
+    tile::tile_remap_xcd<4>(tr_, tn_);
   else {
     tr_ = blockIdx.x;
     tn_ = blockIdx.y;

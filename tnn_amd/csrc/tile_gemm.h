@@ -76,6 +76,27 @@ DEV void tile_remap(int& tm, int& tn) {
   tn = rem / gm;
 }
 
+// XCD-aware variant: the dispatcher round-robins consecutive block ids
+// over the 8 XCDs, so id%8 IS the XCD. Give each XCD a contiguous,
+// grouped stripe of the tile space: panel reuse then hits the XCD's own
+// (non-coherent) 4 MiB L2 instead of only the die-level L3.
+template <int GM>
+DEV void tile_remap_xcd(int& tm, int& tn) {
+  const int nbx = gridDim.x, nby = gridDim.y;
+  const int T = nbx * nby;
+  const int id = blockIdx.y * nbx + blockIdx.x;
+  const int per = (T + 7) >> 3;
+  int lid = (id & 7) * per + (id >> 3);
+  if (lid >= T) lid = id;
+  const int per_group = GM * nby;
+  const int group = lid / per_group;
+  const int first_m = group * GM;
+  const int gm = min(nbx - first_m, GM);
+  const int rem = lid - group * per_group;
+  tm = first_m + rem % gm;
+  tn = rem / gm;
+}
+
 struct WaveCoord {
   int wid, lane, wrow0, wcol0;
   DEV WaveCoord() {
