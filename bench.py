@@ -31,7 +31,11 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=10)
     p.add_argument("--model", default="cifar100_wrn16_8")
     p.add_argument("--batch-per-gpu", type=int, default=256)
-    p.add_argument("--microbatch", type=int, default=64)
+    p.add_argument("--microbatch", type=int, default=32,
+                   help="micro-batch size for the 1F1B schedule: 8 "
+                        "microbatches per pipeline stage (bubble fraction "
+                        "(N-1)/(8N+N-1)); conv M stays >=32k rows so "
+                        "per-kernel efficiency holds")
     p.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
     p.add_argument("--lr", type=float, default=1e-3)
     return p.parse_args()
