@@ -87,15 +87,34 @@ def main():
     x = torch.randn(global_batch, *in_shape, generator=g).to(device)
     y = torch.randint(0, num_classes, (global_batch,), generator=g).to(device)
 
+    # single-GPU: capture the whole train step (grad zero + fwd + loss +
+    # bwd + fused optimizer) as ONE hipGraph and replay it — identical
+    # kernels, no per-launch host dispatch gaps. Disable: TNN_BENCH_GRAPH=0.
+    graphed = (world == 1 and have_gpu
+               and os.environ.get("TNN_BENCH_GRAPH", "1") != "0")
+    if graphed:
+        from tnn_amd.utils.graphstep import GraphedTrainStep
+        try:
+            gstep = GraphedTrainStep(engine.stage, engine.criterion,
+                                     engine.optimizer, x, y)
+            run_step = gstep
+        except Exception as e:  # capture failure: measure eager instead
+            print(f"[bench] graph capture failed ({e}); eager fallback",
+                  file=sys.stderr)
+            graphed = False
+            run_step = lambda: engine.train_batch(x, y)  # noqa: E731
+    else:
+        run_step = lambda: engine.train_batch(x, y)  # noqa: E731
+
     for _ in range(args.warmup):
-        engine.train_batch(x, y)
+        run_step()
 
     comm.barrier()
     if have_gpu:
         torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        engine.train_batch(x, y)
+        run_step()
     comm.barrier()
     if have_gpu:
         torch.cuda.synchronize()
@@ -130,6 +149,7 @@ def main():
                 "global_batch": global_batch,
                 "input": "x".join(map(str, in_shape)),
                 "microbatches": num_micro,
+            "graphed_step": graphed,
                 "parallelism": f"pp{world}" if world > 1 else "single",
             },
         }))

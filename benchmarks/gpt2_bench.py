@@ -27,6 +27,8 @@ def main():
     p.add_argument("--steps", type=int, default=10)
     p.add_argument("--decode-tokens", type=int, default=32)
     p.add_argument("--train", action="store_true", default=True)
+    p.add_argument("--graph", action="store_true",
+                   help="also time the hipGraph-captured whole train step")
     args = p.parse_args()
 
     dev = torch.device("cuda" if torch.cuda.is_available() else "cpu")
@@ -51,6 +53,28 @@ def main():
             opt.step()
             return loss
 
+        # graph capture must see FRESH AccumulateGrad nodes (eager steps
+        # first would pin them to the default stream and crash capture)
+        if args.graph and dev.type == "cuda":
+            from tnn_amd.utils.graphstep import GraphedTrainStep
+            gstep = GraphedTrainStep(model, crit, opt, x, y)
+            for _ in range(3):
+                gstep()
+            torch.cuda.synchronize()
+            t0 = time.perf_counter()
+            for _ in range(args.steps):
+                gloss = gstep()
+            torch.cuda.synchronize()
+            dt = time.perf_counter() - t0
+            toks = args.batch * args.seq_len * args.steps
+            print(f"train[hipGraph]: {args.model} bs={args.batch} "
+                  f"seq={args.seq_len}: {toks / dt:,.0f} tok/s "
+                  f"({dt / args.steps * 1e3:.1f} ms/step, "
+                  f"loss {gloss.item():.3f})")
+            gstep.sync_step_count()
+            opt._step_dev = None  # eager steps resume host step counting
+            del gstep
+
         for _ in range(3):
             step()
         torch.cuda.synchronize() if dev.type == "cuda" else None
@@ -63,6 +87,8 @@ def main():
         print(f"train: {args.model} bs={args.batch} seq={args.seq_len}: "
               f"{toks / dt:,.0f} tok/s ({dt / args.steps * 1e3:.1f} ms/step, "
               f"loss {loss.item():.3f})")
+
+
 
     # ---- greedy decode (reference gpt2_inference loop) ----
     # note: at batch 1 and short sequences decode is kernel-launch-bound on

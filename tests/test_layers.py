@@ -179,3 +179,17 @@ def test_msequential_memory_order_measured_on_gpu():
     # joins still concat in declaration order
     ref = torch.cat([m.branches[0](x), m.branches[1](x)], dim=-1)
     assert torch.allclose(y, ref)
+
+
+def test_linear_qkv_cpu_fallback():
+    import torch
+    from tnn_amd import ops
+    torch.manual_seed(5)
+    x = torch.randn(2, 8, 16)
+    ws = [torch.randn(16, 16) for _ in range(3)]
+    bs = [torch.randn(16) for _ in range(3)]
+    y = ops.linear_qkv(x, *ws, *bs)
+    ref = torch.cat([x @ w + b for w, b in zip(ws, bs)], dim=-1)
+    assert torch.allclose(y, ref, atol=1e-5)
+    o = ops.flash_attention_qkv(y, num_heads=4, causal=True)
+    assert o.shape == (2, 8, 16)

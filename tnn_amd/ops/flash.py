@@ -80,7 +80,14 @@ class _FlashQKV(torch.autograd.Function):
 def flash_attention_qkv(qkv: torch.Tensor, num_heads: int,
                         causal: bool = True) -> torch.Tensor:
     """o [B,S,H*D] from a merged qkv [B,S,3*H*D] (see _FlashQKV)."""
-    return _FlashQKV.apply(qkv, num_heads, causal)
+    if qkv.is_cuda and qkv.dtype == torch.bfloat16:
+        return _FlashQKV.apply(qkv, num_heads, causal)
+    B, S, threeHD = qkv.shape
+    D = threeHD // (3 * num_heads)
+    q, k, v = (t.unflatten(-1, (num_heads, D)).permute(0, 2, 1, 3)
+               for t in qkv.split(num_heads * D, dim=-1))
+    o = _composed(q, k, v, causal).to(qkv.dtype)
+    return o.transpose(1, 2).reshape(B, S, num_heads * D)
 
 
 def _composed(q, k, v, causal):

@@ -349,6 +349,10 @@ class _LinearQKV(torch.autograd.Function):
 
 def linear_qkv(x: torch.Tensor, wq, wk, wv, bq=None, bk=None, bv=None):
     """Merged attention projection: [.., K] -> [.., 3N] (see _LinearQKV)."""
+    if not _use_hip(x):
+        outs = [x @ w + (b if b is not None else 0)
+                for w, b in ((wq, bq), (wk, bk), (wv, bv))]
+        return torch.cat(outs, dim=-1)
     lead = x.shape[:-1]
     x2 = x.reshape(-1, x.shape[-1])
     y = _LinearQKV.apply(x2.contiguous(), wq, wk, wv, bq, bk, bv)
