@@ -766,21 +766,28 @@ __global__ void k_conv_wgrad_vec(const T* __restrict__ X,
   f32x4 acc[FM][FN] = {};
   using VecT = Pack16<T>;
 
-  for (int k0 = m_begin; k0 < m_end; k0 += BK) {
-    // bf16: each thread owns one 8-row group x 4 consecutive m columns --
-    // after the loads an in-thread 8x4 transpose packs 4 m-values per row
-    // into one 8B ds_write_b64 (vs 8x ds_write_b16 per load; LDS
-    // instruction issue was wgrad's post-phase-separation bottleneck:
-    // 48 b16 writes vs 16 MFMAs per thread-chunk). The XOR swizzle flips
-    // byte bits 4-6 only, so the 8B alignment survives. fp32 keeps the
-    // generic scatter.
-    constexpr bool TR = sizeof(T) == 2;
-    // lane mapping: consecutive lanes share the m-quad and span the row
-    // groups: the 16 lanes' 16B global loads form one contiguous 256B run
-    // (the conflict-free same-row/mm-strided write mapping was tried and
-    // lost more to broken global coalescing than its bank relief won)
-    const int a_rr = TR ? (threadIdx.x & (BM / V - 1)) * V : 0;
-    const int a_mm0 = TR ? (threadIdx.x / (BM / V)) * 4 : 0;
+  // bf16: each thread owns one 8-row group x 4 consecutive m columns --
+  // after the loads an in-thread 8x4 transpose packs 4 m-values per row
+  // into one 8B ds_write_b64 (vs 8x ds_write_b16 per load; LDS
+  // instruction issue was wgrad's post-phase-separation bottleneck:
+  // 48 b16 writes vs 16 MFMAs per thread-chunk). The XOR swizzle flips
+  // byte bits 4-6 only, so the 8B alignment survives. fp32 keeps the
+  // generic scatter.
+  //
+  // Software-pipelined (T14 rotate): tile t's registers are written to
+  // LDS, then tile t+1's loads are ISSUED into the same (now dead)
+  // registers before tile t's MFMA phase — the gather latency lands
+  // under the MFMAs instead of being exposed at the head of every
+  // chunk (the attention kernels' TStage split, applied here).
+  constexpr bool TR = sizeof(T) == 2;
+  // lane mapping: consecutive lanes share the m-quad and span the row
+  // groups: the 16 lanes' 16B global loads form one contiguous 256B run
+  // (the conflict-free same-row/mm-strided write mapping was tried and
+  // lost more to broken global coalescing than its bank relief won)
+  const int a_rr = TR ? (threadIdx.x & (BM / V - 1)) * V : 0;
+  const int a_mm0 = TR ? (threadIdx.x / (BM / V)) * 4 : 0;
+  VecT va[RA], vb[RB];
+  auto load_tile = [&](int k0) {
     const T* asrc[RA];
 #pragma unroll
     for (int it = 0; it < RA; ++it) {
@@ -802,7 +809,7 @@ __global__ void k_conv_wgrad_vec(const T* __restrict__ X,
         int ci = imod<POW2>(gr, cs.d_cin);
         int kidx = idiv<POW2>(gr, cs.d_cin);
         int kh, kw;
-    kdecode(kidx, cs, kh, kw);
+        kdecode(kidx, cs, kh, kw);
         int ih = oh * cs.SH - cs.PH + kh;
         int iw = ow * cs.SW - cs.PW + kw;
         if (ih >= 0 && ih < cs.H && iw >= 0 && iw < cs.W)
@@ -821,11 +828,14 @@ __global__ void k_conv_wgrad_vec(const T* __restrict__ X,
                      ? &DY[(int64_t)gm * cs.Cout + n0 + nn]
                      : zero16;
     }
-    VecT va[RA], vb[RB];
 #pragma unroll
     for (int it = 0; it < RA; ++it) va[it] = *(const VecT*)asrc[it];
 #pragma unroll
     for (int it = 0; it < RB; ++it) vb[it] = *(const VecT*)bsrc[it];
+  };
+
+  load_tile(m_begin);
+  for (int k0 = m_begin; k0 < m_end; k0 += BK) {
     if constexpr (TR) {
 #pragma unroll
       for (int j = 0; j < V; ++j) {
@@ -852,6 +862,7 @@ __global__ void k_conv_wgrad_vec(const T* __restrict__ X,
 #pragma unroll
       for (int j = 0; j < V; ++j) Bs[lds_off<T>(nn + j, mm)] = vb[it].e[j];
     }
+    if (k0 + BK < m_end) load_tile(k0 + BK);
     __syncthreads();
     mfma_compute_tile(As, Bs, wc, acc);
     __syncthreads();
