@@ -81,10 +81,11 @@ class GraphedTrainStep:
 
     def _step_body(self):
         self.step_ctr.add_(1)
-        # in-place zeroing (set-to-None would reallocate grads per replay)
-        for p in self.optimizer.params:
-            if p.grad is not None:
-                p.grad.zero_()
+        # in-place zeroing (set-to-None would reallocate grads per replay);
+        # one multi-tensor kernel, not a fill per param (54 launches on WRN)
+        grads = [p.grad for p in self.optimizer.params if p.grad is not None]
+        if grads:
+            torch._foreach_zero_(grads)
         out = self.model(self.static_x)
         loss = self.criterion(out, self.static_y)
         loss.backward()
