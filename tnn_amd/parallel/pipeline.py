@@ -207,7 +207,10 @@ class PipelineEngine:
 
         def recv_forward(idx):
             if self.is_first:
-                return micro_x[idx].to(self.io_dtype)
+                mx = micro_x[idx]
+                # token ids stay integral (embedding input); only float
+                # inputs follow the io dtype
+                return mx.to(self.io_dtype) if mx.is_floating_point() else mx
             with self.profiler.span(EventType.COMMUNICATION, f"recv_act {idx}"):
                 return self._recv_act(mb_size)
 
@@ -308,7 +311,8 @@ class PipelineEngine:
         works = []
         for i in range(M):
             if self.is_first:
-                inp = micro_x[i].to(self.io_dtype)
+                inp = (micro_x[i].to(self.io_dtype)
+                       if micro_x[i].is_floating_point() else micro_x[i])
             else:
                 inp = torch.empty(mb_size, *self.in_shape, device=self.device,
                                   dtype=self.io_dtype)
