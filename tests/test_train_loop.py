@@ -63,3 +63,30 @@ def test_training_config_env(monkeypatch):
     cfg = TrainingConfig.from_env()
     assert cfg.epochs == 3
     assert cfg.num_microbatches == 8
+
+
+def test_trainer_example_entrypoint(tmp_path):
+    """examples/trainer.py end to end on CPU with a tiny synthetic run
+    (the reference examples/trainer.cpp analog must stay runnable)."""
+    import subprocess, sys, os
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    cfg = tmp_path / "cfg.json"
+    cfg.write_text('{"epochs": 1, "batch_size": 16, "learning_rate": 1e-3}')
+    r = subprocess.run(
+        [sys.executable, os.path.join(root, "examples", "trainer.py"),
+         "--model", "mnist_cnn", "--dataset", "synthetic",
+         "--config", str(cfg)],
+        capture_output=True, text=True, timeout=240)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert "epoch" in (r.stdout + r.stderr).lower()
+
+
+def test_inferencer_example_entrypoint():
+    import subprocess, sys, os
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, os.path.join(root, "examples", "inferencer.py"),
+         "--model", "mnist_cnn", "--dataset", "synthetic",
+         "--batch-size", "32"],
+        capture_output=True, text=True, timeout=240)
+    assert r.returncode == 0, r.stderr[-2000:]
