@@ -28,8 +28,13 @@ def main():
     else:
         model = models.create_model(args.model)
     if args.dataset == "synthetic":
-        loader = DataLoaderFactory.create("synthetic_image", shape=(32, 32, 3),
-                                          num_classes=100, num_samples=2048,
+        shapes = {"mnist_cnn": ((28, 28, 1), 10),
+                  "cifar10_resnet9": ((32, 32, 3), 10),
+                  "cifar10_vgg": ((32, 32, 3), 10)}
+        in_shape, classes = shapes.get(args.model, ((32, 32, 3), 100))
+        loader = DataLoaderFactory.create("synthetic_image", shape=in_shape,
+                                          num_classes=classes,
+                                          num_samples=2048,
                                           batch_size=args.batch_size)
     else:
         loader = DataLoaderFactory.create(args.dataset, path=args.data_path,
