@@ -887,6 +887,8 @@ std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
   TORCH_CHECK(attn_view_ok(o, false), "o must be d-contiguous");
   at::Tensor dq, dk, dv;
   if (dq_out.has_value()) {
+    TORCH_CHECK(dk_out.has_value() && dv_out.has_value(),
+                "provide all three grad views or none");
     // caller-provided grad views (e.g. three slices of one merged dQKV
     // buffer: the fused-QKV backward writes it in place, no cat/pad)
     dq = *dq_out; dk = *dk_out; dv = *dv_out;
