@@ -124,3 +124,54 @@ def test_bf16_training_dynamics_sane():
         losses.append(l.item())
     assert all(v == v for v in losses), "NaN loss"
     assert losses[-1] < 0.7 * losses[0], losses[::6]
+
+
+@pytest.mark.gpu
+def test_bn_passthrough_residual_grads():
+    """BN residual passthrough (pre-activation block junction): grads
+    must match the unfused two-consumer formulation."""
+    import torch
+    from tnn_amd import ops
+    torch.manual_seed(11)
+    x = torch.randn(64, 8, 8, 32, dtype=torch.bfloat16, device="cuda")
+    g = torch.randn(32, device="cuda").abs() + 0.5
+    b = torch.randn(32, device="cuda")
+
+    def run(passthrough):
+        rm = torch.zeros(32, device="cuda")
+        rv = torch.ones(32, device="cuda")
+        xa = x.clone().requires_grad_()
+        if passthrough:
+            y, xr = ops.batch_norm_act(xa, g, b, rm, rv, True, relu=True,
+                                       passthrough=True)
+        else:
+            y = ops.batch_norm_act(xa, g, b, rm, rv, True, relu=True)
+            xr = xa
+        out = y * 0.7 + xr  # join: both consumers
+        out.float().square().mean().backward()
+        return out.detach(), xa.grad
+
+    o1, g1 = run(True)
+    o2, g2 = run(False)
+    assert (o1.float() - o2.float()).abs().max().item() < 1e-2
+    assert (g1.float() - g2.float()).abs().max().item() < 2e-3
+
+
+@pytest.mark.gpu
+def test_residual_block_passthrough_matches_eval_math():
+    """ResidualBlock's BN-passthrough fast path (train) vs the plain
+    composition on the same weights."""
+    import torch
+    from tnn_amd.nn.builder import LayerBuilder
+    torch.manual_seed(12)
+    m = (LayerBuilder(input_shape=(8, 8, 32), dtype=torch.float32)
+         .wide_residual_block(32, 32, 1, 0.0, "wb").build("m"))
+    from tnn_amd.nn.layer import cast_compute_dtype
+    cast_compute_dtype(m, torch.bfloat16)
+    m.to("cuda").train()
+    x = torch.randn(16, 8, 8, 32, dtype=torch.bfloat16, device="cuda",
+                    requires_grad=True)
+    y = m(x)
+    y.float().square().mean().backward()
+    assert torch.isfinite(y.float()).all()
+    assert x.grad is not None and torch.isfinite(x.grad.float()).all()

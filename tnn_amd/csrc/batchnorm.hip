@@ -346,8 +346,9 @@ __global__ void k_bn_bwd_apply(const T* __restrict__ x, const T* __restrict__ dy
                                const T* __restrict__ y_relu, const float* mean,
                                const float* invstd, const float* gamma,
                                const float* sum_dy, const float* sum_dy_xhat,
-                               T* __restrict__ dx, int64_t n, int64_t rows,
-                               int cols, float dy_scale) {
+                               T* __restrict__ dx, const T* __restrict__ resid,
+                               int64_t n, int64_t rows, int cols,
+                               float dy_scale) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   const float inv_n = 1.0f / (float)rows;
@@ -358,6 +359,7 @@ __global__ void k_bn_bwd_apply(const T* __restrict__ x, const T* __restrict__ dy
     float xhat = (VecIO<T>::to_f32(x[i]) - mean[c]) * invstd[c];
     float v = gamma[c] * invstd[c] *
               (g - inv_n * (sum_dy[c] + xhat * sum_dy_xhat[c]));
+    if (resid) v += VecIO<T>::to_f32(resid[i]);
     dx[i] = VecIO<T>::from_f32(v);
   }
 }
@@ -398,8 +400,11 @@ __global__ void k_bn_bwd_apply_vec(const T* __restrict__ x,
                                    const float* mean, const float* invstd,
                                    const float* gamma, const float* sum_dy,
                                    const float* sum_dy_xhat,
-                                   T* __restrict__ dx, int64_t nv, int groups,
-                                   int64_t rows, float dy_scale) {
+                                   T* __restrict__ dx,
+                                   const T* __restrict__ resid, int64_t nv,
+                                   int groups, int64_t rows, float dy_scale) {
+  // resid: passthrough-residual grad added into dx (pre-activation
+  // residual blocks route the junction's grad join through bn_bwd)
   constexpr int V = 16 / sizeof(T);
   struct alignas(16) P { T e[16 / sizeof(T)]; };
   const float inv_n = 1.0f / (float)rows;
@@ -411,6 +416,8 @@ __global__ void k_bn_bwd_apply_vec(const T* __restrict__ x,
     P vd = ((const P*)dy)[i];
     P vr;
     if (y_relu) vr = ((const P*)y_relu)[i];
+    P vres;
+    if (resid) vres = ((const P*)resid)[i];
     P o;
 #pragma unroll
     for (int j = 0; j < V; ++j) {
@@ -418,9 +425,10 @@ __global__ void k_bn_bwd_apply_vec(const T* __restrict__ x,
       float g = VecIO<T>::to_f32(vd.e[j]) * dy_scale;
       if (y_relu && VecIO<T>::to_f32(vr.e[j]) <= 0.0f) g = 0.0f;
       const float xhat = (VecIO<T>::to_f32(vx.e[j]) - mean[c]) * invstd[c];
-      o.e[j] = VecIO<T>::from_f32(
-          gamma[c] * invstd[c] *
-          (g - inv_n * (sum_dy[c] + xhat * sum_dy_xhat[c])));
+      float v = gamma[c] * invstd[c] *
+                (g - inv_n * (sum_dy[c] + xhat * sum_dy_xhat[c]));
+      if (resid) v += VecIO<T>::to_f32(vres.e[j]);
+      o.e[j] = VecIO<T>::from_f32(v);
     }
     ((P*)dx)[i] = o;
   }
@@ -645,8 +653,8 @@ void bn_bwd_apply_launch(DT dt, const void* x, const void* dy,
                          const void* y_relu, const float* mean,
                          const float* invstd, const float* gamma,
                          const float* sum_dy, const float* sum_dy_xhat,
-                         void* dx, int64_t rows, int cols, float dy_scale,
-                         hipStream_t s) {
+                         void* dx, const void* resid, int64_t rows, int cols,
+                         float dy_scale, hipStream_t s) {
   int64_t n = rows * cols;
   int blocks = (int)std::min<int64_t>((n + 255) / 256, (int64_t)2048);
   const bool vec = dt == DT::F32 ? bn_vec_ok<float>(x, cols)
@@ -656,27 +664,27 @@ void bn_bwd_apply_launch(DT dt, const void* x, const void* dy,
       hipLaunchKernelGGL(k_bn_bwd_apply_vec<float>, dim3(blocks), dim3(256),
                          0, s, (const float*)x, (const float*)dy,
                          (const float*)y_relu, mean, invstd, gamma, sum_dy,
-                         sum_dy_xhat, (float*)dx, n / 4, cols / 4, rows,
-                         dy_scale);
+                         sum_dy_xhat, (float*)dx, (const float*)resid,
+                         n / 4, cols / 4, rows, dy_scale);
       return;
     }
     hipLaunchKernelGGL(k_bn_bwd_apply<float>, dim3(blocks), dim3(256), 0, s,
                        (const float*)x, (const float*)dy, (const float*)y_relu,
-                       mean, invstd, gamma, sum_dy, sum_dy_xhat, (float*)dx, n,
-                       rows, cols, dy_scale);
+                       mean, invstd, gamma, sum_dy, sum_dy_xhat, (float*)dx,
+                       (const float*)resid, n, rows, cols, dy_scale);
   } else {
     if (vec && ((uintptr_t)dy & 15) == 0) {
       hipLaunchKernelGGL(k_bn_bwd_apply_vec<bf16>, dim3(blocks), dim3(256),
                          0, s, (const bf16*)x, (const bf16*)dy,
                          (const bf16*)y_relu, mean, invstd, gamma, sum_dy,
-                         sum_dy_xhat, (bf16*)dx, n / 8, cols / 8, rows,
-                         dy_scale);
+                         sum_dy_xhat, (bf16*)dx, (const bf16*)resid,
+                         n / 8, cols / 8, rows, dy_scale);
       return;
     }
     hipLaunchKernelGGL(k_bn_bwd_apply<bf16>, dim3(blocks), dim3(256), 0, s,
                        (const bf16*)x, (const bf16*)dy, (const bf16*)y_relu,
-                       mean, invstd, gamma, sum_dy, sum_dy_xhat, (bf16*)dx, n,
-                       rows, cols, dy_scale);
+                       mean, invstd, gamma, sum_dy, sum_dy_xhat, (bf16*)dx,
+                       (const bf16*)resid, n, rows, cols, dy_scale);
   }
 }
 

@@ -546,9 +546,18 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& x, const at::Tensor& dy,
                                const at::Tensor& gamma, const at::Tensor& mean,
                                const at::Tensor& invstd,
                                const c10::optional<at::Tensor>& y_relu,
-                               double dy_scale) {
+                               double dy_scale,
+                               const c10::optional<at::Tensor>& resid =
+                                   c10::nullopt) {
   CHECK_IN(x);
   CHECK_IN(dy);
+  const void* resp = nullptr;
+  if (resid.has_value()) {
+    TORCH_CHECK(resid->is_contiguous() &&
+                resid->scalar_type() == x.scalar_type() &&
+                resid->numel() == x.numel(), "bn_bwd resid mismatch");
+    resp = resid->data_ptr();
+  }
   int C = x.size(-1);
   int64_t rows = x.numel() / C;
   const int64_t wsn = bn_bwd_ws_floats(dt_of(x), x.data_ptr(),
@@ -574,8 +583,8 @@ std::vector<at::Tensor> bn_bwd(const at::Tensor& x, const at::Tensor& dy,
   bn_bwd_apply_launch(dt_of(x), x.data_ptr(), dy.data_ptr(), yr,
                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
                       gamma.data_ptr<float>(), sum_dy.data_ptr<float>(),
-                      sum_dy_xhat.data_ptr<float>(), dx.data_ptr(), rows, C,
-                      (float)dy_scale, cur_stream());
+                      sum_dy_xhat.data_ptr<float>(), dx.data_ptr(), resp,
+                      rows, C, (float)dy_scale, cur_stream());
   // dgamma = sum_dy_xhat, dbeta = sum_dy (fp32, matching fp32 gamma/beta)
   return {dx, sum_dy_xhat, sum_dy};
 }
@@ -1085,7 +1094,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("dropout_p"), py::arg("seed"), py::arg("precomp"),
         py::arg("ctr") = c10::nullopt);
   m.def("bn_fwd_infer", &tnn::bn_fwd_infer);
-  m.def("bn_bwd", &tnn::bn_bwd);
+  m.def("bn_bwd", &tnn::bn_bwd, py::arg("x"), py::arg("dy"),
+        py::arg("gamma"), py::arg("mean"), py::arg("invstd"),
+        py::arg("y_relu"), py::arg("dy_scale"),
+        py::arg("resid") = py::none());
   m.def("maxpool_fwd", &tnn::maxpool_fwd);
   m.def("maxpool_bwd", &tnn::maxpool_bwd);
   m.def("avgpool_fwd", &tnn::avgpool_fwd);

@@ -162,8 +162,9 @@ void bn_bwd_reduce_launch(DT dt, const void* x, const void* dy, const void* y_re
 void bn_bwd_apply_launch(DT dt, const void* x, const void* dy, const void* y_relu,
                          const float* mean, const float* invstd,
                          const float* gamma, const float* sum_dy,
-                         const float* sum_dy_xhat, void* dx, int64_t rows,
-                         int cols, float dy_scale, hipStream_t s);
+                         const float* sum_dy_xhat, void* dx, const void* resid,
+                         int64_t rows, int cols, float dy_scale,
+                         hipStream_t s);
 
 // ---- pool.hip --------------------------------------------------------------
 struct PoolShape {

@@ -24,10 +24,10 @@ class Sequential(Layer):
         super().__init__(name, dtype)
         self.layers = nn.ModuleList(layers or [])
 
-    def forward(self, x):
+    def forward(self, x, start: int = 0):
         from .layers import BatchNorm, Conv2D, Dropout
         n = len(self.layers)
-        i = 0
+        i = start
         while i < n:
             layer = self.layers[i]
             fuse = self.training and x.is_cuda
@@ -118,7 +118,21 @@ class ResidualBlock(Layer):
         self.final_activation = final_activation
 
     def forward(self, x):
-        y = self.main(x)
+        main = self.main
+        from .layers import BatchNorm, Dropout
+        if (x.is_cuda and self.training and isinstance(main, Sequential)
+                and len(main.layers) > 0
+                and isinstance(main.layers[0], BatchNorm)
+                and not (len(main.layers) > 1
+                         and isinstance(main.layers[1], Dropout))):
+            # pre-activation block: the shortcut rides the first BN's
+            # residual passthrough, so x keeps ONE consumer and the
+            # junction grad join runs inside bn_bwd (no autograd add)
+            y0, xr = main.layers[0].forward_res(x)
+            y = main(y0, start=1)
+            s = self.shortcut(xr) if self.shortcut is not None else xr
+            return ops.add_act(y, s, self.final_activation)
+        y = main(x)
         s = self.shortcut(x) if self.shortcut is not None else x
         return ops.add_act(y, s, self.final_activation)
 

@@ -117,6 +117,17 @@ class BatchNorm(Layer):
                                   self.running_mean, self.running_var,
                                   self.training, self.momentum, self.eps, self.relu)
 
+    def forward_res(self, x):
+        """(bn(x), residual passthrough of x): pre-activation residual
+        blocks route their shortcut through the second output so the
+        junction's grad join fuses into bn_bwd (see _BatchNormActRes)."""
+        if x.is_cuda and self.training:
+            return ops.batch_norm_act(x, self.gamma, self.beta,
+                                      self.running_mean, self.running_var,
+                                      True, self.momentum, self.eps,
+                                      self.relu, passthrough=True)
+        return self.forward(x), x
+
     def flops_per_item(self, in_shape):
         return 8 * math.prod(in_shape)
 

@@ -194,6 +194,43 @@ class _BatchNormAct(torch.autograd.Function):
                 None, None)
 
 
+class _BatchNormActRes(torch.autograd.Function):
+    """_BatchNormAct plus a residual PASSTHROUGH output (pre-activation
+    residual blocks: the block input's junction grad join d_x = bn_grad +
+    d_residual runs inside the bn_bwd apply kernel — no autograd fan-in
+    add pass; mirrors _LayerNormRes)."""
+
+    @staticmethod
+    def forward(ctx, x, gamma, beta, running_mean, running_var, momentum,
+                eps, relu, dropout_p, seed, precomp, ctr):
+        ext = _C.ext()
+        y, mean, invstd = ext.bn_fwd_train(x, gamma, beta, running_mean,
+                                           running_var, momentum, eps, relu,
+                                           dropout_p, seed, precomp, ctr)
+        save_y = y if (relu or dropout_p > 0.0) else None
+        ctx.save_for_backward(x, gamma, mean, invstd, save_y)
+        ctx.relu = relu or dropout_p > 0.0
+        ctx.dy_scale = 1.0 / (1.0 - dropout_p) if dropout_p > 0.0 else 1.0
+        ctx.mark_non_differentiable(mean, invstd)
+        ctx.set_materialize_grads(False)
+        return y, mean, invstd, x.view_as(x)
+
+    @staticmethod
+    def backward(ctx, dy, _dm, _di, dres):
+        x, gamma, mean, invstd, y = ctx.saved_tensors
+        ext = _C.ext()
+        if dy is None:
+            return (dres, None, None, None, None, None, None, None, None,
+                    None, None, None)
+        if dres is not None:
+            dres = dres.contiguous()
+        dx, dgamma, dbeta = ext.bn_bwd(x, dy.contiguous(), gamma, mean,
+                                       invstd, y if ctx.relu else None,
+                                       ctx.dy_scale, resid=dres)
+        return (dx, dgamma, dbeta, None, None, None, None, None, None, None,
+                None, None)
+
+
 def batch_norm_act(
     x: torch.Tensor,
     gamma: torch.Tensor,
@@ -206,6 +243,7 @@ def batch_norm_act(
     relu: bool = False,
     dropout_p: float = 0.0,
     precomputed: Optional[torch.Tensor] = None,
+    passthrough: bool = False,
 ):
     """BatchNorm over NHWC channels-last with optional fused ReLU and
     (train-time) fused dropout.
@@ -215,12 +253,20 @@ def batch_norm_act(
     regardless of x dtype (reference batchnorm_layer.cpp:140-148).
     dropout_p > 0 (requires relu) folds the dropout mask + 1/(1-p) scale
     into the BN apply kernel -- one memory pass instead of three.
+    ``passthrough=True`` additionally returns x as a residual-branch
+    alias whose grad joins inside bn_bwd (see _BatchNormActRes).
     """
     C = x.shape[-1]
     n = x.numel() // C
     if _use_hip(x):
         if training:
             seed = _draw_seed() if dropout_p > 0.0 else 0
+            if passthrough:
+                y, _, _, xr = _BatchNormActRes.apply(
+                    x.contiguous(), gamma, beta, running_mean, running_var,
+                    momentum, eps, relu, dropout_p, seed, precomputed,
+                    _graph_ctr)
+                return y, xr
             y, _, _ = _BatchNormAct.apply(x.contiguous(), gamma, beta,
                                           running_mean, running_var, momentum,
                                           eps, relu, dropout_p, seed,
@@ -246,7 +292,8 @@ def batch_norm_act(
         y = F.relu(y)
     if dropout_p > 0.0 and training:
         y = F.dropout(y, dropout_p, training=True)
-    return y.to(x.dtype)
+    y = y.to(x.dtype)
+    return (y, x) if passthrough else y
 
 
 # ---------------------------------------------------------------------------
