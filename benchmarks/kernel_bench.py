@@ -113,8 +113,61 @@ def bench_attn():
                2.5 * fl)
 
 
+def bench_bmm():
+    # materialized-scores attention shapes (per-head batched QK^T)
+    for BH, S, D in [(96, 512, 64), (32, 2048, 128)]:
+        a = torch.randn(BH, S, D, dtype=torch.bfloat16, device=DEV)
+        b = torch.randn(BH, S, D, dtype=torch.bfloat16, device=DEV)
+        fl = 2.0 * BH * S * S * D
+        secs = timeit(lambda: ext.bmm(a, b.transpose(-1, -2)))
+        report("bmm_nt", f"BH{BH} S{S} D{D} (QK^T)", secs, fl)
+
+
+def bench_softmax():
+    from tnn_amd import ops
+    for BH, S in [(96, 512), (32, 2048)]:
+        x = torch.randn(BH, S, S, dtype=torch.bfloat16, device=DEV)
+        by = 2 * x.numel() * 2
+        secs = timeit(lambda: ext.smax_fwd(x, S, 0, 0.125, True))
+        report("smax_fwd", f"BH{BH} S{S} causal", secs, bytes_=by)
+
+
+def bench_ln():
+    for R, C in [(4096, 768), (8192, 1024)]:
+        x = torch.randn(R, C, dtype=torch.bfloat16, device=DEV)
+        g = torch.randn(C, device=DEV)
+        b = torch.randn(C, device=DEV)
+        by = 2 * x.numel() * 2
+        secs = timeit(lambda: ext.ln_fwd(x, g, b, 1e-5))
+        report("ln_fwd", f"{R}x{C}", secs, bytes_=by)
+
+
+def bench_gn():
+    for N, HW, C, G in [(256, 64, 256, 32)]:
+        x = torch.randn(N, HW, C, dtype=torch.bfloat16, device=DEV)
+        g = torch.randn(C, device=DEV)
+        b = torch.randn(C, device=DEV)
+        by = 2 * x.numel() * 2
+        secs = timeit(lambda: ext.gn_fwd(x, g, b, G, 1e-5))
+        report("gn_fwd", f"N{N} HW{HW} C{C} G{G}", secs, bytes_=by)
+
+
+def bench_decode():
+    from tnn_amd import ops
+    BH, cap, D = 12, 1024, 64
+    q = torch.randn(BH, D, dtype=torch.bfloat16, device=DEV)
+    kc = torch.randn(BH, cap, D, dtype=torch.bfloat16, device=DEV)
+    vc = torch.randn_like(kc)
+    secs = timeit(lambda: ops.attention_decode(q, kc, vc, None, cap,
+                                               D ** -0.5))
+    report("attn_decode", f"BH{BH} cap{cap} D{D}", secs,
+           bytes_=2 * BH * cap * D * 2)
+
+
 ALL = {"gemm": bench_gemm, "conv": bench_conv, "bn": bench_bn,
-       "colsum": bench_colsum, "ce": bench_ce, "attn": bench_attn}
+       "colsum": bench_colsum, "ce": bench_ce, "attn": bench_attn,
+       "bmm": bench_bmm, "softmax": bench_softmax, "ln": bench_ln,
+       "gn": bench_gn, "decode": bench_decode}
 
 if __name__ == "__main__":
     which = sys.argv[1:] or list(ALL)
