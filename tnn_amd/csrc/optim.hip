@@ -94,6 +94,10 @@ __global__ void k_adam_mt(const int64_t* __restrict__ desc,
   const bool vec4 = (end - off) % 4 == 0 && ((uintptr_t)p & 15) == 0 &&
                     ((uintptr_t)g & 15) == 0;
   if (vec4) {
+    // 2-deep unroll: two independent float4 groups' loads in flight per
+    // thread (the single-group loop measured 4.3 TB/s on the 5-stream
+    // read/write mix)
+#pragma unroll 2
     for (int64_t i4 = off / 4 + threadIdx.x; i4 * 4 < end; i4 += 256) {
       float4 mi4 = ((float4*)m)[i4];
       float4 vi4 = ((float4*)v)[i4];
