@@ -132,3 +132,25 @@ def test_hwinfo_topology_parse():
     assert hw.xgmi_peers(1) == [0, 2]
     order = hw.pipeline_order()
     assert order == [0, 1, 2]  # chain follows xGMI neighbors
+
+
+def test_visualize_profiler_entrypoint(tmp_path):
+    """visualizers/visualize_profiler.py renders a Profiler chrome trace."""
+    import subprocess, sys, os, time
+    from tnn_amd.utils.profiler import Profiler, EventType
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    prof = Profiler("test")
+    prof.start()
+    with prof.span(EventType.COMPUTE, "fwd"):
+        time.sleep(0.002)
+    with prof.span(EventType.COMMUNICATION, "allreduce"):
+        time.sleep(0.001)
+    prof.stop()
+    trace = tmp_path / "trace.json"
+    prof.export_chrome_trace(str(trace))
+    r = subprocess.run(
+        [sys.executable,
+         os.path.join(root, "visualizers", "visualize_profiler.py"),
+         str(trace)], capture_output=True, text=True, timeout=60)
+    assert r.returncode == 0, r.stderr
+    assert "fwd" in r.stdout and "span" in r.stdout
