@@ -389,6 +389,9 @@ __global__ void k_gemm_nt_z(const T* __restrict__ A, const T* __restrict__ B,
 // by k_splitk_fin_ep with the bias/act/residual epilogue).
 namespace nt32 {
 constexpr int BM2 = 128, BN2 = 64, BK2 = 64;
+#ifndef GEMM_REMAP_GM
+#define GEMM_REMAP_GM 8
+#endif
 
 // C/D map for 32x32x16 (guide: col = lane&31, row = (reg&3) + 8*(reg>>2)
 // + 4*(lane>>5), reg in [0,16))
@@ -409,7 +412,7 @@ __global__ void k_gemm_nt32(const T* __restrict__ A, const T* __restrict__ B,
   // 256 MiB L3 (vocab-head-sized operands measured +9%; L3-resident
   // shapes were flat-to-negative)
   if ((int64_t)M * N + (int64_t)N * K > (int64_t)48 * 1024 * 1024)
-    tile::tile_remap<8>(tm, tn);
+    tile::tile_remap<GEMM_REMAP_GM>(tm, tn);
   else {
     tm = blockIdx.x;
     tn = blockIdx.y;
