@@ -85,9 +85,11 @@ DEV void tile_remap_xcd(int& tm, int& tn) {
   const int nbx = gridDim.x, nby = gridDim.y;
   const int T = nbx * nby;
   const int id = blockIdx.y * nbx + blockIdx.x;
-  const int per = (T + 7) >> 3;
-  int lid = (id & 7) * per + (id >> 3);
-  if (lid >= T) lid = id;
+  // stripe permutation is a bijection only when 8 | T (every CNN-zoo
+  // grid: pow2 tiles); otherwise keep dispatch order — a clamped
+  // fallback here DUPLICATED one tile and dropped another per ragged
+  // stripe, silently corrupting C
+  const int lid = (T & 7) == 0 ? ((id & 7) * (T >> 3) + (id >> 3)) : id;
   const int per_group = GM * nby;
   const int group = lid / per_group;
   const int first_m = group * GM;
